@@ -1,0 +1,116 @@
+"""Batches (ref: src/data/batch.h, corpus_base.h SubBatch/CorpusBatch).
+
+A SubBatch holds the padded word indices and mask for one stream (source or
+target). The reference lays words out time-major ``[T, B]``; on MI355X the
+natural layout for batched GEMM/attention is batch-major ``[B, T]`` —
+we keep ``[B, T]`` tensors and provide the same statistics
+(batch_width/batch_size/words).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+import torch
+
+
+class SubBatch:
+    def __init__(self, indices: torch.Tensor, mask: torch.Tensor):
+        # indices: [B, T] long; mask: [B, T] float (1 for real token incl. EOS)
+        assert indices.shape == mask.shape
+        self.indices = indices
+        self.mask = mask
+
+    @classmethod
+    def from_sentences(cls, sents: Sequence[Sequence[int]], pad_id: int = 0) -> "SubBatch":
+        B = len(sents)
+        T = max((len(s) for s in sents), default=0)
+        idx = torch.full((B, T), pad_id, dtype=torch.long)
+        mask = torch.zeros((B, T), dtype=torch.float32)
+        for b, s in enumerate(sents):
+            if len(s):
+                idx[b, : len(s)] = torch.as_tensor(s, dtype=torch.long)
+                mask[b, : len(s)] = 1.0
+        return cls(idx, mask)
+
+    @property
+    def batch_size(self) -> int:
+        return self.indices.shape[0]
+
+    @property
+    def batch_width(self) -> int:
+        return self.indices.shape[1]
+
+    def batch_words(self) -> int:
+        return int(self.mask.sum().item())
+
+    def to(self, device, non_blocking: bool = True) -> "SubBatch":
+        return SubBatch(self.indices.to(device, non_blocking=non_blocking),
+                        self.mask.to(device, non_blocking=non_blocking))
+
+    def narrow(self, start: int, size: int) -> "SubBatch":
+        return SubBatch(self.indices[start:start + size].clone(),
+                        self.mask[start:start + size].clone())
+
+
+class CorpusBatch:
+    """A tuple of SubBatches; index 0 = source, last = target."""
+
+    def __init__(self, sub_batches: List[SubBatch],
+                 guided_alignment: Optional[torch.Tensor] = None,
+                 data_weights: Optional[torch.Tensor] = None,
+                 sentence_ids: Optional[List[int]] = None):
+        self.sub = sub_batches
+        self.guided_alignment = guided_alignment
+        self.data_weights = data_weights
+        self.sentence_ids = sentence_ids or []
+
+    @property
+    def source(self) -> SubBatch:
+        return self.sub[0]
+
+    @property
+    def target(self) -> SubBatch:
+        return self.sub[-1]
+
+    def size(self) -> int:
+        return self.sub[0].batch_size
+
+    def words(self, stream: int = -1) -> int:
+        return self.sub[stream].batch_words()
+
+    def source_words(self) -> int:
+        return self.sub[0].batch_words()
+
+    def target_words(self) -> int:
+        return self.sub[-1].batch_words()
+
+    def to(self, device, non_blocking: bool = True) -> "CorpusBatch":
+        b = CorpusBatch([sb.to(device, non_blocking) for sb in self.sub],
+                        None if self.guided_alignment is None
+                        else self.guided_alignment.to(device),
+                        None if self.data_weights is None
+                        else self.data_weights.to(device),
+                        self.sentence_ids)
+        return b
+
+    def split(self, n: int) -> List["CorpusBatch"]:
+        """Split into up to n batches along the sentence dimension
+        (ref: CorpusBatch::split used by SyncGraphGroup)."""
+        B = self.size()
+        n = min(n, B)
+        out: List[CorpusBatch] = []
+        base = B // n
+        rem = B % n
+        start = 0
+        for i in range(n):
+            size = base + (1 if i < rem else 0)
+            if size == 0:
+                continue
+            subs = [sb.narrow(start, size) for sb in self.sub]
+            ids = self.sentence_ids[start:start + size] if self.sentence_ids else []
+            dw = None
+            if self.data_weights is not None:
+                dw = self.data_weights[start:start + size].clone()
+            out.append(CorpusBatch(subs, None, dw, ids))
+            start += size
+        return out

@@ -1,0 +1,214 @@
+"""NN building blocks (ref: src/layers/generic.h, embedding.h, output.h,
+models/transformer.h LayerAttention/LayerFFN helpers).
+
+Conventions:
+- weights are stored marian-style as ``[in, out]`` (x @ W + b) so the .npz
+  round-trips bit-for-bit with the reference format;
+- pre/postprocess strings 'd' (dropout), 'a' (residual add), 'n' (LayerNorm)
+  follow --transformer-preprocess/-postprocess.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+
+
+class Dense(nn.Module):
+    """x @ W + b with marian [in, out] weight layout."""
+
+    def __init__(self, dim_in: int, dim_out: int, bias: bool = True):
+        super().__init__()
+        self.W = nn.Parameter(torch.empty(dim_in, dim_out))
+        self.b = nn.Parameter(torch.zeros(dim_out)) if bias else None
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        # marian default init: glorot uniform (graph/node_initializers.cpp)
+        nn.init.xavier_uniform_(self.W)
+        if self.b is not None:
+            nn.init.zeros_(self.b)
+
+    def forward(self, x):
+        y = torch.matmul(x, self.W.to(x.dtype))
+        if self.b is not None:
+            y = y + self.b.to(x.dtype)
+        return y
+
+
+class LayerNorm(nn.Module):
+    """LayerNorm with marian eps and parameter names (scale/bias).
+    Forward dispatches to the fused HIP kernel on device (ops.layer_norm)."""
+
+    def __init__(self, dim: int, eps: float = 1e-9):
+        super().__init__()
+        self.scale = nn.Parameter(torch.ones(dim))
+        self.bias = nn.Parameter(torch.zeros(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return ops.layer_norm(x, self.scale, self.bias, self.eps)
+
+
+def sinusoidal_position_embeddings(length: int, dim: int, start: int = 0,
+                                   device=None, dtype=torch.float32):
+    """ref: graph/node_initializers.cpp sinusoidalPositionEmbeddings
+    (marian uses the Vaswani layout: first half sin, second half cos)."""
+    pos = torch.arange(start, start + length, dtype=torch.float32, device=device)
+    num_timescales = dim // 2
+    log_timescale_increment = math.log(10000.0) / max(1, num_timescales - 1)
+    inv_timescales = torch.exp(torch.arange(num_timescales, dtype=torch.float32,
+                                            device=device)
+                               * -log_timescale_increment)
+    scaled = pos[:, None] * inv_timescales[None, :]
+    emb = torch.cat([torch.sin(scaled), torch.cos(scaled)], dim=1)
+    if dim % 2:
+        emb = torch.nn.functional.pad(emb, (0, 1))
+    return emb.to(dtype)
+
+
+class Embedding(nn.Module):
+    """Word embedding, scaled by sqrt(dim) with sinusoidal positions added
+    (ref: models/transformer.h apply/addPositionalEmbeddings +
+    layers/embedding.cpp)."""
+
+    def __init__(self, vocab_size: int, dim: int, dropout: float = 0.0,
+                 train_positions: bool = False, max_pos: int = 2048):
+        super().__init__()
+        self.E = nn.Parameter(torch.empty(vocab_size, dim))
+        self.dim = dim
+        self.dropout = dropout
+        self.train_positions = train_positions
+        if train_positions:
+            self.Wpos = nn.Parameter(torch.empty(max_pos, dim))
+            nn.init.normal_(self.Wpos, std=dim ** -0.5)
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        nn.init.normal_(self.E, std=self.dim ** -0.5)
+
+    def words(self, indices: torch.Tensor, compute_dtype=torch.float32):
+        """Scaled word embeddings only (marian adds positions AFTER the
+        decoder's target shift, so the two halves are separable)."""
+        x = torch.nn.functional.embedding(indices, self.E.to(compute_dtype))
+        return x * math.sqrt(self.dim)
+
+    def add_positions(self, x: torch.Tensor, start_pos: int = 0):
+        T = x.shape[-2]
+        if self.train_positions:
+            pos = self.Wpos[start_pos:start_pos + T].to(x.dtype)
+        else:
+            pos = sinusoidal_position_embeddings(T, self.dim, start_pos,
+                                                 device=x.device,
+                                                 dtype=x.dtype)
+        return x + pos
+
+    def forward(self, indices: torch.Tensor, start_pos: int = 0,
+                compute_dtype=torch.float32):
+        return self.add_positions(self.words(indices, compute_dtype), start_pos)
+
+
+class PrePostProcessor(nn.Module):
+    """Applies a marian pre/postprocess op string: 'd'ropout, 'a'dd residual,
+    'n'ormalize (ref: transformer.h preProcess/postProcess)."""
+
+    def __init__(self, ops_str: str, dim: int, dropout: float):
+        super().__init__()
+        self.ops_str = ops_str
+        self.dropout = dropout
+        if "n" in ops_str:
+            self.ln = LayerNorm(dim)
+
+    def forward(self, x, residual=None):
+        for op in self.ops_str:
+            if op == "d":
+                if self.dropout > 0 and self.training:
+                    x = torch.nn.functional.dropout(x, self.dropout, True)
+            elif op == "a":
+                if residual is not None:
+                    x = x + residual
+            elif op == "n":
+                x = self.ln(x)
+            else:
+                raise ValueError(f"Unknown pre/postprocess op '{op}'")
+        return x
+
+
+class MultiHeadAttention(nn.Module):
+    """Multi-head attention with marian parameter names Wq/bq..Wo/bo
+    (ref: transformer.h LayerAttention/MultiHead). The core SDPA runs
+    through ops.attention -> fused CDNA4 kernel on device."""
+
+    def __init__(self, dim: int, heads: int, dropout: float = 0.0):
+        super().__init__()
+        assert dim % heads == 0
+        self.dim = dim
+        self.heads = heads
+        self.dh = dim // heads
+        self.Wq = Dense(dim, dim)
+        self.Wk = Dense(dim, dim)
+        self.Wv = Dense(dim, dim)
+        self.Wo = Dense(dim, dim)
+        self.dropout = dropout
+
+    def _split(self, x):
+        B, T, _ = x.shape
+        return x.view(B, T, self.heads, self.dh).transpose(1, 2)  # [B,H,T,dh]
+
+    def _join(self, x):
+        B, H, T, dh = x.shape
+        return x.transpose(1, 2).reshape(B, T, H * dh)
+
+    def forward(self, query, keys, values, mask=None, causal=False,
+                cache: Optional[dict] = None, cache_key: str = ""):
+        """cache: decode-time KV cache dict. For self-attention the new K/V
+        are appended ({cache_key}_k/v); for cross-attention K/V are computed
+        once and reused (ref: transformer.h cache_ keyed tensors)."""
+        q = self._split(self.Wq(query))
+        if cache is not None and cache_key + "_static" in cache:
+            k, v = cache[cache_key + "_k"], cache[cache_key + "_v"]
+        else:
+            k = self._split(self.Wk(keys))
+            v = self._split(self.Wv(values))
+            if cache is not None:
+                if cache_key.endswith("context"):   # cross-attn: static
+                    cache[cache_key + "_k"] = k
+                    cache[cache_key + "_v"] = v
+                    cache[cache_key + "_static"] = True
+                else:                               # self-attn: append
+                    if cache_key + "_k" in cache:
+                        k = torch.cat([cache[cache_key + "_k"], k], dim=2)
+                        v = torch.cat([cache[cache_key + "_v"], v], dim=2)
+                    cache[cache_key + "_k"] = k
+                    cache[cache_key + "_v"] = v
+        drop = self.dropout if self.training else 0.0
+        out = ops.attention(q, k, v, mask, causal=causal, dropout_p=drop)
+        return self.Wo(self._join(out))
+
+
+_ACT = {"relu": torch.nn.functional.relu,
+        "swish": torch.nn.functional.silu,
+        "gelu": torch.nn.functional.gelu}
+
+
+class FFN(nn.Module):
+    """Transformer FFN block, marian names W1/b1/W2/b2
+    (ref: transformer.h LayerFFN)."""
+
+    def __init__(self, dim: int, dim_ffn: int, activation: str = "relu",
+                 dropout: float = 0.0):
+        super().__init__()
+        self.W1 = Dense(dim, dim_ffn)
+        self.W2 = Dense(dim_ffn, dim)
+        self.act = _ACT[activation]
+        self.dropout = dropout
+
+    def forward(self, x):
+        h = self.act(self.W1(x))
+        if self.dropout > 0 and self.training:
+            h = torch.nn.functional.dropout(h, self.dropout, True)
+        return self.W2(h)

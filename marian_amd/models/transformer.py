@@ -1,0 +1,343 @@
+"""Transformer encoder-decoder (ref: src/models/transformer.h,
+models/encoder_decoder.{h,cpp}, models/states.h).
+
+Marian semantics reproduced:
+- embeddings scaled by sqrt(d) + sinusoidal positions;
+- --transformer-preprocess/-postprocess op strings per block (post-norm
+  'dan' default, prenorm via 'n'/'da'+top 'n');
+- decoder input = target shifted right, position 0 a zero embedding
+  (marian's `shift` op pads with zeros — there is no BOS token);
+- decode-time KV cache: self-attn K/V appended per step, encoder-side
+  cross-attn K/V computed once;
+- tied embeddings (--tied-embeddings-all / --tied-embeddings /
+  --tied-embeddings-src) with output projection = Wemb^T;
+- marian .npz parameter naming (encoder_l1_self_Wq ...) via
+  marian_items()/load_marian_items().
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..common.options import Options
+from .layers import (Dense, Embedding, FFN, LayerNorm, MultiHeadAttention,
+                     PrePostProcessor)
+
+NEG_INF = -1e9  # marian uses additive log-masks with large negative value
+
+
+def _log_mask(mask: torch.Tensor, dtype) -> torch.Tensor:
+    """[B, T] 0/1 mask -> [B, 1, 1, T] additive log-mask."""
+    return ((1.0 - mask[:, None, None, :]) * NEG_INF).to(dtype)
+
+
+class EncoderLayer(nn.Module):
+    def __init__(self, opt):
+        super().__init__()
+        d, drop = opt["dim-emb"], opt.get("transformer-dropout", 0.0)
+        self.pre_self = PrePostProcessor(opt["transformer-preprocess"], d, drop)
+        self.post_self = PrePostProcessor(opt["transformer-postprocess"], d, drop)
+        self.pre_ffn = PrePostProcessor(opt["transformer-preprocess"], d, drop)
+        self.post_ffn = PrePostProcessor(opt["transformer-postprocess"], d, drop)
+        self.self_attn = MultiHeadAttention(
+            d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
+        self.ffn = FFN(d, opt["transformer-dim-ffn"],
+                       opt.get("transformer-ffn-activation", "relu"),
+                       opt.get("transformer-dropout-ffn", 0.0))
+
+    def forward(self, x, mask):
+        h = self.pre_self(x)
+        h = self.self_attn(h, h, h, mask=mask)
+        x = self.post_self(h, residual=x)
+        h = self.pre_ffn(x)
+        h = self.ffn(h)
+        return self.post_ffn(h, residual=x)
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, opt):
+        super().__init__()
+        d, drop = opt["dim-emb"], opt.get("transformer-dropout", 0.0)
+        dim_ffn = opt.get("transformer-decoder-dim-ffn", 0) or opt["transformer-dim-ffn"]
+        self.pre_self = PrePostProcessor(opt["transformer-preprocess"], d, drop)
+        self.post_self = PrePostProcessor(opt["transformer-postprocess"], d, drop)
+        self.pre_ctx = PrePostProcessor(opt["transformer-preprocess"], d, drop)
+        self.post_ctx = PrePostProcessor(opt["transformer-postprocess"], d, drop)
+        self.pre_ffn = PrePostProcessor(opt["transformer-preprocess"], d, drop)
+        self.post_ffn = PrePostProcessor(opt["transformer-postprocess"], d, drop)
+        self.self_attn = MultiHeadAttention(
+            d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
+        self.context_attn = MultiHeadAttention(
+            d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
+        self.ffn = FFN(d, dim_ffn, opt.get("transformer-ffn-activation", "relu"),
+                       opt.get("transformer-dropout-ffn", 0.0))
+
+    def forward(self, x, enc_out, src_mask, causal: bool,
+                cache: Optional[dict] = None, layer_id: int = 0):
+        h = self.pre_self(x)
+        h = self.self_attn(h, h, h, causal=causal, cache=cache,
+                           cache_key=f"l{layer_id}_self")
+        x = self.post_self(h, residual=x)
+        h = self.pre_ctx(x)
+        h = self.context_attn(h, enc_out, enc_out, mask=src_mask, cache=cache,
+                              cache_key=f"l{layer_id}_context")
+        x = self.post_ctx(h, residual=x)
+        h = self.pre_ffn(x)
+        h = self.ffn(h)
+        return self.post_ffn(h, residual=x)
+
+
+class TransformerModel(nn.Module):
+    """Encoder-decoder with marian checkpoint compatibility."""
+
+    def __init__(self, options: Options):
+        super().__init__()
+        self.opt = options
+        dims = options.list("dim-vocabs")
+        self.src_vocab_size = int(dims[0])
+        self.trg_vocab_size = int(dims[-1])
+        d = options["dim-emb"]
+        self.dim = d
+        drop = options.get("transformer-dropout", 0.0)
+
+        tied_all = options.get("tied-embeddings-all", False)
+        tied_src = options.get("tied-embeddings-src", False) or tied_all
+        tied_out = options.get("tied-embeddings", False) or tied_all
+        self.tied_all = tied_all
+        self.tied_src = tied_src
+        self.tied_out = tied_out
+        if tied_src and self.src_vocab_size != self.trg_vocab_size:
+            raise ValueError("tied source embeddings require equal vocab sizes")
+
+        self.src_embed = Embedding(
+            self.src_vocab_size, d, drop,
+            options.get("transformer-train-position-embeddings", False))
+        if tied_src:
+            self.trg_embed = self.src_embed
+        else:
+            self.trg_embed = Embedding(
+                self.trg_vocab_size, d, drop,
+                options.get("transformer-train-position-embeddings", False))
+
+        self.post_emb_enc = PrePostProcessor(
+            options.get("transformer-postprocess-emb", "d"), d, drop)
+        self.post_emb_dec = PrePostProcessor(
+            options.get("transformer-postprocess-emb", "d"), d, drop)
+
+        self.encoder = nn.ModuleList(
+            [EncoderLayer(options) for _ in range(options["enc-depth"])])
+        self.decoder = nn.ModuleList(
+            [DecoderLayer(options) for _ in range(options["dec-depth"])])
+
+        top = options.get("transformer-postprocess-top", "")
+        self.post_top_enc = PrePostProcessor(top, d, drop) if top else None
+        self.post_top_dec = PrePostProcessor(top, d, drop) if top else None
+
+        if not tied_out:
+            self.out_W = nn.Parameter(torch.empty(d, self.trg_vocab_size))
+            nn.init.xavier_uniform_(self.out_W)
+        self.out_b = None
+        if not options.get("output-omit-bias", False):
+            self.out_b = nn.Parameter(torch.zeros(self.trg_vocab_size))
+
+        self.compute_dtype = torch.float32
+
+    # ------------------------------------------------------------------
+    def set_compute_dtype(self, dtype):
+        self.compute_dtype = dtype
+
+    # ------------------------------------------------------------------
+    def encode(self, src_indices: torch.Tensor, src_mask: torch.Tensor):
+        dt = self.compute_dtype
+        x = self.src_embed(src_indices, compute_dtype=dt)
+        x = self.post_emb_enc(x)
+        x = x * src_mask.unsqueeze(-1).to(dt)
+        lmask = _log_mask(src_mask, dt)
+        for layer in self.encoder:
+            x = layer(x, lmask)
+        if self.post_top_enc is not None:
+            x = self.post_top_enc(x)
+        return x
+
+    # ------------------------------------------------------------------
+    def decode_train(self, trg_indices, trg_mask, enc_out, src_mask):
+        """Teacher-forced decoder pass. Decoder input is the target shifted
+        right by one with a zero embedding at t=0 (marian `shift`)."""
+        dt = self.compute_dtype
+        emb = self.trg_embed.words(trg_indices, compute_dtype=dt)
+        # marian `shift`: word embeddings move right one step (zero at t=0),
+        # positions are added afterwards, un-shifted
+        emb = torch.cat([torch.zeros_like(emb[:, :1]), emb[:, :-1]], dim=1)
+        emb = self.trg_embed.add_positions(emb, 0)
+        x = self.post_emb_dec(emb)
+        lmask = _log_mask(src_mask, dt)
+        for i, layer in enumerate(self.decoder):
+            x = layer(x, enc_out, lmask, causal=True, layer_id=i)
+        if self.post_top_dec is not None:
+            x = self.post_top_dec(x)
+        return self.output_logits(x)
+
+    # ------------------------------------------------------------------
+    def decode_step(self, prev_indices, enc_out, src_mask, cache: dict,
+                    step: int, shortlist: Optional[torch.Tensor] = None):
+        """One decode step. prev_indices: [B] token ids emitted at step-1
+        (ignored at step 0 -> zero embedding). Returns logits [B, V']."""
+        dt = self.compute_dtype
+        B = enc_out.shape[0]
+        if step == 0:
+            emb = torch.zeros(B, 1, self.dim, device=enc_out.device, dtype=dt)
+        else:
+            emb = self.trg_embed.words(prev_indices.view(B, 1), compute_dtype=dt)
+        # position of the CURRENT step (the word emitted at t-1 sits at t)
+        emb = self.trg_embed.add_positions(emb, step)
+        x = self.post_emb_dec(emb)
+        lmask = _log_mask(src_mask, dt)
+        for i, layer in enumerate(self.decoder):
+            x = layer(x, enc_out, lmask, causal=False, cache=cache, layer_id=i)
+        if self.post_top_dec is not None:
+            x = self.post_top_dec(x)
+        return self.output_logits(x[:, 0], shortlist)
+
+    # ------------------------------------------------------------------
+    def output_logits(self, x, shortlist: Optional[torch.Tensor] = None):
+        """Output projection; tied -> Wemb^T (ref: layers/output.cpp)."""
+        dt = x.dtype
+        if self.tied_out:
+            W = self.trg_embed.E.to(dt).t()  # [d, V]
+        else:
+            W = self.out_W.to(dt)
+        if shortlist is not None:
+            W = W.index_select(1, shortlist)
+        logits = torch.matmul(x, W)
+        if self.out_b is not None:
+            b = self.out_b.to(dt)
+            if shortlist is not None:
+                b = b.index_select(0, shortlist)
+            logits = logits + b
+        return logits
+
+    # ------------------------------------------------------------------
+    def forward(self, batch):
+        """Training criterion forward: returns logits [B, T_trg, V]."""
+        src, trg = batch.source, batch.target
+        enc = self.encode(src.indices, src.mask)
+        return self.decode_train(trg.indices, trg.mask, enc, src.mask)
+
+    # ==================================================================
+    # marian .npz parameter naming
+    # ==================================================================
+    def _name_map(self) -> Dict[str, torch.Tensor]:
+        """marian name -> tensor view (shared storage)."""
+        m: Dict[str, torch.Tensor] = {}
+        if self.tied_all:
+            m["Wemb"] = self.src_embed.E
+        else:
+            m["encoder_Wemb"] = self.src_embed.E
+            if not self.tied_src:
+                m["decoder_Wemb"] = self.trg_embed.E
+        if not self.tied_out:
+            m["decoder_ff_logit_out_W"] = self.out_W
+        if self.out_b is not None:
+            m["decoder_ff_logit_out_b"] = self.out_b
+
+        def attn(prefix: str, a: MultiHeadAttention, post: PrePostProcessor):
+            m[prefix + "_Wq"] = a.Wq.W
+            m[prefix + "_bq"] = a.Wq.b
+            m[prefix + "_Wk"] = a.Wk.W
+            m[prefix + "_bk"] = a.Wk.b
+            m[prefix + "_Wv"] = a.Wv.W
+            m[prefix + "_bv"] = a.Wv.b
+            m[prefix + "_Wo"] = a.Wo.W
+            m[prefix + "_bo"] = a.Wo.b
+            if hasattr(post, "ln"):
+                m[prefix + "_Wo_ln_scale"] = post.ln.scale
+                m[prefix + "_Wo_ln_bias"] = post.ln.bias
+
+        def ffn(prefix: str, f: FFN, post: PrePostProcessor):
+            m[prefix + "_W1"] = f.W1.W
+            m[prefix + "_b1"] = f.W1.b
+            m[prefix + "_W2"] = f.W2.W
+            m[prefix + "_b2"] = f.W2.b
+            if hasattr(post, "ln"):
+                m[prefix + "_ffn_ln_scale"] = post.ln.scale
+                m[prefix + "_ffn_ln_bias"] = post.ln.bias
+
+        for i, layer in enumerate(self.encoder, start=1):
+            attn(f"encoder_l{i}_self", layer.self_attn, layer.post_self)
+            ffn(f"encoder_l{i}_ffn", layer.ffn, layer.post_ffn)
+            # prenorm LNs live in the preprocessors
+            if hasattr(layer.pre_self, "ln"):
+                m[f"encoder_l{i}_self_Wq_ln_scale_pre"] = layer.pre_self.ln.scale
+                m[f"encoder_l{i}_self_Wq_ln_bias_pre"] = layer.pre_self.ln.bias
+            if hasattr(layer.pre_ffn, "ln"):
+                m[f"encoder_l{i}_ffn_W1_ln_scale_pre"] = layer.pre_ffn.ln.scale
+                m[f"encoder_l{i}_ffn_W1_ln_bias_pre"] = layer.pre_ffn.ln.bias
+        for i, layer in enumerate(self.decoder, start=1):
+            attn(f"decoder_l{i}_self", layer.self_attn, layer.post_self)
+            attn(f"decoder_l{i}_context", layer.context_attn, layer.post_ctx)
+            ffn(f"decoder_l{i}_ffn", layer.ffn, layer.post_ffn)
+            if hasattr(layer.pre_self, "ln"):
+                m[f"decoder_l{i}_self_Wq_ln_scale_pre"] = layer.pre_self.ln.scale
+                m[f"decoder_l{i}_self_Wq_ln_bias_pre"] = layer.pre_self.ln.bias
+            if hasattr(layer.pre_ctx, "ln"):
+                m[f"decoder_l{i}_context_Wq_ln_scale_pre"] = layer.pre_ctx.ln.scale
+                m[f"decoder_l{i}_context_Wq_ln_bias_pre"] = layer.pre_ctx.ln.bias
+            if hasattr(layer.pre_ffn, "ln"):
+                m[f"decoder_l{i}_ffn_W1_ln_scale_pre"] = layer.pre_ffn.ln.scale
+                m[f"decoder_l{i}_ffn_W1_ln_bias_pre"] = layer.pre_ffn.ln.bias
+        if self.post_top_enc is not None and hasattr(self.post_top_enc, "ln"):
+            m["encoder_top_ln_scale"] = self.post_top_enc.ln.scale
+            m["encoder_top_ln_bias"] = self.post_top_enc.ln.bias
+        if self.post_top_dec is not None and hasattr(self.post_top_dec, "ln"):
+            m["decoder_top_ln_scale"] = self.post_top_dec.ln.scale
+            m["decoder_top_ln_bias"] = self.post_top_dec.ln.bias
+        return m
+
+    def marian_items(self) -> Dict[str, np.ndarray]:
+        """Export parameters under marian names as float32 numpy arrays.
+        Biases and LN params are stored as [1, d] as marian does."""
+        out = {}
+        for name, t in self._name_map().items():
+            arr = t.detach().cpu().float().numpy()
+            if arr.ndim == 1:
+                arr = arr.reshape(1, -1)
+            out[name] = arr
+        return out
+
+    def load_marian_items(self, items: Dict[str, np.ndarray],
+                          strict: bool = True) -> None:
+        nm = self._name_map()
+        missing = [k for k in nm if k not in items]
+        extra = [k for k in items if k not in nm]
+        if strict and missing:
+            raise KeyError(f"Checkpoint missing parameters: {missing[:8]}...")
+        if strict and extra:
+            raise KeyError(f"Checkpoint has unknown parameters: {extra[:8]}...")
+        with torch.no_grad():
+            for name, t in nm.items():
+                if name not in items:
+                    continue
+                arr = torch.from_numpy(np.ascontiguousarray(items[name]))
+                arr = arr.reshape(t.shape)
+                t.copy_(arr.to(t.dtype))
+
+    # ------------------------------------------------------------------
+    def model_config(self) -> dict:
+        """The embedded special:model.yml config (subset marian stores)."""
+        keys = ["type", "dim-emb", "dim-vocabs", "enc-depth", "dec-depth",
+                "transformer-heads", "transformer-dim-ffn",
+                "transformer-decoder-dim-ffn", "transformer-ffn-depth",
+                "transformer-ffn-activation", "transformer-preprocess",
+                "transformer-postprocess", "transformer-postprocess-emb",
+                "transformer-postprocess-top", "transformer-decoder-autoreg",
+                "transformer-train-position-embeddings",
+                "tied-embeddings", "tied-embeddings-src",
+                "tied-embeddings-all", "output-omit-bias"]
+        cfg = {k: self.opt.get(k) for k in keys if self.opt.has(k)}
+        cfg["version"] = "marian_amd v0.1.0"
+        return cfg

@@ -1,0 +1,300 @@
+"""Optimizers over a bucketed flat parameter store
+(ref: src/optimizers/optimizers.{h,cpp}, training/graph_group_sync.cpp).
+
+MI355X-first design: model parameters live as views into a small number of
+contiguous per-bucket device buffers, and their .grad fields are views into
+matching contiguous gradient buffers — the reference's per-device flat
+param/grad arrangement, restructured into ~bucket_mb chunks so each bucket
+can be reduce-scattered over xGMI as soon as backward produces its last
+gradient (overlap), and so rank r's ZeRO-1 shard is slice r of every bucket.
+
+The sharded optimizer keeps fp32 master weights + Adam state only for its
+shard (ref: SyncGraphGroup sharded optimizer + castOptimizerType).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from .. import ops
+
+
+def _round_up(x: int, m: int) -> int:
+    return (x + m - 1) // m * m
+
+
+@dataclass
+class Bucket:
+    index: int
+    flat_param: torch.Tensor          # [padded]
+    flat_grad: torch.Tensor           # [padded]
+    params: List[Tuple[str, torch.nn.Parameter]] = field(default_factory=list)
+    used: int = 0
+
+    @property
+    def padded(self) -> int:
+        return self.flat_param.numel()
+
+    def shard_view(self, t: torch.Tensor, rank: int, world: int) -> torch.Tensor:
+        s = self.padded // world
+        return t[rank * s:(rank + 1) * s]
+
+
+class ParameterStore:
+    """Flattens a model's parameters into contiguous bucket buffers.
+
+    Parameters are packed in REVERSE of registration order (output side
+    first) so bucket k's gradients finish early in backward — the DDP-style
+    ordering that lets reduce-scatter overlap backward.
+    """
+
+    ALIGN = 256  # elements; keeps shard boundaries aligned
+
+    def __init__(self, model: torch.nn.Module, world_size: int = 1,
+                 bucket_mb: int = 0, grad_dtype: Optional[torch.dtype] = None):
+        named = [(n, p) for n, p in model.named_parameters() if p.requires_grad]
+        seen, uniq = set(), []
+        for n, p in named:
+            if id(p) in seen:
+                continue
+            seen.add(id(p))
+            uniq.append((n, p))
+        uniq.reverse()  # reverse registration order ~ backward readiness
+
+        self.world_size = max(1, world_size)
+        dev = uniq[0][1].device
+        dtype = uniq[0][1].dtype
+        self.device = dev
+        self.dtype = dtype
+        self.grad_dtype = grad_dtype or dtype
+        esize = torch.tensor([], dtype=dtype).element_size()
+        cap = (bucket_mb * 1024 * 1024 // esize) if bucket_mb > 0 else None
+
+        # greedy packing into buckets
+        groups: List[List[Tuple[str, torch.nn.Parameter]]] = [[]]
+        sizes = [0]
+        for n, p in uniq:
+            k = p.numel()
+            if cap and sizes[-1] > 0 and sizes[-1] + k > cap:
+                groups.append([])
+                sizes.append(0)
+            groups[-1].append((n, p))
+            sizes[-1] += k
+
+        self.buckets: List[Bucket] = []
+        self.offsets: Dict[str, Tuple[int, int, int]] = {}  # name -> (bucket, off, numel)
+        align = self.ALIGN * self.world_size
+        with torch.no_grad():
+            for bi, group in enumerate(groups):
+                total = sum(p.numel() for _, p in group)
+                padded = _round_up(max(total, align), align)
+                fp = torch.zeros(padded, device=dev, dtype=dtype)
+                fg = torch.zeros(padded, device=dev, dtype=self.grad_dtype)
+                b = Bucket(bi, fp, fg, group, total)
+                off = 0
+                for n, p in group:
+                    k = p.numel()
+                    fp[off:off + k].copy_(p.detach().reshape(-1))
+                    p.data = fp[off:off + k].view(p.shape)
+                    if p.dtype == self.grad_dtype:
+                        p.grad = fg[off:off + k].view(p.shape)
+                    self.offsets[n] = (bi, off, k)
+                    off += k
+                self.buckets.append(b)
+        self.total_padded = sum(b.padded for b in self.buckets)
+        self.shard_size = self.total_padded // self.world_size
+
+    # ------------------------------------------------------------------
+    def zero_grad(self):
+        for b in self.buckets:
+            b.flat_grad.zero_()
+
+    def grad_finite(self) -> torch.Tensor:
+        """1.0 if every gradient is finite (for dynamic cost scaling)."""
+        ok = torch.ones((), device=self.device, dtype=torch.float32)
+        for b in self.buckets:
+            ok = ok * torch.isfinite(b.flat_grad.float().sum()).float()
+        return ok
+
+    def param_buffers(self) -> List[torch.Tensor]:
+        return [b.flat_param for b in self.buckets]
+
+    def to_dtype_(self, dtype: torch.dtype):
+        if dtype == self.dtype:
+            return
+        self.dtype = dtype
+        for b in self.buckets:
+            b.flat_param = b.flat_param.to(dtype)
+            off = 0
+            for n, p in b.params:
+                k = p.numel()
+                p.data = b.flat_param[off:off + k].view(p.shape)
+                off += k
+
+
+class Sgd:
+    name = "sgd"
+
+    def __init__(self, params: int, device, **kw):
+        pass
+
+    def step(self, master, grad, lr, t):
+        master.add_(grad.float(), alpha=-lr)
+
+    def state_items(self):
+        return {}
+
+    def load_state_items(self, items):
+        pass
+
+
+class Adagrad:
+    name = "adagrad"
+
+    def __init__(self, params: int, device, eps: float = 1e-8, **kw):
+        self.gt = torch.zeros(params, device=device, dtype=torch.float32)
+        self.eps = eps
+
+    def step(self, master, grad, lr, t):
+        g = grad.float()
+        self.gt.addcmul_(g, g)
+        master.addcdiv_(g, self.gt.sqrt().add_(self.eps), value=-lr)
+
+    def state_items(self):
+        return {"adagrad_gt": self.gt}
+
+    def load_state_items(self, items):
+        if "adagrad_gt" in items:
+            self.gt.copy_(items["adagrad_gt"].to(self.gt.device))
+
+
+class Adam:
+    """Adam with bias correction (ref: optimizers.cpp Adam::updateImpl)."""
+
+    name = "adam"
+
+    def __init__(self, params: int, device, beta1: float = 0.9,
+                 beta2: float = 0.999, eps: float = 1e-8, wd: float = 0.0):
+        self.m = torch.zeros(params, device=device, dtype=torch.float32)
+        self.v = torch.zeros(params, device=device, dtype=torch.float32)
+        self.beta1, self.beta2, self.eps, self.wd = beta1, beta2, eps, wd
+
+    def step(self, master, grad, lr, t):
+        ops.adam_step(master, grad, self.m, self.v, lr,
+                      self.beta1, self.beta2, self.eps, self.wd, t)
+
+    def state_items(self):
+        return {"adam_mt": self.m, "adam_vt": self.v}
+
+    def load_state_items(self, items):
+        if "adam_mt" in items:
+            self.m.copy_(items["adam_mt"].to(self.m.device))
+        if "adam_vt" in items:
+            self.v.copy_(items["adam_vt"].to(self.v.device))
+
+
+def create_optimizer_core(options, n_params: int, device):
+    name = options.get("optimizer", "adam")
+    p = [float(x) for x in options.list("optimizer-params")]
+    if name == "adam":
+        kw = {}
+        if len(p) > 0:
+            kw["beta1"] = p[0]
+        if len(p) > 1:
+            kw["beta2"] = p[1]
+        if len(p) > 2:
+            kw["eps"] = p[2]
+        if len(p) > 3:
+            kw["wd"] = p[3]
+        return Adam(n_params, device, **kw)
+    if name == "adagrad":
+        kw = {"eps": p[0]} if p else {}
+        return Adagrad(n_params, device, **kw)
+    if name == "sgd":
+        return Sgd(n_params, device)
+    raise ValueError(f"Unknown optimizer {name}")
+
+
+class ShardedOptimizer:
+    """ZeRO-1 optimizer over rank r's slice of every bucket.
+
+    The shard gradient buffer is ONE contiguous fp-grad-dtype tensor laid out
+    bucket-by-bucket; reduce-scatter writes each bucket's shard directly into
+    its segment. The fp32 master + optimizer state share that layout.
+    (ref: training/graph_group_sync.cpp update + optimizers.cpp)
+    """
+
+    def __init__(self, store: ParameterStore, options, rank: int = 0):
+        self.store = store
+        self.rank = rank
+        self.world = store.world_size
+        dev = store.device
+        self.seg: List[Tuple[int, int]] = []  # per bucket (start, size) in shard buf
+        off = 0
+        for b in store.buckets:
+            s = b.padded // self.world
+            self.seg.append((off, s))
+            off += s
+        self.n = off
+        self.shard_grad = torch.zeros(self.n, device=dev, dtype=store.grad_dtype)
+        self.master = torch.zeros(self.n, device=dev, dtype=torch.float32)
+        with torch.no_grad():
+            for b, (o, s) in zip(store.buckets, self.seg):
+                self.master[o:o + s].copy_(
+                    b.shard_view(b.flat_param, rank, self.world).float())
+        self.core = create_optimizer_core(options, self.n, dev)
+        self.t = 0
+        self.clip_norm = float(options.get("clip-norm", 0.0) or 0.0)
+
+    # ------------------------------------------------------------------
+    def grad_shard_for_bucket(self, bi: int) -> torch.Tensor:
+        o, s = self.seg[bi]
+        return self.shard_grad[o:o + s]
+
+    def local_grad_sq_norm(self) -> torch.Tensor:
+        g = self.shard_grad.float()
+        return (g * g).sum()
+
+    def step(self, lr: float, grad_scale: float = 1.0,
+             grad_norm: Optional[float] = None) -> None:
+        self.t += 1
+        g = self.shard_grad.float()
+        if grad_scale != 1.0:
+            g = g / grad_scale
+        if self.clip_norm > 0 and grad_norm is not None \
+                and grad_norm > self.clip_norm:
+            g = g * (self.clip_norm / grad_norm)
+        self.core.step(self.master, g, lr, self.t)
+        # write updated shard back into each bucket's param slice
+        with torch.no_grad():
+            for b, (o, s) in zip(self.store.buckets, self.seg):
+                b.shard_view(b.flat_param, self.rank, self.world).copy_(
+                    self.master[o:o + s].to(self.store.dtype))
+
+    # ------------------------------------------------------------------
+    # checkpoint (ref: model.npz.optimizer.npz via gatherState/scatterState)
+    def state_items(self) -> Dict[str, np.ndarray]:
+        out = {"master_parameters": self.master.cpu().numpy(),
+               "optimizer_step": np.array([self.t], dtype=np.int64)}
+        for k, v in self.core.state_items().items():
+            out[k] = v.cpu().numpy()
+        return out
+
+    def load_state_items(self, items: Dict[str, np.ndarray]):
+        if "master_parameters" in items:
+            self.master.copy_(torch.from_numpy(
+                np.ascontiguousarray(items["master_parameters"])).to(self.master.device))
+            with torch.no_grad():
+                for b, (o, s) in zip(self.store.buckets, self.seg):
+                    b.shard_view(b.flat_param, self.rank, self.world).copy_(
+                        self.master[o:o + s].to(self.store.dtype))
+        if "optimizer_step" in items:
+            self.t = int(np.asarray(items["optimizer_step"]).reshape(-1)[0])
+        self.core.load_state_items(
+            {k: torch.from_numpy(np.ascontiguousarray(v))
+             for k, v in items.items()
+             if k.startswith(("adam_", "adagrad_"))})

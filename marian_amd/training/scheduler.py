@@ -1,0 +1,168 @@
+"""Scheduler: LR schedule, progress display, save/validate triggers
+(ref: src/training/scheduler.{h,cpp}).
+
+Display line format follows the reference::
+
+  Ep. 1 : Up. 1000 : Sen. 63,284 : Cost 7.20 : Time 12.3s : 31000.12 words/s : gNorm 1.2345
+"""
+from __future__ import annotations
+
+import math
+import time
+from typing import Optional
+
+from ..common.logging_utils import log_info, Timer
+from ..common.options import Options
+from .training_state import SchedulingParameter, TrainingState
+
+
+class Scheduler:
+    def __init__(self, options: Options, state: TrainingState):
+        self.opt = options
+        self.state = state
+        self.disp_freq = SchedulingParameter.parse(options.get("disp-freq", "1000u"))
+        self.save_freq = SchedulingParameter.parse(options.get("save-freq", "10000u"))
+        self.valid_freq = SchedulingParameter.parse(options.get("valid-freq", "10000u"))
+        self.disp_first = int(options.get("disp-first", 0) or 0)
+        self.lr_report = bool(options.get("lr-report", False))
+        self.after_epochs = int(options.get("after-epochs", 0) or 0)
+        self.after_batches = int(options.get("after-batches", 0) or 0)
+        self.early_stopping = int(options.get("early-stopping", 10) or 0)
+        self.base_lr = float(options.get("learn-rate", 1e-4))
+        self.warmup = SchedulingParameter.parse(options.get("lr-warmup", "0"))
+        self.warmup_start_rate = float(options.get("lr-warmup-start-rate", 0.0))
+        inv = options.list("lr-decay-inv-sqrt") or ["0"]
+        self.decay_inv_sqrt = SchedulingParameter.parse(inv[0])
+        self.timer = Timer()
+        self._start_batches = state.batches
+
+    # ------------------------------------------------------------------
+    def learning_rate(self) -> float:
+        """ref: scheduler.h getScheduledLRate: warmup then inv-sqrt decay,
+        both in units of updates (or labels with 't')."""
+        n = self.state.batches + 1
+        lr = self.base_lr
+        warm = self.warmup.n
+        if warm > 0 and n <= warm:
+            s = self.warmup_start_rate
+            progress = n / warm
+            lr = lr * ((1.0 - s) * progress + s) if s else lr * progress
+        if self.decay_inv_sqrt.n > 0:
+            start = self.decay_inv_sqrt.n
+            if n > start:
+                lr = lr * math.sqrt(start / n)
+        lr *= self.state.factor
+        self.state.eta = lr
+        return lr
+
+    # ------------------------------------------------------------------
+    def started(self):
+        log_info("Training started")
+
+    def finished(self):
+        log_info("Training finished")
+
+    # ------------------------------------------------------------------
+    def _hit(self, freq: SchedulingParameter) -> bool:
+        if not freq:
+            return False
+        if freq.unit == "u":
+            return self.state.batches % freq.n == 0
+        if freq.unit == "t":
+            prev = self.state.labels_total - self._last_labels
+            return (self.state.labels_total // freq.n) > ((self.state.labels_total - prev) // freq.n)
+        if freq.unit == "e":
+            return False  # handled at epoch boundary
+        return False
+
+    _last_labels = 0
+
+    # ------------------------------------------------------------------
+    def update(self, cost: float, n_sentences: int, n_labels: int,
+               grad_norm: Optional[float] = None) -> dict:
+        """Record one update; returns dict of due actions."""
+        st = self.state
+        st.batches += 1
+        st.batches_epoch += 1
+        st.samples += n_sentences
+        st.samples_epoch += n_sentences
+        st.labels_total += n_labels
+        st.cost_sum += cost
+        st.cost_count += n_labels
+        st.words_disp += n_labels
+        st.sentences_disp += n_sentences
+
+        actions = {"display": False, "save": False, "validate": False,
+                   "stop": False}
+
+        disp_due = (self.disp_freq and st.batches % max(1, self.disp_freq.n) == 0) \
+            or (st.batches <= self.disp_first)
+        if disp_due:
+            elapsed = self.timer.elapsed()
+            wps = st.words_disp / max(1e-9, elapsed)
+            avg_cost = st.cost_sum / max(1, st.cost_count)
+            line = (f"Ep. {st.epochs} : Up. {st.batches} : Sen. {st.samples:,} : "
+                    f"Cost {avg_cost:.8f} : Time {elapsed:.2f}s : {wps:.2f} words/s")
+            if grad_norm is not None:
+                line += f" : gNorm {grad_norm:.4f}"
+            if self.lr_report:
+                line += f" : L.r. {st.eta:.4e}"
+            log_info(line)
+            st.cost_sum, st.cost_count = 0.0, 0
+            st.words_disp, st.sentences_disp = 0, 0
+            self.timer.reset()
+            actions["display"] = True
+
+        if self.save_freq and self.save_freq.unit == "u" \
+                and st.batches % max(1, self.save_freq.n) == 0:
+            actions["save"] = True
+        if self.valid_freq and self.valid_freq.unit == "u" \
+                and st.batches % max(1, self.valid_freq.n) == 0:
+            actions["validate"] = True
+        if self.after_batches and st.batches >= self.after_batches:
+            actions["stop"] = True
+        if self.early_stopping and st.stalled >= self.early_stopping:
+            log_info("Early stopping after %d stalled validations", st.stalled)
+            actions["stop"] = True
+        return actions
+
+    # ------------------------------------------------------------------
+    def increase_epoch(self):
+        st = self.state
+        log_info("Seen %d samples in epoch %d", st.samples_epoch, st.epochs)
+        st.epochs += 1
+        st.batches_epoch = 0
+        st.samples_epoch = 0
+
+    def keep_going(self) -> bool:
+        st = self.state
+        if self.after_epochs and st.epochs > self.after_epochs:
+            return False
+        if self.after_batches and st.batches >= self.after_batches:
+            return False
+        if self.early_stopping and st.stalled >= self.early_stopping:
+            return False
+        return True
+
+    # ------------------------------------------------------------------
+    def validated(self, metric: str, value: float, lower_is_better: bool = True):
+        """Track best/stalled per metric (ref: validator.h trackBest)."""
+        st = self.state
+        v = st.validators.setdefault(metric, {"best": None, "stalled": 0})
+        best = v["best"]
+        improved = best is None or (value < best if lower_is_better else value > best)
+        if improved:
+            v["best"] = value
+            v["stalled"] = 0
+            if metric == self._first_metric():
+                st.stalled = 0
+        else:
+            v["stalled"] += 1
+            if metric == self._first_metric():
+                st.stalled += 1
+                st.max_stalled = max(st.max_stalled, st.stalled)
+        return improved
+
+    def _first_metric(self):
+        metrics = self.opt.list("valid-metrics") or ["cross-entropy"]
+        return metrics[0]

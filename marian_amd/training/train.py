@@ -1,0 +1,242 @@
+"""Train driver (ref: src/training/training.h Train<GraphGroup>::run,
+command/marian_train.cpp).
+
+Wires corpus -> BatchGenerator -> SyncGraphGroup -> Scheduler, with
+checkpoint/resume (4-file contract), SIGTERM graceful save
+(ref: common/signal_handling.cpp) and validation.
+
+Distributed: one process per GPU (torchrun); rank r consumes batch i with
+i % world == r — the one-process-per-GPU equivalent of the reference's
+batch->split(numDevices) round-robin.
+"""
+from __future__ import annotations
+
+import os
+import signal
+import sys
+from typing import List, Optional
+
+import torch
+
+from ..common.logging_utils import create_loggers, log_info, log_valid
+from ..common.options import Options
+from ..data.batch_generator import BatchGenerator, BatchStats
+from ..data.corpus import Corpus
+from ..data.vocab import Vocab
+from ..models.losses import CrossEntropyCost
+from ..models.model_factory import create_model
+from ..parallel.communicator import dist_env, init_process_group
+from .graph_group import SyncGraphGroup
+from .scheduler import Scheduler
+from .training_state import TrainingState
+
+_sigterm_seen = False
+
+
+def _on_sigterm(signum, frame):
+    global _sigterm_seen
+    _sigterm_seen = True
+    log_info("Received SIGTERM; will save and exit after the current batch")
+
+
+def load_or_create_vocabs(options: Options) -> List[Vocab]:
+    paths = options.list("vocabs")
+    train_sets = options.list("train-sets")
+    if not paths:
+        paths = [ts + ".yml" for ts in train_sets]
+        options.set("vocabs", paths)
+    vocabs = []
+    dim_vocabs = options.list("dim-vocabs") or [0, 0]
+    for i, p in enumerate(paths):
+        max_size = int(dim_vocabs[i]) if i < len(dim_vocabs) else 0
+        if os.path.exists(p):
+            v = Vocab.load(p, max_size=max_size)
+        else:
+            log_info("Building vocabulary from %s", train_sets[i])
+            v = Vocab.create_from_corpus([train_sets[i]], max_size=max_size)
+            v.save(p)
+        vocabs.append(v)
+    return vocabs
+
+
+class Train:
+    def __init__(self, options: Options):
+        self.opt = options
+        create_loggers(options.get("log"), options.get("valid-log"),
+                       options.get("log-level", "info"),
+                       options.get("quiet", False))
+        seed = int(options.get("seed", 0) or 0) or 1234
+        torch.manual_seed(seed)
+
+        self.rank, self.local_rank, self.world = dist_env()
+        cpu = int(options.get("cpu-threads", 0) or 0) > 0 \
+            or not torch.cuda.is_available()
+        if cpu:
+            self.device = torch.device("cpu")
+            nthreads = int(options.get("cpu-threads", 0) or 0)
+            if nthreads > 0:
+                torch.set_num_threads(nthreads)
+        else:
+            self.device = torch.device("cuda", self.local_rank)
+            torch.cuda.set_device(self.device)
+        if self.world > 1:
+            init_process_group(device=self.device)
+
+        self.vocabs = load_or_create_vocabs(options)
+        dims = [len(v) for v in self.vocabs]
+        options.set("dim-vocabs", dims)
+
+        self.model = create_model(options)
+        self.graph_group = SyncGraphGroup(options, self.model, self.device,
+                                          self.rank, self.world)
+        self.state = TrainingState()
+        self.model_path = options.get("model", "model.npz")
+
+        # resume (ref: Train::run restore logic)
+        if not options.get("no-reload", False) and os.path.exists(self.model_path):
+            if self.graph_group.load_checkpoint(self.model_path):
+                prog = self.model_path + ".progress.yml"
+                if os.path.exists(prog):
+                    self.state = TrainingState.load(prog)
+                log_info("Resumed training at epoch %d update %d",
+                         self.state.epochs, self.state.batches)
+
+        self.scheduler = Scheduler(options, self.state)
+
+    # ------------------------------------------------------------------
+    def _batch_generator(self, skip_batches: int = 0):
+        opts = self.opt
+        corpus = Corpus(opts.list("train-sets"), self.vocabs,
+                        max_length=int(opts.get("max-length", 50)),
+                        max_length_crop=bool(opts.get("max-length-crop", False)),
+                        tsv=bool(opts.get("tsv", False)),
+                        tsv_fields=int(opts.get("tsv-fields", 0) or 0),
+                        shuffle_in_ram=bool(opts.get("shuffle-in-ram", False)),
+                        seed=int(opts.get("seed", 0) or 1234))
+        corpus.epoch = self.state.epochs - 1
+        shuffle = opts.get("shuffle", "data") != "none" \
+            and not opts.get("no-shuffle", False)
+        gen = BatchGenerator(
+            corpus.iter_epoch(shuffle=shuffle),
+            mini_batch=int(opts.get("mini-batch", 64)),
+            mini_batch_words=int(opts.get("mini-batch-words", 0) or 0),
+            maxi_batch=int(opts.get("maxi-batch", 100)),
+            maxi_batch_sort=opts.get("maxi-batch-sort", "trg"),
+            mini_batch_fit_table=self._fit_table())
+        return gen
+
+    def _fit_table(self) -> Optional[BatchStats]:
+        if not self.opt.get("mini-batch-fit", False):
+            return None
+        # measured fit table; on CPU or small models fall back to the
+        # token-budget heuristic (collectStats parity is a GPU feature)
+        from .batch_stats_fit import collect_stats
+        try:
+            return collect_stats(self.opt, self.graph_group)
+        except Exception as e:
+            log_info("mini-batch-fit stats collection unavailable (%s); "
+                     "using --mini-batch-words fallback", e)
+            return None
+
+    # ------------------------------------------------------------------
+    def run(self):
+        signal.signal(signal.SIGTERM, _on_sigterm)
+        self.scheduler.started()
+        gg = self.graph_group
+        sched = self.scheduler
+        first_metric_validator = self._make_validators()
+
+        while sched.keep_going() and not _sigterm_seen:
+            gen = self._batch_generator()
+            got_batches = False
+            for i, batch in enumerate(gen):
+                if self.world > 1 and i % self.world != self.rank:
+                    continue
+                got_batches = True
+                gg.set_lr(sched.learning_rate())
+                lv = gg.update(batch)
+                if lv is None:
+                    continue
+                n_sents = getattr(lv, "sentences", batch.size())
+                actions = sched.update(
+                    float(lv.ce_sum) / max(1, lv.label_count),
+                    n_sents, lv.label_count,
+                    grad_norm=getattr(lv, "grad_norm", None))
+                if actions["save"]:
+                    self.save()
+                if actions["validate"]:
+                    self._validate(first_metric_validator)
+                if actions["stop"] or _sigterm_seen:
+                    break
+            if not got_batches:
+                log_info("Corpus produced no batches; check --train-sets")
+                break
+            if not sched.keep_going() or _sigterm_seen:
+                break
+            sched.increase_epoch()
+        self.save()
+        sched.finished()
+
+    # ------------------------------------------------------------------
+    def save(self, suffix: str = ""):
+        path = self.model_path
+        if suffix:
+            base, ext = os.path.splitext(path)
+            path = f"{base}{suffix}{ext}"
+        cfg = self.model.model_config()
+        self.graph_group.save_checkpoint(path, cfg, self.state)
+        if self.rank == 0:
+            # external config (model.npz.yml), ref: io row in SURVEY §2
+            with open(path + ".yml", "w") as f:
+                f.write(self.opt.to_yaml())
+            if not self.opt.get("overwrite", False):
+                itr = f".iter{self.state.batches}"
+                base, ext = os.path.splitext(self.model_path)
+                iter_path = f"{base}{itr}{ext}"
+                from ..common import io_npz
+                io_npz.save_items(iter_path, self.model.marian_items(), cfg)
+            log_info("Saved model to %s", path)
+
+    # ------------------------------------------------------------------
+    def _make_validators(self):
+        if not self.opt.list("valid-sets"):
+            return None
+        from .validators import create_validators
+        return create_validators(self.opt, self.vocabs)
+
+    def _validate(self, validators):
+        if validators is None:
+            return
+        for v in validators:
+            value = v.run(self.model, self.device,
+                          self.graph_group.compute_dtype)
+            improved = self.scheduler.validated(v.name, value,
+                                                v.lower_is_better)
+            st = self.state.validators.get(v.name, {})
+            log_valid(
+                f"Ep. {self.state.epochs} : Up. {self.state.batches} : "
+                f"{v.name} : {value:.6f} : "
+                + ("new best" if improved else
+                   f"stalled {st.get('stalled', 0)} times (last best: "
+                   f"{st.get('best', 0.0):.6f})"))
+            if improved and self.opt.get("keep-best", False) and self.rank == 0:
+                base, ext = os.path.splitext(self.model_path)
+                from ..common import io_npz
+                io_npz.save_items(f"{base}.best-{v.name}{ext}",
+                                  self.model.marian_items(),
+                                  self.model.model_config())
+
+
+def main_train(argv=None):
+    from ..common.config_parser import ConfigParser, MODE_TRAINING
+    parser = ConfigParser(MODE_TRAINING)
+    opts = parser.parse_options(argv if argv is not None else sys.argv[1:])
+    if opts.get("dump-config"):
+        print(parser.dump_config(opts, opts.get("dump-config")))
+        return 0
+    if opts.get("version"):
+        from .. import __version__
+        print(__version__)
+        return 0
+    Train(opts).run()
+    return 0

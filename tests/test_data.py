@@ -1,0 +1,107 @@
+import os
+
+import pytest
+import torch
+
+from marian_amd.data.batch_generator import BatchGenerator, BatchStats
+from marian_amd.data.corpus import Corpus
+from marian_amd.data.synthetic import synthetic_batch, write_synthetic_corpus
+from marian_amd.data.vocab import Vocab
+
+
+@pytest.fixture
+def tiny_corpus(tmp_path):
+    src, trg = str(tmp_path / "c.src"), str(tmp_path / "c.trg")
+    write_synthetic_corpus(src, trg, n_sents=200, vocab_size=50,
+                           min_len=2, max_len=20, seed=7)
+    return src, trg
+
+
+def test_vocab_create_encode_decode(tiny_corpus):
+    src, _ = tiny_corpus
+    v = Vocab.create_from_corpus([src])
+    assert v.token2id["</s>"] == 0
+    assert v.token2id["<unk>"] == 1
+    ids = v.encode("w2 w3 nonexistent-token")
+    assert ids[-1] == v.eos_id
+    assert v.unk_id in ids
+    text = v.decode(v.encode("w2 w3", add_eos=True))
+    assert text == "w2 w3"
+
+
+def test_vocab_save_load_text(tiny_corpus, tmp_path):
+    src, _ = tiny_corpus
+    v = Vocab.create_from_corpus([src])
+    p = str(tmp_path / "v.txt")
+    v.save(p)
+    v2 = Vocab.load(p)
+    assert len(v2) == len(v)
+    assert v2.encode("w2 w3") == v.encode("w2 w3")
+
+
+def test_corpus_max_length_filter(tiny_corpus):
+    src, trg = tiny_corpus
+    v = Vocab.create_from_corpus([src, trg])
+    c = Corpus([src, trg], [v, v], max_length=5, max_length_crop=False)
+    for _, streams in c.iter_epoch(shuffle=False):
+        assert all(len(s) <= 6 for s in streams)  # +1 for EOS
+
+    c2 = Corpus([src, trg], [v, v], max_length=5, max_length_crop=True)
+    n2 = sum(1 for _ in c2.iter_epoch(shuffle=False))
+    n1 = sum(1 for _ in c.iter_epoch(shuffle=False))
+    assert n2 >= n1  # crop keeps all sentences
+
+
+def test_batch_generator_shapes_and_sorting(tiny_corpus):
+    src, trg = tiny_corpus
+    v = Vocab.create_from_corpus([src, trg])
+    c = Corpus([src, trg], [v, v], max_length=30)
+    gen = BatchGenerator(c.iter_epoch(shuffle=False), mini_batch=16,
+                         maxi_batch=10, maxi_batch_sort="trg")
+    batches = list(gen)
+    assert sum(b.size() for b in batches) == 200
+    for b in batches:
+        assert b.source.indices.shape == b.source.mask.shape
+        assert b.target.indices.shape[0] == b.source.indices.shape[0]
+        # mask marks exactly the non-pad positions
+        lens = b.target.mask.sum(1)
+        assert (lens > 0).all()
+
+
+def test_batch_generator_token_sizing(tiny_corpus):
+    src, trg = tiny_corpus
+    v = Vocab.create_from_corpus([src, trg])
+    c = Corpus([src, trg], [v, v], max_length=30)
+    gen = BatchGenerator(c.iter_epoch(shuffle=False), mini_batch=9999,
+                         mini_batch_words=100, maxi_batch=10)
+    for b in gen:
+        assert b.size() * b.target.batch_width <= 120  # soft bound
+
+
+def test_batch_split():
+    b = synthetic_batch(10, 7, 9, 50)
+    parts = b.split(3)
+    assert sum(p.size() for p in parts) == 10
+    assert [p.size() for p in parts] == [4, 3, 3]
+
+
+def test_batch_stats_lookup():
+    st = BatchStats()
+    st.add(10, 10, 128)
+    st.add(20, 20, 64)
+    st.add(30, 30, 32)
+    assert st.find(5, 5) == 128
+    assert st.find(15, 12) == 64
+    assert st.find(25, 25) == 32
+    assert st.find(99, 99) == 32  # falls back to smallest
+
+
+def test_shuffle_deterministic(tiny_corpus):
+    src, trg = tiny_corpus
+    v = Vocab.create_from_corpus([src, trg])
+    c1 = Corpus([src, trg], [v, v], shuffle_in_ram=True, seed=5)
+    c2 = Corpus([src, trg], [v, v], shuffle_in_ram=True, seed=5)
+    l1 = [pos for pos, _ in c1.iter_epoch()]
+    l2 = [pos for pos, _ in c2.iter_epoch()]
+    assert l1 == l2
+    assert l1 != sorted(l1)

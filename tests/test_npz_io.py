@@ -1,0 +1,33 @@
+import numpy as np
+
+from marian_amd.common import io_npz
+
+
+def test_npz_roundtrip_with_config(tmp_path):
+    path = str(tmp_path / "model.npz")
+    items = {
+        "encoder_l1_self_Wq": np.random.rand(8, 8).astype(np.float32),
+        "decoder_ff_logit_out_b": np.random.rand(1, 32).astype(np.float32),
+    }
+    cfg = {"type": "transformer", "dim-emb": 8, "enc-depth": 1}
+    io_npz.save_items(path, items, config=cfg)
+
+    loaded, loaded_cfg = io_npz.load_items(path)
+    assert set(loaded.keys()) == set(items.keys())
+    for k in items:
+        np.testing.assert_array_equal(loaded[k], items[k])
+    assert loaded_cfg == cfg
+
+
+def test_npz_readable_by_plain_numpy(tmp_path):
+    """The special config item must be a plain int8 array inside a standard
+    .npz (the marian on-disk contract)."""
+    path = str(tmp_path / "model.npz")
+    io_npz.save_items(path, {"W": np.zeros((2, 2), np.float32)},
+                      config={"a": 1})
+    with np.load(path) as z:
+        assert "special:model.yml" in z.files
+        raw = z["special:model.yml"]
+        assert raw.dtype == np.int8
+        text = raw.tobytes().rstrip(b"\0").decode()
+        assert "a: 1" in text
