@@ -1,0 +1,164 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: transformer-big en-de synchronous data-parallel
+training (BASELINE.json configs #2/#3), measured as whole-job SOURCE
+tokens/sec.
+
+Single GPU:   python bench.py --gpus 1 --steps 20 --warmup 5
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Synthetic parallel-corpus data, random-init weights (no network for real
+data), bf16 compute with fp32 master (>= reference precision), full training
+step: fwd + bwd + RCCL reduce-scatter + sharded Adam + all-gather.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+from marian_amd.data.synthetic import synthetic_batch
+from marian_amd.models.model_factory import create_model
+from marian_amd.parallel.communicator import dist_env, init_process_group
+from marian_amd.training.graph_group import SyncGraphGroup
+
+MODELS = {
+    "transformer-big": dict(dim=1024, ffn=4096, heads=16, enc=6, dec=6),
+    "transformer-base": dict(dim=512, ffn=2048, heads=8, enc=6, dec=6),
+    "transformer-deep": dict(dim=1024, ffn=4096, heads=16, enc=12, dec=12),
+    "transformer-tiny": dict(dim=128, ffn=256, heads=4, enc=1, dec=1),
+}
+
+
+def build_options(args, world):
+    m = MODELS[args.model]
+    argv = [
+        "--train-sets", "synthetic.src", "synthetic.trg",
+        "--type", "transformer",
+        "--dim-emb", str(m["dim"]),
+        "--transformer-dim-ffn", str(m["ffn"]),
+        "--transformer-heads", str(m["heads"]),
+        "--enc-depth", str(m["enc"]), "--dec-depth", str(m["dec"]),
+        "--transformer-ffn-activation", "relu",
+        "--transformer-dropout", "0.0",
+        "--tied-embeddings-all",
+        "--cost-type", "ce-mean-words",
+        "--label-smoothing", "0.1",
+        "--learn-rate", "0.0002",
+        "--optimizer", "adam", "--optimizer-params", "0.9", "0.98", "1e-9",
+        "--clip-norm", "0.0",
+        "--max-length", str(args.seq_len),
+        "--seed", "1111",
+        "--precision", args.dtype, "float32",
+    ]
+    opts = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    opts.set("dim-vocabs", [args.vocab, args.vocab])
+    return opts
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", default="transformer-big", choices=MODELS)
+    p.add_argument("--batch", type=int, default=0,
+                   help="sentences per GPU (0 = auto)")
+    p.add_argument("--seq-len", type=int, default=64)
+    p.add_argument("--vocab", type=int, default=32000)
+    p.add_argument("--dtype", default="bf16",
+                   choices=["bf16", "fp16", "fp32", "float32", "bfloat16"])
+    args = p.parse_args()
+
+    rank, local_rank, world = dist_env()
+    on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        device = torch.device("cuda", local_rank)
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+    if world > 1:
+        init_process_group(device=device)
+        import torch.distributed as dist
+
+    if args.batch == 0:
+        # per-GPU batch: fill MI355X HBM comfortably; tiny on CPU
+        args.batch = 256 if on_gpu else 8
+        if args.model == "transformer-tiny":
+            args.batch = 64 if on_gpu else 8
+
+    if not on_gpu:
+        args.dtype = "fp32"  # CPU smoke runs use fp32; GPU runs bf16+
+    torch.manual_seed(1111 + rank)
+    opts = build_options(args, world)
+    model = create_model(opts)
+    gg = SyncGraphGroup(opts, model, device, rank, world)
+    gg.set_lr(2e-4)
+
+    batch = synthetic_batch(args.batch, args.seq_len, args.seq_len,
+                            args.vocab, seed=1234 + rank, device=device)
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize(device)
+        if world > 1:
+            import torch.distributed as dist
+            dist.barrier()
+
+    for _ in range(args.warmup):
+        gg.update(batch)
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        gg.update(batch)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], device=device if on_gpu else "cpu",
+                         dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    src_tokens_per_step = args.batch * args.seq_len * world
+    value = src_tokens_per_step * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        dt = {"bf16": "bf16", "bfloat16": "bf16", "fp16": "fp16",
+              "fp32": "fp32", "float32": "fp32"}[args.dtype]
+        print(json.dumps({
+            "metric": "src tokens/sec",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": dt if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world,
+                "seq_len": args.seq_len,
+                "vocab": args.vocab,
+                "parallelism": f"dp{world}",
+            },
+        }))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

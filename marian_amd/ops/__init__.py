@@ -173,14 +173,19 @@ def topk(scores: torch.Tensor, k: int):
 # Fused Adam step (ref: the Element(...) Adam functor + castOptimizerType)
 # ---------------------------------------------------------------------------
 
-def adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd, step):
+def adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd, step,
+              grad_scale: float = 1.0):
+    """One fused Adam update; grad is multiplied by grad_scale first
+    (1/loss-scale x clip factor, folded in to avoid an extra pass)."""
     if _on_gpu(param32):
-        ext = hip_ext()
-        if ext is not None and hasattr(ext, "adam_step"):
-            ext.adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd, step)
-            return
+        ext = hip_ext(required=True)
+        ext.adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd, step,
+                      grad_scale)
+        return
     # torch reference
     g = grad.float()
+    if grad_scale != 1.0:
+        g = g * grad_scale
     m.mul_(beta1).add_(g, alpha=1 - beta1)
     v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
     mh = m / (1 - beta1 ** step)

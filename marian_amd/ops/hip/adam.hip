@@ -1,0 +1,63 @@
+// Fused Adam step for CDNA4 (gfx950).
+// ref semantics: src/optimizers/optimizers.cpp Adam::updateImpl (the
+// Element(...) fused functor). One elementwise pass over the ZeRO-1 shard:
+// fp32 master/state, bf16/fp32 gradient, bias-corrected update.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+using namespace mamd;
+
+namespace {
+
+template <typename GT>
+__global__ void adam_kernel(float* __restrict__ p,
+                            const GT* __restrict__ g,
+                            float* __restrict__ m,
+                            float* __restrict__ v,
+                            long n, float lr, float b1, float b2, float eps,
+                            float wd, float bc1, float bc2, float gscale) {
+  const long i0 = (long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n; i += stride) {
+    const float gi = to_f32(g[i]) * gscale;
+    float mi = b1 * m[i] + (1.f - b1) * gi;
+    float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+    m[i] = mi;
+    v[i] = vi;
+    const float mh = mi / bc1;
+    const float vh = vi / bc2;
+    float pi = p[i];
+    if (wd != 0.f) pi *= (1.f - lr * wd);
+    p[i] = pi - lr * mh / (sqrtf(vh) + eps);
+  }
+}
+
+}  // namespace
+
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, double lr, double b1, double b2, double eps,
+               double wd, int64_t t, double gscale) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(p.is_contiguous() && m.is_contiguous() && v.is_contiguous());
+  auto gc = g.contiguous();
+  const long n = p.numel();
+  const int block = 256;
+  const int grid = (int)std::min<long>((n + block - 1) / block, 4096);
+  const float bc1 = 1.f - powf((float)b1, (float)t);
+  const float bc2 = 1.f - powf((float)b2, (float)t);
+  auto stream = at::cuda::getCurrentHIPStream();
+#define RUN(GT)                                                              \
+  hipLaunchKernelGGL((adam_kernel<GT>), dim3(grid), dim3(block), 0, stream,  \
+                     p.data_ptr<float>(),                                    \
+                     reinterpret_cast<const GT*>(gc.data_ptr()),             \
+                     m.data_ptr<float>(), v.data_ptr<float>(), n, (float)lr, \
+                     (float)b1, (float)b2, (float)eps, (float)wd, bc1, bc2,  \
+                     (float)gscale)
+  if (gc.scalar_type() == torch::kBFloat16) RUN(__hip_bfloat16);
+  else if (gc.scalar_type() == torch::kHalf) RUN(__half);
+  else if (gc.scalar_type() == torch::kFloat32) RUN(float);
+  else TORCH_CHECK(false, "adam: unsupported grad dtype");
+#undef RUN
+}

@@ -1,0 +1,32 @@
+// pybind bindings for the marian_amd gfx950 kernels.
+#include <torch/extension.h>
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
+                                         torch::Tensor beta, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor x, torch::Tensor gamma,
+                                         torch::Tensor mean, torch::Tensor rstd,
+                                         torch::Tensor dy);
+torch::Tensor softmax_fwd(torch::Tensor x);
+torch::Tensor log_softmax_fwd(torch::Tensor x);
+torch::Tensor softmax_bwd(torch::Tensor y, torch::Tensor dy);
+torch::Tensor log_softmax_bwd(torch::Tensor y, torch::Tensor dy);
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels,
+                                  double eps, c10::optional<torch::Tensor> mask);
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
+                     torch::Tensor lse, torch::Tensor dloss, double eps,
+                     c10::optional<torch::Tensor> mask);
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, double lr, double b1, double b2, double eps,
+               double wd, int64_t t, double gscale);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (gfx950)");
+  m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (gfx950)");
+  m.def("softmax_fwd", &softmax_fwd, "row softmax forward");
+  m.def("log_softmax_fwd", &log_softmax_fwd, "row log-softmax forward");
+  m.def("softmax_bwd", &softmax_bwd, "row softmax backward");
+  m.def("log_softmax_bwd", &log_softmax_bwd, "row log-softmax backward");
+  m.def("ce_fwd", &ce_fwd, "fused label-smoothed CE forward");
+  m.def("ce_bwd", &ce_bwd, "fused label-smoothed CE backward");
+  m.def("adam_step", &adam_step, "fused Adam step");
+}

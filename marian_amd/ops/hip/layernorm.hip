@@ -1,0 +1,228 @@
+// LayerNorm forward + backward for CDNA4 (gfx950).
+// ref semantics: src/tensors/gpu/tensor_operators.cu gLNormalization /
+// gLayerNormalizationGrad — rebuilt MI355X-first: one row per 64-lane wave,
+// fp32 math, vectorized 8B/lane loads (guide G13: scalar bf16 is 2-2.5x
+// slower), dgamma/dbeta accumulated in registers across a wave's rows then
+// one atomicAdd pass (memory-bound op: target HBM BW).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+using namespace mamd;
+
+namespace {
+
+template <typename T, int VEC>
+__global__ void ln_fwd_kernel(const T* __restrict__ x,
+                              const T* __restrict__ gamma,
+                              const T* __restrict__ beta,
+                              T* __restrict__ y,
+                              float* __restrict__ mean_out,
+                              float* __restrict__ rstd_out,
+                              int rows, int d, float eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const int row = blockIdx.x * waves_per_block + wid;
+  if (row >= rows) return;
+  const T* xr = x + (size_t)row * d;
+  T* yr = y + (size_t)row * d;
+
+  float s = 0.f, ss = 0.f;
+  for (int c = lane * VEC; c < d; c += WAVE * VEC) {
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float v = to_f32(xr[c + j]);
+      s += v;
+      ss += v * v;
+    }
+  }
+  s = wave_sum(s);
+  ss = wave_sum(ss);
+  const float mu = s / d;
+  const float var = fmaxf(ss / d - mu * mu, 0.f);
+  const float rstd = rsqrtf(var + eps);
+
+  for (int c = lane * VEC; c < d; c += WAVE * VEC) {
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float v = (to_f32(xr[c + j]) - mu) * rstd;
+      yr[c + j] = from_f32<T>(v * to_f32(gamma[c + j]) + to_f32(beta[c + j]));
+    }
+  }
+  if (lane == 0) {
+    mean_out[row] = mu;
+    rstd_out[row] = rstd;
+  }
+}
+
+// backward: each WAVE walks a strided set of rows, keeping per-lane
+// dgamma/dbeta partials for its columns in registers (d <= MAXD), then one
+// atomicAdd per column at the end.
+template <typename T, int VEC, int MAXC>
+__global__ void ln_bwd_kernel(const T* __restrict__ x,
+                              const T* __restrict__ gamma,
+                              const float* __restrict__ mean,
+                              const float* __restrict__ rstd,
+                              const T* __restrict__ dy,
+                              T* __restrict__ dx,
+                              float* __restrict__ dgamma,
+                              float* __restrict__ dbeta,
+                              int rows, int d) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int waves_per_block = blockDim.x / WAVE;
+  const int wave_id = blockIdx.x * waves_per_block + wid;
+  const int n_waves = gridDim.x * waves_per_block;
+
+  float dg[MAXC], db[MAXC];
+#pragma unroll
+  for (int i = 0; i < MAXC; ++i) dg[i] = db[i] = 0.f;
+
+  for (int row = wave_id; row < rows; row += n_waves) {
+    const T* xr = x + (size_t)row * d;
+    const T* dyr = dy + (size_t)row * d;
+    T* dxr = dx + (size_t)row * d;
+    const float mu = mean[row], rs = rstd[row];
+
+    float h1 = 0.f, h2 = 0.f;
+    for (int c = lane * VEC, i = 0; c < d; c += WAVE * VEC) {
+#pragma unroll
+      for (int j = 0; j < VEC; ++j, ++i) {
+        const float xh = (to_f32(xr[c + j]) - mu) * rs;
+        const float g = to_f32(dyr[c + j]);
+        const float gg = g * to_f32(gamma[c + j]);
+        h1 += gg;
+        h2 += gg * xh;
+        dg[i] += g * xh;
+        db[i] += g;
+      }
+    }
+    h1 = wave_sum(h1) / d;
+    h2 = wave_sum(h2) / d;
+    for (int c = lane * VEC; c < d; c += WAVE * VEC) {
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        const float xh = (to_f32(xr[c + j]) - mu) * rs;
+        const float gg = to_f32(dyr[c + j]) * to_f32(gamma[c + j]);
+        dxr[c + j] = from_f32<T>(rs * (gg - h1 - xh * h2));
+      }
+    }
+  }
+  for (int c = lane * VEC, i = 0; c < d; c += WAVE * VEC) {
+#pragma unroll
+    for (int j = 0; j < VEC; ++j, ++i) {
+      atomicAdd(&dgamma[c + j], dg[i]);
+      atomicAdd(&dbeta[c + j], db[i]);
+    }
+  }
+}
+
+constexpr int kBlock = 256;
+
+template <typename T>
+void ln_fwd_launch(const torch::Tensor& x, const torch::Tensor& g,
+                   const torch::Tensor& b, torch::Tensor& y,
+                   torch::Tensor& mean, torch::Tensor& rstd, double eps) {
+  const int d = x.size(-1);
+  const int rows = x.numel() / d;
+  TORCH_CHECK(d % 4 == 0, "layer_norm: dim must be divisible by 4, got ", d);
+  const int waves = kBlock / WAVE;
+  const int grid = (rows + waves - 1) / waves;
+  auto stream = at::cuda::getCurrentHIPStream();
+  hipLaunchKernelGGL((ln_fwd_kernel<T, 4>), dim3(grid), dim3(kBlock), 0,
+                     stream,
+                     reinterpret_cast<const T*>(x.data_ptr()),
+                     reinterpret_cast<const T*>(g.data_ptr()),
+                     reinterpret_cast<const T*>(b.data_ptr()),
+                     reinterpret_cast<T*>(y.data_ptr()),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     rows, d, (float)eps);
+}
+
+template <typename T>
+void ln_bwd_launch(const torch::Tensor& x, const torch::Tensor& g,
+                   const torch::Tensor& mean, const torch::Tensor& rstd,
+                   const torch::Tensor& dy, torch::Tensor& dx,
+                   torch::Tensor& dgamma, torch::Tensor& dbeta) {
+  const int d = x.size(-1);
+  const int rows = x.numel() / d;
+  TORCH_CHECK(d % 4 == 0 && d <= 64 * 64,
+              "layer_norm bwd: dim must be /4 and <= 4096, got ", d);
+  const int waves = kBlock / WAVE;
+  // enough waves to fill the chip (256 CUs x 8 XCDs -> >> 256 workgroups),
+  // but not more than rows
+  int grid = (rows + waves - 1) / waves;
+  grid = std::min(grid, 4096);
+  auto stream = at::cuda::getCurrentHIPStream();
+#define LAUNCH_BWD(MAXC)                                                     \
+  hipLaunchKernelGGL((ln_bwd_kernel<T, 4, MAXC>), dim3(grid), dim3(kBlock),  \
+                     0, stream, reinterpret_cast<const T*>(x.data_ptr()),    \
+                     reinterpret_cast<const T*>(g.data_ptr()),               \
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),         \
+                     reinterpret_cast<const T*>(dy.data_ptr()),              \
+                     reinterpret_cast<T*>(dx.data_ptr()),                    \
+                     dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),      \
+                     rows, d)
+  const int per_lane = (d + WAVE - 1) / WAVE;  // columns per lane
+  if (per_lane <= 8) {
+    LAUNCH_BWD(8);
+  } else if (per_lane <= 16) {
+    LAUNCH_BWD(16);
+  } else if (per_lane <= 32) {
+    LAUNCH_BWD(32);
+  } else {
+    LAUNCH_BWD(64);
+  }
+#undef LAUNCH_BWD
+}
+
+}  // namespace
+
+// returns (y, mean, rstd)
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
+                                         torch::Tensor beta, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  auto y = torch::empty_like(x);
+  const int d = x.size(-1);
+  const int rows = x.numel() / d;
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({rows}, opts);
+  auto rstd = torch::empty({rows}, opts);
+  auto g = gamma.contiguous(), b = beta.contiguous();
+  if (x.scalar_type() == torch::kBFloat16) {
+    ln_fwd_launch<__hip_bfloat16>(x, g, b, y, mean, rstd, eps);
+  } else if (x.scalar_type() == torch::kHalf) {
+    ln_fwd_launch<__half>(x, g, b, y, mean, rstd, eps);
+  } else if (x.scalar_type() == torch::kFloat32) {
+    ln_fwd_launch<float>(x, g, b, y, mean, rstd, eps);
+  } else {
+    TORCH_CHECK(false, "layernorm: unsupported dtype");
+  }
+  return {y, mean, rstd};
+}
+
+// returns (dx, dgamma_f32, dbeta_f32)
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor x, torch::Tensor gamma,
+                                         torch::Tensor mean, torch::Tensor rstd,
+                                         torch::Tensor dy) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  auto dyc = dy.contiguous();
+  auto dx = torch::empty_like(x);
+  const int d = x.size(-1);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto dgamma = torch::zeros({d}, opts);
+  auto dbeta = torch::zeros({d}, opts);
+  auto g = gamma.contiguous();
+  if (x.scalar_type() == torch::kBFloat16) {
+    ln_bwd_launch<__hip_bfloat16>(x, g, mean, rstd, dyc, dx, dgamma, dbeta);
+  } else if (x.scalar_type() == torch::kHalf) {
+    ln_bwd_launch<__half>(x, g, mean, rstd, dyc, dx, dgamma, dbeta);
+  } else if (x.scalar_type() == torch::kFloat32) {
+    ln_bwd_launch<float>(x, g, mean, rstd, dyc, dx, dgamma, dbeta);
+  } else {
+    TORCH_CHECK(false, "layernorm: unsupported dtype");
+  }
+  return {dx, dgamma, dbeta};
+}

@@ -183,9 +183,10 @@ class Adam:
         self.v = torch.zeros(params, device=device, dtype=torch.float32)
         self.beta1, self.beta2, self.eps, self.wd = beta1, beta2, eps, wd
 
-    def step(self, master, grad, lr, t):
+    def step(self, master, grad, lr, t, grad_scale: float = 1.0):
         ops.adam_step(master, grad, self.m, self.v, lr,
-                      self.beta1, self.beta2, self.eps, self.wd, t)
+                      self.beta1, self.beta2, self.eps, self.wd, t,
+                      grad_scale)
 
     def state_items(self):
         return {"adam_mt": self.m, "adam_vt": self.v}
@@ -262,13 +263,19 @@ class ShardedOptimizer:
     def step(self, lr: float, grad_scale: float = 1.0,
              grad_norm: Optional[float] = None) -> None:
         self.t += 1
-        g = self.shard_grad.float()
-        if grad_scale != 1.0:
-            g = g / grad_scale
+        mult = 1.0 / grad_scale
         if self.clip_norm > 0 and grad_norm is not None \
                 and grad_norm > self.clip_norm:
-            g = g * (self.clip_norm / grad_norm)
-        self.core.step(self.master, g, lr, self.t)
+            mult *= self.clip_norm / grad_norm
+        if isinstance(self.core, Adam):
+            # single fused pass: scaling folded into the Adam kernel
+            self.core.step(self.master, self.shard_grad, lr, self.t,
+                           grad_scale=mult)
+        else:
+            g = self.shard_grad.float()
+            if mult != 1.0:
+                g = g * mult
+            self.core.step(self.master, g, lr, self.t)
         # write updated shard back into each bucket's param slice
         with torch.no_grad():
             for b, (o, s) in zip(self.store.buckets, self.seg):
