@@ -52,11 +52,11 @@ class Options:
         return float(self[key])
 
     def str(self, key: str) -> str:
-        v = self[key]
+        v = self.get(key)
         return "" if v is None else str(v)
 
     def list(self, key: str) -> list:
-        v = self[key]
+        v = self.get(key)
         if v is None:
             return []
         if isinstance(v, (list, tuple)):

@@ -63,7 +63,7 @@ def _on_gpu(t: torch.Tensor) -> bool:
 def layer_norm(x: torch.Tensor, scale: torch.Tensor, bias: torch.Tensor,
                eps: float = 1e-9) -> torch.Tensor:
     if _on_gpu(x):
-        from .layernorm import LayerNormFn
+        from ._layernorm import LayerNormFn
         return LayerNormFn.apply(x, scale, bias, eps)
     return torch.nn.functional.layer_norm(
         x.float(), (x.shape[-1],), scale.float(), bias.float(), eps).to(x.dtype)
@@ -84,7 +84,7 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     The 1/sqrt(Dh) scale is applied here.
     """
     if _on_gpu(q) and dropout_p == 0.0:
-        from .attention import fused_attention
+        from ._attention import fused_attention
         out = fused_attention(q, k, v, mask, causal)
         if out is not None:
             return out
@@ -112,14 +112,14 @@ def attention_torch(q, k, v, mask=None, causal=False, dropout_p=0.0):
 
 def softmax(x: torch.Tensor, dim: int = -1) -> torch.Tensor:
     if _on_gpu(x) and dim in (-1, x.dim() - 1):
-        from .softmax import SoftmaxFn
+        from ._softmax import SoftmaxFn
         return SoftmaxFn.apply(x)
     return torch.softmax(x.float(), dim=dim).to(x.dtype)
 
 
 def log_softmax(x: torch.Tensor, dim: int = -1) -> torch.Tensor:
     if _on_gpu(x) and dim in (-1, x.dim() - 1):
-        from .softmax import LogSoftmaxFn
+        from ._softmax import LogSoftmaxFn
         return LogSoftmaxFn.apply(x)
     return torch.log_softmax(x.float(), dim=dim).to(x.dtype)
 
@@ -138,7 +138,7 @@ def cross_entropy(logits: torch.Tensor, labels: torch.Tensor,
     reference's label-smoothing formulation (layers/loss.cpp).
     """
     if _on_gpu(logits):
-        from .cross_entropy import CrossEntropyFn
+        from ._cross_entropy import CrossEntropyFn
         return CrossEntropyFn.apply(logits, labels, label_smoothing, mask)
     return cross_entropy_torch(logits, labels, label_smoothing, mask)
 

@@ -1,0 +1,308 @@
+// Fused flash-style attention forward for CDNA4 (gfx950), bf16, Dh = 64.
+//
+// ref semantics: the reference composes attention from ProdBatched + mask +
+// gSoftmax + ProdBatched (src/models/transformer.h LayerAttention +
+// src/tensors/gpu/prod.cpp). Rebuilt here as ONE kernel per the MI355X
+// guide's transposed-fragment recipe:
+//
+//  - S^T = mfma_f32_32x32x16_bf16(A=K-tile, B=Q^T)  -> lane holds one
+//    q-COLUMN of S^T, so the softmax row-reduce (over keys) is 16 in-lane
+//    values + one __shfl_xor(32) with the partner half: no LDS round-trip.
+//  - online softmax (running m, l per q) entirely in registers; the O
+//    accumulator is kept TRANSPOSED (O^T fragments), so the per-q rescale
+//    factor is a per-lane scalar multiply.
+//  - P (C-layout) is repacked to the MFMA B-operand layout with
+//    v_cvt_pk_bf16_f32 + __builtin_amdgcn_permlane32_swap (guide T12).
+//  - V is staged through LDS TRANSPOSED (V^T image) so the PV A-fragment is
+//    one 16-byte ds_read per 8 keys.
+//  - masks are PREFIX masks (NMT pads at the end): kv_len per batch entry +
+//    optional causal, evaluated in-register - no mask tensor traffic.
+//
+// One wave (64 lanes) per 32-row q-block; grid = (ceil(Tq/32), B*H).
+// Saves per-row logsumexp for a later fused backward.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+using namespace mamd;
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+namespace {
+
+constexpr float NEG_BIG = -1e30f;
+
+// C-fragment row for reg r, half hi (mfma_f32_32x32x16_bf16):
+// row = (r&3) + 8*(r>>2) + 4*hi
+__device__ __forceinline__ int crow(int r, int hi) {
+  return (r & 3) + 8 * (r >> 2) + 4 * hi;
+}
+
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+// Repack a 32x32 S^T C-fragment (16 f32, already exp()ed) into the two
+// B-operand bf16x8 fragments of the PV mfma (k-chunks 0-15 and 16-31).
+// See guide T12: r = permlane32_swap(cvtpk(p[4i],p[4i+1]), cvtpk(p[4i+2],p[4i+3]))
+__device__ __forceinline__ void pack_p_to_b(const float p[16], bf16x8 out[2]) {
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {   // k-chunk: regs 8c..8c+7 = k rows 16c..16c+15
+    unsigned w[4];
+    // pairs of consecutive k rows within this chunk:
+    // regs (8c+0,8c+1) -> rows (16c+0,1)+4hi ; (8c+2,8c+3) -> (16c+2,3)+4hi
+    // regs (8c+4,8c+5) -> rows (16c+8,9)+4hi ; (8c+6,8c+7) -> (16c+10,11)+4hi
+    unsigned a01 = cvt_pk_bf16(p[8 * c + 0], p[8 * c + 1]);  // k 0,1 | 4,5
+    unsigned a23 = cvt_pk_bf16(p[8 * c + 2], p[8 * c + 3]);  // k 2,3 | 6,7
+    unsigned a89 = cvt_pk_bf16(p[8 * c + 4], p[8 * c + 5]);  // k 8,9 | 12,13
+    unsigned aAB = cvt_pk_bf16(p[8 * c + 6], p[8 * c + 7]);  // k 10,11 | 14,15
+    {
+      auto r = __builtin_amdgcn_permlane32_swap(a01, a89, false, false);
+      w[0] = r[0];  // lane<32: k0,1 ; lane>=32: k8,9
+      w[2] = r[1];  // lane<32: k4,5 ; lane>=32: k12,13
+    }
+    {
+      auto r = __builtin_amdgcn_permlane32_swap(a23, aAB, false, false);
+      w[1] = r[0];  // lane<32: k2,3 ; lane>=32: k10,11
+      w[3] = r[1];  // lane<32: k6,7 ; lane>=32: k14,15
+    }
+    // B fragment k-order: dwords k(0,1),(2,3),(4,5),(6,7) (+8 on upper half)
+    union {
+      unsigned u[4];
+      bf16x8 v;
+    } cvt;
+    cvt.u[0] = w[0];
+    cvt.u[1] = w[1];
+    cvt.u[2] = w[2];
+    cvt.u[3] = w[3];
+    out[c] = cvt.v;
+  }
+}
+
+// LDS V^T image: [D][KVBLK] with row pad to dodge bank conflicts.
+constexpr int KVBLK = 32;
+constexpr int QBLK = 32;
+constexpr int VT_PAD = 8;  // elements
+
+template <int D, bool CAUSAL, bool HAS_LEN>
+__global__ __launch_bounds__(64)
+void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
+                     const __hip_bfloat16* __restrict__ K,
+                     const __hip_bfloat16* __restrict__ V,
+                     const int* __restrict__ kv_lens,  // [B] or null
+                     __hip_bfloat16* __restrict__ O,
+                     float* __restrict__ LSE,          // [B*H*Tq] or null
+                     int B, int H, int Tq, int Tk, float scale) {
+  __shared__ __hip_bfloat16 vt[D][KVBLK + VT_PAD];
+  __shared__ float obounce[QBLK][D + 4];
+
+  const int lane = threadIdx.x;
+  const int hi = lane >> 5;     // half id
+  const int lq = lane & 31;     // q column this lane owns in S^T fragments
+  const int qtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / H;
+  const int q0 = qtile * QBLK;
+  if (q0 >= Tq) return;
+  const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
+
+  const size_t qoff = (size_t)bh * Tq * D;
+  const size_t koff = (size_t)bh * Tk * D;
+
+  // ---- load Q^T B-fragments (kept for the whole kernel) -----------------
+  // B-fragment for chunk c (d = 16c + (hi)*8 + j): lane needs
+  // Q[q0+lq][16c + hi*8 + j], j=0..7 -> one 16-byte load per chunk.
+  bf16x8 qfrag[D / 16];
+  {
+    const int qrow = min(q0 + lq, Tq - 1);
+    const __hip_bfloat16* qp = Q + qoff + (size_t)qrow * D;
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c)
+      qfrag[c] = *reinterpret_cast<const bf16x8*>(qp + 16 * c + 8 * hi);
+    // fold the 1/sqrt(d) scale into Q once
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        qfrag[c][j] = (__bf16)((float)qfrag[c][j] * scale);
+  }
+
+  // ---- running softmax state (per q = lq, identical in both halves) -----
+  float m_run = NEG_BIG;
+  float l_run = 0.f;
+  f32x16 ot[D / 32];  // O^T accumulators: d-tiles of 32 rows x 32 q cols
+#pragma unroll
+  for (int dt = 0; dt < D / 32; ++dt) ot[dt] = (f32x16)0.f;
+
+  const int kv_end = CAUSAL ? min(kv_len, q0 + QBLK) : kv_len;
+
+  for (int k0 = 0; k0 < kv_end; k0 += KVBLK) {
+    // ---- stage V^T tile into LDS (transpose in flight) ------------------
+    // read 8 consecutive k-rows x D cols coalesced; each lane holds 8 d's
+    // of one k; scatter-write transposed.
+    {
+      const int rows_per_pass = 64 / (D / 8);  // 8 lanes per row at D=64
+      for (int kk = 0; kk < KVBLK; kk += rows_per_pass) {
+        const int krow = k0 + kk + lane / (D / 8);
+        const int dcol = (lane % (D / 8)) * 8;
+        bf16x8 vv = (krow < kv_len)
+            ? *reinterpret_cast<const bf16x8*>(V + koff + (size_t)krow * D + dcol)
+            : (bf16x8)(__bf16)0.f;
+        const int klocal = kk + lane / (D / 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          vt[dcol + j][klocal] = (__hip_bfloat16)(float)vv[j];
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);  // lgkm+vm drain before ds_read
+    __builtin_amdgcn_s_barrier();
+
+    // ---- S^T = K . Q^T --------------------------------------------------
+    f32x16 st = (f32x16)0.f;
+    {
+      // A-fragment: lane holds K[k0 + lq][d = 16c + hi*8 + j] (row = key)
+      const int krow = k0 + lq;
+      const __hip_bfloat16* kp = K + koff + (size_t)min(krow, Tk - 1) * D;
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp + 16 * c + 8 * hi);
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            kf, qfrag[c], st, 0, 0, 0);
+      }
+    }
+
+    // ---- mask + per-q max of this tile ---------------------------------
+    float p[16];
+    float tile_max = NEG_BIG;
+    const int q_abs = q0 + lq;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int k_abs = k0 + crow(r, hi);
+      float v = st[r];
+      const bool dead = (k_abs >= kv_len) || (CAUSAL && k_abs > q_abs);
+      v = dead ? NEG_BIG : v;
+      p[r] = v;
+      tile_max = fmaxf(tile_max, v);
+    }
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32));
+
+    // ---- online rescale -------------------------------------------------
+    const float m_new = fmaxf(m_run, tile_max);
+    float alpha = (m_new > NEG_BIG * 0.5f) ? __expf(m_run - m_new) : 1.f;
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      p[r] = (p[r] <= NEG_BIG * 0.5f) ? 0.f : __expf(p[r] - m_new);
+      psum += p[r];
+    }
+    psum += __shfl_xor(psum, 32);
+    l_run = l_run * alpha + psum;
+    m_run = m_new;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
+
+    // ---- O^T += V^T . P^T ----------------------------------------------
+    bf16x8 pb[2];
+    pack_p_to_b(p, pb);
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt) {
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        // A-fragment: lane holds V^T[32*dt + lq][k = 16c + hi*8 + j]
+        const __hip_bfloat16* vp = &vt[32 * dt + lq][16 * c + 8 * hi];
+        bf16x8 va = *reinterpret_cast<const bf16x8*>(vp);
+        ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            va, pb[c], ot[dt], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_barrier();  // protect LDS V^T before next stage
+  }
+
+  // ---- epilogue: O = O^T / l, via LDS bounce ---------------------------
+  const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+  for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      obounce[lq][32 * dt + crow(r, hi)] = ot[dt][r] * inv_l;
+  __builtin_amdgcn_s_barrier();
+
+  // write rows coalesced: each lane writes 8 floats -> bf16x8 chunks
+  {
+    const int lanes_per_row = D / 8;          // 8 at D=64
+    const int rows_per_pass = 64 / lanes_per_row;
+    for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
+      const int qrow = q0 + qq + lane / lanes_per_row;
+      const int dcol = (lane % lanes_per_row) * 8;
+      if (qrow < Tq) {
+        bf16x8 out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          out[j] = (__bf16)obounce[qq + lane / lanes_per_row][dcol + j];
+        *reinterpret_cast<bf16x8*>(O + qoff + (size_t)qrow * D + dcol) = out;
+      }
+    }
+  }
+  if (LSE != nullptr && hi == 0) {
+    const int qrow = q0 + lq;
+    if (qrow < Tq)
+      LSE[(size_t)bh * Tq + qrow] = m_run + __logf(fmaxf(l_run, 1e-30f));
+  }
+}
+
+}  // namespace
+
+// q,k,v: [B, H, T, D] bf16 contiguous. kv_lens: int32 [B] or undefined.
+// returns (o, lse)
+std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
+                                         torch::Tensor v,
+                                         c10::optional<torch::Tensor> kv_lens,
+                                         bool causal, double scale,
+                                         bool need_lse) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
+  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
+  const int B = qc.size(0), H = qc.size(1), Tq = qc.size(2), D = qc.size(3);
+  const int Tk = kc.size(2);
+  TORCH_CHECK(kc.size(0) == B && kc.size(1) == H && kc.size(3) == D);
+  TORCH_CHECK(D == 64, "attention_fwd: only Dh=64 supported (got ", D, ")");
+  auto o = torch::empty_like(qc);
+  torch::Tensor lse;
+  float* lse_ptr = nullptr;
+  if (need_lse) {
+    lse = torch::empty({B, H, Tq}, qc.options().dtype(torch::kFloat32));
+    lse_ptr = lse.data_ptr<float>();
+  } else {
+    lse = torch::Tensor();
+  }
+  const int* lens_ptr = nullptr;
+  torch::Tensor lensc;
+  if (kv_lens.has_value() && kv_lens->defined()) {
+    lensc = kv_lens->to(torch::kInt32).contiguous();
+    TORCH_CHECK(lensc.numel() == B);
+    lens_ptr = lensc.data_ptr<int>();
+  }
+  dim3 grid((Tq + 31) / 32, B * H);
+  auto stream = at::cuda::getCurrentHIPStream();
+#define RUN(CAUSAL, HASLEN)                                                  \
+  hipLaunchKernelGGL((attn_fwd_kernel<64, CAUSAL, HASLEN>), grid, dim3(64),  \
+                     0, stream,                                              \
+                     reinterpret_cast<const __hip_bfloat16*>(qc.data_ptr()), \
+                     reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()), \
+                     reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()), \
+                     lens_ptr,                                               \
+                     reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),        \
+                     lse_ptr, B, H, Tq, Tk, (float)scale)
+  if (causal && lens_ptr) RUN(true, true);
+  else if (causal) RUN(true, false);
+  else if (lens_ptr) RUN(false, true);
+  else RUN(false, false);
+#undef RUN
+  return {o, lse};
+}

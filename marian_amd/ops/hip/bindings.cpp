@@ -18,6 +18,11 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, double lr, double b1, double b2, double eps,
                double wd, int64_t t, double gscale);
+std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
+                                         torch::Tensor v,
+                                         c10::optional<torch::Tensor> kv_lens,
+                                         bool causal, double scale,
+                                         bool need_lse);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (gfx950)");
@@ -29,4 +34,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_fwd", &ce_fwd, "fused label-smoothed CE forward");
   m.def("ce_bwd", &ce_bwd, "fused label-smoothed CE backward");
   m.def("adam_step", &adam_step, "fused Adam step");
+  m.def("attention_fwd", &attention_fwd,
+        "fused flash-style attention forward (bf16, Dh=64)");
 }
