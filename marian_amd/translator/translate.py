@@ -1,0 +1,186 @@
+"""Translate driver + service (ref: src/translator/translator.h
+Translate<BeamSearch> / TranslateService, output_collector.{h,cpp},
+output_printer.{h,cpp}, command/marian_decoder.cpp).
+"""
+from __future__ import annotations
+
+import os
+import sys
+from typing import Iterator, List, Optional
+
+import torch
+
+from ..common import io_npz
+from ..common.logging_utils import create_loggers, log_info
+from ..common.options import Options
+from ..data.batch_generator import BatchGenerator
+from ..data.corpus import TextInput
+from ..data.vocab import Vocab
+from ..models.model_factory import create_model
+from .beam_search import BeamSearch, ModelScorer, SentenceResult
+
+
+def load_model_for_inference(path: str, options: Options, device):
+    """Load a .npz checkpoint; the embedded special:model.yml config wins
+    over CLI architecture flags unless --ignore-model-config."""
+    items, cfg = io_npz.load_items(path)
+    opts = options.clone()
+    if cfg and not options.get("ignore-model-config", False):
+        for k, v in cfg.items():
+            if k != "version":
+                opts.set(k, v)
+    model = create_model(opts)
+    model.load_marian_items(items, strict=False)
+    model = model.to(device)
+    prec = (options.list("precision") or ["float32"])[0]
+    if prec in ("float16", "fp16"):
+        model = model.half()
+        model.set_compute_dtype(torch.float16)
+    elif prec in ("bfloat16", "bf16"):
+        model = model.to(torch.bfloat16)
+        model.set_compute_dtype(torch.bfloat16)
+    model.eval()
+    return model
+
+
+class OutputPrinter:
+    """Formats hypotheses (ref: translator/output_printer.{h,cpp})."""
+
+    def __init__(self, options: Options, vocab: Vocab):
+        self.vocab = vocab
+        self.n_best = bool(options.get("n-best", False))
+        self.word_scores = bool(options.get("word-scores", False))
+
+    def text(self, hyp) -> str:
+        return self.vocab.decode(hyp.tokens, remove_eos=True)
+
+    def print(self, sentence_id: int, result: SentenceResult) -> str:
+        if self.n_best:
+            lines = []
+            for h in result.nbest:
+                lines.append(f"{sentence_id} ||| {self.text(h)} ||| "
+                             f"F0= {h.score:.6f} ||| {h.normalized:.6f}")
+            return "\n".join(lines)
+        return self.text(result.best)
+
+
+class OutputCollector:
+    """Restores input order for out-of-order batch completion
+    (ref: translator/output_collector.{h,cpp})."""
+
+    def __init__(self, out_stream):
+        self.out = out_stream
+        self.next_id = 0
+        self.pending = {}
+
+    def write(self, sentence_id: int, text: str):
+        self.pending[sentence_id] = text
+        while self.next_id in self.pending:
+            print(self.pending.pop(self.next_id), file=self.out)
+            self.next_id += 1
+
+
+class Translator:
+    """Shared core used by marian-decoder and marian-server."""
+
+    def __init__(self, options: Options):
+        self.opt = options
+        self.device = (torch.device("cuda", 0) if torch.cuda.is_available()
+                       and int(options.get("cpu-threads", 0) or 0) == 0
+                       else torch.device("cpu"))
+        vocab_paths = options.list("vocabs")
+        if not vocab_paths:
+            raise ValueError("marian-decoder needs --vocabs")
+        self.src_vocab = Vocab.load(vocab_paths[0])
+        self.trg_vocab = Vocab.load(vocab_paths[-1])
+        self.models = [load_model_for_inference(m, options, self.device)
+                       for m in options.list("models")]
+        scorers = [ModelScorer(m, self.device) for m in self.models]
+        shortlist_gen = None
+        sl = options.list("shortlist")
+        if sl:
+            from ..data.shortlist import LexicalShortlistGenerator
+            shortlist_gen = LexicalShortlistGenerator(
+                sl[0], self.src_vocab, self.trg_vocab,
+                first_num=int(sl[1]) if len(sl) > 1 else 100,
+                best_num=int(sl[2]) if len(sl) > 2 else 100)
+        self.search = BeamSearch(
+            scorers,
+            beam_size=int(options.get("beam-size", 12)),
+            normalize=float(options.get("normalize", 0.0) or 0.0),
+            word_penalty=float(options.get("word-penalty", 0.0) or 0.0),
+            allow_unk=bool(options.get("allow-unk", False)),
+            n_best=self._nbest_size(),
+            max_length_factor=float(options.get("max-length-factor", 3.0)),
+            max_length=int(options.get("max-length", 1000)),
+            eos_id=self.trg_vocab.eos_id, unk_id=self.trg_vocab.unk_id,
+            weights=[float(w) for w in options.list("weights")] or None,
+            shortlist_generator=shortlist_gen)
+
+    def _nbest_size(self):
+        if self.opt.get("n-best", False):
+            return int(self.opt.get("beam-size", 12))
+        return 1
+
+    # ------------------------------------------------------------------
+    def translate_lines(self, lines: List[str]) -> List[SentenceResult]:
+        ti = TextInput(iter(lines), self.src_vocab,
+                       max_length=int(self.opt.get("max-length", 1000)),
+                       max_length_crop=True)
+        gen = BatchGenerator(
+            iter(ti),
+            mini_batch=int(self.opt.get("mini-batch", 1) or 1),
+            mini_batch_words=int(self.opt.get("mini-batch-words", 0) or 0),
+            maxi_batch=int(self.opt.get("maxi-batch", 1) or 1),
+            maxi_batch_sort=self.opt.get("maxi-batch-sort", "none"),
+            prefetch=False)
+        results: dict = {}
+        for batch in gen:
+            out = self.search.search(batch, self.device)
+            for sid, res in zip(batch.sentence_ids, out):
+                results[sid] = res
+        return [results[i] for i in range(len(results))]
+
+
+class TranslateService:
+    """In-process string->string translation (marian-server backend)."""
+
+    def __init__(self, options: Options):
+        self.tr = Translator(options)
+        self.printer = OutputPrinter(options, self.tr.trg_vocab)
+
+    def run(self, text: str) -> str:
+        lines = text.split("\n")
+        results = self.tr.translate_lines(lines)
+        return "\n".join(self.printer.print(i, r)
+                         for i, r in enumerate(results))
+
+
+def main_decoder(argv=None):
+    from ..common.config_parser import ConfigParser, MODE_TRANSLATION
+    parser = ConfigParser(MODE_TRANSLATION)
+    opts = parser.parse_options(argv if argv is not None else sys.argv[1:])
+    if opts.get("dump-config"):
+        print(parser.dump_config(opts, opts.get("dump-config")))
+        return 0
+    create_loggers(opts.get("log"), None, opts.get("log-level", "info"),
+                   opts.get("quiet", False) or opts.get("quiet-translation", False))
+    tr = Translator(opts)
+    printer = OutputPrinter(opts, tr.trg_vocab)
+
+    inputs = opts.list("input") or ["stdin"]
+    if inputs[0] == "stdin":
+        lines = [l.rstrip("\n") for l in sys.stdin]
+    else:
+        with open(inputs[0], "r", encoding="utf-8") as f:
+            lines = [l.rstrip("\n") for l in f]
+
+    out_path = opts.get("output", "stdout")
+    out = sys.stdout if out_path == "stdout" else open(out_path, "w")
+    collector = OutputCollector(out)
+    results = tr.translate_lines(lines)
+    for i, r in enumerate(results):
+        collector.write(i, printer.print(i, r))
+    if out is not sys.stdout:
+        out.close()
+    return 0
