@@ -1,12 +1,10 @@
-"""Fused CDNA4 attention dispatch.
+"""Fused CDNA4 attention dispatch + autograd.
 
-The hand-written flash-style MFMA kernel (ops/hip/attention.hip) covers
-bf16 Dh=64 with prefix-length masks (NMT's only mask shape) + causal.
+The hand-written flash-style MFMA kernels (ops/hip/attention.hip) cover
+bf16 Dh=64 with prefix-length masks (NMT's only mask shape) + causal, for
+both training (fused fwd+bwd with saved logsumexp) and inference.
 Unsupported shapes/dtypes return None and the caller runs the batched-GEMM
 composite (hipBLASLt GEMMs + the HIP softmax kernel).
-
-Until the fused backward lands, the fused path serves no-grad (decode /
-validation) calls only; training uses the composite.
 """
 from __future__ import annotations
 
@@ -34,22 +32,48 @@ def mask_to_lens(mask: Optional[torch.Tensor], B: int, Tk: int):
     return lens.to(torch.int32)
 
 
+class AttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, lens, causal):
+        ext = hip_ext(required=True)
+        scale = q.shape[-1] ** -0.5
+        qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
+        o, lse = ext.attention_fwd(qc, kc, vc, lens, causal, scale, True)
+        ctx.save_for_backward(qc, kc, vc, o, lse,
+                              lens if lens is not None else torch.Tensor())
+        ctx.causal = causal
+        ctx.scale = scale
+        ctx.has_lens = lens is not None
+        return o
+
+    @staticmethod
+    def backward(ctx, d_o):
+        ext = hip_ext(required=True)
+        q, k, v, o, lse, lens = ctx.saved_tensors
+        dq, dk32, dv32 = ext.attention_bwd(
+            q, k, v, o, d_o.contiguous(), lse,
+            lens if ctx.has_lens else None, ctx.causal, ctx.scale)
+        return dq, dk32.to(q.dtype), dv32.to(q.dtype), None, None
+
+
 def fused_attention(q, k, v, mask, causal) -> Optional[torch.Tensor]:
     ext = hip_ext(required=True)
     if not hasattr(ext, "attention_fwd"):
         return None
     if q.dtype != torch.bfloat16 or q.shape[-1] != 64:
         return None
-    if torch.is_grad_enabled() and (q.requires_grad or k.requires_grad
-                                    or v.requires_grad):
-        return None  # fused backward not wired yet -> composite
     B, H, Tq, D = q.shape
     Tk = k.shape[2]
+    if k.shape[1] != H:
+        return None
     lens = mask_to_lens(mask, B, Tk)
     if lens is None:
         return None
+    lens = lens if lens.numel() else None
+    if torch.is_grad_enabled() and (q.requires_grad or k.requires_grad
+                                    or v.requires_grad):
+        return AttentionFn.apply(q, k, v, lens, causal)
     scale = D ** -0.5
-    o, _lse = ext.attention_fwd(q, k, v,
-                                lens if lens.numel() else None,
-                                causal, scale, False)
+    o, _lse = ext.attention_fwd(q.contiguous(), k.contiguous(),
+                                v.contiguous(), lens, causal, scale, False)
     return o

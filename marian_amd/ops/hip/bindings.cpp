@@ -23,6 +23,11 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                                          c10::optional<torch::Tensor> kv_lens,
                                          bool causal, double scale,
                                          bool need_lse);
+std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
+                                         torch::Tensor v, torch::Tensor o,
+                                         torch::Tensor d_o, torch::Tensor lse,
+                                         c10::optional<torch::Tensor> kv_lens,
+                                         bool causal, double scale);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (gfx950)");
@@ -36,4 +41,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step", &adam_step, "fused Adam step");
   m.def("attention_fwd", &attention_fwd,
         "fused flash-style attention forward (bf16, Dh=64)");
+  m.def("attention_bwd", &attention_bwd,
+        "fused flash-style attention backward (bf16, Dh=64)");
 }

@@ -29,6 +29,31 @@ template <> __device__ __forceinline__ __half from_f32<__half>(float x) {
   return __float2half(x);
 }
 
+// ---- vectorized 4-element access (8B/lane for bf16/f16, 16B for f32) -----
+// hipcc does not auto-vectorize bf16 scalar loads (guide G13); these struct
+// copies compile to global_load_dwordx2/x4.
+template <typename T>
+struct alignas(4 * sizeof(T)) Pack4 {
+  T v[4];
+};
+
+template <typename T>
+__device__ __forceinline__ void load4_f32(const T* __restrict__ p,
+                                          float o[4]) {
+  Pack4<T> t = *reinterpret_cast<const Pack4<T>*>(p);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) o[j] = to_f32(t.v[j]);
+}
+
+template <typename T>
+__device__ __forceinline__ void store4_f32(T* __restrict__ p,
+                                           const float o[4]) {
+  Pack4<T> t;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) t.v[j] = from_f32<T>(o[j]);
+  *reinterpret_cast<Pack4<T>*>(p) = t;
+}
+
 // ---- wave reductions (64-wide) -------------------------------------------
 __device__ __forceinline__ float wave_sum(float v) {
 #pragma unroll

@@ -31,7 +31,24 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
     const T* xr = logits + (size_t)row * cols;
     const int64_t label = labels[row];
     float m = -INFINITY, s = 0.f, xsum = 0.f, xlab = 0.f;
-    for (int c = threadIdx.x; c < cols; c += blockDim.x) {
+    const int cols4 = (cols % 4 == 0) ? cols : 0;  // unaligned rows -> scalar path
+    for (int c = threadIdx.x * 4; c < cols4; c += blockDim.x * 4) {
+      float v4[4];
+      load4_f32(xr + c, v4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float v = v4[j];
+        xsum += v;
+        if (c + j == (int)label) xlab = v;
+        if (v > m) {
+          s = s * __expf(m - v) + 1.f;
+          m = v;
+        } else {
+          s += __expf(v - m);
+        }
+      }
+    }
+    for (int c = cols4 + threadIdx.x; c < cols; c += blockDim.x) {
       const float v = to_f32(xr[c]);
       xsum += v;
       if (c == (int)label) xlab = v;
@@ -73,11 +90,23 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits,
     const float l = lse[row];
     const float w = (mask ? mask[row] : 1.f) * dloss[row];
     const float smooth = eps / cols;
-    for (int c = threadIdx.x; c < cols; c += blockDim.x) {
+    const int cols4 = (cols % 4 == 0) ? cols : 0;  // unaligned rows -> scalar path
+    for (int c = threadIdx.x * 4; c < cols4; c += blockDim.x * 4) {
+      float v4[4];
+      load4_f32(xr + c, v4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float p = __expf(v4[j] - l);
+        p -= (c + j == (int)label) ? (1.f - eps) : 0.f;
+        v4[j] = (p - smooth) * w;
+      }
+      store4_f32(dxr + c, v4);
+    }
+    for (int c = cols4 + threadIdx.x; c < cols; c += blockDim.x) {
       float p = __expf(to_f32(xr[c]) - l);
       p -= (c == (int)label) ? (1.f - eps) : 0.f;
       p -= smooth;
-      dxr[c] = from_f32<T>(p * w);
+      dxr[c] = from_f32<T>((p)*w);
     }
   }
 }

@@ -20,9 +20,24 @@ __global__ void softmax_fwd_kernel(const T* __restrict__ x,
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + (size_t)row * cols;
     T* yr = y + (size_t)row * cols;
-    // online max+sum in one pass over the row
+    // online max+sum in one pass over the row (8B/lane vectorized)
     float m = -INFINITY, s = 0.f;
-    for (int c = threadIdx.x; c < cols; c += blockDim.x) {
+    const int cols4 = (cols % 4 == 0) ? cols : 0;  // unaligned rows -> scalar path
+    for (int c = threadIdx.x * 4; c < cols4; c += blockDim.x * 4) {
+      float v4[4];
+      load4_f32(xr + c, v4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float v = v4[j];
+        if (v > m) {
+          s = s * __expf(m - v) + 1.f;
+          m = v;
+        } else {
+          s += __expf(v - m);
+        }
+      }
+    }
+    for (int c = cols4 + threadIdx.x; c < cols; c += blockDim.x) {
       const float v = to_f32(xr[c]);
       if (v > m) {
         s = s * __expf(m - v) + 1.f;
@@ -37,7 +52,15 @@ __global__ void softmax_fwd_kernel(const T* __restrict__ x,
     const float gs = block_sum(s_scaled, smem);
     const float inv = 1.f / gs;
     const float lse = gm + __logf(gs);
-    for (int c = threadIdx.x; c < cols; c += blockDim.x) {
+    for (int c = threadIdx.x * 4; c < cols4; c += blockDim.x * 4) {
+      float v4[4];
+      load4_f32(xr + c, v4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        v4[j] = LOG ? (v4[j] - lse) : __expf(v4[j] - gm) * inv;
+      store4_f32(yr + c, v4);
+    }
+    for (int c = cols4 + threadIdx.x; c < cols; c += blockDim.x) {
       const float v = to_f32(xr[c]);
       yr[c] = from_f32<T>(LOG ? (v - lse) : __expf(v - gm) * inv);
     }
@@ -55,15 +78,32 @@ __global__ void softmax_bwd_kernel(const T* __restrict__ y,
     const T* dyr = dy + (size_t)row * cols;
     T* dxr = dx + (size_t)row * cols;
     float acc = 0.f;
-    for (int c = threadIdx.x; c < cols; c += blockDim.x) {
+    const int cols4 = (cols % 4 == 0) ? cols : 0;  // unaligned rows -> scalar path
+    for (int c = threadIdx.x * 4; c < cols4; c += blockDim.x * 4) {
+      float g4[4], y4[4];
+      load4_f32(dyr + c, g4);
+      load4_f32(yr + c, y4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc += LOG ? g4[j] : g4[j] * y4[j];
+    }
+    for (int c = cols4 + threadIdx.x; c < cols; c += blockDim.x) {
       const float g = to_f32(dyr[c]);
       acc += LOG ? g : g * to_f32(yr[c]);
     }
     const float tot = block_sum(acc, smem);
-    for (int c = threadIdx.x; c < cols; c += blockDim.x) {
+    // softmax: dx = (dy - sum(dy*y)) * y ; logsoftmax: dx = dy - exp(y)*sum(dy)
+    for (int c = threadIdx.x * 4; c < cols4; c += blockDim.x * 4) {
+      float g4[4], y4[4];
+      load4_f32(dyr + c, g4);
+      load4_f32(yr + c, y4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        g4[j] = LOG ? (g4[j] - __expf(y4[j]) * tot) : (g4[j] - tot) * y4[j];
+      store4_f32(dxr + c, g4);
+    }
+    for (int c = cols4 + threadIdx.x; c < cols; c += blockDim.x) {
       const float g = to_f32(dyr[c]);
       const float yv = to_f32(yr[c]);
-      // softmax: dx = (dy - sum(dy*y)) * y ; logsoftmax: dx = dy - exp(y)*sum(dy)
       dxr[c] = from_f32<T>(LOG ? (g - __expf(yv) * tot) : (g - tot) * yv);
     }
   }

@@ -234,3 +234,79 @@ def test_attention_fused_dispatch_inference():
         got = ops.attention(q, k, v, mask)
         want = ops.attention_torch(q, k, v, mask)
     assert (got.float() - want.float()).abs().max().item() < 3e-2
+
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("shape", [(2, 4, 32, 32), (2, 3, 64, 64),
+                                   (1, 2, 70, 50)])
+def test_attention_bwd_fused(ext, causal, shape):
+    """Fused attention backward vs fp32 autograd reference."""
+    torch.manual_seed(21)
+    B, H, Tq, Tk = shape
+    D = 64
+    from marian_amd.ops._attention import AttentionFn
+    q = torch.randn(B, H, Tq, D, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, H, Tk, D, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, H, Tk, D, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    lens = torch.randint(max(1, Tk // 2), Tk + 1, (B,), device=DEV,
+                         dtype=torch.int32)
+    o = AttentionFn.apply(q, k, v, lens, causal)
+    dy = torch.randn_like(o)
+    o.backward(dy)
+
+    # fp32 reference
+    qr = q.detach().float().clone().requires_grad_()
+    kr = k.detach().float().clone().requires_grad_()
+    vr = v.detach().float().clone().requires_grad_()
+    s = (qr @ kr.transpose(-1, -2)) * (D ** -0.5)
+    idx = torch.arange(Tk, device=DEV)
+    s = s.masked_fill(idx[None, None, None, :] >= lens[:, None, None, None],
+                      float("-inf"))
+    if causal:
+        cm = torch.ones(Tq, Tk, device=DEV, dtype=torch.bool).tril(0)
+        s = s.masked_fill(~cm[None, None], float("-inf"))
+    all_dead = torch.isinf(s).all(dim=-1)
+    p = torch.softmax(s, dim=-1)
+    p = torch.where(all_dead[..., None], torch.zeros_like(p), p)
+    oref = p @ vr
+    oref.backward(dy.float())
+
+    for got, want, name in [(q.grad, qr.grad, "dq"), (k.grad, kr.grad, "dk"),
+                            (v.grad, vr.grad, "dv")]:
+        err = (got.float() - want).abs().max().item()
+        ref = want.abs().max().item()
+        assert err < 0.05 * max(ref, 1.0), (name, err, ref)
+
+
+def test_attention_training_dispatch_matches_composite():
+    """ops.attention grad path: fused vs composite end-to-end gradients."""
+    from marian_amd import ops
+    torch.manual_seed(22)
+    B, H, T, D = 2, 4, 64, 64
+    q0 = torch.randn(B, H, T, D, device=DEV, dtype=torch.bfloat16)
+    k0 = torch.randn_like(q0)
+    v0 = torch.randn_like(q0)
+    mask01 = (torch.arange(T, device=DEV)[None, :]
+              < torch.tensor([40, 64], device=DEV)[:, None]).float()
+    mask = (1.0 - mask01[:, None, None, :]) * -1e9
+
+    def run(use_fused):
+        q = q0.clone().requires_grad_()
+        k = k0.clone().requires_grad_()
+        v = v0.clone().requires_grad_()
+        if use_fused:
+            out = ops.attention(q, k, v, mask)
+        else:
+            out = ops.attention_torch(q, k, v, mask)
+        out.float().pow(2).sum().backward()
+        return out.detach(), q.grad, k.grad, v.grad
+
+    o1, dq1, dk1, dv1 = run(True)
+    o2, dq2, dk2, dv2 = run(False)
+    for a, b in [(o1, o2), (dq1, dq2), (dk1, dk2), (dv1, dv2)]:
+        err = (a.float() - b.float()).abs().max().item()
+        scale = b.float().abs().max().item()
+        assert err < 0.06 * max(scale, 1.0), (err, scale)
