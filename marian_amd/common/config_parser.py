@@ -280,7 +280,7 @@ class ConfigParser:
             a("valid-translation-output", str, None, "Path to store validation translations", "validation")
 
         # ---- translation (ref: addOptionsTranslation) ----
-        if self.mode in (D, S, E):
+        if self.mode in (D, S, E, MODE_SERVER):
             a("input", str, ["stdin"], "Paths to input files", "translation", nargs="+")
             a("output", str, "stdout", "Path to output file", "translation")
             a("vocabs", str, [], "Paths to vocabulary files", "translation", nargs="+")
@@ -297,7 +297,7 @@ class ConfigParser:
             a("tsv", "bool", False, "Tab-separated input", "translation")
             a("tsv-fields", int, 0, "Number of TSV fields", "translation")
             a("data-threads", int, 8, "Data reading threads", "translation")
-        if self.mode in (D,):
+        if self.mode in (D, MODE_SERVER):
             a("beam-size", int, 12, "Beam size", "translation")
             a("normalize", float, 0.0, "Divide score by length^alpha", "translation")
             a("word-penalty", float, 0.0, "Subtract penalty*length from score", "translation")

@@ -5,8 +5,6 @@ Implemented on fastapi/uvicorn (wheelhouse) with a plain-websocket endpoint
 at '/translate' mirroring the reference's message-in -> translation-out
 protocol, plus an HTTP POST /translate convenience endpoint.
 """
-from __future__ import annotations
-
 import sys
 
 from ..common.logging_utils import create_loggers, log_info
@@ -14,14 +12,18 @@ from ..common.options import Options
 from .translate import TranslateService
 
 
+from pydantic import BaseModel
+
+
+class TranslateRequest(BaseModel):
+    text: str
+
+
 def build_app(service: TranslateService):
     from fastapi import FastAPI, WebSocket, WebSocketDisconnect
-    from pydantic import BaseModel
 
     app = FastAPI(title="marian_amd server")
-
-    class Req(BaseModel):
-        text: str
+    Req = TranslateRequest
 
     @app.websocket("/translate")
     async def ws_translate(ws: WebSocket):
