@@ -164,7 +164,8 @@ class MultiHeadAttention(nn.Module):
         return x.transpose(1, 2).reshape(B, T, H * dh)
 
     def forward(self, query, keys, values, mask=None, causal=False,
-                cache: Optional[dict] = None, cache_key: str = ""):
+                cache: Optional[dict] = None, cache_key: str = "",
+                lens: Optional[torch.Tensor] = None):
         """cache: decode-time KV cache dict. For self-attention the new K/V
         are appended ({cache_key}_k/v); for cross-attention K/V are computed
         once and reused (ref: transformer.h cache_ keyed tensors)."""
@@ -186,7 +187,8 @@ class MultiHeadAttention(nn.Module):
                     cache[cache_key + "_k"] = k
                     cache[cache_key + "_v"] = v
         drop = self.dropout if self.training else 0.0
-        out = ops.attention(q, k, v, mask, causal=causal, dropout_p=drop)
+        out = ops.attention(q, k, v, mask, causal=causal, dropout_p=drop,
+                            lens=lens)
         return self.Wo(self._join(out))
 
 

@@ -50,9 +50,9 @@ class EncoderLayer(nn.Module):
                        opt.get("transformer-ffn-activation", "relu"),
                        opt.get("transformer-dropout-ffn", 0.0))
 
-    def forward(self, x, mask):
+    def forward(self, x, mask, lens=None):
         h = self.pre_self(x)
-        h = self.self_attn(h, h, h, mask=mask)
+        h = self.self_attn(h, h, h, mask=mask, lens=lens)
         x = self.post_self(h, residual=x)
         h = self.pre_ffn(x)
         h = self.ffn(h)
@@ -78,14 +78,15 @@ class DecoderLayer(nn.Module):
                        opt.get("transformer-dropout-ffn", 0.0))
 
     def forward(self, x, enc_out, src_mask, causal: bool,
-                cache: Optional[dict] = None, layer_id: int = 0):
+                cache: Optional[dict] = None, layer_id: int = 0,
+                src_lens=None):
         h = self.pre_self(x)
         h = self.self_attn(h, h, h, causal=causal, cache=cache,
                            cache_key=f"l{layer_id}_self")
         x = self.post_self(h, residual=x)
         h = self.pre_ctx(x)
         h = self.context_attn(h, enc_out, enc_out, mask=src_mask, cache=cache,
-                              cache_key=f"l{layer_id}_context")
+                              cache_key=f"l{layer_id}_context", lens=src_lens)
         x = self.post_ctx(h, residual=x)
         h = self.pre_ffn(x)
         h = self.ffn(h)
@@ -158,8 +159,9 @@ class TransformerModel(nn.Module):
         x = self.post_emb_enc(x)
         x = x * src_mask.unsqueeze(-1).to(dt)
         lmask = _log_mask(src_mask, dt)
+        lens = src_mask.sum(dim=1).to(torch.int32)
         for layer in self.encoder:
-            x = layer(x, lmask)
+            x = layer(x, lmask, lens=lens)
         if self.post_top_enc is not None:
             x = self.post_top_enc(x)
         return x
@@ -176,8 +178,10 @@ class TransformerModel(nn.Module):
         emb = self.trg_embed.add_positions(emb, 0)
         x = self.post_emb_dec(emb)
         lmask = _log_mask(src_mask, dt)
+        lens = src_mask.sum(dim=1).to(torch.int32)
         for i, layer in enumerate(self.decoder):
-            x = layer(x, enc_out, lmask, causal=True, layer_id=i)
+            x = layer(x, enc_out, lmask, causal=True, layer_id=i,
+                      src_lens=lens)
         if self.post_top_dec is not None:
             x = self.post_top_dec(x)
         return self.output_logits(x)
@@ -197,8 +201,10 @@ class TransformerModel(nn.Module):
         emb = self.trg_embed.add_positions(emb, step)
         x = self.post_emb_dec(emb)
         lmask = _log_mask(src_mask, dt)
+        lens = src_mask.sum(dim=1).to(torch.int32)
         for i, layer in enumerate(self.decoder):
-            x = layer(x, enc_out, lmask, causal=False, cache=cache, layer_id=i)
+            x = layer(x, enc_out, lmask, causal=False, cache=cache, layer_id=i,
+                      src_lens=lens)
         if self.post_top_dec is not None:
             x = self.post_top_dec(x)
         return self.output_logits(x[:, 0], shortlist)

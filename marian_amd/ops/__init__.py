@@ -76,16 +76,19 @@ def layer_norm(x: torch.Tensor, scale: torch.Tensor, bias: torch.Tensor,
 
 def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
               mask: Optional[torch.Tensor] = None,
-              causal: bool = False, dropout_p: float = 0.0) -> torch.Tensor:
+              causal: bool = False, dropout_p: float = 0.0,
+              lens: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Scaled dot-product attention.
 
     q: [B, H, Tq, Dh]; k/v: [B, H, Tk, Dh] (H may be 1 on k/v for MQA).
     mask: [B, 1, 1, Tk] or [B, 1, Tq, Tk] additive log-mask (0 / -inf).
+    lens: optional int32 [B] prefix key lengths equivalent to mask (NMT
+    masks are prefix masks); lets the fused kernel skip mask analysis.
     The 1/sqrt(Dh) scale is applied here.
     """
     if _on_gpu(q) and dropout_p == 0.0:
         from ._attention import fused_attention
-        out = fused_attention(q, k, v, mask, causal)
+        out = fused_attention(q, k, v, mask, causal, lens=lens)
         if out is not None:
             return out
     return attention_torch(q, k, v, mask, causal, dropout_p)

@@ -8,11 +8,14 @@ composite (hipBLASLt GEMMs + the HIP softmax kernel).
 """
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
 
 from . import hip_ext
+
+_DISABLED = os.environ.get("MAMD_NO_FUSED_ATTN", "") == "1"
 
 
 def mask_to_lens(mask: Optional[torch.Tensor], B: int, Tk: int):
@@ -56,7 +59,12 @@ class AttentionFn(torch.autograd.Function):
         return dq, dk32.to(q.dtype), dv32.to(q.dtype), None, None
 
 
-def fused_attention(q, k, v, mask, causal) -> Optional[torch.Tensor]:
+def fused_attention(q, k, v, mask, causal,
+                    lens: Optional[torch.Tensor] = None) -> Optional[torch.Tensor]:
+    """lens: pre-computed per-batch prefix lengths (int32 [B]); when given,
+    the mask tensor is ignored (no host-syncing prefix verification)."""
+    if _DISABLED:
+        return None
     ext = hip_ext(required=True)
     if not hasattr(ext, "attention_fwd"):
         return None
@@ -66,10 +74,13 @@ def fused_attention(q, k, v, mask, causal) -> Optional[torch.Tensor]:
     Tk = k.shape[2]
     if k.shape[1] != H:
         return None
-    lens = mask_to_lens(mask, B, Tk)
     if lens is None:
-        return None
-    lens = lens if lens.numel() else None
+        lens = mask_to_lens(mask, B, Tk)  # host-syncs; slow path
+        if lens is None:
+            return None
+        lens = lens if lens.numel() else None
+    elif lens.numel() == 0:
+        lens = None
     if torch.is_grad_enabled() and (q.requires_grad or k.requires_grad
                                     or v.requires_grad):
         return AttentionFn.apply(q, k, v, lens, causal)

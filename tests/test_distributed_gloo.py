@@ -1,0 +1,159 @@
+"""Multi-process data-parallel correctness on CPU (gloo, world_size=2).
+
+Verifies the exact logic the RCCL path uses on the GPU node: bucketed
+gradient reduction, ZeRO-1 sharded optimizer, param all-gather, and the
+global cost scaling — by checking 2-rank training equals single-process
+training on the concatenated batch.
+"""
+import os
+import pickle
+import tempfile
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from marian_amd.common.options import Options
+
+
+def tiny_options():
+    return {
+        "type": "transformer", "dim-emb": 32, "enc-depth": 1, "dec-depth": 1,
+        "transformer-heads": 2, "transformer-dim-ffn": 64,
+        "transformer-preprocess": "", "transformer-postprocess": "dan",
+        "transformer-postprocess-emb": "d", "transformer-postprocess-top": "",
+        "transformer-ffn-activation": "relu", "transformer-dropout": 0.0,
+        "dim-vocabs": [43, 43], "tied-embeddings-all": True,
+        "transformer-train-position-embeddings": False,
+        "transformer-decoder-dim-ffn": 0,
+        "precision": ["float32", "float32"],
+        "cost-type": "ce-mean-words", "label-smoothing": 0.1,
+        "optimizer": "adam", "optimizer-params": [0.9, 0.98, 1e-9],
+        "learn-rate": 0.01, "clip-norm": 1.0,
+        "amd-comm-bucket-mb": 1,   # force multiple buckets
+    }
+
+
+def _worker(rank, world, port, result_dir, n_steps, optimizer="adam"):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    from marian_amd.data.synthetic import synthetic_batch
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.training.graph_group import SyncGraphGroup
+
+    torch.manual_seed(123)  # same init on both ranks (broadcast also runs)
+    o = tiny_options()
+    o["optimizer"] = optimizer
+    opts = Options(o)
+    model = create_model(opts)
+    gg = SyncGraphGroup(opts, model, torch.device("cpu"), rank, world)
+    gg.set_lr(0.01)
+    for step in range(n_steps):
+        batch = synthetic_batch(6, 9, 9, 43, seed=1000 + step * world + rank)
+        lv = gg.update(batch)
+        assert lv is not None
+    # dump flat params + a probe loss
+    params = torch.cat([b.flat_param for b in gg.store.buckets]).numpy()
+    with open(os.path.join(result_dir, f"rank{rank}.pkl"), "wb") as f:
+        pickle.dump({"params": params,
+                     "ce": float(lv.ce_sum), "labels": lv.label_count}, f)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_two_rank_matches_single_process(tmp_path):
+    world, port, n_steps = 2, 29531, 4
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker,
+                         args=(r, world, port, str(tmp_path), n_steps, "sgd"))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+
+    import numpy as np
+    r0 = pickle.load(open(tmp_path / "rank0.pkl", "rb"))
+    r1 = pickle.load(open(tmp_path / "rank1.pkl", "rb"))
+    # both ranks end with identical parameters (all-gather correctness)
+    np.testing.assert_allclose(r0["params"], r1["params"], atol=1e-6)
+
+    # single-process reference: same two batches per step concatenated
+    from marian_amd.data.synthetic import synthetic_batch
+    from marian_amd.data.batch import CorpusBatch, SubBatch
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.training.graph_group import SyncGraphGroup
+
+    torch.manual_seed(123)
+    o = tiny_options()
+    o["optimizer"] = "sgd"
+    opts = Options(o)
+    model = create_model(opts)
+    gg = SyncGraphGroup(opts, model, torch.device("cpu"))
+    gg.set_lr(0.01)
+    for step in range(n_steps):
+        b0 = synthetic_batch(6, 9, 9, 43, seed=1000 + step * world + 0)
+        b1 = synthetic_batch(6, 9, 9, 43, seed=1000 + step * world + 1)
+        subs = [SubBatch(torch.cat([a.indices, b.indices]),
+                         torch.cat([a.mask, b.mask]))
+                for a, b in zip(b0.sub, b1.sub)]
+        gg.update(CorpusBatch(subs))
+    ref = torch.cat([b.flat_param for b in gg.store.buckets]).numpy()
+    np.testing.assert_allclose(r0["params"], ref, atol=2e-5)
+
+
+def _worker_ckpt(rank, world, port, result_dir):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from marian_amd.data.synthetic import synthetic_batch
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.training.graph_group import SyncGraphGroup
+
+    torch.manual_seed(5)
+    opts = Options(tiny_options())
+    model = create_model(opts)
+    gg = SyncGraphGroup(opts, model, torch.device("cpu"), rank, world)
+    gg.set_lr(0.01)
+    for step in range(2):
+        gg.update(synthetic_batch(4, 6, 6, 43, seed=step * world + rank))
+    path = os.path.join(result_dir, "model.npz")
+    gg.save_checkpoint(path, {"dim-emb": 32})
+    dist.barrier()
+    # reload into a fresh group and verify optimizer state round-trips
+    torch.manual_seed(99)
+    model2 = create_model(opts)
+    gg2 = SyncGraphGroup(opts, model2, torch.device("cpu"), rank, world)
+    assert gg2.load_checkpoint(path)
+    assert gg2.optimizer.t == gg.optimizer.t
+    d = (gg2.optimizer.master - gg.optimizer.master).abs().max()
+    assert float(d) < 1e-6
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_distributed_checkpoint_shards(tmp_path):
+    world, port = 2, 29541
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker_ckpt, args=(r, world, port, str(tmp_path)))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    # optimizer npz holds both ranks' shards
+    from marian_amd.common import io_npz
+    items, _ = io_npz.load_items(str(tmp_path / "model.npz.optimizer.npz"))
+    assert any(k.startswith("rank0_") for k in items)
+    assert any(k.startswith("rank1_") for k in items)
