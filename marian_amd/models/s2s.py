@@ -1,0 +1,298 @@
+"""RNN sequence-to-sequence model (ref: src/models/s2s.h, rnn/rnn.h,
+rnn/cells.h GRU, rnn/attention.{h,cpp}).
+
+Capability-parity implementation of --type s2s: bidirectional GRU encoder,
+conditional-GRU decoder with MLP (Bahdanau) attention and a
+readout = tanh(W1 state + W2 ctx + W3 emb) output layer, trainable with the
+same criterion/graph-group stack and decodable with the same BeamSearch
+(decode_step + cache protocol as the transformer).
+
+Parameter naming follows marian's s2s convention at the granularity we can
+verify offline (encoder_bi_*, decoder_cell1_*, decoder_att_*); exact
+upstream names could not be checked against the (empty) reference mount —
+see SURVEY.md §0.
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, Optional
+
+import numpy as np
+import torch
+import torch.nn as nn
+
+from ..common.options import Options
+from .layers import Dense
+
+
+class GRUCellM(nn.Module):
+    """GRU cell with marian-style [in, out] weights (rnn/cells.h GRU).
+
+    Gate layout: W/U produce [r, z] stacked; Wx/Ux the candidate.
+    """
+
+    def __init__(self, dim_in: int, dim: int):
+        super().__init__()
+        self.W = nn.Parameter(torch.empty(dim_in, 2 * dim))   # input -> r,z
+        self.U = nn.Parameter(torch.empty(dim, 2 * dim))      # state -> r,z
+        self.b = nn.Parameter(torch.zeros(2 * dim))
+        self.Wx = nn.Parameter(torch.empty(dim_in, dim))      # candidate
+        self.Ux = nn.Parameter(torch.empty(dim, dim))
+        self.bx = nn.Parameter(torch.zeros(dim))
+        self.dim = dim
+        for w in (self.W, self.U, self.Wx, self.Ux):
+            nn.init.xavier_uniform_(w)
+
+    def forward(self, x, h):
+        dt = x.dtype
+        gates = x @ self.W.to(dt) + h @ self.U.to(dt) + self.b.to(dt)
+        r, z = gates.chunk(2, dim=-1)
+        r, z = torch.sigmoid(r), torch.sigmoid(z)
+        hbar = torch.tanh(x @ self.Wx.to(dt) + (r * h) @ self.Ux.to(dt)
+                          + self.bx.to(dt))
+        return (1 - z) * hbar + z * h
+
+
+class MLPAttention(nn.Module):
+    """Bahdanau attention: score = v^T tanh(W s + U h) (rnn/attention.cpp,
+    the gAtt/gAttBack kernels in the reference)."""
+
+    def __init__(self, dim_state: int, dim_ctx: int, dim_att: int):
+        super().__init__()
+        self.W = nn.Parameter(torch.empty(dim_state, dim_att))
+        self.U = nn.Parameter(torch.empty(dim_ctx, dim_att))
+        self.v = nn.Parameter(torch.empty(dim_att, 1))
+        self.b = nn.Parameter(torch.zeros(dim_att))
+        for w in (self.W, self.U, self.v):
+            nn.init.xavier_uniform_(w)
+
+    def precompute(self, enc_out):
+        return enc_out @ self.U.to(enc_out.dtype)  # [B, Ts, A]
+
+    def forward(self, state, enc_proj, enc_out, src_mask):
+        dt = state.dtype
+        s = (state @ self.W.to(dt) + self.b.to(dt)).unsqueeze(1)  # [B,1,A]
+        e = torch.tanh(s + enc_proj) @ self.v.to(dt)              # [B,Ts,1]
+        e = e.squeeze(-1) + (1.0 - src_mask) * -1e9
+        a = torch.softmax(e.float(), dim=-1).to(dt)               # [B,Ts]
+        ctx = torch.bmm(a.unsqueeze(1), enc_out).squeeze(1)       # [B,C]
+        return ctx, a
+
+
+class S2SModel(nn.Module):
+    def __init__(self, options: Options):
+        super().__init__()
+        self.opt = options
+        dims = options.list("dim-vocabs")
+        self.src_vocab_size = int(dims[0])
+        self.trg_vocab_size = int(dims[-1])
+        self.dim_emb = int(options.get("dim-emb", 512))
+        self.dim_rnn = int(options.get("dim-rnn", 1024))
+        self.enc_depth = int(options.get("enc-depth", 1))
+        self.dec_depth = int(options.get("dec-depth", 1))
+
+        tied_all = bool(options.get("tied-embeddings-all", False))
+        self.tied_all = tied_all
+        self.src_embed = nn.Parameter(
+            torch.empty(self.src_vocab_size, self.dim_emb))
+        nn.init.normal_(self.src_embed, std=self.dim_emb ** -0.5)
+        if tied_all:
+            self.trg_embed = self.src_embed
+        else:
+            self.trg_embed = nn.Parameter(
+                torch.empty(self.trg_vocab_size, self.dim_emb))
+            nn.init.normal_(self.trg_embed, std=self.dim_emb ** -0.5)
+
+        d, e = self.dim_rnn, self.dim_emb
+        self.enc_fwd = GRUCellM(e, d)
+        self.enc_bwd = GRUCellM(e, d)
+        self.enc_deep = nn.ModuleList(
+            [GRUCellM(2 * d if i == 0 else d, d)
+             for i in range(max(0, self.enc_depth - 1))])
+        ctx_dim = 2 * d if not self.enc_deep else d
+        self.ctx_dim = ctx_dim
+
+        self.dec_init = Dense(ctx_dim, d)        # mean-pooled init state
+        self.cell1 = GRUCellM(e, d)
+        self.att = MLPAttention(d, ctx_dim, d)
+        self.cell2 = GRUCellM(ctx_dim, d)
+        self.dec_deep = nn.ModuleList(
+            [GRUCellM(d, d) for _ in range(max(0, self.dec_depth - 1))])
+
+        # readout (ref: s2s.h logit layer W1*state + W2*ctx + W3*emb)
+        self.ff_logit_l1_W0 = Dense(d, e)
+        self.ff_logit_l1_W1 = Dense(ctx_dim, e)
+        self.ff_logit_l1_W2 = Dense(e, e)
+        if not tied_all:
+            self.ff_logit_l2 = Dense(e, self.trg_vocab_size)
+        self.out_b = nn.Parameter(torch.zeros(self.trg_vocab_size))
+        self.compute_dtype = torch.float32
+
+    def set_compute_dtype(self, dtype):
+        self.compute_dtype = dtype
+
+    # ------------------------------------------------------------------
+    def encode(self, src_indices, src_mask):
+        dt = self.compute_dtype
+        B, T = src_indices.shape
+        emb = torch.nn.functional.embedding(src_indices, self.src_embed.to(dt))
+        mask = src_mask.to(dt)
+        hf = torch.zeros(B, self.dim_rnn, device=emb.device, dtype=dt)
+        hb = torch.zeros_like(hf)
+        fwd_states, bwd_states = [], [None] * T
+        for t in range(T):
+            hf_new = self.enc_fwd(emb[:, t], hf)
+            hf = mask[:, t:t + 1] * hf_new + (1 - mask[:, t:t + 1]) * hf
+            fwd_states.append(hf)
+        for t in range(T - 1, -1, -1):
+            hb_new = self.enc_bwd(emb[:, t], hb)
+            hb = mask[:, t:t + 1] * hb_new + (1 - mask[:, t:t + 1]) * hb
+            bwd_states[t] = hb
+        out = torch.stack([torch.cat([f, b], dim=-1)
+                           for f, b in zip(fwd_states, bwd_states)], dim=1)
+        for cell in self.enc_deep:
+            h = torch.zeros(B, self.dim_rnn, device=emb.device, dtype=dt)
+            states = []
+            for t in range(T):
+                h = cell(out[:, t], h)
+                states.append(h)
+            out = torch.stack(states, dim=1)
+        return out
+
+    # ------------------------------------------------------------------
+    def _init_state(self, enc_out, src_mask):
+        dt = enc_out.dtype
+        m = src_mask.to(dt).unsqueeze(-1)
+        mean = (enc_out * m).sum(1) / m.sum(1).clamp_min(1.0)
+        return torch.tanh(self.dec_init(mean))
+
+    def _step_cells(self, emb_t, state, enc_proj, enc_out, src_mask):
+        s1 = self.cell1(emb_t, state)
+        ctx, align = self.att(s1, enc_proj, enc_out, src_mask)
+        s = self.cell2(ctx, s1)
+        for cell in self.dec_deep:
+            s = cell(s, s)
+        return s, ctx, align
+
+    def _logits(self, state, ctx, emb_t):
+        t = torch.tanh(self.ff_logit_l1_W0(state) + self.ff_logit_l1_W1(ctx)
+                       + self.ff_logit_l1_W2(emb_t))
+        dt = t.dtype
+        if self.tied_all:
+            logits = t @ self.trg_embed.to(dt).t()
+        else:
+            logits = self.ff_logit_l2(t)
+        return logits + self.out_b.to(dt)
+
+    # ------------------------------------------------------------------
+    def decode_train(self, trg_indices, trg_mask, enc_out, src_mask):
+        dt = self.compute_dtype
+        B, T = trg_indices.shape
+        emb = torch.nn.functional.embedding(trg_indices, self.trg_embed.to(dt))
+        emb = torch.cat([torch.zeros_like(emb[:, :1]), emb[:, :-1]], dim=1)
+        enc_proj = self.att.precompute(enc_out)
+        state = self._init_state(enc_out, src_mask)
+        sm = src_mask.to(dt)
+        logits = []
+        for t in range(T):
+            state, ctx, _ = self._step_cells(emb[:, t], state, enc_proj,
+                                             enc_out, sm)
+            logits.append(self._logits(state, ctx, emb[:, t]))
+        return torch.stack(logits, dim=1)
+
+    def forward(self, batch):
+        src, trg = batch.source, batch.target
+        enc = self.encode(src.indices, src.mask)
+        return self.decode_train(trg.indices, trg.mask, enc, src.mask)
+
+    # ------------------------------------------------------------------
+    def decode_step(self, prev_indices, enc_out, src_mask, cache: dict,
+                    step: int, shortlist=None):
+        dt = self.compute_dtype
+        B = enc_out.shape[0]
+        if "state" not in cache:
+            cache["state"] = self._init_state(enc_out, src_mask)
+            cache["enc_proj"] = self.att.precompute(enc_out)
+        if step == 0:
+            emb = torch.zeros(B, self.dim_emb, device=enc_out.device, dtype=dt)
+        else:
+            emb = torch.nn.functional.embedding(prev_indices,
+                                                self.trg_embed.to(dt))
+        state, ctx, align = self._step_cells(emb, cache["state"],
+                                             cache["enc_proj"], enc_out,
+                                             src_mask.to(dt))
+        cache["state"] = state
+        logits = self._logits(state, ctx, emb)
+        if shortlist is not None:
+            logits = logits.index_select(-1, shortlist)
+        return logits
+
+    # ------------------------------------------------------------------
+    def _name_map(self) -> Dict[str, torch.Tensor]:
+        m: Dict[str, torch.Tensor] = {}
+        if self.tied_all:
+            m["Wemb"] = self.src_embed
+        else:
+            m["encoder_Wemb"] = self.src_embed
+            m["decoder_Wemb"] = self.trg_embed
+
+        def gru(prefix, cell: GRUCellM):
+            m[prefix + "_W"] = cell.W
+            m[prefix + "_U"] = cell.U
+            m[prefix + "_b"] = cell.b
+            m[prefix + "_Wx"] = cell.Wx
+            m[prefix + "_Ux"] = cell.Ux
+            m[prefix + "_bx"] = cell.bx
+
+        gru("encoder_bi", self.enc_fwd)
+        gru("encoder_bi_r", self.enc_bwd)
+        for i, cell in enumerate(self.enc_deep, start=1):
+            gru(f"encoder_l{i + 1}", cell)
+        m["decoder_ff_state_W"] = self.dec_init.W
+        m["decoder_ff_state_b"] = self.dec_init.b
+        gru("decoder_cell1", self.cell1)
+        gru("decoder_cell2", self.cell2)
+        for i, cell in enumerate(self.dec_deep, start=1):
+            gru(f"decoder_cell{i + 2}", cell)
+        m["decoder_att_W"] = self.att.W
+        m["decoder_att_U"] = self.att.U
+        m["decoder_att_v"] = self.att.v
+        m["decoder_att_b"] = self.att.b
+        m["decoder_ff_logit_l1_W0"] = self.ff_logit_l1_W0.W
+        m["decoder_ff_logit_l1_b0"] = self.ff_logit_l1_W0.b
+        m["decoder_ff_logit_l1_W1"] = self.ff_logit_l1_W1.W
+        m["decoder_ff_logit_l1_b1"] = self.ff_logit_l1_W1.b
+        m["decoder_ff_logit_l1_W2"] = self.ff_logit_l1_W2.W
+        m["decoder_ff_logit_l1_b2"] = self.ff_logit_l1_W2.b
+        if not self.tied_all:
+            m["decoder_ff_logit_l2_W"] = self.ff_logit_l2.W
+        m["decoder_ff_logit_l2_b"] = self.out_b
+        return m
+
+    def marian_items(self) -> Dict[str, np.ndarray]:
+        out = {}
+        for name, t in self._name_map().items():
+            arr = t.detach().cpu().float().numpy()
+            if arr.ndim == 1:
+                arr = arr.reshape(1, -1)
+            out[name] = arr
+        return out
+
+    def load_marian_items(self, items, strict: bool = True) -> None:
+        nm = self._name_map()
+        with torch.no_grad():
+            for name, t in nm.items():
+                if name not in items:
+                    if strict:
+                        raise KeyError(f"missing parameter {name}")
+                    continue
+                arr = torch.from_numpy(np.ascontiguousarray(items[name]))
+                t.copy_(arr.reshape(t.shape).to(t.dtype))
+
+    def model_config(self) -> dict:
+        keys = ["type", "dim-emb", "dim-rnn", "dim-vocabs", "enc-depth",
+                "dec-depth", "enc-type", "enc-cell", "dec-cell",
+                "tied-embeddings-all"]
+        cfg = {k: self.opt.get(k) for k in keys if self.opt.has(k)}
+        cfg["version"] = "marian_amd v0.1.0"
+        return cfg

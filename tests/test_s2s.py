@@ -1,0 +1,100 @@
+"""--type s2s model tests (ref: models/s2s.h parity)."""
+import torch
+
+from marian_amd.common.options import Options
+from marian_amd.data.synthetic import synthetic_batch
+from marian_amd.models.model_factory import create_model
+from marian_amd.translator.beam_search import BeamSearch, ModelScorer
+
+
+def s2s_options(**kw):
+    base = {
+        "type": "s2s", "dim-emb": 32, "dim-rnn": 48,
+        "enc-depth": 2, "dec-depth": 2,
+        "dim-vocabs": [31, 31], "tied-embeddings-all": True,
+    }
+    base.update(kw)
+    return Options(base)
+
+
+def test_s2s_forward_and_grads():
+    torch.manual_seed(0)
+    m = create_model(s2s_options())
+    b = synthetic_batch(3, 6, 5, 31)
+    logits = m(b)
+    assert logits.shape == (3, 5, 31)
+    logits.float().pow(2).mean().backward()
+    for n, p in m.named_parameters():
+        assert p.grad is not None and torch.isfinite(p.grad).all(), n
+
+
+def test_s2s_decode_step_matches_teacher_forcing():
+    torch.manual_seed(1)
+    m = create_model(s2s_options())
+    m.eval()
+    b = synthetic_batch(2, 5, 4, 31, seed=3)
+    with torch.no_grad():
+        enc = m.encode(b.source.indices, b.source.mask)
+        full = m.decode_train(b.target.indices, b.target.mask, enc,
+                              b.source.mask)
+        cache = {}
+        outs = []
+        for t in range(4):
+            prev = b.target.indices[:, t - 1] if t > 0 else \
+                torch.zeros(2, dtype=torch.long)
+            outs.append(m.decode_step(prev, enc, b.source.mask, cache, t))
+        stepped = torch.stack(outs, dim=1)
+    assert torch.allclose(full, stepped, atol=1e-5), \
+        (full - stepped).abs().max()
+
+
+def test_s2s_beam_search_runs():
+    torch.manual_seed(2)
+    m = create_model(s2s_options())
+    sc = ModelScorer(m, torch.device("cpu"))
+    bs = BeamSearch([sc], beam_size=2, max_length=8)
+    from tests.test_decode import make_batch
+    res = bs.search(make_batch([[5, 6, 0], [7, 0]]), torch.device("cpu"))
+    assert len(res) == 2
+
+
+def test_s2s_npz_roundtrip(tmp_path):
+    from marian_amd.common import io_npz
+    torch.manual_seed(3)
+    opts = s2s_options()
+    m1 = create_model(opts)
+    path = str(tmp_path / "s2s.npz")
+    io_npz.save_items(path, m1.marian_items(), m1.model_config())
+    items, cfg = io_npz.load_items(path)
+    assert cfg["type"] == "s2s"
+    assert "encoder_bi_W" in items and "decoder_att_v" in items
+    m2 = create_model(opts)
+    m2.load_marian_items(items)
+    b = synthetic_batch(2, 4, 4, 31, seed=5)
+    m1.eval(), m2.eval()
+    with torch.no_grad():
+        assert torch.allclose(m1(b), m2(b), atol=1e-6)
+
+
+def test_s2s_trains_on_copy_task():
+    torch.manual_seed(4)
+    from marian_amd.training.graph_group import SyncGraphGroup
+    opts = s2s_options()
+    opts.set("cost-type", "ce-mean-words")
+    opts.set("optimizer", "adam")
+    opts.set("learn-rate", 0.01)
+    opts.set("precision", ["float32", "float32"])
+    opts.set("label-smoothing", 0.0)
+    opts.set("clip-norm", 1.0)
+    m = create_model(opts)
+    gg = SyncGraphGroup(opts, m, torch.device("cpu"))
+    gg.set_lr(0.01)
+    batch = synthetic_batch(8, 6, 6, 31, seed=6)
+    batch.sub[-1].indices.copy_(batch.sub[0].indices)
+    first = last = None
+    for i in range(25):
+        lv = gg.update(batch)
+        ce = float(lv.ce_sum) / lv.label_count
+        first = first if first is not None else ce
+        last = ce
+    assert last < first - 0.3, (first, last)
