@@ -53,10 +53,10 @@ class AttentionFn(torch.autograd.Function):
     def backward(ctx, d_o):
         ext = hip_ext(required=True)
         q, k, v, o, lse, lens = ctx.saved_tensors
-        dq, dk32, dv32 = ext.attention_bwd(
+        dq, dk, dv = ext.attention_bwd(
             q, k, v, o, d_o.contiguous(), lse,
             lens if ctx.has_lens else None, ctx.causal, ctx.scale)
-        return dq, dk32.to(q.dtype), dv32.to(q.dtype), None, None
+        return dq, dk, dv, None, None
 
 
 def fused_attention(q, k, v, mask, causal,

@@ -308,64 +308,48 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
 }
 
 // ===========================================================================
-// Fused attention BACKWARD (flash-style, bf16, Dh=64).
+// Fused attention BACKWARD (flash-attn-2 style split, bf16, Dh=64).
 //
-// Per 32-row q-tile workgroup (1 wave), looping over 32-key kv tiles:
-//   recompute S in BOTH fragment orientations by swapping the mfma operands
-//   (S^T = mfma(K,Q^T) gives q-per-lane for dQ; S = mfma(Q,K^T) gives
-//   k-per-lane for dK/dV) — cheaper than cross-lane transposes;
-//   dP   = mfma(dO, V^T), dP^T = mfma(V, dO^T)     (same fragments swapped)
-//   dS   = P o (dP - D) * scale  (D = rowsum(dO o O), precomputed per q)
-//   dQ^T += mfma(K^T, pack(dS^T))                   (K^T staged in LDS)
-//   dK    = mfma(pack(dS^T_from_S), Q from Q^T LDS) -> atomicAdd fp32 scratch
-//   dV    = mfma(pack(P^T_from_S), dO from dO^T LDS) -> atomicAdd
-// dK/dV accumulate across q-tiles via fp32 scratch (few-way contention:
-// Tq/32 workgroups per (b,h)).
+// Two kernels, no atomics:
+//  - attn_bwd_dq_kernel:  parallel over q-tiles, loops kv; dQ^T accumulates
+//    in registers. Uses the "q per lane" orientation (S^T = mfma(K, Q^T),
+//    dP^T = mfma(V, dO^T)); dS^T packed via cvt_pk+permlane32_swap feeds
+//    dQ^T += mfma(K^T, dS^T) with K^T staged transposed in LDS.
+//  - attn_bwd_dkv_kernel: parallel over kv-tiles, loops q; dK/dV accumulate
+//    in registers. Uses the "k per lane" orientation (S = mfma(Q, K^T),
+//    dP = mfma(dO, V^T)); packed P^T / dS^T feed
+//    dV += mfma(P^T, dO), dK += mfma(dS^T, Q) with Q^T/dO^T staged in LDS.
+// D = rowsum(dO o O) is precomputed by the host wrapper.
 // ===========================================================================
 
 namespace {
 
-// pack a C-fragment's 16 regs (rows over regs) into two A/B-operand bf16x8
-// fragments (rows 8-packed per lane) — same dance as pack_p_to_b.
-__device__ __forceinline__ void pack_c_to_op(const float p[16], bf16x8 out[2]) {
-  pack_p_to_b(p, out);
-}
-
 template <int D, bool CAUSAL, bool HAS_LEN>
 __global__ __launch_bounds__(64)
-void attn_bwd_kernel(const __hip_bfloat16* __restrict__ Q,
-                     const __hip_bfloat16* __restrict__ K,
-                     const __hip_bfloat16* __restrict__ V,
-                     const __hip_bfloat16* __restrict__ O,
-                     const __hip_bfloat16* __restrict__ dO,
-                     const float* __restrict__ LSE,
-                     const int* __restrict__ kv_lens,
-                     __hip_bfloat16* __restrict__ dQ,
-                     float* __restrict__ dK32,
-                     float* __restrict__ dV32,
-                     int B, int H, int Tq, int Tk, float scale) {
-  __shared__ __hip_bfloat16 qt[D][QBLK + VT_PAD];   // Q^T image
-  __shared__ __hip_bfloat16 dot[D][QBLK + VT_PAD];  // dO^T image
-  __shared__ __hip_bfloat16 kt[D][KVBLK + VT_PAD];  // K^T image (per tile)
-  __shared__ float lse_s[QBLK];
-  __shared__ float dsum_s[QBLK];
+void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
+                        const __hip_bfloat16* __restrict__ K,
+                        const __hip_bfloat16* __restrict__ V,
+                        const __hip_bfloat16* __restrict__ dO,
+                        const float* __restrict__ LSE,
+                        const float* __restrict__ Dsum,  // [B*H*Tq]
+                        const int* __restrict__ kv_lens,
+                        __hip_bfloat16* __restrict__ dQ,
+                        int B, int H, int Tq, int Tk, float scale) {
+  __shared__ __hip_bfloat16 kt[D][KVBLK + VT_PAD];  // K^T image per tile
+  __shared__ float dq_b[QBLK][D + 4];
 
   const int lane = threadIdx.x;
   const int hi = lane >> 5;
   const int lq = lane & 31;
-  const int qtile = blockIdx.x;
   const int bh = blockIdx.y;
   const int b = bh / H;
-  const int q0 = qtile * QBLK;
+  const int q0 = blockIdx.x * QBLK;
   if (q0 >= Tq) return;
   const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
-
   const size_t qoff = (size_t)bh * Tq * D;
   const size_t koff = (size_t)bh * Tk * D;
 
-  // ---- prologue: Q / dO fragments + transposed LDS images + D[q] ---------
-  bf16x8 qfrag[D / 16];   // Q[q=lq][d-chunk]  (UNscaled)
-  bf16x8 dofrag[D / 16];  // dO[q=lq][d-chunk] (zeroed for q >= Tq)
+  bf16x8 qfrag[D / 16], dofrag[D / 16];
   {
     const bool qlive = (q0 + lq) < Tq;
     const int qrow = qlive ? (q0 + lq) : (Tq - 1);
@@ -378,69 +362,17 @@ void attn_bwd_kernel(const __hip_bfloat16* __restrict__ Q,
       if (!qlive) dv = (bf16x8)(__bf16)0.f;
       dofrag[c] = dv;
     }
-    // stage Q^T and dO^T images (8 rows per pass, transpose in flight)
-    const int rows_per_pass = 64 / (D / 8);
-    for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
-      const int qrow2 = q0 + qq + lane / (D / 8);
-      const int dcol = (lane % (D / 8)) * 8;
-      const bool live = qrow2 < Tq;
-      const int qr = live ? qrow2 : (Tq - 1);
-      bf16x8 vq = *reinterpret_cast<const bf16x8*>(Q + qoff + (size_t)qr * D + dcol);
-      bf16x8 vd = *reinterpret_cast<const bf16x8*>(dO + qoff + (size_t)qr * D + dcol);
-      if (!live) {
-        vq = (bf16x8)(__bf16)0.f;
-        vd = (bf16x8)(__bf16)0.f;
-      }
-      const int qlocal = qq + lane / (D / 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        qt[dcol + j][qlocal] = (__hip_bfloat16)(float)vq[j];
-        dot[dcol + j][qlocal] = (__hip_bfloat16)(float)vd[j];
-      }
-    }
-    // D[q] = rowsum(dO o O); lane pair (2 lanes per q-row)
-    const int qrow3 = q0 + lane / 2;
-    float acc = 0.f;
-    if (qrow3 < Tq) {
-      const __hip_bfloat16* op = O + qoff + (size_t)qrow3 * D + (lane & 1) * (D / 2);
-      const __hip_bfloat16* dp2 = dO + qoff + (size_t)qrow3 * D + (lane & 1) * (D / 2);
-      for (int j = 0; j < D / 2; j += 8) {
-        bf16x8 ov = *reinterpret_cast<const bf16x8*>(op + j);
-        bf16x8 dv = *reinterpret_cast<const bf16x8*>(dp2 + j);
-#pragma unroll
-        for (int u = 0; u < 8; ++u) acc += (float)ov[u] * (float)dv[u];
-      }
-    }
-    acc += __shfl_xor(acc, 1);
-    if ((lane & 1) == 0) dsum_s[lane / 2] = acc;
-    // lse per q row
-    if (hi == 0) {
-      const int qr = q0 + lq;
-      lse_s[lq] = (qr < Tq) ? LSE[(size_t)bh * Tq + qr] : 0.f;
-    }
   }
-  __builtin_amdgcn_s_waitcnt(0);
-  __builtin_amdgcn_s_barrier();
+  const int q_abs = q0 + lq;
+  const float lse_lane = (q_abs < Tq) ? LSE[(size_t)bh * Tq + q_abs] : 0.f;
+  const float dsum_lane = (q_abs < Tq) ? Dsum[(size_t)bh * Tq + q_abs] : 0.f;
 
-  // reg copies of lse/D indexed by C-fragment row (for the S orientation)
-  float lse_row[16], dsum_row[16];
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    lse_row[r] = lse_s[crow(r, hi)];
-    dsum_row[r] = dsum_s[crow(r, hi)];
-  }
-  const float lse_lane = lse_s[lq];   // for the S^T orientation (q per lane)
-  const float dsum_lane = dsum_s[lq];
-
-  f32x16 dqt[D / 32];  // dQ^T accumulators
+  f32x16 dqt[D / 32];
 #pragma unroll
   for (int dt2 = 0; dt2 < D / 32; ++dt2) dqt[dt2] = (f32x16)0.f;
 
   const int kv_end = CAUSAL ? min(kv_len, q0 + QBLK) : kv_len;
-  const int q_abs = q0 + lq;
-
   for (int k0 = 0; k0 < kv_end; k0 += KVBLK) {
-    // ---- load K/V fragments (row k = lq, contiguous d) + stage K^T ------
     bf16x8 kfrag[D / 16], vfrag[D / 16];
     {
       const int krow = min(k0 + lq, Tk - 1);
@@ -467,131 +399,217 @@ void attn_bwd_kernel(const __hip_bfloat16* __restrict__ Q,
     __builtin_amdgcn_s_waitcnt(0);
     __builtin_amdgcn_s_barrier();
 
-    // =================== orientation 1: S^T (q per lane) =================
-    {
-      f32x16 st = (f32x16)0.f;   // S^T = K.Q^T
-      f32x16 dpt = (f32x16)0.f;  // dP^T = V.dO^T
+    f32x16 st = (f32x16)0.f;
+    f32x16 dpt = (f32x16)0.f;
 #pragma unroll
-      for (int c = 0; c < D / 16; ++c) {
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag[c], qfrag[c], st, 0, 0, 0);
-        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag[c], dofrag[c], dpt, 0, 0, 0);
-      }
-      float ds[16];
+    for (int c = 0; c < D / 16; ++c) {
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag[c], qfrag[c], st, 0, 0, 0);
+      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag[c], dofrag[c], dpt, 0, 0, 0);
+    }
+    float ds[16];
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int k_abs = k0 + crow(r, hi);
-        const bool dead = (k_abs >= kv_len) || (CAUSAL && k_abs > q_abs)
-                          || (q_abs >= Tq);
-        const float pval = dead ? 0.f : __expf(st[r] * scale - lse_lane);
-        ds[r] = pval * (dpt[r] - dsum_lane) * scale;
-      }
-      bf16x8 dst_op[2];
-      pack_c_to_op(ds, dst_op);  // dS^T chunks [16k x 32q]
-      // dQ^T[dtile] += mfma(A=K^T from LDS, B=dS^T)
+    for (int r = 0; r < 16; ++r) {
+      const int k_abs = k0 + crow(r, hi);
+      const bool dead = (k_abs >= kv_len) || (CAUSAL && k_abs > q_abs)
+                        || (q_abs >= Tq);
+      const float pval = dead ? 0.f : __expf(st[r] * scale - lse_lane);
+      ds[r] = pval * (dpt[r] - dsum_lane) * scale;
+    }
+    bf16x8 dst_op[2];
+    pack_p_to_b(ds, dst_op);
 #pragma unroll
-      for (int dt2 = 0; dt2 < D / 32; ++dt2) {
+    for (int dt2 = 0; dt2 < D / 32; ++dt2) {
 #pragma unroll
-        for (int c = 0; c < 2; ++c) {
-          const __hip_bfloat16* kp2 = &kt[32 * dt2 + lq][16 * c + 8 * hi];
-          bf16x8 ka = *reinterpret_cast<const bf16x8*>(kp2);
-          dqt[dt2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, dst_op[c],
-                                                             dqt[dt2], 0, 0, 0);
-        }
+      for (int c = 0; c < 2; ++c) {
+        bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+            &kt[32 * dt2 + lq][16 * c + 8 * hi]);
+        dqt[dt2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, dst_op[c],
+                                                           dqt[dt2], 0, 0, 0);
       }
     }
-
-    // =================== orientation 2: S (k per lane) ===================
-    {
-      f32x16 s = (f32x16)0.f;   // S = Q.K^T
-      f32x16 dp = (f32x16)0.f;  // dP = dO.V^T
-#pragma unroll
-      for (int c = 0; c < D / 16; ++c) {
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfrag[c], kfrag[c], s, 0, 0, 0);
-        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofrag[c], vfrag[c], dp, 0, 0, 0);
-      }
-      const int k_lane = k0 + lq;
-      float pv[16], ds[16];
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int q_r = q0 + crow(r, hi);
-        const bool dead = (k_lane >= kv_len) || (CAUSAL && k_lane > q_r)
-                          || (q_r >= Tq);
-        const float pval = dead ? 0.f : __expf(s[r] * scale - lse_row[r]);
-        pv[r] = pval;
-        ds[r] = pval * (dp[r] - dsum_row[r]) * scale;
-      }
-      bf16x8 p_op[2], ds_op[2];
-      pack_c_to_op(pv, p_op);   // P^T chunks  [16q x 32k] -> A for dV
-      pack_c_to_op(ds, ds_op);  // dS^T chunks [16q x 32k] -> A for dK
-      // NOTE on orientation: packing the S-fragment's q-regs gives operands
-      // whose ROW dim is q; used as A in mfma(A, B) they compute
-      // C[k?][..]... — see the A-operand note: A[row=l&31][...]: here the
-      // per-lane dim is k (l&31 = k-col of S) so A = P^T [32k x 16q]. OK.
-      f32x16 dv_acc = (f32x16)0.f;
-      f32x16 dk_acc = (f32x16)0.f;
-#pragma unroll
-      for (int dt2 = 0; dt2 < D / 32; ++dt2) {
-        dv_acc = (f32x16)0.f;
-        dk_acc = (f32x16)0.f;
-#pragma unroll
-        for (int c = 0; c < 2; ++c) {
-          // B = dO chunk [16q x 32d] from dO^T image; B = Q chunk likewise
-          const __hip_bfloat16* dp2 = &dot[32 * dt2 + lq][16 * c + 8 * hi];
-          const __hip_bfloat16* qp2 = &qt[32 * dt2 + lq][16 * c + 8 * hi];
-          // CAREFUL: B operand wants B[q-row][d-col=l&31]; our images are
-          // [d][q], so a 16B read at row (32*dt2+lq) gives d per LANE and
-          // q packed — that IS the B layout with col=lane=d. ok.
-          bf16x8 dob = *reinterpret_cast<const bf16x8*>(dp2);
-          bf16x8 qb = *reinterpret_cast<const bf16x8*>(qp2);
-          dv_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(p_op[c], dob,
-                                                           dv_acc, 0, 0, 0);
-          dk_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ds_op[c], qb,
-                                                           dk_acc, 0, 0, 0);
-        }
-        // C fragments: [32k x 32d]: lane = d-col, regs = k-rows
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int k_r = k0 + crow(r, hi);
-          if (k_r < kv_len) {
-            const size_t base = koff + (size_t)k_r * D + 32 * dt2 + lq;
-            atomicAdd(&dV32[base], dv_acc[r]);
-            atomicAdd(&dK32[base], dk_acc[r]);
-          }
-        }
-      }
-    }
-    __builtin_amdgcn_s_barrier();  // protect kt before next tile's staging
+    __builtin_amdgcn_s_barrier();
   }
 
-  // ---- epilogue: dQ = dQ^T^T via LDS bounce (reuse qt image space) -------
-  __builtin_amdgcn_s_barrier();
-  __shared__ float dq_b[QBLK][D + 4];
+  // epilogue: dQ via LDS bounce
 #pragma unroll
   for (int dt2 = 0; dt2 < D / 32; ++dt2)
 #pragma unroll
     for (int r = 0; r < 16; ++r)
       dq_b[lq][32 * dt2 + crow(r, hi)] = dqt[dt2][r];
   __builtin_amdgcn_s_barrier();
-  {
-    const int lanes_per_row = D / 8;
-    const int rows_per_pass = 64 / lanes_per_row;
-    for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
-      const int qrow = q0 + qq + lane / lanes_per_row;
-      const int dcol = (lane % lanes_per_row) * 8;
-      if (qrow < Tq) {
-        bf16x8 out;
+  const int lanes_per_row = D / 8;
+  const int rows_per_pass = 64 / lanes_per_row;
+  for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
+    const int qrow = q0 + qq + lane / lanes_per_row;
+    const int dcol = (lane % lanes_per_row) * 8;
+    if (qrow < Tq) {
+      bf16x8 out;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          out[j] = (__bf16)dq_b[qq + lane / lanes_per_row][dcol + j];
-        *reinterpret_cast<bf16x8*>(dQ + qoff + (size_t)qrow * D + dcol) = out;
+      for (int j = 0; j < 8; ++j)
+        out[j] = (__bf16)dq_b[qq + lane / lanes_per_row][dcol + j];
+      *reinterpret_cast<bf16x8*>(dQ + qoff + (size_t)qrow * D + dcol) = out;
+    }
+  }
+}
+
+template <int D, bool CAUSAL, bool HAS_LEN>
+__global__ __launch_bounds__(64)
+void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
+                         const __hip_bfloat16* __restrict__ K,
+                         const __hip_bfloat16* __restrict__ V,
+                         const __hip_bfloat16* __restrict__ dO,
+                         const float* __restrict__ LSE,
+                         const float* __restrict__ Dsum,
+                         const int* __restrict__ kv_lens,
+                         __hip_bfloat16* __restrict__ dK,
+                         __hip_bfloat16* __restrict__ dV,
+                         int B, int H, int Tq, int Tk, float scale) {
+  __shared__ __hip_bfloat16 qt[D][QBLK + VT_PAD];   // Q^T image per q-tile
+  __shared__ __hip_bfloat16 dot[D][QBLK + VT_PAD];  // dO^T image per q-tile
+  __shared__ float kv_b[KVBLK][D + 4];
+
+  const int lane = threadIdx.x;
+  const int hi = lane >> 5;
+  const int lq = lane & 31;
+  const int bh = blockIdx.y;
+  const int b = bh / H;
+  const int k0 = blockIdx.x * KVBLK;
+  if (k0 >= Tk) return;
+  const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
+  const size_t qoff = (size_t)bh * Tq * D;
+  const size_t koff = (size_t)bh * Tk * D;
+
+  bf16x8 kfrag[D / 16], vfrag[D / 16];
+  {
+    const int krow = min(k0 + lq, Tk - 1);
+    const __hip_bfloat16* kp = K + koff + (size_t)krow * D;
+    const __hip_bfloat16* vp = V + koff + (size_t)krow * D;
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c) {
+      kfrag[c] = *reinterpret_cast<const bf16x8*>(kp + 16 * c + 8 * hi);
+      vfrag[c] = *reinterpret_cast<const bf16x8*>(vp + 16 * c + 8 * hi);
+    }
+  }
+  const int k_lane = k0 + lq;
+
+  f32x16 dk_acc[D / 32], dv_acc[D / 32];
+#pragma unroll
+  for (int dt2 = 0; dt2 < D / 32; ++dt2) {
+    dk_acc[dt2] = (f32x16)0.f;
+    dv_acc[dt2] = (f32x16)0.f;
+  }
+
+  const int q_start = CAUSAL ? (k0 / QBLK) * QBLK : 0;
+  for (int q0 = q_start; q0 < Tq; q0 += QBLK) {
+    // per-q-tile fragments + transposed images
+    bf16x8 qfrag[D / 16], dofrag[D / 16];
+    {
+      const bool qlive = (q0 + lq) < Tq;
+      const int qrow = qlive ? (q0 + lq) : (Tq - 1);
+      const __hip_bfloat16* qp = Q + qoff + (size_t)qrow * D;
+      const __hip_bfloat16* dp = dO + qoff + (size_t)qrow * D;
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        qfrag[c] = *reinterpret_cast<const bf16x8*>(qp + 16 * c + 8 * hi);
+        bf16x8 dv = *reinterpret_cast<const bf16x8*>(dp + 16 * c + 8 * hi);
+        if (!qlive) dv = (bf16x8)(__bf16)0.f;
+        dofrag[c] = dv;
+      }
+      const int rows_per_pass = 64 / (D / 8);
+      for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
+        const int qrow2 = q0 + qq + lane / (D / 8);
+        const int dcol = (lane % (D / 8)) * 8;
+        const bool live = qrow2 < Tq;
+        const int qr = live ? qrow2 : (Tq - 1);
+        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Q + qoff + (size_t)qr * D + dcol);
+        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dO + qoff + (size_t)qr * D + dcol);
+        if (!live) {
+          vq = (bf16x8)(__bf16)0.f;
+          vd = (bf16x8)(__bf16)0.f;
+        }
+        const int qlocal = qq + lane / (D / 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          qt[dcol + j][qlocal] = (__hip_bfloat16)(float)vq[j];
+          dot[dcol + j][qlocal] = (__hip_bfloat16)(float)vd[j];
+        }
       }
     }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_s_barrier();
+
+    f32x16 s = (f32x16)0.f;
+    f32x16 dp = (f32x16)0.f;
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c) {
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfrag[c], kfrag[c], s, 0, 0, 0);
+      dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofrag[c], vfrag[c], dp, 0, 0, 0);
+    }
+    float pv[16], ds[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int q_r = q0 + crow(r, hi);
+      const bool qlive = q_r < Tq;
+      const float lse_r = qlive ? LSE[(size_t)bh * Tq + q_r] : 0.f;
+      const float dsum_r = qlive ? Dsum[(size_t)bh * Tq + q_r] : 0.f;
+      const bool dead = (k_lane >= kv_len) || (CAUSAL && k_lane > q_r) || !qlive;
+      const float pval = dead ? 0.f : __expf(s[r] * scale - lse_r);
+      pv[r] = pval;
+      ds[r] = pval * (dp[r] - dsum_r) * scale;
+    }
+    bf16x8 p_op[2], ds_op[2];
+    pack_p_to_b(pv, p_op);
+    pack_p_to_b(ds, ds_op);
+#pragma unroll
+    for (int dt2 = 0; dt2 < D / 32; ++dt2) {
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        bf16x8 dob = *reinterpret_cast<const bf16x8*>(
+            &dot[32 * dt2 + lq][16 * c + 8 * hi]);
+        bf16x8 qb = *reinterpret_cast<const bf16x8*>(
+            &qt[32 * dt2 + lq][16 * c + 8 * hi]);
+        dv_acc[dt2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(p_op[c], dob,
+                                                              dv_acc[dt2], 0, 0, 0);
+        dk_acc[dt2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ds_op[c], qb,
+                                                              dk_acc[dt2], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: dV then dK via the same LDS bounce ([32k x 32d] frags:
+  // lane = d-col, regs = k-rows)
+  for (int which = 0; which < 2; ++which) {
+    __hip_bfloat16* out_base = (which == 0) ? dV : dK;
+    f32x16* acc = (which == 0) ? dv_acc : dk_acc;
+#pragma unroll
+    for (int dt2 = 0; dt2 < D / 32; ++dt2)
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        kv_b[crow(r, hi)][32 * dt2 + lq] = acc[dt2][r];
+    __builtin_amdgcn_s_barrier();
+    const int lanes_per_row = D / 8;
+    const int rows_per_pass = 64 / lanes_per_row;
+    for (int kk = 0; kk < KVBLK; kk += rows_per_pass) {
+      const int krow = k0 + kk + lane / lanes_per_row;
+      const int dcol = (lane % lanes_per_row) * 8;
+      if (krow < Tk) {
+        bf16x8 out;
+        const bool live = krow < kv_len;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          out[j] = live ? (__bf16)kv_b[kk + lane / lanes_per_row][dcol + j]
+                        : (__bf16)0.f;
+        *reinterpret_cast<bf16x8*>(out_base + koff + (size_t)krow * D + dcol) = out;
+      }
+    }
+    __builtin_amdgcn_s_barrier();
   }
 }
 
 }  // namespace
 
-// returns (dq bf16, dk_f32, dv_f32)
+// returns (dq, dk, dv) - all bf16
 std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v, torch::Tensor o,
                                          torch::Tensor d_o, torch::Tensor lse,
@@ -604,32 +622,48 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
   const int Tk = kc.size(2);
   TORCH_CHECK(D == 64, "attention_bwd: only Dh=64 supported");
   auto dq = torch::empty_like(qc);
-  auto dk32 = torch::zeros({B, H, Tk, D}, qc.options().dtype(torch::kFloat32));
-  auto dv32 = torch::zeros({B, H, Tk, D}, qc.options().dtype(torch::kFloat32));
+  auto dk = torch::empty_like(kc);
+  auto dv = torch::empty_like(vc);
+  // D = rowsum(dO o O) per query row, fp32
+  auto dsum = (doc.to(torch::kFloat32) * oc.to(torch::kFloat32))
+                  .sum(-1).contiguous();
   const int* lens_ptr = nullptr;
   torch::Tensor lensc;
   if (kv_lens.has_value() && kv_lens->defined()) {
     lensc = kv_lens->to(torch::kInt32).contiguous();
     lens_ptr = lensc.data_ptr<int>();
   }
-  dim3 grid((Tq + 31) / 32, B * H);
   auto stream = at::cuda::getCurrentHIPStream();
+  dim3 gq((Tq + 31) / 32, B * H);
+  dim3 gk((Tk + 31) / 32, B * H);
 #define RUNB(CAUSAL, HASLEN)                                                  \
-  hipLaunchKernelGGL((attn_bwd_kernel<64, CAUSAL, HASLEN>), grid, dim3(64),   \
-                     0, stream,                                               \
-                     reinterpret_cast<const __hip_bfloat16*>(qc.data_ptr()),  \
-                     reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),  \
-                     reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),  \
-                     reinterpret_cast<const __hip_bfloat16*>(oc.data_ptr()),  \
-                     reinterpret_cast<const __hip_bfloat16*>(doc.data_ptr()), \
-                     lse.data_ptr<float>(), lens_ptr,                         \
-                     reinterpret_cast<__hip_bfloat16*>(dq.data_ptr()),        \
-                     dk32.data_ptr<float>(), dv32.data_ptr<float>(),          \
-                     B, H, Tq, Tk, (float)scale)
+  do {                                                                        \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<64, CAUSAL, HASLEN>), gq,          \
+                       dim3(64), 0, stream,                                   \
+                       reinterpret_cast<const __hip_bfloat16*>(qc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(doc.data_ptr()),\
+                       lse.data_ptr<float>(), dsum.data_ptr<float>(),         \
+                       lens_ptr,                                              \
+                       reinterpret_cast<__hip_bfloat16*>(dq.data_ptr()),      \
+                       B, H, Tq, Tk, (float)scale);                           \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, CAUSAL, HASLEN>), gk,         \
+                       dim3(64), 0, stream,                                   \
+                       reinterpret_cast<const __hip_bfloat16*>(qc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(doc.data_ptr()),\
+                       lse.data_ptr<float>(), dsum.data_ptr<float>(),         \
+                       lens_ptr,                                              \
+                       reinterpret_cast<__hip_bfloat16*>(dk.data_ptr()),      \
+                       reinterpret_cast<__hip_bfloat16*>(dv.data_ptr()),      \
+                       B, H, Tq, Tk, (float)scale);                           \
+  } while (0)
   if (causal && lens_ptr) RUNB(true, true);
   else if (causal) RUNB(true, false);
   else if (lens_ptr) RUNB(false, true);
   else RUNB(false, false);
 #undef RUNB
-  return {dq, dk32, dv32};
+  return {dq, dk, dv};
 }

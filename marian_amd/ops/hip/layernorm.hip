@@ -123,7 +123,9 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
   }
 }
 
-// column-sum the per-wave partials: dgamma[c] = sum_w scratch[w][c]
+// column-sum the per-wave partials: dgamma[c] = sum_w scratch[w][c].
+// 2-D grid: blockIdx.x tiles columns, blockIdx.y tiles row stripes, so the
+// sum is latency-parallel; stripes combine with a handful of atomics.
 __global__ void ln_bwd_reduce_kernel(const float* __restrict__ dg_scratch,
                                      const float* __restrict__ db_scratch,
                                      float* __restrict__ dgamma,
@@ -131,13 +133,22 @@ __global__ void ln_bwd_reduce_kernel(const float* __restrict__ dg_scratch,
                                      int n_waves, int d) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= d) return;
+  const int stripes = gridDim.y;
+  const int per = (n_waves + stripes - 1) / stripes;
+  const int w0 = blockIdx.y * per;
+  const int w1 = min(w0 + per, n_waves);
   float sg = 0.f, sb = 0.f;
-  for (int w = 0; w < n_waves; ++w) {
+  for (int w = w0; w < w1; ++w) {
     sg += dg_scratch[(size_t)w * d + c];
     sb += db_scratch[(size_t)w * d + c];
   }
-  dgamma[c] = sg;
-  dbeta[c] = sb;
+  if (stripes == 1) {
+    dgamma[c] = sg;
+    dbeta[c] = sb;
+  } else {
+    atomicAdd(&dgamma[c], sg);
+    atomicAdd(&dbeta[c], sb);
+  }
 }
 
 constexpr int kBlock = 256;
@@ -202,7 +213,9 @@ void ln_bwd_launch(const torch::Tensor& x, const torch::Tensor& g,
 #undef LAUNCH_BWD
   // waves with no rows still write their (zero) register partials, so the
   // whole scratch is defined.
-  hipLaunchKernelGGL(ln_bwd_reduce_kernel, dim3((d + 255) / 256), dim3(256),
+  const int stripes = std::max(1, std::min(n_waves / 64, 64));
+  hipLaunchKernelGGL(ln_bwd_reduce_kernel,
+                     dim3((d + 255) / 256, stripes), dim3(256),
                      0, stream, dg_scratch.data_ptr<float>(),
                      db_scratch.data_ptr<float>(), dgamma.data_ptr<float>(),
                      dbeta.data_ptr<float>(), n_waves, d);

@@ -250,6 +250,11 @@ class ShardedOptimizer:
         self.core = create_optimizer_core(options, self.n, dev)
         self.t = 0
         self.clip_norm = float(options.get("clip-norm", 0.0) or 0.0)
+        # --exponential-smoothing: Polyak average of the shard
+        # (ref: optimizers/exponential_smoothing + SyncGraphGroup swapParams)
+        self.smoothing = float(options.get("exponential-smoothing", 0.0) or 0.0)
+        self.smooth = (self.master.clone() if self.smoothing > 0 else None)
+        self._swapped = None
 
     # ------------------------------------------------------------------
     def grad_shard_for_bucket(self, bi: int) -> torch.Tensor:
@@ -276,6 +281,10 @@ class ShardedOptimizer:
             if mult != 1.0:
                 g = g * mult
             self.core.step(self.master, g, lr, self.t)
+        if self.smooth is not None:
+            # marian caps the effective decay early in training
+            decay = min(self.smoothing, (self.t + 1.0) / (self.t + 10.0))
+            self.smooth.mul_(decay).add_(self.master, alpha=1.0 - decay)
         # write updated shard back into each bucket's param slice
         with torch.no_grad():
             for b, (o, s) in zip(self.store.buckets, self.seg):
@@ -283,10 +292,46 @@ class ShardedOptimizer:
                     self.master[o:o + s].to(self.store.dtype))
 
     # ------------------------------------------------------------------
+    # --exponential-smoothing swap (ref: communicator swapParams)
+    def swap_in_smoothed(self, comm=None):
+        """Write the smoothed shard into the live params (then all-gather
+        via the given communicator); swap_out_smoothed restores."""
+        if self.smooth is None or self._swapped is not None:
+            return
+        self._swapped = self.master.clone()
+        with torch.no_grad():
+            for b, (o, s) in zip(self.store.buckets, self.seg):
+                b.shard_view(b.flat_param, self.rank, self.world).copy_(
+                    self.smooth[o:o + s].to(self.store.dtype))
+            if comm is not None:
+                for b in self.store.buckets:
+                    comm.all_gather_bucket(
+                        b.flat_param,
+                        b.shard_view(b.flat_param, self.rank, self.world))
+                comm.wait_all()
+
+    def swap_out_smoothed(self, comm=None):
+        if self._swapped is None:
+            return
+        with torch.no_grad():
+            for b, (o, s) in zip(self.store.buckets, self.seg):
+                b.shard_view(b.flat_param, self.rank, self.world).copy_(
+                    self._swapped[o:o + s].to(self.store.dtype))
+            if comm is not None:
+                for b in self.store.buckets:
+                    comm.all_gather_bucket(
+                        b.flat_param,
+                        b.shard_view(b.flat_param, self.rank, self.world))
+                comm.wait_all()
+        self._swapped = None
+
+    # ------------------------------------------------------------------
     # checkpoint (ref: model.npz.optimizer.npz via gatherState/scatterState)
     def state_items(self) -> Dict[str, np.ndarray]:
         out = {"master_parameters": self.master.cpu().numpy(),
                "optimizer_step": np.array([self.t], dtype=np.int64)}
+        if self.smooth is not None:
+            out["exp_smoothing"] = self.smooth.cpu().numpy()
         for k, v in self.core.state_items().items():
             out[k] = v.cpu().numpy()
         return out
