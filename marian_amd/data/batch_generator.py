@@ -35,7 +35,7 @@ class BatchGenerator:
 
     # ------------------------------------------------------------------
     def _sort_key(self, tup: SentenceTuple):
-        _, streams = tup
+        streams = tup[1]
         if self.maxi_batch_sort == "src":
             return (len(streams[0]), len(streams[-1]))
         return (len(streams[-1]), len(streams[0]))
@@ -46,7 +46,28 @@ class BatchGenerator:
         for s in range(n_streams):
             pad = self.pad_ids[s] if self.pad_ids else 0
             subs.append(SubBatch.from_sentences([t[1][s] for t in tuples], pad_id=pad))
-        return CorpusBatch(subs, sentence_ids=[t[0] for t in tuples])
+        ga = dw = None
+        extras = [t[2] if len(t) > 2 else {} for t in tuples]
+        if any("align" in e for e in extras):
+            import torch
+            from .alignment import alignments_to_tensor
+            ga = alignments_to_tensor([e.get("align", []) for e in extras],
+                                      subs[0].batch_width,
+                                      subs[-1].batch_width)
+        if any("weight" in e for e in extras):
+            import torch
+            ws = []
+            T = subs[-1].batch_width
+            per_word = any(isinstance(e.get("weight"), list) for e in extras)
+            for e in extras:
+                w = e.get("weight", 1.0)
+                if per_word:
+                    w = (w if isinstance(w, list) else [w] * T)
+                    w = (w + [1.0] * T)[:T]
+                ws.append(w)
+            dw = torch.tensor(ws, dtype=torch.float32)
+        return CorpusBatch(subs, guided_alignment=ga, data_weights=dw,
+                           sentence_ids=[t[0] for t in tuples])
 
     def _emit_from_pool(self, pool: List[SentenceTuple]) -> Iterator[CorpusBatch]:
         if self.maxi_batch_sort != "none":

@@ -12,7 +12,8 @@ from typing import Iterator, List, Optional, Sequence, Tuple
 
 from .vocab import Vocab
 
-SentenceTuple = Tuple[int, List[List[int]]]  # (corpus position id, [stream ids])
+# (corpus position id, [stream ids], extras {"align": pairs, "weight": float})
+SentenceTuple = Tuple[int, List[List[int]], dict]
 
 
 def _open(path: str):
@@ -27,8 +28,12 @@ class Corpus:
     def __init__(self, paths: Sequence[str], vocabs: Sequence[Vocab],
                  max_length: int = 50, max_length_crop: bool = False,
                  tsv: bool = False, tsv_fields: int = 0,
-                 shuffle_in_ram: bool = False, seed: int = 1234):
+                 shuffle_in_ram: bool = False, seed: int = 1234,
+                 align_path: Optional[str] = None,
+                 weight_path: Optional[str] = None):
         assert len(paths) >= 1
+        self.align_path = align_path     # --guided-alignment (pharaoh file)
+        self.weight_path = weight_path   # --data-weighting
         self.paths = list(paths)
         self.vocabs = list(vocabs)
         self.max_length = max_length
@@ -71,6 +76,15 @@ class Corpus:
         else:
             source = enumerate(self._raw_lines())
 
+        aligns = weights = None
+        if self.align_path:
+            from .alignment import parse_alignment
+            with _open(self.align_path) as f:
+                aligns = [parse_alignment(l) for l in f]
+        if self.weight_path:
+            with _open(self.weight_path) as f:
+                weights = [[float(x) for x in l.split()] for l in f]
+
         n_emitted = 0
         for pos, rows in source:
             tup = []
@@ -90,7 +104,13 @@ class Corpus:
             n_emitted += 1
             if n_emitted <= skip:
                 continue
-            yield (pos, tup)
+            extras = {}
+            if aligns is not None and pos < len(aligns):
+                extras["align"] = aligns[pos]
+            if weights is not None and pos < len(weights):
+                w = weights[pos]
+                extras["weight"] = w[0] if len(w) == 1 else w
+            yield (pos, tup, extras)
         self.epoch += 1
 
 
@@ -110,4 +130,4 @@ class TextInput:
             ids = self.vocab.encode(line.rstrip("\n"), add_eos=True)
             if len(ids) > self.max_length + 1 and self.max_length_crop:
                 ids = ids[: self.max_length] + [self.vocab.eos_id]
-            yield (i, [ids])
+            yield (i, [ids], {})

@@ -34,10 +34,13 @@ class Dense(nn.Module):
             nn.init.zeros_(self.b)
 
     def forward(self, x):
-        y = torch.matmul(x, self.W.to(x.dtype))
-        if self.b is not None:
-            y = y + self.b.to(x.dtype)
-        return y
+        W = self.W.to(x.dtype)
+        if self.b is None:
+            return torch.matmul(x, W)
+        # addmm routes to hipBLASLt's fused bias epilogue
+        shp = x.shape[:-1] + (W.shape[1],)
+        return torch.addmm(self.b.to(x.dtype), x.reshape(-1, W.shape[0]),
+                           W).view(shp)
 
 
 class LayerNorm(nn.Module):
@@ -165,16 +168,40 @@ class MultiHeadAttention(nn.Module):
 
     def forward(self, query, keys, values, mask=None, causal=False,
                 cache: Optional[dict] = None, cache_key: str = "",
-                lens: Optional[torch.Tensor] = None):
+                lens: Optional[torch.Tensor] = None,
+                collect: Optional[dict] = None):
         """cache: decode-time KV cache dict. For self-attention the new K/V
         are appended ({cache_key}_k/v); for cross-attention K/V are computed
         once and reused (ref: transformer.h cache_ keyed tensors)."""
-        q = self._split(self.Wq(query))
+        dt = query.dtype
+        fuse_qkv = keys is query and values is query and cache is None
+        if fuse_qkv:
+            # one [d, 3d] GEMM for Q/K/V (the cat is 6 MB vs three GEMM
+            # launches); split()s are views the fused kernel takes strided
+            W = torch.cat([self.Wq.W, self.Wk.W, self.Wv.W], dim=1).to(dt)
+            bias = torch.cat([self.Wq.b, self.Wk.b, self.Wv.b]).to(dt)
+            shp = query.shape[:-1] + (3 * self.dim,)
+            qkv = torch.addmm(bias, query.reshape(-1, self.dim), W).view(shp)
+            qh, kh, vh = qkv.split(self.dim, dim=-1)
+            q = self._split(qh)
+            k = self._split(kh)
+            v = self._split(vh)
+        else:
+            q = self._split(self.Wq(query))
         if cache is not None and cache_key + "_static" in cache:
             k, v = cache[cache_key + "_k"], cache[cache_key + "_v"]
-        else:
-            k = self._split(self.Wk(keys))
-            v = self._split(self.Wv(values))
+        elif not fuse_qkv:
+            if keys is values:
+                W = torch.cat([self.Wk.W, self.Wv.W], dim=1).to(dt)
+                bias = torch.cat([self.Wk.b, self.Wv.b]).to(dt)
+                shp = keys.shape[:-1] + (2 * self.dim,)
+                kv = torch.addmm(bias, keys.reshape(-1, self.dim), W).view(shp)
+                kh, vh = kv.split(self.dim, dim=-1)
+                k = self._split(kh)
+                v = self._split(vh)
+            else:
+                k = self._split(self.Wk(keys))
+                v = self._split(self.Wv(values))
             if cache is not None:
                 if cache_key.endswith("context"):   # cross-attn: static
                     cache[cache_key + "_k"] = k
@@ -187,8 +214,24 @@ class MultiHeadAttention(nn.Module):
                     cache[cache_key + "_k"] = k
                     cache[cache_key + "_v"] = v
         drop = self.dropout if self.training else 0.0
-        out = ops.attention(q, k, v, mask, causal=causal, dropout_p=drop,
-                            lens=lens)
+        if collect is not None:
+            # composite path that also returns head-averaged probs
+            # (guided alignment / --alignment output need them)
+            scale = q.shape[-1] ** -0.5
+            scores = torch.matmul(q, k.transpose(-1, -2)) * scale
+            if causal:
+                Tq, Tk = scores.shape[-2], scores.shape[-1]
+                cm = torch.ones(Tq, Tk, device=scores.device,
+                                dtype=torch.bool).tril(Tk - Tq)
+                scores = scores.masked_fill(~cm, float("-inf"))
+            if mask is not None:
+                scores = scores + mask
+            probs = torch.softmax(scores.float(), dim=-1)
+            collect["probs"] = probs.mean(dim=1)  # head-averaged [B,Tq,Tk]
+            out = torch.matmul(probs.to(q.dtype), v)
+        else:
+            out = ops.attention(q, k, v, mask, causal=causal, dropout_p=drop,
+                                lens=lens)
         return self.Wo(self._join(out))
 
 

@@ -79,14 +79,15 @@ class DecoderLayer(nn.Module):
 
     def forward(self, x, enc_out, src_mask, causal: bool,
                 cache: Optional[dict] = None, layer_id: int = 0,
-                src_lens=None):
+                src_lens=None, collect: Optional[dict] = None):
         h = self.pre_self(x)
         h = self.self_attn(h, h, h, causal=causal, cache=cache,
                            cache_key=f"l{layer_id}_self")
         x = self.post_self(h, residual=x)
         h = self.pre_ctx(x)
         h = self.context_attn(h, enc_out, enc_out, mask=src_mask, cache=cache,
-                              cache_key=f"l{layer_id}_context", lens=src_lens)
+                              cache_key=f"l{layer_id}_context", lens=src_lens,
+                              collect=collect)
         x = self.post_ctx(h, residual=x)
         h = self.pre_ffn(x)
         h = self.ffn(h)
@@ -147,6 +148,12 @@ class TransformerModel(nn.Module):
             self.out_b = nn.Parameter(torch.zeros(self.trg_vocab_size))
 
         self.compute_dtype = torch.float32
+        self.collect_cross_attn = False   # guided alignment training
+        self.last_cross_attn = None
+
+    def _guided_layer(self) -> int:
+        g = self.opt.get("transformer-guided-alignment-layer", "last")
+        return len(self.decoder) - 1 if g == "last" else int(g) - 1
 
     # ------------------------------------------------------------------
     def set_compute_dtype(self, dtype):
@@ -179,9 +186,14 @@ class TransformerModel(nn.Module):
         x = self.post_emb_dec(emb)
         lmask = _log_mask(src_mask, dt)
         lens = src_mask.sum(dim=1).to(torch.int32)
+        ga_layer = self._guided_layer() if self.collect_cross_attn else -1
+        self.last_cross_attn = None
         for i, layer in enumerate(self.decoder):
+            collect = {} if i == ga_layer else None
             x = layer(x, enc_out, lmask, causal=True, layer_id=i,
-                      src_lens=lens)
+                      src_lens=lens, collect=collect)
+            if collect is not None:
+                self.last_cross_attn = collect.get("probs")
         if self.post_top_dec is not None:
             x = self.post_top_dec(x)
         return self.output_logits(x)
@@ -202,9 +214,15 @@ class TransformerModel(nn.Module):
         x = self.post_emb_dec(emb)
         lmask = _log_mask(src_mask, dt)
         lens = src_mask.sum(dim=1).to(torch.int32)
+        ga_layer = len(self.decoder) - 1
         for i, layer in enumerate(self.decoder):
+            collect = {} if (cache is not None
+                            and cache.get("collect_alignment")) \
+                and i == ga_layer else None
             x = layer(x, enc_out, lmask, causal=False, cache=cache, layer_id=i,
-                      src_lens=lens)
+                      src_lens=lens, collect=collect)
+            if collect is not None:
+                cache.setdefault("alignments", []).append(collect["probs"])
         if self.post_top_dec is not None:
             x = self.post_top_dec(x)
         return self.output_logits(x[:, 0], shortlist)

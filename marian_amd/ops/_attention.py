@@ -38,11 +38,11 @@ def mask_to_lens(mask: Optional[torch.Tensor], B: int, Tk: int):
 class AttentionFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, lens, causal):
+        # strided [B,H,T,D] views pass straight through to the kernels
         ext = hip_ext(required=True)
         scale = q.shape[-1] ** -0.5
-        qc, kc, vc = q.contiguous(), k.contiguous(), v.contiguous()
-        o, lse = ext.attention_fwd(qc, kc, vc, lens, causal, scale, True)
-        ctx.save_for_backward(qc, kc, vc, o, lse,
+        o, lse = ext.attention_fwd(q, k, v, lens, causal, scale, True)
+        ctx.save_for_backward(q, k, v, o, lse,
                               lens if lens is not None else torch.Tensor())
         ctx.causal = causal
         ctx.scale = scale
@@ -54,7 +54,7 @@ class AttentionFn(torch.autograd.Function):
         ext = hip_ext(required=True)
         q, k, v, o, lse, lens = ctx.saved_tensors
         dq, dk, dv = ext.attention_bwd(
-            q, k, v, o, d_o.contiguous(), lse,
+            q, k, v, o, d_o, lse,
             lens if ctx.has_lens else None, ctx.causal, ctx.scale)
         return dq, dk, dv, None, None
 
@@ -85,6 +85,5 @@ def fused_attention(q, k, v, mask, causal,
                                     or v.requires_grad):
         return AttentionFn.apply(q, k, v, lens, causal)
     scale = D ** -0.5
-    o, _lse = ext.attention_fwd(q.contiguous(), k.contiguous(),
-                                v.contiguous(), lens, causal, scale, False)
+    o, _lse = ext.attention_fwd(q, k, v, lens, causal, scale, False)
     return o

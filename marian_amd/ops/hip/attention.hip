@@ -83,6 +83,13 @@ __device__ __forceinline__ void pack_p_to_b(const float p[16], bf16x8 out[2]) {
   }
 }
 
+// Element strides for a [B, H, T, D] tensor view (innermost dim must be
+// contiguous). Lets [B,T,H*D] activations feed the kernels through permuted
+// VIEWS - no transpose copies around attention.
+struct TStride {
+  long b, h, s;  // batch, head, sequence strides (elements)
+};
+
 // LDS V^T image: [D][KVBLK] with row pad to dodge bank conflicts.
 constexpr int KVBLK = 32;
 constexpr int QBLK = 32;
@@ -96,6 +103,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
                      const int* __restrict__ kv_lens,  // [B] or null
                      __hip_bfloat16* __restrict__ O,
                      float* __restrict__ LSE,          // [B*H*Tq] or null
+                     TStride qs, TStride ks, TStride os,
                      int B, int H, int Tq, int Tk, float scale) {
   __shared__ __hip_bfloat16 vt[D][KVBLK + VT_PAD];
   __shared__ float obounce[QBLK][D + 4];
@@ -110,8 +118,11 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
   if (q0 >= Tq) return;
   const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
 
-  const size_t qoff = (size_t)bh * Tq * D;
-  const size_t koff = (size_t)bh * Tk * D;
+  const int h = bh % H;
+  const __hip_bfloat16* Qb = Q + (size_t)b * qs.b + (size_t)h * qs.h;
+  const __hip_bfloat16* Kb = K + (size_t)b * ks.b + (size_t)h * ks.h;
+  const __hip_bfloat16* Vb = V + (size_t)b * ks.b + (size_t)h * ks.h;
+  __hip_bfloat16* Ob = O + (size_t)b * os.b + (size_t)h * os.h;
 
   // ---- load Q^T B-fragments (kept for the whole kernel) -----------------
   // B-fragment for chunk c (d = 16c + (hi)*8 + j): lane needs
@@ -119,7 +130,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
   bf16x8 qfrag[D / 16];
   {
     const int qrow = min(q0 + lq, Tq - 1);
-    const __hip_bfloat16* qp = Q + qoff + (size_t)qrow * D;
+    const __hip_bfloat16* qp = Qb + (size_t)qrow * qs.s;
 #pragma unroll
     for (int c = 0; c < D / 16; ++c)
       qfrag[c] = *reinterpret_cast<const bf16x8*>(qp + 16 * c + 8 * hi);
@@ -150,7 +161,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
         const int krow = k0 + kk + lane / (D / 8);
         const int dcol = (lane % (D / 8)) * 8;
         bf16x8 vv = (krow < kv_len)
-            ? *reinterpret_cast<const bf16x8*>(V + koff + (size_t)krow * D + dcol)
+            ? *reinterpret_cast<const bf16x8*>(Vb + (size_t)krow * ks.s + dcol)
             : (bf16x8)(__bf16)0.f;
         const int klocal = kk + lane / (D / 8);
 #pragma unroll
@@ -166,7 +177,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
     {
       // A-fragment: lane holds K[k0 + lq][d = 16c + hi*8 + j] (row = key)
       const int krow = k0 + lq;
-      const __hip_bfloat16* kp = K + koff + (size_t)min(krow, Tk - 1) * D;
+      const __hip_bfloat16* kp = Kb + (size_t)min(krow, Tk - 1) * ks.s;
 #pragma unroll
       for (int c = 0; c < D / 16; ++c) {
         bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp + 16 * c + 8 * hi);
@@ -245,7 +256,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           out[j] = (__bf16)obounce[qq + lane / lanes_per_row][dcol + j];
-        *reinterpret_cast<bf16x8*>(O + qoff + (size_t)qrow * D + dcol) = out;
+        *reinterpret_cast<bf16x8*>(Ob + (size_t)qrow * os.s + dcol) = out;
       }
     }
   }
@@ -258,8 +269,19 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
 
 }  // namespace
 
-// q,k,v: [B, H, T, D] bf16 contiguous. kv_lens: int32 [B] or undefined.
-// returns (o, lse)
+// Strided-view support: a [B,H,T,D] view whose last dim is contiguous and
+// 16B-aligned passes through un-copied (e.g. a permute of [B,T,H*D]).
+static bool view_ok(const torch::Tensor& t) {
+  return t.stride(3) == 1 && (t.stride(2) % 8 == 0) && (t.stride(1) % 8 == 0)
+      && (t.stride(0) % 8 == 0);
+}
+
+static TStride tstride(const torch::Tensor& t) {
+  return TStride{t.stride(0), t.stride(1), t.stride(2)};
+}
+
+// q,k,v: [B, H, T, D] bf16 (any strided view, D contiguous).
+// returns (o [B,H,T,D] view with BSHD storage, lse)
 std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v,
                                          c10::optional<torch::Tensor> kv_lens,
@@ -267,12 +289,20 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                                          bool need_lse) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4);
-  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
+  auto qc = view_ok(q) ? q : q.contiguous();
+  auto kc = view_ok(k) ? k : k.contiguous();
+  auto vc = (v.strides() == kc.strides() && view_ok(v)) ? v : v.contiguous();
+  if (vc.strides() != kc.strides()) {  // kernels share k/v strides
+    kc = kc.contiguous();
+    vc = vc.contiguous();
+  }
   const int B = qc.size(0), H = qc.size(1), Tq = qc.size(2), D = qc.size(3);
   const int Tk = kc.size(2);
   TORCH_CHECK(kc.size(0) == B && kc.size(1) == H && kc.size(3) == D);
   TORCH_CHECK(D == 64, "attention_fwd: only Dh=64 supported (got ", D, ")");
-  auto o = torch::empty_like(qc);
+  // O allocated in [B, T, H, D] storage, returned as a [B,H,T,D] view so
+  // the caller's head-join is a free reshape
+  auto o = torch::empty({B, Tq, H, (int)D}, qc.options()).permute({0, 2, 1, 3});
   torch::Tensor lse;
   float* lse_ptr = nullptr;
   if (need_lse) {
@@ -298,7 +328,8 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                      reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()), \
                      lens_ptr,                                               \
                      reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),        \
-                     lse_ptr, B, H, Tq, Tk, (float)scale)
+                     lse_ptr, tstride(qc), tstride(kc), tstride(o),          \
+                     B, H, Tq, Tk, (float)scale)
   if (causal && lens_ptr) RUN(true, true);
   else if (causal) RUN(true, false);
   else if (lens_ptr) RUN(false, true);
@@ -334,6 +365,7 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
                         const float* __restrict__ Dsum,  // [B*H*Tq]
                         const int* __restrict__ kv_lens,
                         __hip_bfloat16* __restrict__ dQ,
+                        TStride qs, TStride ks, TStride dos, TStride dqs,
                         int B, int H, int Tq, int Tk, float scale) {
   __shared__ __hip_bfloat16 kt[D][KVBLK + VT_PAD];  // K^T image per tile
   __shared__ float dq_b[QBLK][D + 4];
@@ -346,15 +378,19 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
   const int q0 = blockIdx.x * QBLK;
   if (q0 >= Tq) return;
   const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
-  const size_t qoff = (size_t)bh * Tq * D;
-  const size_t koff = (size_t)bh * Tk * D;
+  const int h = bh % H;
+  const __hip_bfloat16* Qb = Q + (size_t)b * qs.b + (size_t)h * qs.h;
+  const __hip_bfloat16* Kb = K + (size_t)b * ks.b + (size_t)h * ks.h;
+  const __hip_bfloat16* Vb = V + (size_t)b * ks.b + (size_t)h * ks.h;
+  const __hip_bfloat16* dOb = dO + (size_t)b * dos.b + (size_t)h * dos.h;
+  __hip_bfloat16* dQb = dQ + (size_t)b * dqs.b + (size_t)h * dqs.h;
 
   bf16x8 qfrag[D / 16], dofrag[D / 16];
   {
     const bool qlive = (q0 + lq) < Tq;
     const int qrow = qlive ? (q0 + lq) : (Tq - 1);
-    const __hip_bfloat16* qp = Q + qoff + (size_t)qrow * D;
-    const __hip_bfloat16* dp = dO + qoff + (size_t)qrow * D;
+    const __hip_bfloat16* qp = Qb + (size_t)qrow * qs.s;
+    const __hip_bfloat16* dp = dOb + (size_t)qrow * dos.s;
 #pragma unroll
     for (int c = 0; c < D / 16; ++c) {
       qfrag[c] = *reinterpret_cast<const bf16x8*>(qp + 16 * c + 8 * hi);
@@ -376,8 +412,8 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
     bf16x8 kfrag[D / 16], vfrag[D / 16];
     {
       const int krow = min(k0 + lq, Tk - 1);
-      const __hip_bfloat16* kp = K + koff + (size_t)krow * D;
-      const __hip_bfloat16* vp = V + koff + (size_t)krow * D;
+      const __hip_bfloat16* kp = Kb + (size_t)krow * ks.s;
+      const __hip_bfloat16* vp = Vb + (size_t)krow * ks.s;
 #pragma unroll
       for (int c = 0; c < D / 16; ++c) {
         kfrag[c] = *reinterpret_cast<const bf16x8*>(kp + 16 * c + 8 * hi);
@@ -388,7 +424,7 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
         const int krow2 = k0 + kk + lane / (D / 8);
         const int dcol = (lane % (D / 8)) * 8;
         bf16x8 kv = (krow2 < kv_len)
-            ? *reinterpret_cast<const bf16x8*>(K + koff + (size_t)krow2 * D + dcol)
+            ? *reinterpret_cast<const bf16x8*>(Kb + (size_t)krow2 * ks.s + dcol)
             : (bf16x8)(__bf16)0.f;
         const int klocal = kk + lane / (D / 8);
 #pragma unroll
@@ -447,7 +483,7 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         out[j] = (__bf16)dq_b[qq + lane / lanes_per_row][dcol + j];
-      *reinterpret_cast<bf16x8*>(dQ + qoff + (size_t)qrow * D + dcol) = out;
+      *reinterpret_cast<bf16x8*>(dQb + (size_t)qrow * dqs.s + dcol) = out;
     }
   }
 }
@@ -463,6 +499,7 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
                          const int* __restrict__ kv_lens,
                          __hip_bfloat16* __restrict__ dK,
                          __hip_bfloat16* __restrict__ dV,
+                         TStride qs, TStride ks, TStride dos, TStride dks,
                          int B, int H, int Tq, int Tk, float scale) {
   __shared__ __hip_bfloat16 qt[D][QBLK + VT_PAD];   // Q^T image per q-tile
   __shared__ __hip_bfloat16 dot[D][QBLK + VT_PAD];  // dO^T image per q-tile
@@ -476,14 +513,19 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
   const int k0 = blockIdx.x * KVBLK;
   if (k0 >= Tk) return;
   const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
-  const size_t qoff = (size_t)bh * Tq * D;
-  const size_t koff = (size_t)bh * Tk * D;
+  const int h = bh % H;
+  const __hip_bfloat16* Qb = Q + (size_t)b * qs.b + (size_t)h * qs.h;
+  const __hip_bfloat16* Kb = K + (size_t)b * ks.b + (size_t)h * ks.h;
+  const __hip_bfloat16* Vb = V + (size_t)b * ks.b + (size_t)h * ks.h;
+  const __hip_bfloat16* dOb = dO + (size_t)b * dos.b + (size_t)h * dos.h;
+  __hip_bfloat16* dKb = dK + (size_t)b * dks.b + (size_t)h * dks.h;
+  __hip_bfloat16* dVb = dV + (size_t)b * dks.b + (size_t)h * dks.h;
 
   bf16x8 kfrag[D / 16], vfrag[D / 16];
   {
     const int krow = min(k0 + lq, Tk - 1);
-    const __hip_bfloat16* kp = K + koff + (size_t)krow * D;
-    const __hip_bfloat16* vp = V + koff + (size_t)krow * D;
+    const __hip_bfloat16* kp = Kb + (size_t)krow * ks.s;
+    const __hip_bfloat16* vp = Vb + (size_t)krow * ks.s;
 #pragma unroll
     for (int c = 0; c < D / 16; ++c) {
       kfrag[c] = *reinterpret_cast<const bf16x8*>(kp + 16 * c + 8 * hi);
@@ -506,8 +548,8 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
     {
       const bool qlive = (q0 + lq) < Tq;
       const int qrow = qlive ? (q0 + lq) : (Tq - 1);
-      const __hip_bfloat16* qp = Q + qoff + (size_t)qrow * D;
-      const __hip_bfloat16* dp = dO + qoff + (size_t)qrow * D;
+      const __hip_bfloat16* qp = Qb + (size_t)qrow * qs.s;
+      const __hip_bfloat16* dp = dOb + (size_t)qrow * dos.s;
 #pragma unroll
       for (int c = 0; c < D / 16; ++c) {
         qfrag[c] = *reinterpret_cast<const bf16x8*>(qp + 16 * c + 8 * hi);
@@ -521,8 +563,8 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
         const int dcol = (lane % (D / 8)) * 8;
         const bool live = qrow2 < Tq;
         const int qr = live ? qrow2 : (Tq - 1);
-        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Q + qoff + (size_t)qr * D + dcol);
-        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dO + qoff + (size_t)qr * D + dcol);
+        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (size_t)qr * qs.s + dcol);
+        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (size_t)qr * dos.s + dcol);
         if (!live) {
           vq = (bf16x8)(__bf16)0.f;
           vd = (bf16x8)(__bf16)0.f;
@@ -580,7 +622,7 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
   // epilogue: dV then dK via the same LDS bounce ([32k x 32d] frags:
   // lane = d-col, regs = k-rows)
   for (int which = 0; which < 2; ++which) {
-    __hip_bfloat16* out_base = (which == 0) ? dV : dK;
+    __hip_bfloat16* out_base = (which == 0) ? dVb : dKb;
     f32x16* acc = (which == 0) ? dv_acc : dk_acc;
 #pragma unroll
     for (int dt2 = 0; dt2 < D / 32; ++dt2)
@@ -600,7 +642,7 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
         for (int j = 0; j < 8; ++j)
           out[j] = live ? (__bf16)kv_b[kk + lane / lanes_per_row][dcol + j]
                         : (__bf16)0.f;
-        *reinterpret_cast<bf16x8*>(out_base + koff + (size_t)krow * D + dcol) = out;
+        *reinterpret_cast<bf16x8*>(out_base + (size_t)krow * dks.s + dcol) = out;
       }
     }
     __builtin_amdgcn_s_barrier();
@@ -609,21 +651,29 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
 
 }  // namespace
 
-// returns (dq, dk, dv) - all bf16
+// returns (dq, dk, dv) - all bf16, allocated in [B,T,H,D] storage and
+// returned as [B,H,T,D] views (caller reshapes for free)
 std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v, torch::Tensor o,
                                          torch::Tensor d_o, torch::Tensor lse,
                                          c10::optional<torch::Tensor> kv_lens,
                                          bool causal, double scale) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
-  auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
-  auto oc = o.contiguous(), doc = d_o.contiguous();
+  auto qc = view_ok(q) ? q : q.contiguous();
+  auto kc = view_ok(k) ? k : k.contiguous();
+  auto vc = (v.strides() == kc.strides() && view_ok(v)) ? v : v.contiguous();
+  if (vc.strides() != kc.strides()) {
+    kc = kc.contiguous();
+    vc = vc.contiguous();
+  }
+  auto oc = view_ok(o) ? o : o.contiguous();
+  auto doc = view_ok(d_o) ? d_o : d_o.contiguous();
   const int B = qc.size(0), H = qc.size(1), Tq = qc.size(2), D = qc.size(3);
   const int Tk = kc.size(2);
   TORCH_CHECK(D == 64, "attention_bwd: only Dh=64 supported");
-  auto dq = torch::empty_like(qc);
-  auto dk = torch::empty_like(kc);
-  auto dv = torch::empty_like(vc);
+  auto dq = torch::empty({B, Tq, H, (int)D}, qc.options()).permute({0, 2, 1, 3});
+  auto dk = torch::empty({B, Tk, H, (int)D}, qc.options()).permute({0, 2, 1, 3});
+  auto dv = torch::empty({B, Tk, H, (int)D}, qc.options()).permute({0, 2, 1, 3});
   // D = rowsum(dO o O) per query row, fp32
   auto dsum = (doc.to(torch::kFloat32) * oc.to(torch::kFloat32))
                   .sum(-1).contiguous();
@@ -647,6 +697,7 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
                        lse.data_ptr<float>(), dsum.data_ptr<float>(),         \
                        lens_ptr,                                              \
                        reinterpret_cast<__hip_bfloat16*>(dq.data_ptr()),      \
+                       tstride(qc), tstride(kc), tstride(doc), tstride(dq),   \
                        B, H, Tq, Tk, (float)scale);                           \
     hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, CAUSAL, HASLEN>), gk,         \
                        dim3(64), 0, stream,                                   \
@@ -658,6 +709,7 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
                        lens_ptr,                                              \
                        reinterpret_cast<__hip_bfloat16*>(dk.data_ptr()),      \
                        reinterpret_cast<__hip_bfloat16*>(dv.data_ptr()),      \
+                       tstride(qc), tstride(kc), tstride(doc), tstride(dk),   \
                        B, H, Tq, Tk, (float)scale);                           \
   } while (0)
   if (causal && lens_ptr) RUNB(true, true);

@@ -89,6 +89,12 @@ class SyncGraphGroup:
         self.criterion = CrossEntropyCost(
             float(options.get("label-smoothing", 0.0) or 0.0),
             options.get("cost-type", "ce-sum"))
+        ga = options.get("guided-alignment", "none")
+        self.guided_alignment = (ga not in (None, "none", ""))
+        self.guided_weight = float(options.get("guided-alignment-weight", 0.1))
+        self.guided_cost = options.get("guided-alignment-cost", "ce")
+        if self.guided_alignment and hasattr(self.model, "collect_cross_attn"):
+            self.model.collect_cross_attn = True
         self.delay = max(1, int(float(options.get("optimizer-delay", 1) or 1)))
         self.scaler = CostScaler(options.list("cost-scaling"),
                                  self.compute_dtype == torch.float16)
@@ -151,6 +157,14 @@ class SyncGraphGroup:
         lv = self.criterion(logits, batch)
         # backprop the SUM; cost-type scaling happens at optimizer time
         bloss = lv.ce_sum
+        if self.guided_alignment and batch.guided_alignment is not None \
+                and getattr(self.model, "last_cross_attn", None) is not None:
+            from ..data.alignment import guided_alignment_loss
+            ga_loss = guided_alignment_loss(
+                self.model.last_cross_attn,
+                batch.guided_alignment.to(self.device),
+                batch.target.mask, self.guided_cost)
+            bloss = bloss + self.guided_weight * ga_loss * max(1, lv.label_count)
         if self.scaler.enabled and self.scaler.scale != 1.0:
             bloss = bloss * self.scaler.scale
         bloss.backward()

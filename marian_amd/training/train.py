@@ -106,13 +106,16 @@ class Train:
     # ------------------------------------------------------------------
     def _batch_generator(self, skip_batches: int = 0):
         opts = self.opt
+        ga = opts.get("guided-alignment", "none")
         corpus = Corpus(opts.list("train-sets"), self.vocabs,
                         max_length=int(opts.get("max-length", 50)),
                         max_length_crop=bool(opts.get("max-length-crop", False)),
                         tsv=bool(opts.get("tsv", False)),
                         tsv_fields=int(opts.get("tsv-fields", 0) or 0),
                         shuffle_in_ram=bool(opts.get("shuffle-in-ram", False)),
-                        seed=int(opts.get("seed", 0) or 1234))
+                        seed=int(opts.get("seed", 0) or 1234),
+                        align_path=None if ga in (None, "none", "") else ga,
+                        weight_path=opts.get("data-weighting"))
         corpus.epoch = self.state.epochs - 1
         shuffle = opts.get("shuffle", "data") != "none" \
             and not opts.get("no-shuffle", False)

@@ -43,7 +43,7 @@ def test_corpus_max_length_filter(tiny_corpus):
     src, trg = tiny_corpus
     v = Vocab.create_from_corpus([src, trg])
     c = Corpus([src, trg], [v, v], max_length=5, max_length_crop=False)
-    for _, streams in c.iter_epoch(shuffle=False):
+    for _, streams, _x in c.iter_epoch(shuffle=False):
         assert all(len(s) <= 6 for s in streams)  # +1 for EOS
 
     c2 = Corpus([src, trg], [v, v], max_length=5, max_length_crop=True)
@@ -101,7 +101,7 @@ def test_shuffle_deterministic(tiny_corpus):
     v = Vocab.create_from_corpus([src, trg])
     c1 = Corpus([src, trg], [v, v], shuffle_in_ram=True, seed=5)
     c2 = Corpus([src, trg], [v, v], shuffle_in_ram=True, seed=5)
-    l1 = [pos for pos, _ in c1.iter_epoch()]
-    l2 = [pos for pos, _ in c2.iter_epoch()]
+    l1 = [t[0] for t in c1.iter_epoch()]
+    l2 = [t[0] for t in c2.iter_epoch()]
     assert l1 == l2
     assert l1 != sorted(l1)

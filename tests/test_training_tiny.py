@@ -107,3 +107,25 @@ def test_loss_decreases(workdir):
     tr.run()
     after = probe_ce()
     assert after < before - 0.5, (before, after)
+
+
+def test_guided_alignment_and_data_weighting(workdir):
+    """--guided-alignment + --data-weighting end to end (tiny)."""
+    tmp_path, src, trg = workdir
+    n = sum(1 for _ in open(src))
+    align = tmp_path / "align.txt"
+    weights = tmp_path / "weights.txt"
+    with open(align, "w") as fa, open(weights, "w") as fw:
+        for line in open(src):
+            L = len(line.split())
+            fa.write(" ".join(f"{i}-{i}" for i in range(L)) + "\n")
+            fw.write("1.5\n")
+    opts = _opts(tmp_path, src, src, extra=[
+        "--guided-alignment", str(align),
+        "--guided-alignment-weight", "0.2",
+        "--data-weighting", str(weights),
+        "--after-batches", "10"])
+    tr = Train(opts)
+    assert tr.graph_group.guided_alignment
+    tr.run()
+    assert tr.state.batches == 10
