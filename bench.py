@@ -90,7 +90,7 @@ def main():
 
     if args.batch == 0:
         # per-GPU batch: fill MI355X HBM comfortably; tiny on CPU
-        args.batch = 256 if on_gpu else 8
+        args.batch = 1024 if on_gpu else 8
         if args.model == "transformer-tiny":
             args.batch = 64 if on_gpu else 8
 

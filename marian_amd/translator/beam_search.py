@@ -47,7 +47,8 @@ class BeamSearch:
                  n_best: int = 1, max_length_factor: float = 3.0,
                  max_length: int = 256, eos_id: int = 0, unk_id: int = 1,
                  weights: Optional[List[float]] = None,
-                 shortlist_generator=None):
+                 shortlist_generator=None, collect_alignment: bool = False):
+        self.collect_alignment = collect_alignment
         self.scorers = scorers if isinstance(scorers, (list, tuple)) else [scorers]
         self.K = beam_size
         self.alpha = normalize
@@ -83,11 +84,14 @@ class BeamSearch:
 
         # encoder pass per scorer; expand to beam immediately
         states = []
-        for sc in self.scorers:
+        for si, sc in enumerate(self.scorers):
             enc_out, src_mask = sc.start(src)
             enc_out = enc_out.repeat_interleave(K, dim=0)
             src_mask = src_mask.repeat_interleave(K, dim=0)
-            states.append({"enc": enc_out, "mask": src_mask, "cache": {}})
+            cache = {}
+            if self.collect_alignment and si == 0:
+                cache["collect_alignment"] = True
+            states.append({"enc": enc_out, "mask": src_mask, "cache": cache})
 
         max_len = min(self.max_length,
                       int(self.max_length_factor * src.indices.shape[1]) + 1)
@@ -108,6 +112,7 @@ class BeamSearch:
                                device=device)
         fin_count = torch.zeros(B, dtype=torch.long, device=device)
 
+        align_steps: List[torch.Tensor] = []
         V = None
         for step in range(max_len):
             # ensemble step: weighted sum of log-probs
@@ -117,6 +122,10 @@ class BeamSearch:
                 lp = ops.log_softmax(logits.float())
                 logprobs = w * lp if logprobs is None else logprobs + w * lp
             V = logprobs.shape[-1]
+            if self.collect_alignment:
+                al = states[0]["cache"].pop("alignments", None)
+                align_steps.append(al[-1].squeeze(1).cpu()
+                                   if al else None)  # [B*K, Tsrc]
             if not self.allow_unk and shortlist is None:
                 logprobs[:, self.unk_id] = NEG_INF
             # dead beams must not propose continuations
@@ -185,18 +194,20 @@ class BeamSearch:
         for b in range(B):
             hyps: List[Hypothesis] = []
             for c in range(int(fin_count_h[b])):
-                toks = self._backtrace(bp, tk, b, int(fin_step_h[b, c]),
-                                       int(fin_beam_h[b, c]))
+                toks, al = self._backtrace(bp, tk, b, int(fin_step_h[b, c]),
+                                           int(fin_beam_h[b, c]),
+                                           align_steps, K)
                 score = float(fin_scores_h[b, c])
                 hyps.append(Hypothesis(
-                    tokens=toks, score=score,
+                    tokens=toks, score=score, alignment=al,
                     normalized=self._normalized(score, int(fin_len_h[b, c]))))
             if not hyps:
                 # no EOS reached: take the best alive beam at the last step
                 kk = int(scores_h[b].argmax())
-                toks = self._backtrace(bp, tk, b, len(tok_steps) - 1, kk)
+                toks, al = self._backtrace(bp, tk, b, len(tok_steps) - 1, kk,
+                                           align_steps, K)
                 score = float(scores_h[b, kk])
-                hyps.append(Hypothesis(tokens=toks, score=score,
+                hyps.append(Hypothesis(tokens=toks, score=score, alignment=al,
                                        normalized=self._normalized(
                                            score, max(len(toks), 1))))
             hyps.sort(key=lambda h: h.normalized, reverse=True)
@@ -205,17 +216,26 @@ class BeamSearch:
 
     # ------------------------------------------------------------------
     @staticmethod
-    def _backtrace(bp, tk, b, step, beam) -> List[int]:
+    def _backtrace(bp, tk, b, step, beam, align_steps=None, K=1):
         toks = []
+        aligns = []
         k = beam
         for t in range(step, -1, -1):
             toks.append(int(tk[t, b, k]))
-            k = int(bp[t, b, k])
+            prev_k = int(bp[t, b, k])
+            if align_steps and t < len(align_steps) \
+                    and align_steps[t] is not None:
+                # the alignment at step t was produced under the PREVIOUS
+                # beam layout: row = b*K + back-pointer beam
+                aligns.append(align_steps[t][b * K + prev_k])
+            k = prev_k
         toks.reverse()
-        # strip the final EOS if present (printer adds nothing)
+        aligns.reverse()
         if toks and toks[-1] == 0:
             toks = toks[:-1]
-        return toks
+            if aligns:
+                aligns = aligns[:-1]
+        return toks, (aligns if aligns else None)
 
 
 class ModelScorer:

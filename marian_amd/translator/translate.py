@@ -50,18 +50,39 @@ class OutputPrinter:
         self.vocab = vocab
         self.n_best = bool(options.get("n-best", False))
         self.word_scores = bool(options.get("word-scores", False))
+        self.alignment = options.get("alignment")
 
     def text(self, hyp) -> str:
         return self.vocab.decode(hyp.tokens, remove_eos=True)
+
+    def _align_str(self, h) -> str:
+        if not self.alignment or h.alignment is None:
+            return ""
+        import torch
+        from ..data.alignment import alignment_to_string
+        att = torch.stack([a for a in h.alignment])  # [T_trg, T_src]
+        mode = self.alignment
+        if mode in ("soft", "hard"):
+            return alignment_to_string(att, mode=mode)
+        try:
+            return alignment_to_string(att, threshold=float(mode), mode="t")
+        except ValueError:
+            return alignment_to_string(att, mode="hard")
 
     def print(self, sentence_id: int, result: SentenceResult) -> str:
         if self.n_best:
             lines = []
             for h in result.nbest:
-                lines.append(f"{sentence_id} ||| {self.text(h)} ||| "
-                             f"F0= {h.score:.6f} ||| {h.normalized:.6f}")
+                line = f"{sentence_id} ||| {self.text(h)}"
+                if self.alignment:
+                    line += f" ||| {self._align_str(h)}"
+                line += f" ||| F0= {h.score:.6f} ||| {h.normalized:.6f}"
+                lines.append(line)
             return "\n".join(lines)
-        return self.text(result.best)
+        out = self.text(result.best)
+        if self.alignment:
+            out += f" ||| {self._align_str(result.best)}"
+        return out
 
 
 class OutputCollector:
@@ -115,7 +136,8 @@ class Translator:
             max_length=int(options.get("max-length", 1000)),
             eos_id=self.trg_vocab.eos_id, unk_id=self.trg_vocab.unk_id,
             weights=[float(w) for w in options.list("weights")] or None,
-            shortlist_generator=shortlist_gen)
+            shortlist_generator=shortlist_gen,
+            collect_alignment=bool(options.get("alignment")))
 
     def _nbest_size(self):
         if self.opt.get("n-best", False):

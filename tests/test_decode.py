@@ -200,3 +200,21 @@ def test_server_app(trained_model_dir):
         ws.send_text("w5 w6")
         reply = ws.receive_text()
         assert isinstance(reply, str)
+
+
+def test_decoder_alignment_output(trained_model_dir, tmp_path):
+    d = trained_model_dir
+    inp = d / "in2.txt"
+    inp.write_text("w5 w6 w7\n")
+    out = d / "out2.txt"
+    from marian_amd.translator.translate import main_decoder
+    main_decoder(["--models", str(d / "model.npz"),
+                  "--vocabs", str(d / "v.yml"), str(d / "v.yml"),
+                  "--input", str(inp), "--output", str(out),
+                  "--beam-size", "2", "--cpu-threads", "1",
+                  "--alignment", "hard", "--max-length", "20", "--quiet"])
+    line = out.read_text().splitlines()[0]
+    assert "|||" in line
+    align = line.split("|||")[1].strip()
+    if align:  # model may emit empty translation; pairs look like "0-0"
+        assert all("-" in p for p in align.split())
