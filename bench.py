@@ -56,7 +56,9 @@ def build_options(args, world):
         "--clip-norm", "0.0",
         "--max-length", str(args.seq_len),
         "--seed", "1111",
-        "--precision", args.dtype, "float32",
+        "--precision",
+        {"fp8": "float8", "float8": "float8"}.get(args.dtype, args.dtype),
+        "float32",
     ]
     opts = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
     opts.set("dim-vocabs", [args.vocab, args.vocab])
@@ -74,7 +76,8 @@ def main():
     p.add_argument("--seq-len", type=int, default=64)
     p.add_argument("--vocab", type=int, default=32000)
     p.add_argument("--dtype", default="bf16",
-                   choices=["bf16", "fp16", "fp32", "float32", "bfloat16"])
+                   choices=["bf16", "fp16", "fp32", "float32", "bfloat16",
+                            "fp8", "float8"])
     args = p.parse_args()
 
     rank, local_rank, world = dist_env()
@@ -135,7 +138,8 @@ def main():
 
     if rank == 0:
         dt = {"bf16": "bf16", "bfloat16": "bf16", "fp16": "fp16",
-              "fp32": "fp32", "float32": "fp32"}[args.dtype]
+              "fp32": "fp32", "float32": "fp32", "fp8": "fp8",
+              "float8": "fp8"}[args.dtype]
         print(json.dumps({
             "metric": "src tokens/sec",
             "value": round(value, 2),

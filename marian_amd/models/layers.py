@@ -34,13 +34,7 @@ class Dense(nn.Module):
             nn.init.zeros_(self.b)
 
     def forward(self, x):
-        W = self.W.to(x.dtype)
-        if self.b is None:
-            return torch.matmul(x, W)
-        # addmm routes to hipBLASLt's fused bias epilogue
-        shp = x.shape[:-1] + (W.shape[1],)
-        return torch.addmm(self.b.to(x.dtype), x.reshape(-1, W.shape[0]),
-                           W).view(shp)
+        return ops.linear(x, self.W, self.b)
 
 
 class LayerNorm(nn.Module):
@@ -178,10 +172,9 @@ class MultiHeadAttention(nn.Module):
         if fuse_qkv:
             # one [d, 3d] GEMM for Q/K/V (the cat is 6 MB vs three GEMM
             # launches); split()s are views the fused kernel takes strided
-            W = torch.cat([self.Wq.W, self.Wk.W, self.Wv.W], dim=1).to(dt)
-            bias = torch.cat([self.Wq.b, self.Wk.b, self.Wv.b]).to(dt)
-            shp = query.shape[:-1] + (3 * self.dim,)
-            qkv = torch.addmm(bias, query.reshape(-1, self.dim), W).view(shp)
+            W = torch.cat([self.Wq.W, self.Wk.W, self.Wv.W], dim=1)
+            bias = torch.cat([self.Wq.b, self.Wk.b, self.Wv.b])
+            qkv = ops.linear(query, W, bias)
             qh, kh, vh = qkv.split(self.dim, dim=-1)
             q = self._split(qh)
             k = self._split(kh)
@@ -192,10 +185,9 @@ class MultiHeadAttention(nn.Module):
             k, v = cache[cache_key + "_k"], cache[cache_key + "_v"]
         elif not fuse_qkv:
             if keys is values:
-                W = torch.cat([self.Wk.W, self.Wv.W], dim=1).to(dt)
-                bias = torch.cat([self.Wk.b, self.Wv.b]).to(dt)
-                shp = keys.shape[:-1] + (2 * self.dim,)
-                kv = torch.addmm(bias, keys.reshape(-1, self.dim), W).view(shp)
+                W = torch.cat([self.Wk.W, self.Wv.W], dim=1)
+                bias = torch.cat([self.Wk.b, self.Wv.b])
+                kv = ops.linear(keys, W, bias)
                 kh, vh = kv.split(self.dim, dim=-1)
                 k = self._split(kh)
                 v = self._split(vh)

@@ -73,6 +73,8 @@ class SyncGraphGroup:
 
         prec = options.list("precision") or ["float32"]
         self.compute_dtype = _DTYPES.get(prec[0], torch.float32)
+        from ..ops.fp8 import enable_fp8
+        enable_fp8(prec[0] in ("float8", "fp8") and device.type == "cuda")
         self.model = model.to(device)
         if self.compute_dtype != torch.float32:
             self.model = self.model.to(self.compute_dtype)
