@@ -146,18 +146,29 @@ class BeamSearch:
 
             is_eos = (real_words == self.eos_id) & (top_scores > NEG_INF / 2)
 
-            # record finished hypotheses (EOS chosen)
-            n_new = is_eos.sum(dim=1)
-            if int(n_new.sum()) > 0:
-                for b in torch.nonzero(n_new, as_tuple=False).flatten().tolist():
-                    for kk in torch.nonzero(is_eos[b], as_tuple=False).flatten().tolist():
-                        c = int(fin_count[b])
-                        if c < fin_scores.shape[1]:
-                            fin_scores[b, c] = top_scores[b, kk]
-                            fin_len[b, c] = step + 1
-                            fin_step[b, c] = step
-                            fin_beam[b, c] = kk
-                            fin_count[b] += 1
+            # record finished hypotheses (EOS chosen) - fully on device:
+            # slot for the j-th EOS of row b = fin_count[b] + rank_j
+            cap = fin_scores.shape[1]
+            eos_rank = torch.cumsum(is_eos.long(), dim=1) - 1
+            slot = fin_count[:, None] + eos_rank
+            valid = is_eos & (slot < cap) & (slot >= 0)
+            slot_c = slot.clamp(0, cap - 1)
+            kidx = torch.arange(K, device=device)[None, :].expand(B, K)
+            fin_scores.scatter_(1, slot_c,
+                                torch.where(valid, top_scores,
+                                            fin_scores.gather(1, slot_c)))
+            fin_len.scatter_(1, slot_c,
+                             torch.where(valid,
+                                         torch.full_like(slot_c, step + 1),
+                                         fin_len.gather(1, slot_c)))
+            fin_step.scatter_(1, slot_c,
+                              torch.where(valid,
+                                          torch.full_like(slot_c, step),
+                                          fin_step.gather(1, slot_c)))
+            fin_beam.scatter_(1, slot_c,
+                              torch.where(valid, kidx,
+                                          fin_beam.gather(1, slot_c)))
+            fin_count = fin_count + valid.sum(dim=1)
 
             # EOS-ed candidates die; their slot keeps -inf so other hyps win
             scores = torch.where(is_eos, torch.full_like(top_scores, NEG_INF),
