@@ -218,3 +218,26 @@ def test_decoder_alignment_output(trained_model_dir, tmp_path):
     align = line.split("|||")[1].strip()
     if align:  # model may emit empty translation; pairs look like "0-0"
         assert all("-" in p for p in align.split())
+
+
+def test_embedder(trained_model_dir, capsys):
+    d = trained_model_dir
+    inp = d / "emb_in.txt"
+    inp.write_text("w5 w6 w7\nw8 w9\n")
+    from marian_amd.translator.embedder import main_embedder
+    main_embedder(["--models", str(d / "model.npz"),
+                   "--vocabs", str(d / "v.yml"),
+                   "--input", str(inp), "--cpu-threads", "1", "--quiet"])
+    out = capsys.readouterr().out.strip().splitlines()
+    assert len(out) == 2
+    assert len(out[0].split()) == 64  # dim-emb of the tiny model
+
+    sim_in2 = d / "emb_in2.txt"
+    sim_in2.write_text("w5 w6 w7\nw20 w21\n")
+    main_embedder(["--models", str(d / "model.npz"),
+                   "--vocabs", str(d / "v.yml"),
+                   "--input", str(inp), str(sim_in2),
+                   "--compute-similarity", "--cpu-threads", "1", "--quiet"])
+    sims = [float(x) for x in capsys.readouterr().out.strip().splitlines()]
+    assert len(sims) == 2
+    assert sims[0] > 0.99  # identical sentence

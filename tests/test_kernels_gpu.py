@@ -367,3 +367,69 @@ def test_fp8_training_step(ext):
         assert ce < first - 0.2, (first, ce)
     finally:
         enable_fp8(False)
+
+
+def test_marian_train_cli_on_gpu(ext, tmp_path):
+    """Full marian-train CLI path on the GPU (bf16, checkpoints, resume)."""
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    src = str(tmp_path / "c.src")
+    write_synthetic_corpus(src, str(tmp_path / "c.trg"), 400, 200, 3, 16,
+                           seed=5)
+    argv = ["--train-sets", src, src,
+            "--vocabs", str(tmp_path / "v.yml"), str(tmp_path / "v.yml"),
+            "--model", str(tmp_path / "model.npz"),
+            "--dim-emb", "256", "--enc-depth", "2", "--dec-depth", "2",
+            "--transformer-heads", "4", "--transformer-dim-ffn", "512",
+            "--mini-batch", "64", "--learn-rate", "0.005",
+            "--after-batches", "30", "--seed", "3",
+            "--disp-freq", "10u", "--save-freq", "30u", "--max-length", "20",
+            "--tied-embeddings-all", "--cost-type", "ce-mean-words",
+            "--precision", "bfloat16", "float32", "--overwrite", "--quiet"]
+    opts = ConfigParser(MODE_TRAINING).parse_options(argv)
+    tr = Train(opts)
+    assert tr.device.type == "cuda"
+    tr.run()
+    assert tr.state.batches == 30
+    import os
+    assert os.path.exists(str(tmp_path / "model.npz"))
+    # resume on GPU
+    opts2 = ConfigParser(MODE_TRAINING).parse_options(
+        argv[:-1] + ["--after-batches", "35", "--quiet"])
+    tr2 = Train(opts2)
+    assert tr2.state.batches == 30
+    tr2.run()
+    assert tr2.state.batches == 35
+
+
+def test_decoder_cli_on_gpu(ext, tmp_path):
+    """marian-decoder on GPU with the fused decode path."""
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    from marian_amd.translator.translate import main_decoder
+    src = str(tmp_path / "c.src")
+    write_synthetic_corpus(src, str(tmp_path / "c.trg"), 300, 100, 3, 10,
+                           seed=11)
+    argv = ["--train-sets", src, src,
+            "--vocabs", str(tmp_path / "v.yml"), str(tmp_path / "v.yml"),
+            "--model", str(tmp_path / "model.npz"),
+            "--dim-emb", "128", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "256",
+            "--mini-batch", "64", "--learn-rate", "0.01",
+            "--after-batches", "80", "--seed", "3", "--max-length", "20",
+            "--tied-embeddings-all", "--cost-type", "ce-mean-words",
+            "--precision", "bfloat16", "float32", "--overwrite", "--quiet",
+            "--disp-freq", "40u", "--save-freq", "80u"]
+    Train(ConfigParser(MODE_TRAINING).parse_options(argv)).run()
+    inp = tmp_path / "in.txt"
+    inp.write_text("w5 w6 w7\nw8 w9 w10 w11\n")
+    out = tmp_path / "out.txt"
+    main_decoder(["--models", str(tmp_path / "model.npz"),
+                  "--vocabs", str(tmp_path / "v.yml"), str(tmp_path / "v.yml"),
+                  "--input", str(inp), "--output", str(out),
+                  "--beam-size", "4", "--mini-batch", "2", "--quiet",
+                  "--max-length", "20", "--precision", "bfloat16"])
+    lines = out.read_text().splitlines()
+    assert len(lines) == 2
