@@ -60,7 +60,8 @@ class ConfigParser:
 
     def _declare_all(self):
         a = self._add
-        T, D, S, E = MODE_TRAINING, MODE_TRANSLATION, MODE_SCORING, MODE_SERVER
+        T, D, S = MODE_TRAINING, MODE_TRANSLATION, MODE_SCORING
+        E = MODE_EMBEDDING
 
         # ---- general (ref: config_parser.cpp addOptionsGeneral) ----
         a("config", str, None, "Paths to config file(s) in YAML format",
