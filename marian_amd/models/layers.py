@@ -200,11 +200,29 @@ class MultiHeadAttention(nn.Module):
                     cache[cache_key + "_v"] = v
                     cache[cache_key + "_static"] = True
                 else:                               # self-attn: append
-                    if cache_key + "_k" in cache:
-                        k = torch.cat([cache[cache_key + "_k"], k], dim=2)
-                        v = torch.cat([cache[cache_key + "_v"], v], dim=2)
-                    cache[cache_key + "_k"] = k
-                    cache[cache_key + "_v"] = v
+                    # preallocated KV buffer: in-place write + slice view
+                    # (no per-step torch.cat; ref keeps append semantics)
+                    step = cache.get("step", 0)
+                    max_len = cache.get("max_len", 0)
+                    if max_len and cache_key + "_buf_k" not in cache:
+                        B_, H_, _, D_ = k.shape
+                        cache[cache_key + "_buf_k"] = torch.zeros(
+                            B_, H_, max_len, D_, device=k.device, dtype=k.dtype)
+                        cache[cache_key + "_buf_v"] = torch.zeros_like(
+                            cache[cache_key + "_buf_k"])
+                    if max_len:
+                        bk = cache[cache_key + "_buf_k"]
+                        bv = cache[cache_key + "_buf_v"]
+                        bk[:, :, step:step + k.shape[2]] = k
+                        bv[:, :, step:step + v.shape[2]] = v
+                        k = bk[:, :, :step + 1]
+                        v = bv[:, :, :step + 1]
+                    else:
+                        if cache_key + "_k" in cache:
+                            k = torch.cat([cache[cache_key + "_k"], k], dim=2)
+                            v = torch.cat([cache[cache_key + "_v"], v], dim=2)
+                        cache[cache_key + "_k"] = k
+                        cache[cache_key + "_v"] = v
         drop = self.dropout if self.training else 0.0
         if collect is not None:
             # composite path that also returns head-averaged probs

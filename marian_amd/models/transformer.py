@@ -212,6 +212,7 @@ class TransformerModel(nn.Module):
         # position of the CURRENT step (the word emitted at t-1 sits at t)
         emb = self.trg_embed.add_positions(emb, step)
         x = self.post_emb_dec(emb)
+        cache["step"] = step
         lmask = _log_mask(src_mask, dt)
         lens = src_mask.sum(dim=1).to(torch.int32)
         ga_layer = len(self.decoder) - 1

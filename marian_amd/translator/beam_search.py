@@ -95,6 +95,8 @@ class BeamSearch:
 
         max_len = min(self.max_length,
                       int(self.max_length_factor * src.indices.shape[1]) + 1)
+        for st in states:
+            st["cache"]["max_len"] = max_len
 
         scores = torch.full((B, K), NEG_INF, device=device)
         scores[:, 0] = 0.0  # only beam 0 live at step 0
