@@ -213,6 +213,20 @@ class TransformerModel(nn.Module):
         emb = self.trg_embed.add_positions(emb, step)
         x = self.post_emb_dec(emb)
         cache["step"] = step
+        # all layers' self-attn KV in ONE buffer so the beam reorder is a
+        # single gather; per-layer views are refreshed each step (the beam
+        # search replaces the backing tensor on reorder)
+        max_len = cache.get("max_len", 0)
+        if max_len:
+            L = len(self.decoder)
+            if "self_kv" not in cache:
+                H = self.decoder[0].self_attn.heads
+                dh = self.decoder[0].self_attn.dh
+                cache["self_kv"] = torch.zeros(2, L, B, H, max_len, dh,
+                                               device=enc_out.device, dtype=dt)
+            for i in range(len(self.decoder)):
+                cache[f"l{i}_self_buf_k"] = cache["self_kv"][0, i]
+                cache[f"l{i}_self_buf_v"] = cache["self_kv"][1, i]
         lmask = _log_mask(src_mask, dt)
         lens = src_mask.sum(dim=1).to(torch.int32)
         ga_layer = len(self.decoder) - 1

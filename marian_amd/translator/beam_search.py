@@ -179,18 +179,31 @@ class BeamSearch:
             tok_steps.append(real_words)
             alive = (scores > NEG_INF / 2).reshape(B * K)
 
-            stop = bool((fin_count >= self.n_best).all()) or not bool(alive.any())
-            if stop:
-                break
+            # host-sync the stop flag only every few steps (decode is
+            # launch-bound; each sync costs more than the extra steps)
+            if (step & 3) == 3 or step == max_len - 1:
+                done = (fin_count >= self.n_best).all() | ~alive.any()
+                if bool(done):
+                    break
 
-            # reorder cache state by selected back-pointers
+            # reorder cache state by selected back-pointers.
+            # cross-attn ("context"/static) KV is identical across the
+            # beams of a sentence (it comes from the shared encoder pass),
+            # so only self-attn state moves; the transformer packs it in
+            # ONE "self_kv" tensor -> a single gather.
             flat_sel = (torch.arange(B, device=device)[:, None] * K
                         + beam_idx).reshape(-1)
             for st in states:
                 cache = st["cache"]
+                if "self_kv" in cache:
+                    cache["self_kv"] = cache["self_kv"].index_select(
+                        2, flat_sel)
                 for key, t in list(cache.items()):
-                    if isinstance(t, torch.Tensor):
-                        cache[key] = t.index_select(0, flat_sel)
+                    if not isinstance(t, torch.Tensor) or key == "self_kv" \
+                            or "_buf_" in key or "context" in key \
+                            or "static" in key:
+                        continue
+                    cache[key] = t.index_select(0, flat_sel)
             prev_tokens = real_words.reshape(B * K)
 
         # ---- harvest: reconstruct token sequences from backpointers -------
