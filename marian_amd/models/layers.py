@@ -121,7 +121,18 @@ class PrePostProcessor(nn.Module):
             self.ln = LayerNorm(dim)
 
     def forward(self, x, residual=None):
-        for op in self.ops_str:
+        ops_str = self.ops_str
+        # 'an' tail with no active dropout: fused residual-add + LN kernel
+        if residual is not None and ops_str.endswith("an") \
+                and not (self.dropout > 0 and self.training) and x.is_cuda:
+            for op in ops_str[:-2]:
+                if op == "d":
+                    pass  # inactive
+                elif op == "n":
+                    x = self.ln(x)
+            return ops.layer_norm_residual(x, residual, self.ln.scale,
+                                           self.ln.bias, self.ln.eps)
+        for op in ops_str:
             if op == "d":
                 if self.dropout > 0 and self.training:
                     x = torch.nn.functional.dropout(x, self.dropout, True)

@@ -87,6 +87,18 @@ def layer_norm(x: torch.Tensor, scale: torch.Tensor, bias: torch.Tensor,
         x.float(), (x.shape[-1],), scale.float(), bias.float(), eps).to(x.dtype)
 
 
+def layer_norm_residual(x: torch.Tensor, res: torch.Tensor,
+                        scale: torch.Tensor, bias: torch.Tensor,
+                        eps: float = 1e-9) -> torch.Tensor:
+    """LN(x + res) with the add fused into the LN kernel (the marian 'dan'
+    postprocess epilogue)."""
+    if _on_gpu(x) and x.shape == res.shape:
+        from ._layernorm import LayerNormResFn
+        return LayerNormResFn.apply(x.contiguous(), res.contiguous(),
+                                    scale, bias, eps)
+    return layer_norm(x + res, scale, bias, eps)
+
+
 # ---------------------------------------------------------------------------
 # Attention  (ref: composite ProdBatched+mask+gSoftmax+ProdBatched;
 #             rebuilt as a fused flash-style CDNA4 kernel)

@@ -3,6 +3,10 @@
 
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
                                          torch::Tensor beta, double eps);
+std::vector<torch::Tensor> layernorm_res_fwd(torch::Tensor x,
+                                             torch::Tensor res,
+                                             torch::Tensor gamma,
+                                             torch::Tensor beta, double eps);
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor x, torch::Tensor gamma,
                                          torch::Tensor mean, torch::Tensor rstd,
                                          torch::Tensor dy);
@@ -31,6 +35,8 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (gfx950)");
+  m.def("layernorm_res_fwd", &layernorm_res_fwd,
+        "fused residual-add + LayerNorm forward");
   m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (gfx950)");
   m.def("softmax_fwd", &softmax_fwd, "row softmax forward");
   m.def("log_softmax_fwd", &log_softmax_fwd, "row log-softmax forward");

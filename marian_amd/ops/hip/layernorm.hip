@@ -13,11 +13,16 @@ using namespace mamd;
 
 namespace {
 
-template <typename T, int VEC>
+// HAS_RES: y = LN(x + res), with the sum also written to sum_out (the
+// backward normalizes against the summed input) - fuses the marian "dan"
+// postprocess's residual add into the LN pass.
+template <typename T, int VEC, bool HAS_RES>
 __global__ void ln_fwd_kernel(const T* __restrict__ x,
+                              const T* __restrict__ res,
                               const T* __restrict__ gamma,
                               const T* __restrict__ beta,
                               T* __restrict__ y,
+                              T* __restrict__ sum_out,
                               float* __restrict__ mean_out,
                               float* __restrict__ rstd_out,
                               int rows, int d, float eps) {
@@ -27,13 +32,19 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x,
   const int row = blockIdx.x * waves_per_block + wid;
   if (row >= rows) return;
   const T* xr = x + (size_t)row * d;
+  const T* rr = HAS_RES ? res + (size_t)row * d : nullptr;
   T* yr = y + (size_t)row * d;
+  T* sr = HAS_RES ? sum_out + (size_t)row * d : nullptr;
 
   float s = 0.f, ss = 0.f;
   for (int c = lane * VEC; c < d; c += WAVE * VEC) {
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       float v = to_f32(xr[c + j]);
+      if (HAS_RES) {
+        v += to_f32(rr[c + j]);
+        sr[c + j] = from_f32<T>(v);
+      }
       s += v;
       ss += v * v;
     }
@@ -44,10 +55,14 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x,
   const float var = fmaxf(ss / d - mu * mu, 0.f);
   const float rstd = rsqrtf(var + eps);
 
+  // second pass re-reads the (L1/L2-resident) inputs - a register cache
+  // would be runtime-indexed and spill to scratch (guide rule 20)
   for (int c = lane * VEC; c < d; c += WAVE * VEC) {
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
-      float v = (to_f32(xr[c + j]) - mu) * rstd;
+      float v = to_f32(xr[c + j]);
+      if (HAS_RES) v = to_f32(sr[c + j]);
+      v = (v - mu) * rstd;
       yr[c + j] = from_f32<T>(v * to_f32(gamma[c + j]) + to_f32(beta[c + j]));
     }
   }
@@ -154,8 +169,10 @@ __global__ void ln_bwd_reduce_kernel(const float* __restrict__ dg_scratch,
 constexpr int kBlock = 256;
 
 template <typename T>
-void ln_fwd_launch(const torch::Tensor& x, const torch::Tensor& g,
+void ln_fwd_launch(const torch::Tensor& x, const torch::Tensor* res,
+                   const torch::Tensor& g,
                    const torch::Tensor& b, torch::Tensor& y,
+                   torch::Tensor* sum_out,
                    torch::Tensor& mean, torch::Tensor& rstd, double eps) {
   const int d = x.size(-1);
   const int rows = x.numel() / d;
@@ -163,14 +180,27 @@ void ln_fwd_launch(const torch::Tensor& x, const torch::Tensor& g,
   const int waves = kBlock / WAVE;
   const int grid = (rows + waves - 1) / waves;
   auto stream = at::cuda::getCurrentHIPStream();
-  hipLaunchKernelGGL((ln_fwd_kernel<T, 4>), dim3(grid), dim3(kBlock), 0,
-                     stream,
-                     reinterpret_cast<const T*>(x.data_ptr()),
-                     reinterpret_cast<const T*>(g.data_ptr()),
-                     reinterpret_cast<const T*>(b.data_ptr()),
-                     reinterpret_cast<T*>(y.data_ptr()),
-                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     rows, d, (float)eps);
+  if (res) {
+    hipLaunchKernelGGL((ln_fwd_kernel<T, 4, true>), dim3(grid), dim3(kBlock),
+                       0, stream,
+                       reinterpret_cast<const T*>(x.data_ptr()),
+                       reinterpret_cast<const T*>(res->data_ptr()),
+                       reinterpret_cast<const T*>(g.data_ptr()),
+                       reinterpret_cast<const T*>(b.data_ptr()),
+                       reinterpret_cast<T*>(y.data_ptr()),
+                       reinterpret_cast<T*>(sum_out->data_ptr()),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       rows, d, (float)eps);
+  } else {
+    hipLaunchKernelGGL((ln_fwd_kernel<T, 4, false>), dim3(grid), dim3(kBlock),
+                       0, stream,
+                       reinterpret_cast<const T*>(x.data_ptr()), nullptr,
+                       reinterpret_cast<const T*>(g.data_ptr()),
+                       reinterpret_cast<const T*>(b.data_ptr()),
+                       reinterpret_cast<T*>(y.data_ptr()), nullptr,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       rows, d, (float)eps);
+  }
 }
 
 template <typename T>
@@ -235,15 +265,43 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor gamma,
   auto rstd = torch::empty({rows}, opts);
   auto g = gamma.contiguous(), b = beta.contiguous();
   if (x.scalar_type() == torch::kBFloat16) {
-    ln_fwd_launch<__hip_bfloat16>(x, g, b, y, mean, rstd, eps);
+    ln_fwd_launch<__hip_bfloat16>(x, nullptr, g, b, y, nullptr, mean, rstd, eps);
   } else if (x.scalar_type() == torch::kHalf) {
-    ln_fwd_launch<__half>(x, g, b, y, mean, rstd, eps);
+    ln_fwd_launch<__half>(x, nullptr, g, b, y, nullptr, mean, rstd, eps);
   } else if (x.scalar_type() == torch::kFloat32) {
-    ln_fwd_launch<float>(x, g, b, y, mean, rstd, eps);
+    ln_fwd_launch<float>(x, nullptr, g, b, y, nullptr, mean, rstd, eps);
   } else {
     TORCH_CHECK(false, "layernorm: unsupported dtype");
   }
   return {y, mean, rstd};
+}
+
+// y = LN(x + res); returns (y, sum, mean, rstd)
+std::vector<torch::Tensor> layernorm_res_fwd(torch::Tensor x,
+                                             torch::Tensor res,
+                                             torch::Tensor gamma,
+                                             torch::Tensor beta, double eps) {
+  TORCH_CHECK(x.is_cuda());
+  auto xc = x.contiguous();
+  auto rc = res.contiguous();
+  auto y = torch::empty_like(xc);
+  auto sum = torch::empty_like(xc);
+  const int d = xc.size(-1);
+  const int rows = xc.numel() / d;
+  auto opts = xc.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({rows}, opts);
+  auto rstd = torch::empty({rows}, opts);
+  auto g = gamma.contiguous(), b = beta.contiguous();
+  if (xc.scalar_type() == torch::kBFloat16) {
+    ln_fwd_launch<__hip_bfloat16>(xc, &rc, g, b, y, &sum, mean, rstd, eps);
+  } else if (xc.scalar_type() == torch::kHalf) {
+    ln_fwd_launch<__half>(xc, &rc, g, b, y, &sum, mean, rstd, eps);
+  } else if (xc.scalar_type() == torch::kFloat32) {
+    ln_fwd_launch<float>(xc, &rc, g, b, y, &sum, mean, rstd, eps);
+  } else {
+    TORCH_CHECK(false, "layernorm: unsupported dtype");
+  }
+  return {y, sum, mean, rstd};
 }
 
 // returns (dx, dgamma_f32, dbeta_f32)

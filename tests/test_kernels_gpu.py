@@ -433,3 +433,29 @@ def test_decoder_cli_on_gpu(ext, tmp_path):
                   "--max-length", "20", "--precision", "bfloat16"])
     lines = out.read_text().splitlines()
     assert len(lines) == 2
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_layernorm_residual_fused(ext, dtype):
+    from marian_amd import ops
+    torch.manual_seed(41)
+    B, T, d = 4, 9, 512
+    x = torch.randn(B, T, d, device=DEV, dtype=dtype, requires_grad=True)
+    r = torch.randn(B, T, d, device=DEV, dtype=dtype, requires_grad=True)
+    g = (torch.rand(d, device=DEV, dtype=dtype) + 0.5).requires_grad_()
+    b = torch.randn(d, device=DEV, dtype=dtype, requires_grad=True)
+    y = ops.layer_norm_residual(x, r, g, b, 1e-9)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xr = x.detach().float().clone().requires_grad_()
+    rr = r.detach().float().clone().requires_grad_()
+    gr = g.detach().float().clone().requires_grad_()
+    br = b.detach().float().clone().requires_grad_()
+    yr = torch.nn.functional.layer_norm(xr + rr, (d,), gr, br, 1e-9)
+    yr.backward(dy.float())
+    tol = 1e-4 if dtype == torch.float32 else 5e-2
+    assert _rel_err(y, yr) < tol
+    assert _rel_err(x.grad, xr.grad) < tol
+    assert _rel_err(r.grad, rr.grad) < tol
+    assert _rel_err(g.grad, gr.grad) < tol
