@@ -30,7 +30,9 @@ class Corpus:
                  tsv: bool = False, tsv_fields: int = 0,
                  shuffle_in_ram: bool = False, seed: int = 1234,
                  align_path: Optional[str] = None,
-                 weight_path: Optional[str] = None):
+                 weight_path: Optional[str] = None,
+                 right_left: bool = False):
+        self.right_left = right_left  # --right-left: reverse target side
         assert len(paths) >= 1
         self.align_path = align_path     # --guided-alignment (pharaoh file)
         self.weight_path = weight_path   # --data-weighting
@@ -101,6 +103,9 @@ class Corpus:
                 tup.append(ids)
             if not ok or any(len(t) == 0 for t in tup):
                 continue
+            if self.right_left and len(tup) > 1:
+                # reverse word order of the target, EOS stays last
+                tup[-1] = tup[-1][-2::-1] + [tup[-1][-1]]
             n_emitted += 1
             if n_emitted <= skip:
                 continue

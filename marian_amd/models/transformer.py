@@ -150,6 +150,19 @@ class TransformerModel(nn.Module):
         self.compute_dtype = torch.float32
         self.collect_cross_attn = False   # guided alignment training
         self.last_cross_attn = None
+        self.gradient_checkpointing = bool(
+            options.get("gradient-checkpointing", False))
+        if options.get("transformer-depth-scaling", False):
+            # depth-scaled init (ref: transformer.h DepthScaling): scale
+            # residual-path weights by 1/sqrt(layer_index)
+            with torch.no_grad():
+                for li, layer in enumerate(self.encoder, start=1):
+                    for m in (layer.self_attn.Wo, layer.ffn.W2):
+                        m.W.mul_(li ** -0.5)
+                for li, layer in enumerate(self.decoder, start=1):
+                    for m in (layer.self_attn.Wo, layer.context_attn.Wo,
+                              layer.ffn.W2):
+                        m.W.mul_(li ** -0.5)
 
     def _guided_layer(self) -> int:
         g = self.opt.get("transformer-guided-alignment-layer", "last")
@@ -167,8 +180,14 @@ class TransformerModel(nn.Module):
         x = x * src_mask.unsqueeze(-1).to(dt)
         lmask = _log_mask(src_mask, dt)
         lens = src_mask.sum(dim=1).to(torch.int32)
+        ckpt = self.gradient_checkpointing and self.training \
+            and torch.is_grad_enabled()
         for layer in self.encoder:
-            x = layer(x, lmask, lens=lens)
+            if ckpt:
+                x = torch.utils.checkpoint.checkpoint(
+                    layer, x, lmask, lens, use_reentrant=False)
+            else:
+                x = layer(x, lmask, lens=lens)
         if self.post_top_enc is not None:
             x = self.post_top_enc(x)
         return x
@@ -188,10 +207,18 @@ class TransformerModel(nn.Module):
         lens = src_mask.sum(dim=1).to(torch.int32)
         ga_layer = self._guided_layer() if self.collect_cross_attn else -1
         self.last_cross_attn = None
+        ckpt = self.gradient_checkpointing and self.training \
+            and torch.is_grad_enabled()
         for i, layer in enumerate(self.decoder):
             collect = {} if i == ga_layer else None
-            x = layer(x, enc_out, lmask, causal=True, layer_id=i,
-                      src_lens=lens, collect=collect)
+            if ckpt and collect is None:
+                x = torch.utils.checkpoint.checkpoint(
+                    lambda x_, e_, m_, l_, i_=i: self.decoder[i_](
+                        x_, e_, m_, causal=True, layer_id=i_, src_lens=l_),
+                    x, enc_out, lmask, lens, use_reentrant=False)
+            else:
+                x = layer(x, enc_out, lmask, causal=True, layer_id=i,
+                          src_lens=lens, collect=collect)
             if collect is not None:
                 self.last_cross_attn = collect.get("probs")
         if self.post_top_dec is not None:

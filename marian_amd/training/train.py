@@ -115,7 +115,8 @@ class Train:
                         shuffle_in_ram=bool(opts.get("shuffle-in-ram", False)),
                         seed=int(opts.get("seed", 0) or 1234),
                         align_path=None if ga in (None, "none", "") else ga,
-                        weight_path=opts.get("data-weighting"))
+                        weight_path=opts.get("data-weighting"),
+                        right_left=bool(opts.get("right-left", False)))
         corpus.epoch = self.state.epochs - 1
         shuffle = opts.get("shuffle", "data") != "none" \
             and not opts.get("no-shuffle", False)

@@ -47,7 +47,17 @@ class BeamSearch:
                  n_best: int = 1, max_length_factor: float = 3.0,
                  max_length: int = 256, eos_id: int = 0, unk_id: int = 1,
                  weights: Optional[List[float]] = None,
-                 shortlist_generator=None, collect_alignment: bool = False):
+                 shortlist_generator=None, collect_alignment: bool = False,
+                 output_sampling=None):
+        # --output-sampling [full eps | topk k] (ref: beam_search sampling)
+        self.sampling = None
+        if output_sampling:
+            kind = output_sampling[0] if isinstance(output_sampling, list) \
+                else str(output_sampling)
+            arg = (output_sampling[1]
+                   if isinstance(output_sampling, list)
+                   and len(output_sampling) > 1 else None)
+            self.sampling = (str(kind), arg)
         self.collect_alignment = collect_alignment
         self.scorers = scorers if isinstance(scorers, (list, tuple)) else [scorers]
         self.K = beam_size
@@ -134,11 +144,30 @@ class BeamSearch:
             logprobs = torch.where(alive[:, None], logprobs,
                                    torch.full_like(logprobs, NEG_INF))
 
-            cand = scores.view(B * K, 1) + logprobs          # [B*K, V]
-            cand = cand.view(B, K * V)
-            top_scores, top_idx = ops.topk(cand, K)          # [B, K]
-            beam_idx = top_idx // V                          # [B, K]
-            word_idx = top_idx % V
+            if self.sampling is not None and K == 1:
+                kind, arg = self.sampling
+                lp = logprobs
+                if kind == "topk":
+                    kk = int(arg or 10)
+                    tv, ti = torch.topk(lp, kk, dim=-1)
+                    pick = torch.multinomial(torch.softmax(tv, -1), 1)
+                    word = ti.gather(1, pick)                # [B*K, 1]
+                else:  # full (optional temperature arg)
+                    temp = float(arg or 1.0)
+                    pick = torch.multinomial(
+                        torch.softmax(lp / temp, -1), 1)
+                    word = pick
+                samp_scores = lp.gather(1, word)
+                top_scores = (scores.view(B * K, 1) + samp_scores).view(B, K)
+                top_idx = word.view(B, K)  # beam 0 implied
+                beam_idx = torch.zeros_like(top_idx)
+                word_idx = top_idx
+            else:
+                cand = scores.view(B * K, 1) + logprobs      # [B*K, V]
+                cand = cand.view(B, K * V)
+                top_scores, top_idx = ops.topk(cand, K)      # [B, K]
+                beam_idx = top_idx // V                      # [B, K]
+                word_idx = top_idx % V
 
             # map shortlisted ids back to real vocab ids for EOS test
             if shortlist is not None:

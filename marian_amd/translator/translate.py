@@ -137,7 +137,8 @@ class Translator:
             eos_id=self.trg_vocab.eos_id, unk_id=self.trg_vocab.unk_id,
             weights=[float(w) for w in options.list("weights")] or None,
             shortlist_generator=shortlist_gen,
-            collect_alignment=bool(options.get("alignment")))
+            collect_alignment=bool(options.get("alignment")),
+            output_sampling=options.list("output-sampling") or None)
 
     def _nbest_size(self):
         if self.opt.get("n-best", False):

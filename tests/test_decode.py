@@ -241,3 +241,29 @@ def test_embedder(trained_model_dir, capsys):
     sims = [float(x) for x in capsys.readouterr().out.strip().splitlines()]
     assert len(sims) == 2
     assert sims[0] > 0.99  # identical sentence
+
+
+def test_output_sampling(trained_model_dir):
+    """--output-sampling produces valid (varying) output with beam 1."""
+    d = trained_model_dir
+    import torch
+    from marian_amd.translator.beam_search import BeamSearch, ModelScorer
+    from marian_amd.translator.translate import load_model_for_inference
+    from marian_amd.common.options import Options
+    m = load_model_for_inference(str(d / "model.npz"), Options({}),
+                                 torch.device("cpu"))
+    sc = ModelScorer(m, torch.device("cpu"))
+    torch.manual_seed(1)
+    bs = BeamSearch([sc], beam_size=1, max_length=10,
+                    output_sampling=["full", "1.5"])
+    batch = make_batch([[5, 6, 7, 0]])
+    outs = set()
+    for _ in range(5):
+        r = bs.search(batch, torch.device("cpu"))[0]
+        outs.add(tuple(r.best.tokens))
+    assert len(outs) >= 2  # sampling varies
+
+    bs2 = BeamSearch([sc], beam_size=1, max_length=10,
+                     output_sampling=["topk", "5"])
+    r = bs2.search(batch, torch.device("cpu"))[0]
+    assert isinstance(r.best.tokens, list)

@@ -129,3 +129,50 @@ def test_guided_alignment_and_data_weighting(workdir):
     assert tr.graph_group.guided_alignment
     tr.run()
     assert tr.state.batches == 10
+
+
+def test_gradient_checkpointing(workdir):
+    tmp_path, src, trg = workdir
+    opts = _opts(tmp_path, src, src, extra=["--gradient-checkpointing",
+                                            "--after-batches", "5"])
+    tr = Train(opts)
+    assert tr.model.gradient_checkpointing
+    tr.run()
+    assert tr.state.batches == 5
+
+
+def test_right_left(workdir):
+    tmp_path, src, trg = workdir
+    from marian_amd.data.corpus import Corpus
+    from marian_amd.data.vocab import Vocab
+    v = Vocab.create_from_corpus([src])
+    c1 = Corpus([src, src], [v, v], right_left=False)
+    c2 = Corpus([src, src], [v, v], right_left=True)
+    t1 = next(iter(c1.iter_epoch(shuffle=False)))
+    t2 = next(iter(c2.iter_epoch(shuffle=False)))
+    assert t2[1][-1][:-1] == t1[1][-1][-2::-1]  # reversed, EOS last
+    assert t2[1][-1][-1] == 0
+
+
+def test_exponential_smoothing(workdir):
+    tmp_path, src, trg = workdir
+    opts = _opts(tmp_path, src, src, extra=["--exponential-smoothing", "0.999",
+                                            "--after-batches", "5"])
+    tr = Train(opts)
+    opt = tr.graph_group.optimizer
+    assert opt.smooth is not None
+    tr.run()
+    # smoothed differs from raw but is close
+    d = (opt.smooth - opt.master).abs().max()
+    assert 0 < float(d) < 1.0
+    # swap in/out restores exactly
+    import torch
+    before = opt.param_shard_snapshot = [b.flat_param.clone()
+                                         for b in tr.graph_group.store.buckets]
+    opt.swap_in_smoothed()
+    changed = any(not torch.equal(a, b.flat_param) for a, b in
+                  zip(before, tr.graph_group.store.buckets))
+    assert changed
+    opt.swap_out_smoothed()
+    for a, b in zip(before, tr.graph_group.store.buckets):
+        assert torch.equal(a, b.flat_param)
