@@ -278,3 +278,55 @@ class FFN(nn.Module):
         if self.dropout > 0 and self.training:
             h = torch.nn.functional.dropout(h, self.dropout, True)
         return self.W2(h)
+
+
+class AverageAttention(nn.Module):
+    """Average Attention Network decoder self-attention replacement
+    (ref: models/transformer.h DecoderLayerAAN,
+    --transformer-decoder-autoreg average-attention).
+
+    h_t = FFN(cumulative-mean(y_{<=t})); gated combine with the input."""
+
+    def __init__(self, dim: int, dim_aan: int, activation: str = "swish",
+                 depth: int = 2, gated: bool = True, dropout: float = 0.0):
+        super().__init__()
+        self.gated = gated
+        self.W1 = Dense(dim, dim_aan)
+        self.W2 = Dense(dim_aan, dim)
+        self.act = _ACT[activation]
+        self.depth = depth
+        if gated:
+            self.Wi = Dense(2 * dim, 2 * dim)
+        self.dropout = dropout
+
+    def _ffn(self, a):
+        if self.depth >= 2:
+            h = self.act(self.W1(a))
+            if self.dropout > 0 and self.training:
+                h = torch.nn.functional.dropout(h, self.dropout, True)
+            return self.W2(h)
+        return a
+
+    def _combine(self, y, h):
+        if not self.gated:
+            return h
+        g = torch.sigmoid(self.Wi(torch.cat([y, h], dim=-1)))
+        i, f = g.chunk(2, dim=-1)
+        return i * y + f * h
+
+    def forward(self, y, cache=None, cache_key: str = ""):
+        """y: [B, T, d] (training, T>1) or [B, 1, d] (decode with cache)."""
+        if cache is None:
+            T = y.shape[1]
+            csum = torch.cumsum(y.float(), dim=1)
+            denom = torch.arange(1, T + 1, device=y.device,
+                                 dtype=torch.float32).view(1, T, 1)
+            a = (csum / denom).to(y.dtype)
+        else:
+            step = cache.get("step", 0)
+            key = cache_key + "_aan_sum"
+            if key not in cache:
+                cache[key] = torch.zeros_like(y)
+            cache[key] = cache[key] + y
+            a = (cache[key] / (step + 1)).to(y.dtype)
+        return self._combine(y, self._ffn(a))

@@ -70,8 +70,17 @@ class DecoderLayer(nn.Module):
         self.post_ctx = PrePostProcessor(opt["transformer-postprocess"], d, drop)
         self.pre_ffn = PrePostProcessor(opt["transformer-preprocess"], d, drop)
         self.post_ffn = PrePostProcessor(opt["transformer-postprocess"], d, drop)
-        self.self_attn = MultiHeadAttention(
-            d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
+        self.autoreg = opt.get("transformer-decoder-autoreg", "self-attention")
+        if self.autoreg == "average-attention":
+            from .layers import AverageAttention
+            self.self_attn = AverageAttention(
+                d, opt.get("transformer-dim-aan", 2048),
+                opt.get("transformer-aan-activation", "swish"),
+                opt.get("transformer-ffn-depth-aan", 2),
+                not opt.get("transformer-aan-nogate", False), drop)
+        else:
+            self.self_attn = MultiHeadAttention(
+                d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
         self.context_attn = MultiHeadAttention(
             d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
         self.ffn = FFN(d, dim_ffn, opt.get("transformer-ffn-activation", "relu"),
@@ -81,8 +90,11 @@ class DecoderLayer(nn.Module):
                 cache: Optional[dict] = None, layer_id: int = 0,
                 src_lens=None, collect: Optional[dict] = None):
         h = self.pre_self(x)
-        h = self.self_attn(h, h, h, causal=causal, cache=cache,
-                           cache_key=f"l{layer_id}_self")
+        if self.autoreg == "average-attention":
+            h = self.self_attn(h, cache=cache, cache_key=f"l{layer_id}_self")
+        else:
+            h = self.self_attn(h, h, h, causal=causal, cache=cache,
+                               cache_key=f"l{layer_id}_self")
         x = self.post_self(h, residual=x)
         h = self.pre_ctx(x)
         h = self.context_attn(h, enc_out, enc_out, mask=src_mask, cache=cache,
@@ -344,7 +356,20 @@ class TransformerModel(nn.Module):
                 m[f"encoder_l{i}_ffn_W1_ln_scale_pre"] = layer.pre_ffn.ln.scale
                 m[f"encoder_l{i}_ffn_W1_ln_bias_pre"] = layer.pre_ffn.ln.bias
         for i, layer in enumerate(self.decoder, start=1):
-            attn(f"decoder_l{i}_self", layer.self_attn, layer.post_self)
+            if layer.autoreg == "average-attention":
+                a = layer.self_attn
+                m[f"decoder_l{i}_aan_W1"] = a.W1.W
+                m[f"decoder_l{i}_aan_b1"] = a.W1.b
+                m[f"decoder_l{i}_aan_W2"] = a.W2.W
+                m[f"decoder_l{i}_aan_b2"] = a.W2.b
+                if a.gated:
+                    m[f"decoder_l{i}_aan_Wi"] = a.Wi.W
+                    m[f"decoder_l{i}_aan_bi"] = a.Wi.b
+                if hasattr(layer.post_self, "ln"):
+                    m[f"decoder_l{i}_aan_ln_scale"] = layer.post_self.ln.scale
+                    m[f"decoder_l{i}_aan_ln_bias"] = layer.post_self.ln.bias
+            else:
+                attn(f"decoder_l{i}_self", layer.self_attn, layer.post_self)
             attn(f"decoder_l{i}_context", layer.context_attn, layer.post_ctx)
             ffn(f"decoder_l{i}_ffn", layer.ffn, layer.post_ffn)
             if hasattr(layer.pre_self, "ln"):

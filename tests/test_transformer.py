@@ -130,3 +130,38 @@ def test_grad_flows_everywhere():
     for n, p in m.named_parameters():
         assert p.grad is not None, n
         assert torch.isfinite(p.grad).all(), n
+
+
+def test_average_attention_decoder():
+    """--transformer-decoder-autoreg average-attention: train + cached
+    decode equivalence."""
+    torch.manual_seed(7)
+    opts = tiny_options(**{"transformer-decoder-autoreg": "average-attention",
+                           "transformer-dim-aan": 64,
+                           "transformer-aan-activation": "swish",
+                           "transformer-ffn-depth-aan": 2,
+                           "transformer-aan-nogate": False})
+    m = create_model(opts)
+    b = synthetic_batch(2, 5, 6, 37)
+    logits = m(b)
+    assert logits.shape == (2, 6, 37)
+    logits.float().pow(2).mean().backward()
+    items = m.marian_items()
+    assert "decoder_l1_aan_W1" in items and "decoder_l1_aan_Wi" in items
+
+    m.eval()
+    for p in m.parameters():
+        p.grad = None
+    with torch.no_grad():
+        enc = m.encode(b.source.indices, b.source.mask)
+        full = m.decode_train(b.target.indices, b.target.mask, enc,
+                              b.source.mask)
+        cache = {}
+        outs = []
+        for t in range(6):
+            prev = b.target.indices[:, t - 1] if t > 0 else \
+                torch.zeros(2, dtype=torch.long)
+            outs.append(m.decode_step(prev, enc, b.source.mask, cache, t))
+        stepped = torch.stack(outs, dim=1)
+    assert torch.allclose(full, stepped, atol=1e-4), \
+        (full - stepped).abs().max()
