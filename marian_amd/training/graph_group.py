@@ -185,6 +185,15 @@ class SyncGraphGroup:
     def _apply_update(self) -> LossValue:
         opt = self.optimizer
         if self._overlap:
+            # safety: a bucket whose params produced no grads this step
+            # (frozen/unused parameter) never fired its hook - reduce it
+            # synchronously so the shard still sees zeros/partial grads
+            for bi, pending in self._bucket_pending.items():
+                if pending > 0:
+                    b = self.store.buckets[bi]
+                    self.comm.reduce_scatter_bucket(
+                        b.flat_grad, opt.grad_shard_for_bucket(bi),
+                        async_op=True)
             self.comm.wait_all()
         else:
             for bi, b in enumerate(self.store.buckets):

@@ -75,7 +75,9 @@ class Scheduler:
             return False  # handled at epoch boundary
         return False
 
-    _last_labels = 0
+    _disp_labels = 0
+    _save_labels = 0
+    _valid_labels = 0
 
     # ------------------------------------------------------------------
     def update(self, cost: float, n_sentences: int, n_labels: int,
@@ -95,7 +97,21 @@ class Scheduler:
         actions = {"display": False, "save": False, "validate": False,
                    "stop": False}
 
-        disp_due = (self.disp_freq and st.batches % max(1, self.disp_freq.n) == 0) \
+        def due(freq: SchedulingParameter, last_attr: str) -> bool:
+            """updates ('u') and target-label ('t') schedule units."""
+            if not freq:
+                return False
+            if freq.unit == "u":
+                return st.batches % max(1, freq.n) == 0
+            if freq.unit == "t":
+                last = getattr(self, last_attr)
+                if st.labels_total // freq.n > last // freq.n:
+                    setattr(self, last_attr, st.labels_total)
+                    return True
+                return False
+            return False  # 'e' handled at epoch boundary
+
+        disp_due = due(self.disp_freq, "_disp_labels") \
             or (st.batches <= self.disp_first)
         if disp_due:
             elapsed = self.timer.elapsed()
@@ -113,11 +129,9 @@ class Scheduler:
             self.timer.reset()
             actions["display"] = True
 
-        if self.save_freq and self.save_freq.unit == "u" \
-                and st.batches % max(1, self.save_freq.n) == 0:
+        if due(self.save_freq, "_save_labels"):
             actions["save"] = True
-        if self.valid_freq and self.valid_freq.unit == "u" \
-                and st.batches % max(1, self.valid_freq.n) == 0:
+        if due(self.valid_freq, "_valid_labels"):
             actions["validate"] = True
         if self.after_batches and st.batches >= self.after_batches:
             actions["stop"] = True
@@ -127,6 +141,16 @@ class Scheduler:
         return actions
 
     # ------------------------------------------------------------------
+    def epoch_actions(self) -> dict:
+        """'e'-unit save/valid triggers, checked at epoch boundaries."""
+        st = self.state
+        out = {"save": False, "validate": False}
+        for freq, key in ((self.save_freq, "save"),
+                          (self.valid_freq, "validate")):
+            if freq and freq.unit == "e" and st.epochs % max(1, freq.n) == 0:
+                out[key] = True
+        return out
+
     def increase_epoch(self):
         st = self.state
         log_info("Seen %d samples in epoch %d", st.samples_epoch, st.epochs)

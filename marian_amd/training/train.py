@@ -177,6 +177,11 @@ class Train:
                 break
             if not sched.keep_going() or _sigterm_seen:
                 break
+            ea = sched.epoch_actions()
+            if ea["save"]:
+                self.save()
+            if ea["validate"]:
+                self._validate(first_metric_validator)
             sched.increase_epoch()
         self.save()
         sched.finished()
