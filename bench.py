@@ -19,6 +19,15 @@ import os
 import sys
 import time
 
+# hipBLASLt GEMM autotuning: ship pre-tuned tables for the bench shapes
+# (profiles/tunableop, +6-7% on transformer-big); env must be set before
+# torch initializes. Users override by exporting the vars themselves.
+_ROOT = os.path.dirname(os.path.abspath(__file__))
+os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                      os.path.join(_ROOT, "profiles", "tunableop", "tuned.csv"))
+
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
