@@ -39,6 +39,31 @@ def _on_sigterm(signum, frame):
     log_info("Received SIGTERM; will save and exit after the current batch")
 
 
+def _train_spm(path: str, corpus: str, options: Options, vocab_size: int):
+    """Train a SentencePiece model (ref: data/sentencepiece_vocab.cpp — the
+    reference trains .spm vocabs on the fly when the file is missing)."""
+    import sentencepiece as spm
+    log_info("Training SentencePiece vocabulary %s from %s", path, corpus)
+    args = {
+        "input": corpus,
+        "model_prefix": path[:-4] if path.endswith(".spm") else path,
+        "vocab_size": vocab_size or 8000,
+        "input_sentence_size": int(options.get("sentencepiece-max-lines",
+                                               2000000) or 2000000),
+        "shuffle_input_sentence": True,
+        "hard_vocab_limit": False,
+    }
+    extra = options.get("sentencepiece-options")
+    if extra:
+        for kv in str(extra).split("--"):
+            kv = kv.strip()
+            if "=" in kv:
+                k, v = kv.split("=", 1)
+                args[k.strip()] = v.strip()
+    spm.SentencePieceTrainer.train(**args)
+    os.replace(args["model_prefix"] + ".model", path)
+
+
 def load_or_create_vocabs(options: Options) -> List[Vocab]:
     paths = options.list("vocabs")
     train_sets = options.list("train-sets")
@@ -51,6 +76,10 @@ def load_or_create_vocabs(options: Options) -> List[Vocab]:
         max_size = int(dim_vocabs[i]) if i < len(dim_vocabs) else 0
         if os.path.exists(p):
             v = Vocab.load(p, max_size=max_size)
+        elif p.endswith(".spm") or p.endswith(".model"):
+            _train_spm(p, train_sets[min(i, len(train_sets) - 1)], options,
+                       max_size)
+            v = Vocab.load(p)
         else:
             log_info("Building vocabulary from %s", train_sets[i])
             v = Vocab.create_from_corpus([train_sets[i]], max_size=max_size)

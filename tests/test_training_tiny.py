@@ -176,3 +176,42 @@ def test_exponential_smoothing(workdir):
     opt.swap_out_smoothed()
     for a, b in zip(before, tr.graph_group.store.buckets):
         assert torch.equal(a, b.flat_param)
+
+
+def test_sentencepiece_vocab_training_and_use(tmp_path):
+    """.spm vocab trained on the fly + used for training and decode."""
+    import string
+    src = tmp_path / "real.src"
+    # text with subword structure
+    words = ["hello", "world", "translation", "quality", "system",
+             "neural", "machine", "attention", "encoder", "decoder"]
+    import random
+    rng = random.Random(3)
+    with open(src, "w") as f:
+        for _ in range(300):
+            f.write(" ".join(rng.choices(words, k=rng.randint(3, 8))) + "\n")
+    argv = ["--train-sets", str(src), str(src),
+            "--vocabs", str(tmp_path / "v.spm"), str(tmp_path / "v.spm"),
+            "--model", str(tmp_path / "model.npz"),
+            "--dim-vocabs", "200", "200",
+            "--dim-emb", "32", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "64",
+            "--mini-batch", "16", "--after-batches", "8", "--cpu-threads", "1",
+            "--max-length", "40", "--tied-embeddings-all", "--quiet",
+            "--overwrite"]
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    opts = ConfigParser(MODE_TRAINING).parse_options(argv)
+    tr = Train(opts)
+    tr.run()
+    assert os.path.exists(tmp_path / "v.spm")
+    # decode with the spm vocab
+    from marian_amd.translator.translate import main_decoder
+    inp = tmp_path / "in.txt"
+    inp.write_text("hello world\n")
+    out = tmp_path / "out.txt"
+    main_decoder(["--models", str(tmp_path / "model.npz"),
+                  "--vocabs", str(tmp_path / "v.spm"), str(tmp_path / "v.spm"),
+                  "--input", str(inp), "--output", str(out),
+                  "--beam-size", "2", "--cpu-threads", "1",
+                  "--max-length", "20", "--quiet"])
+    assert out.exists()
