@@ -81,7 +81,9 @@ class BeamSearch:
 
     # ------------------------------------------------------------------
     @torch.no_grad()
-    def search(self, batch, device) -> List[SentenceResult]:
+    def search(self, batch, device, prefix=None) -> List[SentenceResult]:
+        """prefix: optional (ids [B, Tp], mask [B, Tp]) forcing the first
+        target tokens (--force-decode)."""
         src = batch.source.to(device)
         B = src.indices.shape[0]
         K = self.K
@@ -144,6 +146,14 @@ class BeamSearch:
             logprobs = torch.where(alive[:, None], logprobs,
                                    torch.full_like(logprobs, NEG_INF))
 
+            forced_now = None
+            if prefix is not None:
+                pids, pmask = prefix
+                if step < pids.shape[1]:
+                    active = pmask[:, step] > 0          # [B]
+                    if bool(active.any()):
+                        forced_now = (pids[:, step].to(device),
+                                      active.to(device))
             if self.sampling is not None and K == 1:
                 kind, arg = self.sampling
                 lp = logprobs
@@ -168,6 +178,18 @@ class BeamSearch:
                 top_scores, top_idx = ops.topk(cand, K)      # [B, K]
                 beam_idx = top_idx // V                      # [B, K]
                 word_idx = top_idx % V
+            if forced_now is not None:
+                ftok, factive = forced_now
+                # forced rows: every beam continues itself with the forced
+                # token; score += that token's logprob
+                lp_forced = logprobs.view(B, K, V).gather(
+                    2, ftok.view(B, 1, 1).expand(B, K, 1)).squeeze(-1)
+                kidx0 = torch.arange(K, device=device)[None, :].expand(B, K)
+                beam_idx = torch.where(factive[:, None], kidx0, beam_idx)
+                word_idx = torch.where(factive[:, None],
+                                       ftok[:, None].expand(B, K), word_idx)
+                top_scores = torch.where(factive[:, None],
+                                         scores + lp_forced, top_scores)
 
             # map shortlisted ids back to real vocab ids for EOS test
             if shortlist is not None:

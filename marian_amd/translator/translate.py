@@ -147,6 +147,15 @@ class Translator:
 
     # ------------------------------------------------------------------
     def translate_lines(self, lines: List[str]) -> List[SentenceResult]:
+        # --force-decode: tab-separated input "source<TAB>target-prefix"
+        force = bool(self.opt.get("force-decode", False))
+        prefixes = {}
+        if force:
+            split = [l.split("\t") for l in lines]
+            lines = [p[0] for p in split]
+            for i, p in enumerate(split):
+                if len(p) > 1 and p[1]:
+                    prefixes[i] = self.trg_vocab.encode(p[1], add_eos=False)
         ti = TextInput(iter(lines), self.src_vocab,
                        max_length=int(self.opt.get("max-length", 1000)),
                        max_length_crop=True)
@@ -159,7 +168,19 @@ class Translator:
             prefetch=False)
         results: dict = {}
         for batch in gen:
-            out = self.search.search(batch, self.device)
+            prefix = None
+            if force and any(sid in prefixes for sid in batch.sentence_ids):
+                import torch as _t
+                plist = [prefixes.get(sid, []) for sid in batch.sentence_ids]
+                Tp = max(len(x) for x in plist)
+                pids = _t.zeros(len(plist), Tp, dtype=_t.long)
+                pmask = _t.zeros(len(plist), Tp)
+                for i, x in enumerate(plist):
+                    if x:
+                        pids[i, :len(x)] = _t.tensor(x)
+                        pmask[i, :len(x)] = 1.0
+                prefix = (pids.to(self.device), pmask.to(self.device))
+            out = self.search.search(batch, self.device, prefix=prefix)
             for sid, res in zip(batch.sentence_ids, out):
                 results[sid] = res
         return [results[i] for i in range(len(results))]

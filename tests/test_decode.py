@@ -267,3 +267,18 @@ def test_output_sampling(trained_model_dir):
                      output_sampling=["topk", "5"])
     r = bs2.search(batch, torch.device("cpu"))[0]
     assert isinstance(r.best.tokens, list)
+
+
+def test_force_decode(trained_model_dir):
+    d = trained_model_dir
+    inp = d / "force_in.txt"
+    inp.write_text("w5 w6 w7\tw9 w9\n")
+    out = d / "force_out.txt"
+    from marian_amd.translator.translate import main_decoder
+    main_decoder(["--models", str(d / "model.npz"),
+                  "--vocabs", str(d / "v.yml"), str(d / "v.yml"),
+                  "--input", str(inp), "--output", str(out),
+                  "--beam-size", "2", "--cpu-threads", "1",
+                  "--force-decode", "--max-length", "20", "--quiet"])
+    line = out.read_text().splitlines()[0]
+    assert line.startswith("w9 w9"), line  # forced prefix respected
