@@ -95,8 +95,10 @@ constexpr int KVBLK = 32;
 constexpr int QBLK = 32;
 constexpr int VT_PAD = 8;  // elements
 
-template <int D, bool CAUSAL, bool HAS_LEN>
-__global__ __launch_bounds__(64)
+// Forward kernel: NW waves per workgroup, each owning one 32-row q-tile of
+// the same (b,h); the KVBLK=64 K/V^T LDS stage is shared by all waves.
+template <int D, int NW, bool CAUSAL, bool HAS_LEN>
+__global__ __launch_bounds__(64 * NW)
 void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
                      const __hip_bfloat16* __restrict__ K,
                      const __hip_bfloat16* __restrict__ V,
@@ -105,148 +107,157 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
                      float* __restrict__ LSE,          // [B*H*Tq] or null
                      TStride qs, TStride ks, TStride os,
                      int B, int H, int Tq, int Tk, float scale) {
-  __shared__ __hip_bfloat16 vt[D][KVBLK + VT_PAD];
-  __shared__ float obounce[QBLK][D + 4];
+  constexpr int KVB = 64;                       // kv tile (2 x 32-row mfma)
+  __shared__ __hip_bfloat16 vt[D][KVB + VT_PAD];  // V^T image (shared)
+  __shared__ float obounce[NW][QBLK][D + 4];
 
-  const int lane = threadIdx.x;
-  const int hi = lane >> 5;     // half id
-  const int lq = lane & 31;     // q column this lane owns in S^T fragments
-  const int qtile = blockIdx.x;
+  const int lane = threadIdx.x & 63;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int hi = lane >> 5;
+  const int lq = lane & 31;
   const int bh = blockIdx.y;
   const int b = bh / H;
-  const int q0 = qtile * QBLK;
-  if (q0 >= Tq) return;
+  const int h = bh % H;
+  const int q0 = (blockIdx.x * NW + wid) * QBLK;
+  const bool wave_live = q0 < Tq;
   const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
 
-  const int h = bh % H;
   const __hip_bfloat16* Qb = Q + (size_t)b * qs.b + (size_t)h * qs.h;
   const __hip_bfloat16* Kb = K + (size_t)b * ks.b + (size_t)h * ks.h;
   const __hip_bfloat16* Vb = V + (size_t)b * ks.b + (size_t)h * ks.h;
   __hip_bfloat16* Ob = O + (size_t)b * os.b + (size_t)h * os.h;
 
-  // ---- load Q^T B-fragments (kept for the whole kernel) -----------------
-  // B-fragment for chunk c (d = 16c + (hi)*8 + j): lane needs
-  // Q[q0+lq][16c + hi*8 + j], j=0..7 -> one 16-byte load per chunk.
+  // Q^T B-fragments, scaled once
   bf16x8 qfrag[D / 16];
   {
-    const int qrow = min(q0 + lq, Tq - 1);
+    const int qrow = wave_live ? min(q0 + lq, Tq - 1) : 0;
     const __hip_bfloat16* qp = Qb + (size_t)qrow * qs.s;
 #pragma unroll
-    for (int c = 0; c < D / 16; ++c)
+    for (int c = 0; c < D / 16; ++c) {
       qfrag[c] = *reinterpret_cast<const bf16x8*>(qp + 16 * c + 8 * hi);
-    // fold the 1/sqrt(d) scale into Q once
-#pragma unroll
-    for (int c = 0; c < D / 16; ++c)
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         qfrag[c][j] = (__bf16)((float)qfrag[c][j] * scale);
+    }
   }
 
-  // ---- running softmax state (per q = lq, identical in both halves) -----
   float m_run = NEG_BIG;
   float l_run = 0.f;
-  f32x16 ot[D / 32];  // O^T accumulators: d-tiles of 32 rows x 32 q cols
+  f32x16 ot[D / 32];
 #pragma unroll
   for (int dt = 0; dt < D / 32; ++dt) ot[dt] = (f32x16)0.f;
 
-  const int kv_end = CAUSAL ? min(kv_len, q0 + QBLK) : kv_len;
+  // shared-staging loop bound: the LAST wave's causal frontier
+  const int q_hi = min((blockIdx.x * NW + NW) * QBLK, Tq);
+  const int kv_end_blk = CAUSAL ? min(kv_len, q_hi) : kv_len;
+  const int kv_end_own = CAUSAL ? min(kv_len, q0 + QBLK) : kv_len;
+  const int q_abs = q0 + lq;
 
-  for (int k0 = 0; k0 < kv_end; k0 += KVBLK) {
-    // ---- stage V^T tile into LDS (transpose in flight) ------------------
-    // read 8 consecutive k-rows x D cols coalesced; each lane holds 8 d's
-    // of one k; scatter-write transposed.
+  for (int k0 = 0; k0 < kv_end_blk; k0 += KVB) {
+    // ---- cooperative V^T staging (all NW*64 threads) --------------------
     {
-      const int rows_per_pass = 64 / (D / 8);  // 8 lanes per row at D=64
-      for (int kk = 0; kk < KVBLK; kk += rows_per_pass) {
-        const int krow = k0 + kk + lane / (D / 8);
-        const int dcol = (lane % (D / 8)) * 8;
-        bf16x8 vv = (krow < kv_len)
-            ? *reinterpret_cast<const bf16x8*>(Vb + (size_t)krow * ks.s + dcol)
-            : (bf16x8)(__bf16)0.f;
-        const int klocal = kk + lane / (D / 8);
+      const int rows_per_pass = (NW * 64) / (D / 8);
+      for (int kk = 0; kk < KVB; kk += rows_per_pass) {
+        const int krow = k0 + kk + tid / (D / 8);
+        const int dcol = (tid % (D / 8)) * 8;
+        if (kk + tid / (D / 8) < KVB) {
+          bf16x8 vv = (krow < kv_len)
+              ? *reinterpret_cast<const bf16x8*>(Vb + (size_t)krow * ks.s + dcol)
+              : (bf16x8)(__bf16)0.f;
+          const int klocal = kk + tid / (D / 8);
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          vt[dcol + j][klocal] = (__hip_bfloat16)(float)vv[j];
+          for (int j = 0; j < 8; ++j)
+            vt[dcol + j][klocal] = (__hip_bfloat16)(float)vv[j];
+        }
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);  // lgkm+vm drain before ds_read
+    __builtin_amdgcn_s_waitcnt(0);
     __builtin_amdgcn_s_barrier();
 
-    // ---- S^T = K . Q^T --------------------------------------------------
-    f32x16 st = (f32x16)0.f;
-    {
-      // A-fragment: lane holds K[k0 + lq][d = 16c + hi*8 + j] (row = key)
-      const int krow = k0 + lq;
-      const __hip_bfloat16* kp = Kb + (size_t)min(krow, Tk - 1) * ks.s;
+    if (wave_live && k0 < kv_end_own) {
+      // ---- S^T = K.Q^T for both 32-row k-halves -------------------------
+      f32x16 st0 = (f32x16)0.f, st1 = (f32x16)0.f;
 #pragma unroll
       for (int c = 0; c < D / 16; ++c) {
-        bf16x8 kf = *reinterpret_cast<const bf16x8*>(kp + 16 * c + 8 * hi);
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            kf, qfrag[c], st, 0, 0, 0);
+        const int kr0 = min(k0 + lq, Tk - 1);
+        const int kr1 = min(k0 + 32 + lq, Tk - 1);
+        bf16x8 kf0 = *reinterpret_cast<const bf16x8*>(
+            Kb + (size_t)kr0 * ks.s + 16 * c + 8 * hi);
+        bf16x8 kf1 = *reinterpret_cast<const bf16x8*>(
+            Kb + (size_t)kr1 * ks.s + 16 * c + 8 * hi);
+        st0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf0, qfrag[c], st0, 0, 0, 0);
+        st1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf1, qfrag[c], st1, 0, 0, 0);
+      }
+
+      // ---- mask + tile max ----------------------------------------------
+      float p[32];
+      float tile_max = NEG_BIG;
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int k_abs = k0 + 32 * half + crow(r, hi);
+          float v = half ? st1[r] : st0[r];
+          const bool dead = (k_abs >= kv_len) || (CAUSAL && k_abs > q_abs);
+          v = dead ? NEG_BIG : v;
+          p[16 * half + r] = v;
+          tile_max = fmaxf(tile_max, v);
+        }
+      }
+      tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32));
+
+      // ---- online rescale -----------------------------------------------
+      const float m_new = fmaxf(m_run, tile_max);
+      const float alpha = (m_new > NEG_BIG * 0.5f) ? __expf(m_run - m_new) : 1.f;
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 32; ++r) {
+        p[r] = (p[r] <= NEG_BIG * 0.5f) ? 0.f : __expf(p[r] - m_new);
+        psum += p[r];
+      }
+      psum += __shfl_xor(psum, 32);
+      l_run = l_run * alpha + psum;
+      m_run = m_new;
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
+
+      // ---- O^T += V^T . P^T over 4 k-chunks -----------------------------
+      bf16x8 pb0[2], pb1[2];
+      pack_p_to_b(p, pb0);
+      pack_p_to_b(p + 16, pb1);
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt) {
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          bf16x8 va0 = *reinterpret_cast<const bf16x8*>(
+              &vt[32 * dt + lq][16 * c + 8 * hi]);
+          bf16x8 va1 = *reinterpret_cast<const bf16x8*>(
+              &vt[32 * dt + lq][32 + 16 * c + 8 * hi]);
+          ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va0, pb0[c],
+                                                           ot[dt], 0, 0, 0);
+          ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va1, pb1[c],
+                                                           ot[dt], 0, 0, 0);
+        }
       }
     }
+    __builtin_amdgcn_s_barrier();
+  }
 
-    // ---- mask + per-q max of this tile ---------------------------------
-    float p[16];
-    float tile_max = NEG_BIG;
-    const int q_abs = q0 + lq;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int k_abs = k0 + crow(r, hi);
-      float v = st[r];
-      const bool dead = (k_abs >= kv_len) || (CAUSAL && k_abs > q_abs);
-      v = dead ? NEG_BIG : v;
-      p[r] = v;
-      tile_max = fmaxf(tile_max, v);
-    }
-    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32));
-
-    // ---- online rescale -------------------------------------------------
-    const float m_new = fmaxf(m_run, tile_max);
-    float alpha = (m_new > NEG_BIG * 0.5f) ? __expf(m_run - m_new) : 1.f;
-    float psum = 0.f;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      p[r] = (p[r] <= NEG_BIG * 0.5f) ? 0.f : __expf(p[r] - m_new);
-      psum += p[r];
-    }
-    psum += __shfl_xor(psum, 32);
-    l_run = l_run * alpha + psum;
-    m_run = m_new;
+  // ---- epilogue ----------------------------------------------------------
+  if (wave_live) {
+    const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
 #pragma unroll
     for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) ot[dt][r] *= alpha;
-
-    // ---- O^T += V^T . P^T ----------------------------------------------
-    bf16x8 pb[2];
-    pack_p_to_b(p, pb);
-#pragma unroll
-    for (int dt = 0; dt < D / 32; ++dt) {
-#pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        // A-fragment: lane holds V^T[32*dt + lq][k = 16c + hi*8 + j]
-        const __hip_bfloat16* vp = &vt[32 * dt + lq][16 * c + 8 * hi];
-        bf16x8 va = *reinterpret_cast<const bf16x8*>(vp);
-        ot[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            va, pb[c], ot[dt], 0, 0, 0);
-      }
-    }
-    __builtin_amdgcn_s_barrier();  // protect LDS V^T before next stage
+      for (int r = 0; r < 16; ++r)
+        obounce[wid][lq][32 * dt + crow(r, hi)] = ot[dt][r] * inv_l;
   }
-
-  // ---- epilogue: O = O^T / l, via LDS bounce ---------------------------
-  const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
-#pragma unroll
-  for (int dt = 0; dt < D / 32; ++dt)
-#pragma unroll
-    for (int r = 0; r < 16; ++r)
-      obounce[lq][32 * dt + crow(r, hi)] = ot[dt][r] * inv_l;
   __builtin_amdgcn_s_barrier();
-
-  // write rows coalesced: each lane writes 8 floats -> bf16x8 chunks
-  {
-    const int lanes_per_row = D / 8;          // 8 at D=64
+  if (wave_live) {
+    const int lanes_per_row = D / 8;
     const int rows_per_pass = 64 / lanes_per_row;
     for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
       const int qrow = q0 + qq + lane / lanes_per_row;
@@ -255,15 +266,15 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
         bf16x8 out;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          out[j] = (__bf16)obounce[qq + lane / lanes_per_row][dcol + j];
+          out[j] = (__bf16)obounce[wid][qq + lane / lanes_per_row][dcol + j];
         *reinterpret_cast<bf16x8*>(Ob + (size_t)qrow * os.s + dcol) = out;
       }
     }
-  }
-  if (LSE != nullptr && hi == 0) {
-    const int qrow = q0 + lq;
-    if (qrow < Tq)
-      LSE[(size_t)bh * Tq + qrow] = m_run + __logf(fmaxf(l_run, 1e-30f));
+    if (LSE != nullptr && hi == 0) {
+      const int qrow = q0 + lq;
+      if (qrow < Tq)
+        LSE[(size_t)bh * Tq + qrow] = m_run + __logf(fmaxf(l_run, 1e-30f));
+    }
   }
 }
 
@@ -318,11 +329,13 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
     TORCH_CHECK(lensc.numel() == B);
     lens_ptr = lensc.data_ptr<int>();
   }
-  dim3 grid((Tq + 31) / 32, B * H);
   auto stream = at::cuda::getCurrentHIPStream();
-#define RUN(CAUSAL, HASLEN)                                                  \
-  hipLaunchKernelGGL((attn_fwd_kernel<64, CAUSAL, HASLEN>), grid, dim3(64),  \
-                     0, stream,                                              \
+  const int nqt = (Tq + 31) / 32;
+  const int NW = (nqt % 4 == 0) ? 4 : (nqt % 2 == 0 ? 2 : 1);
+  dim3 grid(nqt / NW, B * H);
+#define RUN1(NWV, CAUSAL, HASLEN)                                            \
+  hipLaunchKernelGGL((attn_fwd_kernel<64, NWV, CAUSAL, HASLEN>), grid,       \
+                     dim3(64 * NWV), 0, stream,                              \
                      reinterpret_cast<const __hip_bfloat16*>(qc.data_ptr()), \
                      reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()), \
                      reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()), \
@@ -330,11 +343,18 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                      reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),        \
                      lse_ptr, tstride(qc), tstride(kc), tstride(o),          \
                      B, H, Tq, Tk, (float)scale)
+#define RUN(CAUSAL, HASLEN)                                                  \
+  do {                                                                       \
+    if (NW == 4) RUN1(4, CAUSAL, HASLEN);                                    \
+    else if (NW == 2) RUN1(2, CAUSAL, HASLEN);                               \
+    else RUN1(1, CAUSAL, HASLEN);                                            \
+  } while (0)
   if (causal && lens_ptr) RUN(true, true);
   else if (causal) RUN(true, false);
   else if (lens_ptr) RUN(false, true);
   else RUN(false, false);
 #undef RUN
+#undef RUN1
   return {o, lse};
 }
 
