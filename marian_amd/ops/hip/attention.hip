@@ -154,26 +154,45 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
   const int kv_end_own = CAUSAL ? min(kv_len, q0 + QBLK) : kv_len;
   const int q_abs = q0 + lq;
 
+  // async-STAGE split (guide T14): each thread keeps its share of the NEXT
+  // V tile in registers; global loads are issued right after the barrier so
+  // HBM latency hides under the tile's MFMA compute, and the ds_write of
+  // the prefetched tile happens at the top of the next iteration.
+  constexpr int RPP = (NW * 64) / (D / 8);  // staged rows per pass
+  constexpr int NPASS = (KVB + RPP - 1) / RPP;
+  bf16x8 vld[NPASS];
+  const int stage_row = tid / (D / 8);      // k-row this thread stages
+  const int stage_col = (tid % (D / 8)) * 8;
+
+  auto issue_v_loads = [&](int k0) {
+#pragma unroll
+    for (int ps = 0; ps < NPASS; ++ps) {
+      const int kl = ps * RPP + stage_row;
+      const int krow = k0 + kl;
+      vld[ps] = (kl < KVB && krow < kv_len)
+          ? *reinterpret_cast<const bf16x8*>(Vb + (size_t)krow * ks.s + stage_col)
+          : (bf16x8)(__bf16)0.f;
+    }
+  };
+  issue_v_loads(0);
+
   for (int k0 = 0; k0 < kv_end_blk; k0 += KVB) {
-    // ---- cooperative V^T staging (all NW*64 threads) --------------------
+    // ---- write the prefetched V^T tile (transpose in flight) -----------
     {
-      const int rows_per_pass = (NW * 64) / (D / 8);
-      for (int kk = 0; kk < KVB; kk += rows_per_pass) {
-        const int krow = k0 + kk + tid / (D / 8);
-        const int dcol = (tid % (D / 8)) * 8;
-        if (kk + tid / (D / 8) < KVB) {
-          bf16x8 vv = (krow < kv_len)
-              ? *reinterpret_cast<const bf16x8*>(Vb + (size_t)krow * ks.s + dcol)
-              : (bf16x8)(__bf16)0.f;
-          const int klocal = kk + tid / (D / 8);
+#pragma unroll
+      for (int ps = 0; ps < NPASS; ++ps) {
+        const int kl = ps * RPP + stage_row;
+        if (kl < KVB) {
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            vt[dcol + j][klocal] = (__hip_bfloat16)(float)vv[j];
+            vt[stage_col + j][kl] = (__hip_bfloat16)(float)vld[ps][j];
         }
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_s_waitcnt(0xc07f);  // lgkmcnt(0) only: keep loads live
     __builtin_amdgcn_s_barrier();
+    if (k0 + KVB < kv_end_blk)
+      issue_v_loads(k0 + KVB);  // in flight under this tile's compute
 
     if (wave_live && k0 < kv_end_own) {
       // ---- S^T = K.Q^T for both 32-row k-halves -------------------------
