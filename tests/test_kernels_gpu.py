@@ -459,3 +459,35 @@ def test_layernorm_residual_fused(ext, dtype):
     assert _rel_err(x.grad, xr.grad) < tol
     assert _rel_err(r.grad, rr.grad) < tol
     assert _rel_err(g.grad, gr.grad) < tol
+
+
+def test_mini_batch_fit_collect_stats(ext):
+    """--mini-batch-fit measured grid on the GPU (ref: collectStats)."""
+    from marian_amd.common.options import Options
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.training.graph_group import SyncGraphGroup
+    from marian_amd.training.batch_stats_fit import collect_stats
+    opts = Options({
+        "type": "transformer", "dim-emb": 128, "enc-depth": 1, "dec-depth": 1,
+        "transformer-heads": 2, "transformer-dim-ffn": 256,
+        "transformer-preprocess": "", "transformer-postprocess": "dan",
+        "transformer-postprocess-emb": "d", "transformer-postprocess-top": "",
+        "transformer-ffn-activation": "relu", "transformer-dropout": 0.0,
+        "dim-vocabs": [2000, 2000], "tied-embeddings-all": True,
+        "transformer-train-position-embeddings": False,
+        "transformer-decoder-dim-ffn": 0,
+        "precision": ["bfloat16", "float32"],
+        "cost-type": "ce-mean-words", "label-smoothing": 0.1,
+        "optimizer": "adam", "learn-rate": 1e-3, "clip-norm": 0.0,
+        "max-length": 30, "mini-batch-fit-step": 10,
+        "workspace": 4000,  # 4 GB budget
+    })
+    torch.manual_seed(3)
+    model = create_model(opts)
+    gg = SyncGraphGroup(opts, model, torch.device(DEV))
+    stats = collect_stats(opts, gg)
+    # fit table covers the length grid and batch sizes are monotone-ish
+    b10 = stats.find(10, 10)
+    b30 = stats.find(30, 30)
+    assert b10 >= b30 >= 1
+    assert b10 > 8  # a 4GB budget fits far more than 8 tiny sentences
