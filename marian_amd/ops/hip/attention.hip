@@ -394,8 +394,8 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
 
 namespace {
 
-template <int D, bool CAUSAL, bool HAS_LEN>
-__global__ __launch_bounds__(64)
+template <int D, int NW, bool CAUSAL, bool HAS_LEN>
+__global__ __launch_bounds__(64 * NW)
 void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
                         const __hip_bfloat16* __restrict__ K,
                         const __hip_bfloat16* __restrict__ V,
@@ -406,16 +406,18 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
                         __hip_bfloat16* __restrict__ dQ,
                         TStride qs, TStride ks, TStride dos, TStride dqs,
                         int B, int H, int Tq, int Tk, float scale) {
-  __shared__ __hip_bfloat16 kt[D][KVBLK + VT_PAD];  // K^T image per tile
-  __shared__ float dq_b[QBLK][D + 4];
+  __shared__ __hip_bfloat16 kt[D][KVBLK + VT_PAD];  // K^T image (shared)
+  __shared__ float dq_b[NW][QBLK][D + 4];
 
-  const int lane = threadIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
   const int hi = lane >> 5;
   const int lq = lane & 31;
   const int bh = blockIdx.y;
   const int b = bh / H;
-  const int q0 = blockIdx.x * QBLK;
-  if (q0 >= Tq) return;
+  const int q0 = (blockIdx.x * NW + wid) * QBLK;
+  const bool wave_live = q0 < Tq;
   const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
   const int h = bh % H;
   const __hip_bfloat16* Qb = Q + (size_t)b * qs.b + (size_t)h * qs.h;
@@ -439,15 +441,42 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
     }
   }
   const int q_abs = q0 + lq;
-  const float lse_lane = (q_abs < Tq) ? LSE[(size_t)bh * Tq + q_abs] : 0.f;
-  const float dsum_lane = (q_abs < Tq) ? Dsum[(size_t)bh * Tq + q_abs] : 0.f;
+  const float lse_lane = (wave_live && q_abs < Tq)
+      ? LSE[(size_t)bh * Tq + q_abs] : 0.f;
+  const float dsum_lane = (wave_live && q_abs < Tq)
+      ? Dsum[(size_t)bh * Tq + q_abs] : 0.f;
 
   f32x16 dqt[D / 32];
 #pragma unroll
   for (int dt2 = 0; dt2 < D / 32; ++dt2) dqt[dt2] = (f32x16)0.f;
 
+  const int q_hi = min((blockIdx.x * NW + NW) * QBLK, Tq);
+  const int kv_end_blk = CAUSAL ? min(kv_len, q_hi) : kv_len;
   const int kv_end = CAUSAL ? min(kv_len, q0 + QBLK) : kv_len;
-  for (int k0 = 0; k0 < kv_end; k0 += KVBLK) {
+  for (int k0 = 0; k0 < kv_end_blk; k0 += KVBLK) {
+    {
+      // cooperative K^T staging with all NW*64 threads
+      const int rows_per_pass = (NW * 64) / (D / 8);
+      for (int kk = 0; kk < KVBLK; kk += rows_per_pass) {
+        const int kl = kk + tid / (D / 8);
+        const int krow2 = k0 + kl;
+        const int dcol = (tid % (D / 8)) * 8;
+        if (kl < KVBLK) {
+          bf16x8 kv = (krow2 < kv_len)
+              ? *reinterpret_cast<const bf16x8*>(Kb + (size_t)krow2 * ks.s + dcol)
+              : (bf16x8)(__bf16)0.f;
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            kt[dcol + j][kl] = (__hip_bfloat16)(float)kv[j];
+        }
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_s_barrier();
+    if (!wave_live || k0 >= kv_end) {
+      __builtin_amdgcn_s_barrier();
+      continue;
+    }
     bf16x8 kfrag[D / 16], vfrag[D / 16];
     {
       const int krow = min(k0 + lq, Tk - 1);
@@ -458,21 +487,7 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
         kfrag[c] = *reinterpret_cast<const bf16x8*>(kp + 16 * c + 8 * hi);
         vfrag[c] = *reinterpret_cast<const bf16x8*>(vp + 16 * c + 8 * hi);
       }
-      const int rows_per_pass = 64 / (D / 8);
-      for (int kk = 0; kk < KVBLK; kk += rows_per_pass) {
-        const int krow2 = k0 + kk + lane / (D / 8);
-        const int dcol = (lane % (D / 8)) * 8;
-        bf16x8 kv = (krow2 < kv_len)
-            ? *reinterpret_cast<const bf16x8*>(Kb + (size_t)krow2 * ks.s + dcol)
-            : (bf16x8)(__bf16)0.f;
-        const int klocal = kk + lane / (D / 8);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          kt[dcol + j][klocal] = (__hip_bfloat16)(float)kv[j];
-      }
     }
-    __builtin_amdgcn_s_waitcnt(0);
-    __builtin_amdgcn_s_barrier();
 
     f32x16 st = (f32x16)0.f;
     f32x16 dpt = (f32x16)0.f;
@@ -505,30 +520,34 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
     __builtin_amdgcn_s_barrier();
   }
 
-  // epilogue: dQ via LDS bounce
+  // epilogue: dQ via per-wave LDS bounce slice
+  if (wave_live) {
 #pragma unroll
-  for (int dt2 = 0; dt2 < D / 32; ++dt2)
+    for (int dt2 = 0; dt2 < D / 32; ++dt2)
 #pragma unroll
-    for (int r = 0; r < 16; ++r)
-      dq_b[lq][32 * dt2 + crow(r, hi)] = dqt[dt2][r];
+      for (int r = 0; r < 16; ++r)
+        dq_b[wid][lq][32 * dt2 + crow(r, hi)] = dqt[dt2][r];
+  }
   __builtin_amdgcn_s_barrier();
-  const int lanes_per_row = D / 8;
-  const int rows_per_pass = 64 / lanes_per_row;
-  for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
-    const int qrow = q0 + qq + lane / lanes_per_row;
-    const int dcol = (lane % lanes_per_row) * 8;
-    if (qrow < Tq) {
-      bf16x8 out;
+  if (wave_live) {
+    const int lanes_per_row = D / 8;
+    const int rows_per_pass = 64 / lanes_per_row;
+    for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
+      const int qrow = q0 + qq + lane / lanes_per_row;
+      const int dcol = (lane % lanes_per_row) * 8;
+      if (qrow < Tq) {
+        bf16x8 out;
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        out[j] = (__bf16)dq_b[qq + lane / lanes_per_row][dcol + j];
-      *reinterpret_cast<bf16x8*>(dQb + (size_t)qrow * dqs.s + dcol) = out;
+        for (int j = 0; j < 8; ++j)
+          out[j] = (__bf16)dq_b[wid][qq + lane / lanes_per_row][dcol + j];
+        *reinterpret_cast<bf16x8*>(dQb + (size_t)qrow * dqs.s + dcol) = out;
+      }
     }
   }
 }
 
-template <int D, bool CAUSAL, bool HAS_LEN>
-__global__ __launch_bounds__(64)
+template <int D, int NW, bool CAUSAL, bool HAS_LEN>
+__global__ __launch_bounds__(64 * NW)
 void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
                          const __hip_bfloat16* __restrict__ K,
                          const __hip_bfloat16* __restrict__ V,
@@ -540,17 +559,19 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
                          __hip_bfloat16* __restrict__ dV,
                          TStride qs, TStride ks, TStride dos, TStride dks,
                          int B, int H, int Tq, int Tk, float scale) {
-  __shared__ __hip_bfloat16 qt[D][QBLK + VT_PAD];   // Q^T image per q-tile
-  __shared__ __hip_bfloat16 dot[D][QBLK + VT_PAD];  // dO^T image per q-tile
-  __shared__ float kv_b[KVBLK][D + 4];
+  __shared__ __hip_bfloat16 qt[D][QBLK + VT_PAD];   // Q^T image (shared)
+  __shared__ __hip_bfloat16 dot[D][QBLK + VT_PAD];  // dO^T image (shared)
+  __shared__ float kv_b[NW][KVBLK][D + 4];
 
-  const int lane = threadIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
   const int hi = lane >> 5;
   const int lq = lane & 31;
   const int bh = blockIdx.y;
   const int b = bh / H;
-  const int k0 = blockIdx.x * KVBLK;
-  if (k0 >= Tk) return;
+  const int k0 = (blockIdx.x * NW + wid) * KVBLK;
+  const bool wave_live = k0 < Tk;
   const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
   const int h = bh % H;
   const __hip_bfloat16* Qb = Q + (size_t)b * qs.b + (size_t)h * qs.h;
@@ -580,9 +601,42 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
     dv_acc[dt2] = (f32x16)0.f;
   }
 
-  const int q_start = CAUSAL ? (k0 / QBLK) * QBLK : 0;
-  for (int q0 = q_start; q0 < Tq; q0 += QBLK) {
-    // per-q-tile fragments + transposed images
+  // causal: the FIRST wave's kv tile bounds the shared q loop start
+  const int k0_first = (blockIdx.x * NW) * KVBLK;
+  const int q_start_blk = CAUSAL ? (k0_first / QBLK) * QBLK : 0;
+  const int q_start_own = CAUSAL ? (k0 / QBLK) * QBLK : 0;
+  for (int q0 = q_start_blk; q0 < Tq; q0 += QBLK) {
+    {
+      // cooperative Q^T/dO^T staging with all NW*64 threads
+      const int rows_per_pass = (NW * 64) / (D / 8);
+      for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
+        const int qlocal = qq + tid / (D / 8);
+        const int qrow2 = q0 + qlocal;
+        const int dcol = (tid % (D / 8)) * 8;
+        if (qlocal < QBLK) {
+          const bool live = qrow2 < Tq;
+          const int qr = live ? qrow2 : (Tq - 1);
+          bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (size_t)qr * qs.s + dcol);
+          bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (size_t)qr * dos.s + dcol);
+          if (!live) {
+            vq = (bf16x8)(__bf16)0.f;
+            vd = (bf16x8)(__bf16)0.f;
+          }
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            qt[dcol + j][qlocal] = (__hip_bfloat16)(float)vq[j];
+            dot[dcol + j][qlocal] = (__hip_bfloat16)(float)vd[j];
+          }
+        }
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __builtin_amdgcn_s_barrier();
+    if (!wave_live || q0 < q_start_own) {
+      __builtin_amdgcn_s_barrier();
+      continue;
+    }
+    // per-q-tile fragments (direct loads)
     bf16x8 qfrag[D / 16], dofrag[D / 16];
     {
       const bool qlive = (q0 + lq) < Tq;
@@ -596,28 +650,7 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
         if (!qlive) dv = (bf16x8)(__bf16)0.f;
         dofrag[c] = dv;
       }
-      const int rows_per_pass = 64 / (D / 8);
-      for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
-        const int qrow2 = q0 + qq + lane / (D / 8);
-        const int dcol = (lane % (D / 8)) * 8;
-        const bool live = qrow2 < Tq;
-        const int qr = live ? qrow2 : (Tq - 1);
-        bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (size_t)qr * qs.s + dcol);
-        bf16x8 vd = *reinterpret_cast<const bf16x8*>(dOb + (size_t)qr * dos.s + dcol);
-        if (!live) {
-          vq = (bf16x8)(__bf16)0.f;
-          vd = (bf16x8)(__bf16)0.f;
-        }
-        const int qlocal = qq + lane / (D / 8);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          qt[dcol + j][qlocal] = (__hip_bfloat16)(float)vq[j];
-          dot[dcol + j][qlocal] = (__hip_bfloat16)(float)vd[j];
-        }
-      }
     }
-    __builtin_amdgcn_s_waitcnt(0);
-    __builtin_amdgcn_s_barrier();
 
     f32x16 s = (f32x16)0.f;
     f32x16 dp = (f32x16)0.f;
@@ -658,30 +691,34 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
     __builtin_amdgcn_s_barrier();
   }
 
-  // epilogue: dV then dK via the same LDS bounce ([32k x 32d] frags:
-  // lane = d-col, regs = k-rows)
+  // epilogue: dV then dK via per-wave LDS bounce slices
   for (int which = 0; which < 2; ++which) {
     __hip_bfloat16* out_base = (which == 0) ? dVb : dKb;
     f32x16* acc = (which == 0) ? dv_acc : dk_acc;
+    if (wave_live) {
 #pragma unroll
-    for (int dt2 = 0; dt2 < D / 32; ++dt2)
+      for (int dt2 = 0; dt2 < D / 32; ++dt2)
 #pragma unroll
-      for (int r = 0; r < 16; ++r)
-        kv_b[crow(r, hi)][32 * dt2 + lq] = acc[dt2][r];
+        for (int r = 0; r < 16; ++r)
+          kv_b[wid][crow(r, hi)][32 * dt2 + lq] = acc[dt2][r];
+    }
     __builtin_amdgcn_s_barrier();
-    const int lanes_per_row = D / 8;
-    const int rows_per_pass = 64 / lanes_per_row;
-    for (int kk = 0; kk < KVBLK; kk += rows_per_pass) {
-      const int krow = k0 + kk + lane / lanes_per_row;
-      const int dcol = (lane % lanes_per_row) * 8;
-      if (krow < Tk) {
-        bf16x8 out;
-        const bool live = krow < kv_len;
+    if (wave_live) {
+      const int lanes_per_row = D / 8;
+      const int rows_per_pass = 64 / lanes_per_row;
+      for (int kk = 0; kk < KVBLK; kk += rows_per_pass) {
+        const int krow = k0 + kk + lane / lanes_per_row;
+        const int dcol = (lane % lanes_per_row) * 8;
+        if (krow < Tk) {
+          bf16x8 out;
+          const bool live = krow < kv_len;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          out[j] = live ? (__bf16)kv_b[kk + lane / lanes_per_row][dcol + j]
-                        : (__bf16)0.f;
-        *reinterpret_cast<bf16x8*>(out_base + (size_t)krow * dks.s + dcol) = out;
+          for (int j = 0; j < 8; ++j)
+            out[j] = live
+                ? (__bf16)kv_b[wid][kk + lane / lanes_per_row][dcol + j]
+                : (__bf16)0.f;
+          *reinterpret_cast<bf16x8*>(out_base + (size_t)krow * dks.s + dcol) = out;
+        }
       }
     }
     __builtin_amdgcn_s_barrier();
@@ -723,12 +760,16 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
     lens_ptr = lensc.data_ptr<int>();
   }
   auto stream = at::cuda::getCurrentHIPStream();
-  dim3 gq((Tq + 31) / 32, B * H);
-  dim3 gk((Tk + 31) / 32, B * H);
-#define RUNB(CAUSAL, HASLEN)                                                  \
+  const int nqt = (Tq + 31) / 32;
+  const int nkt = (Tk + 31) / 32;
+  const int NWQ = (nqt % 4 == 0) ? 4 : (nqt % 2 == 0 ? 2 : 1);
+  const int NWK = (nkt % 4 == 0) ? 4 : (nkt % 2 == 0 ? 2 : 1);
+  dim3 gq(nqt / NWQ, B * H);
+  dim3 gk(nkt / NWK, B * H);
+#define RUNB1(NQ, NK, CAUSAL, HASLEN)                                         \
   do {                                                                        \
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<64, CAUSAL, HASLEN>), gq,          \
-                       dim3(64), 0, stream,                                   \
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<64, NQ, CAUSAL, HASLEN>), gq,      \
+                       dim3(64 * NQ), 0, stream,                              \
                        reinterpret_cast<const __hip_bfloat16*>(qc.data_ptr()),\
                        reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),\
                        reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),\
@@ -738,8 +779,8 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
                        reinterpret_cast<__hip_bfloat16*>(dq.data_ptr()),      \
                        tstride(qc), tstride(kc), tstride(doc), tstride(dq),   \
                        B, H, Tq, Tk, (float)scale);                           \
-    hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, CAUSAL, HASLEN>), gk,         \
-                       dim3(64), 0, stream,                                   \
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<64, NK, CAUSAL, HASLEN>), gk,     \
+                       dim3(64 * NK), 0, stream,                              \
                        reinterpret_cast<const __hip_bfloat16*>(qc.data_ptr()),\
                        reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),\
                        reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),\
@@ -751,10 +792,24 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
                        tstride(qc), tstride(kc), tstride(doc), tstride(dk),   \
                        B, H, Tq, Tk, (float)scale);                           \
   } while (0)
+#define RUNB2(NQ, CAUSAL, HASLEN)                                             \
+  do {                                                                        \
+    if (NWK == 4) RUNB1(NQ, 4, CAUSAL, HASLEN);                               \
+    else if (NWK == 2) RUNB1(NQ, 2, CAUSAL, HASLEN);                          \
+    else RUNB1(NQ, 1, CAUSAL, HASLEN);                                        \
+  } while (0)
+#define RUNB(CAUSAL, HASLEN)                                                  \
+  do {                                                                        \
+    if (NWQ == 4) RUNB2(4, CAUSAL, HASLEN);                                   \
+    else if (NWQ == 2) RUNB2(2, CAUSAL, HASLEN);                              \
+    else RUNB2(1, CAUSAL, HASLEN);                                            \
+  } while (0)
   if (causal && lens_ptr) RUNB(true, true);
   else if (causal) RUNB(true, false);
   else if (lens_ptr) RUNB(false, true);
   else RUNB(false, false);
 #undef RUNB
+#undef RUNB2
+#undef RUNB1
   return {dq, dk, dv};
 }
