@@ -74,6 +74,7 @@ class BeamSearch:
 
     # ------------------------------------------------------------------
     def _normalized(self, score: float, length: int) -> float:
+        length = max(length, 1)
         s = score - self.word_penalty * length
         if self.alpha > 0:
             s = s / (length ** self.alpha)
