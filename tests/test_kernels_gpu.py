@@ -491,3 +491,39 @@ def test_mini_batch_fit_collect_stats(ext):
     b30 = stats.find(30, 30)
     assert b10 >= b30 >= 1
     assert b10 > 8  # a 4GB budget fits far more than 8 tiny sentences
+
+
+def test_fp16_training_with_cost_scaling(ext):
+    """--precision float16 + dynamic cost scaling trains on GPU."""
+    from marian_amd.common.options import Options
+    from marian_amd.data.synthetic import synthetic_batch
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.training.graph_group import SyncGraphGroup
+    opts = Options({
+        "type": "transformer", "dim-emb": 128, "enc-depth": 1, "dec-depth": 1,
+        "transformer-heads": 2, "transformer-dim-ffn": 256,
+        "transformer-preprocess": "", "transformer-postprocess": "dan",
+        "transformer-postprocess-emb": "d", "transformer-postprocess-top": "",
+        "transformer-ffn-activation": "relu", "transformer-dropout": 0.0,
+        "dim-vocabs": [1000, 1000], "tied-embeddings-all": True,
+        "transformer-train-position-embeddings": False,
+        "transformer-decoder-dim-ffn": 0,
+        "precision": ["float16", "float32"],
+        "cost-scaling": ["8", "100", "2", "1"],
+        "cost-type": "ce-mean-words", "label-smoothing": 0.1,
+        "optimizer": "adam", "learn-rate": 5e-3, "clip-norm": 1.0,
+    })
+    torch.manual_seed(17)
+    model = create_model(opts)
+    gg = SyncGraphGroup(opts, model, torch.device(DEV))
+    assert gg.scaler.enabled and gg.scaler.scale == 256.0
+    gg.set_lr(5e-3)
+    batch = synthetic_batch(16, 10, 10, 1000, device=DEV)
+    batch.sub[-1].indices.copy_(batch.sub[0].indices)
+    first = None
+    for i in range(15):
+        lv = gg.update(batch)
+        ce = float(lv.ce_sum) / lv.label_count
+        first = first if first is not None else ce
+    assert ce < first - 0.2, (first, ce)
+    assert torch.isfinite(gg.optimizer.master).all()
