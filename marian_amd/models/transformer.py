@@ -147,6 +147,17 @@ class TransformerModel(nn.Module):
             [EncoderLayer(options) for _ in range(options["enc-depth"])])
         self.decoder = nn.ModuleList(
             [DecoderLayer(options) for _ in range(options["dec-depth"])])
+        # --transformer-tied-layers: entry i (1-based) = layer whose
+        # parameters layer i reuses (universal-transformer style sharing)
+        tied_layers = [int(x) for x in
+                       options.list("transformer-tied-layers") or []]
+        for i, src in enumerate(tied_layers):
+            if 1 <= src <= len(self.encoder) and src - 1 != i \
+                    and i < len(self.encoder):
+                self.encoder[i] = self.encoder[src - 1]
+            if 1 <= src <= len(self.decoder) and src - 1 != i \
+                    and i < len(self.decoder):
+                self.decoder[i] = self.decoder[src - 1]
 
         top = options.get("transformer-postprocess-top", "")
         self.post_top_enc = PrePostProcessor(top, d, drop) if top else None

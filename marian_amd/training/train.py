@@ -127,6 +127,13 @@ class Train:
                 prog = self.model_path + ".progress.yml"
                 if os.path.exists(prog):
                     self.state = TrainingState.load(prog)
+                if options.get("valid-reset-all", False):
+                    self.state.validators = {}
+                    self.state.stalled = 0
+                elif options.get("valid-reset-stalled", False):
+                    self.state.stalled = 0
+                    for v in self.state.validators.values():
+                        v["stalled"] = 0
                 log_info("Resumed training at epoch %d update %d",
                          self.state.epochs, self.state.batches)
 

@@ -165,3 +165,16 @@ def test_average_attention_decoder():
         stepped = torch.stack(outs, dim=1)
     assert torch.allclose(full, stepped, atol=1e-4), \
         (full - stepped).abs().max()
+
+
+def test_tied_layers():
+    """--transformer-tied-layers shares layer parameters."""
+    opts = tiny_options(**{"transformer-tied-layers": [1, 1]})
+    m = create_model(opts)
+    assert m.encoder[1] is m.encoder[0]
+    assert m.decoder[1] is m.decoder[0]
+    b = synthetic_batch(2, 4, 4, 37)
+    logits = m(b)
+    logits.float().pow(2).mean().backward()
+    # shared params got gradients once
+    assert m.encoder[0].self_attn.Wq.W.grad is not None
