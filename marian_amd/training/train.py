@@ -199,8 +199,7 @@ class Train:
                     continue
                 n_sents = getattr(lv, "sentences", batch.size())
                 actions = sched.update(
-                    float(lv.ce_sum) / max(1, lv.label_count),
-                    n_sents, lv.label_count,
+                    float(lv.ce_sum), n_sents, lv.label_count,
                     grad_norm=getattr(lv, "grad_norm", None))
                 if actions["save"]:
                     self.save()
