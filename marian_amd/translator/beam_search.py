@@ -141,11 +141,16 @@ class BeamSearch:
                 al = states[0]["cache"].pop("alignments", None)
                 align_steps.append(al[-1].squeeze(1).cpu()
                                    if al else None)  # [B*K, Tsrc]
-            if not self.allow_unk and shortlist is None:
-                logprobs[:, self.unk_id] = NEG_INF
-            # dead beams must not propose continuations
-            logprobs = torch.where(alive[:, None], logprobs,
-                                   torch.full_like(logprobs, NEG_INF))
+            fused_topk = (self.sampling is None and logprobs.is_cuda)
+            if fused_topk:
+                from ..ops import hip_ext
+                fused_topk = hasattr(hip_ext(required=True), "beam_topk")
+            if not fused_topk:
+                if not self.allow_unk and shortlist is None:
+                    logprobs[:, self.unk_id] = NEG_INF
+                # dead beams must not propose continuations
+                logprobs = torch.where(alive[:, None], logprobs,
+                                       torch.full_like(logprobs, NEG_INF))
 
             forced_now = None
             if prefix is not None:
@@ -173,6 +178,17 @@ class BeamSearch:
                 top_idx = word.view(B, K)  # beam 0 implied
                 beam_idx = torch.zeros_like(top_idx)
                 word_idx = top_idx
+            elif fused_topk:
+                # fused HIP kernel: path-score add + dead-beam skip + UNK
+                # suppression + top-k without materializing [B, K*V]
+                # (ref: nth_element.cu NthElementGPU)
+                from ..ops import hip_ext
+                sup = -1 if (self.allow_unk or shortlist is not None) \
+                    else self.unk_id
+                top_scores, top_idx = hip_ext().beam_topk(
+                    logprobs, scores.view(B * K), K, K, sup)
+                beam_idx = top_idx // V
+                word_idx = top_idx % V
             else:
                 cand = scores.view(B * K, 1) + logprobs      # [B*K, V]
                 cand = cand.view(B, K * V)

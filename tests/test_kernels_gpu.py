@@ -635,3 +635,27 @@ def test_transformer_base_converges_T128(ext):
         lv = gg.update(batch)
         ces.append(float(lv.ce_sum) / lv.label_count)
     assert ces[-1] < ces[0] - 2.0, (ces[0], ces[-1])
+
+
+def test_beam_topk_kernel(ext):
+    """Fused beam top-k vs torch reference incl. dead beams + suppression."""
+    torch.manual_seed(51)
+    B, NB, V, K = 5, 4, 32000, 4
+    lp = torch.randn(B * NB, V, device=DEV)
+    scores = torch.randn(B * NB, device=DEV)
+    scores[1] = -1e9  # dead beam
+    v, i = ext.beam_topk(lp, scores, NB, K, 1)  # suppress UNK=1
+    # reference
+    cand = (lp + scores[:, None]).clone()
+    cand[:, 1] = -1e30
+    cand[1, :] = -1e30
+    cand = cand.view(B, NB * V)
+    rv, ri = torch.topk(cand, K, dim=-1)
+    assert torch.allclose(v, rv, atol=1e-5), (v - rv).abs().max()
+    assert torch.equal(i, ri)
+    # bf16 input path
+    v2, i2 = ext.beam_topk(lp.to(torch.bfloat16), scores, NB, K, -1)
+    cand2 = (lp.to(torch.bfloat16).float() + scores[:, None])
+    cand2[1, :] = -1e30
+    rv2, ri2 = torch.topk(cand2.view(B, NB * V), K, dim=-1)
+    assert torch.equal(i2, ri2)
