@@ -102,6 +102,9 @@ def main():
                                     0.0, 5, 1.0),
               10, bytes_moved=n * 26)
 
+    if which in ("fp8",):
+        bench_fp8()
+
     if which in ("softmax", "all"):
         R, C = 32768, 32000
         x = torch.randn(R, C, device=dev, dtype=torch.bfloat16)
@@ -111,3 +114,26 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def bench_fp8():
+    """Is torch._scaled_mm actually hitting fp8 MFMA rate on gfx950?"""
+    import torch
+    dev = "cuda"
+    for (M, K, N) in [(32768, 1024, 1024), (32768, 1024, 4096),
+                      (32768, 1024, 32000), (8192, 8192, 8192)]:
+        a = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+        b = torch.randn(K, N, device=dev, dtype=torch.bfloat16)
+        flops = 2.0 * M * K * N
+        bench(f"bf16 matmul {M}x{K}x{N}", lambda: a @ b, 20, flops=flops)
+        a8 = a.to(torch.float8_e4m3fn)
+        b8 = b.t().contiguous().t().to(torch.float8_e4m3fn)
+        sa = torch.tensor(1.0, device=dev)
+        sb = torch.tensor(1.0, device=dev)
+        bench(f"fp8 scaled_mm {M}x{K}x{N}",
+              lambda: torch._scaled_mm(a8, b8, scale_a=sa, scale_b=sb,
+                                       out_dtype=torch.bfloat16),
+              20, flops=flops)
+        from marian_amd.ops.fp8 import fp8_linear
+        bench(f"fp8_linear(+quant) {M}x{K}x{N}",
+              lambda: fp8_linear(a, b), 20, flops=flops)
