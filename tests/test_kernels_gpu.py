@@ -573,70 +573,6 @@ def test_transformer_base_T128_fused_grads_match_composite(ext):
         assert err < 0.05 * max(scale, 1e-3) + 1e-3, (n, err, scale)
 
 
-def test_fp16_training_with_cost_scaling(ext):
-    """--precision float16 + dynamic cost scaling trains on GPU."""
-    from marian_amd.common.options import Options
-    from marian_amd.data.synthetic import synthetic_batch
-    from marian_amd.models.model_factory import create_model
-    from marian_amd.training.graph_group import SyncGraphGroup
-    opts = Options({
-        "type": "transformer", "dim-emb": 128, "enc-depth": 1, "dec-depth": 1,
-        "transformer-heads": 2, "transformer-dim-ffn": 256,
-        "transformer-preprocess": "", "transformer-postprocess": "dan",
-        "transformer-postprocess-emb": "d", "transformer-postprocess-top": "",
-        "transformer-ffn-activation": "relu", "transformer-dropout": 0.0,
-        "dim-vocabs": [1000, 1000], "tied-embeddings-all": True,
-        "transformer-train-position-embeddings": False,
-        "transformer-decoder-dim-ffn": 0,
-        "precision": ["float16", "float32"],
-        "cost-scaling": ["8", "100", "2", "1"],
-        "cost-type": "ce-mean-words", "label-smoothing": 0.1,
-        "optimizer": "adam", "learn-rate": 5e-3, "clip-norm": 1.0,
-    })
-    torch.manual_seed(17)
-    model = create_model(opts)
-    gg = SyncGraphGroup(opts, model, torch.device(DEV))
-    assert gg.scaler.enabled and gg.scaler.scale == 256.0
-    gg.set_lr(5e-3)
-    batch = synthetic_batch(16, 10, 10, 1000, device=DEV)
-    batch.sub[-1].indices.copy_(batch.sub[0].indices)
-    first = None
-    for i in range(15):
-        lv = gg.update(batch)
-        ce = float(lv.ce_sum) / lv.label_count
-        first = first if first is not None else ce
-    assert ce < first - 0.2, (first, ce)
-    assert torch.isfinite(gg.optimizer.master).all()
-
-
-def test_transformer_base_converges_T128(ext):
-    """transformer-base (6/6, d512) memorizes a repeated copy batch on GPU
-    at T=128 — exercises the NW=4 multi-wave attention fwd/bwd variants
-    end-to-end through training."""
-    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
-    from marian_amd.data.synthetic import synthetic_batch
-    from marian_amd.models.model_factory import create_model
-    from marian_amd.training.graph_group import SyncGraphGroup
-    argv = ["--train-sets", "a", "b", "--task", "transformer-base",
-            "--dim-vocabs", "500", "500", "--seed", "1",
-            "--learn-rate", "0.001", "--mini-batch-fit", "false",
-            "--max-length", "128",
-            "--precision", "bfloat16", "float32"]
-    opts = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
-    opts.set("dim-vocabs", [500, 500])
-    torch.manual_seed(1)
-    model = create_model(opts)
-    gg = SyncGraphGroup(opts, model, torch.device(DEV))
-    batch = synthetic_batch(16, 128, 128, 500, seed=3, device=DEV)
-    batch.sub[-1].indices.copy_(batch.sub[0].indices)
-    ces = []
-    for i in range(40):
-        gg.set_lr(0.001 * min(1.0, (i + 1) / 20))
-        lv = gg.update(batch)
-        ces.append(float(lv.ce_sum) / lv.label_count)
-    assert ces[-1] < ces[0] - 2.0, (ces[0], ces[-1])
-
-
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
