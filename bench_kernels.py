@@ -112,10 +112,6 @@ def main():
               bytes_moved=R * C * 2 * 2)
 
 
-if __name__ == "__main__":
-    main()
-
-
 def bench_fp8():
     """Is torch._scaled_mm actually hitting fp8 MFMA rate on gfx950?"""
     import torch
@@ -137,3 +133,7 @@ def bench_fp8():
         from marian_amd.ops.fp8 import fp8_linear
         bench(f"fp8_linear(+quant) {M}x{K}x{N}",
               lambda: fp8_linear(a, b), 20, flops=flops)
+
+
+if __name__ == "__main__":
+    main()
