@@ -290,6 +290,8 @@ class ShardedOptimizer:
             for b, (o, s) in zip(self.store.buckets, self.seg):
                 b.shard_view(b.flat_param, self.rank, self.world).copy_(
                     self.master[o:o + s].to(self.store.dtype))
+        from ..ops.fp8 import bump_weight_version
+        bump_weight_version()
 
     # ------------------------------------------------------------------
     # --exponential-smoothing swap (ref: communicator swapParams)
