@@ -573,7 +573,7 @@ def test_transformer_base_T128_fused_grads_match_composite(ext):
         a, b = g_f[n], g_c[n]
         scale = b.abs().max().item()
         err = (a - b).abs().max().item()
-        assert err < 0.12 * max(scale, 1e-3) + 1e-3, (n, err, scale)
+        assert err < 0.12 * max(scale, 1e-3) + 5e-3, (n, err, scale)  # abs floor: K-bias grads are analytically ~0 (softmax shift invariance) -> noise-vs-noise
 
 
 def test_beam_topk_kernel(ext):
