@@ -62,49 +62,36 @@ def _quantize_x(t: torch.Tensor):
 
 
 def _quantize_w(W: torch.Tensor):
-    """Version-cached fp8 weight in BOTH layouts:
-    - column-major [K,M] for the forward (B operand);
-    - row-major [K,M] whose .t() view is the column-major W^T the dgrad
-      GEMM needs."""
+    """Version-cached, column-major fp8 weight."""
     key = id(W)
     hit = _wcache.get(key)
     if hit is not None and hit[0] == _weight_version:
-        return hit[1], hit[2], hit[3]
+        return hit[1], hit[2]
     scale = _scale_of(W)
-    w8_row = (W * scale.reciprocal()).to(torch.float8_e4m3fn).contiguous()
-    w8_col = w8_row.t().contiguous().t()  # column-major for hipBLASLt
-    _wcache[key] = (_weight_version, w8_col, w8_row, scale)
-    return w8_col, w8_row, scale
+    w8 = (W * scale.reciprocal()).to(torch.float8_e4m3fn)
+    w8 = w8.t().contiguous().t()  # column-major for hipBLASLt
+    _wcache[key] = (_weight_version, w8, scale)
+    return w8, scale
 
 
 class Fp8LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, W, b):
         x8, sx = _quantize_x(x)
-        w8_col, w8_row, sw = _quantize_w(W)
-        y = torch._scaled_mm(x8, w8_col, scale_a=sx, scale_b=sw,
+        w8, sw = _quantize_w(W)
+        y = torch._scaled_mm(x8, w8, scale_a=sx, scale_b=sw,
                              bias=b.to(torch.bfloat16) if b is not None else None,
                              out_dtype=torch.bfloat16)
-        # save the fp8 activations for the wgrad GEMM (also halves the
-        # autograd memory vs saving bf16 x)
-        ctx.save_for_backward(x8, sx, W)
+        ctx.save_for_backward(x, W)
         ctx.has_bias = b is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x8, sx, W = ctx.saved_tensors
+        x, W = ctx.saved_tensors
         dyc = dy.contiguous()
-        dy8, sdy = _quantize_x(dyc)
-        _w8_col, w8_row, sw = _quantize_w(W)
-        # dgrad: dy [R,M] @ W^T [M,K]; w8_row.t() IS column-major W^T
-        dx = torch._scaled_mm(dy8, w8_row.t(), scale_a=sdy, scale_b=sw,
-                              out_dtype=torch.bfloat16)
-        # wgrad: x^T [K,R] @ dy [R,M]; fp8 transpose copies are cheap
-        xT8 = x8.t().contiguous()
-        dy8_col = dy8.t().contiguous().t()
-        dW = torch._scaled_mm(xT8, dy8_col, scale_a=sx, scale_b=sdy,
-                              out_dtype=torch.bfloat16)
+        dx = dyc @ W.t()
+        dW = x.t() @ dyc
         db = dyc.sum(0) if ctx.has_bias else None
         return dx, dW, db
 
