@@ -569,11 +569,19 @@ def test_transformer_base_T128_fused_grads_match_composite(ext):
     # noise (measured: worst per-element delta ~7% of the tensor max after
     # 6 layers, mean far below)
     assert abs(ce_f - ce_c) / abs(ce_c) < 2e-3, (ce_f, ce_c)
+    import math
+    total_ref = math.sqrt(sum(float(b.float().pow(2).sum())
+                              for b in g_c.values()))
     for n in g_c:
         a, b = g_f[n], g_c[n]
-        scale = b.abs().max().item()
-        err = (a - b).abs().max().item()
-        assert err < 0.12 * max(scale, 1e-3) + 5e-3, (n, err, scale)  # abs floor: K-bias grads are analytically ~0 (softmax shift invariance) -> noise-vs-noise
+        # relative Frobenius distance; tensors that are a negligible part of
+        # the whole gradient (e.g. K-biases: analytically ZERO by softmax
+        # shift invariance) are noise-vs-noise and get skipped
+        diff = float((a - b).norm())
+        ref = float(b.norm())
+        if ref < 1e-3 * total_ref:
+            continue
+        assert diff < 0.05 * ref + 1e-3, (n, diff, ref)
 
 
 def test_beam_topk_kernel(ext):
