@@ -127,9 +127,87 @@ class BeamSearch:
                                device=device)
         fin_count = torch.zeros(B, dtype=torch.long, device=device)
 
+        # HIP-graph fast path: whole decode step captured once, replayed
+        # per step (single transformer scorer, bf16/Dh=64, no alignment/
+        # sampling/shortlist/prefix)
+        graphed = None
+        if (device.type == "cuda" and len(self.scorers) == 1
+                and self.sampling is None and shortlist is None
+                and prefix is None and not self.collect_alignment):
+            model = getattr(self.scorers[0], "model", None)
+            from ..models.transformer import TransformerModel
+            if isinstance(model, TransformerModel) \
+                    and model.compute_dtype == torch.bfloat16 \
+                    and model.decoder and model.decoder[0].autoreg \
+                    == "self-attention" \
+                    and model.dim // model.decoder[0].self_attn.heads == 64:
+                try:
+                    from .graph_decode import GraphedDecodeStep
+                    sup = -1 if self.allow_unk else self.unk_id
+                    graphed = GraphedDecodeStep(
+                        model, states[0]["enc"], states[0]["mask"],
+                        B * K, max_len, K, sup)
+                    graphed.capture()
+                    if not getattr(BeamSearch, "_graph_logged", False):
+                        from ..common.logging_utils import log_info
+                        log_info("beam search: HIP-graph decode step active")
+                        BeamSearch._graph_logged = True
+                except Exception as e:
+                    if not getattr(BeamSearch, "_graph_warned", False):
+                        from ..common.logging_utils import log_info
+                        log_info("HIP-graph decode unavailable (%s); eager", e)
+                        BeamSearch._graph_warned = True
+                    graphed = None
+
+        flat_sel = (torch.arange(B, device=device)[:, None] * K
+                    + torch.arange(K, device=device)[None, :]).reshape(-1)
         align_steps: List[torch.Tensor] = []
         V = None
         for step in range(max_len):
+            if graphed is not None:
+                V = self.scorers[0].model.trg_vocab_size
+                top_scores, top_idx = graphed.run_step(
+                    step, prev_tokens, flat_sel, scores.view(B * K))
+                beam_idx = top_idx // V
+                word_idx = top_idx % V
+                real_words = word_idx
+                is_eos = (real_words == self.eos_id) \
+                    & (top_scores > NEG_INF / 2)
+                cap = fin_scores.shape[1]
+                eos_rank = torch.cumsum(is_eos.long(), dim=1) - 1
+                slot = fin_count[:, None] + eos_rank
+                valid = is_eos & (slot < cap) & (slot >= 0)
+                slot_c = slot.clamp(0, cap - 1)
+                kidx = torch.arange(K, device=device)[None, :].expand(B, K)
+                fin_scores.scatter_(1, slot_c,
+                                    torch.where(valid, top_scores,
+                                                fin_scores.gather(1, slot_c)))
+                fin_len.scatter_(1, slot_c,
+                                 torch.where(valid,
+                                             torch.full_like(slot_c, step + 1),
+                                             fin_len.gather(1, slot_c)))
+                fin_step.scatter_(1, slot_c,
+                                  torch.where(valid,
+                                              torch.full_like(slot_c, step),
+                                              fin_step.gather(1, slot_c)))
+                fin_beam.scatter_(1, slot_c,
+                                  torch.where(valid, kidx,
+                                              fin_beam.gather(1, slot_c)))
+                fin_count = fin_count + valid.sum(dim=1)
+                scores = torch.where(is_eos,
+                                     torch.full_like(top_scores, NEG_INF),
+                                     top_scores)
+                bp_steps.append(beam_idx)
+                tok_steps.append(real_words)
+                alive = (scores > NEG_INF / 2).reshape(B * K)
+                if (step & 3) == 3 or step == max_len - 1:
+                    done = (fin_count >= self.n_best).all() | ~alive.any()
+                    if bool(done):
+                        break
+                flat_sel = (torch.arange(B, device=device)[:, None] * K
+                            + beam_idx).reshape(-1)
+                prev_tokens = real_words.reshape(B * K)
+                continue
             # ensemble step: weighted sum of log-probs
             logprobs = None
             for w, sc, st in zip(self.weights, self.scorers, states):
