@@ -144,10 +144,19 @@ class BeamSearch:
                 try:
                     from .graph_decode import GraphedDecodeStep
                     sup = -1 if self.allow_unk else self.unk_id
-                    graphed = GraphedDecodeStep(
-                        model, states[0]["enc"], states[0]["mask"],
-                        B * K, max_len, K, sup)
-                    graphed.capture()
+                    Ts = states[0]["enc"].shape[1]
+                    key = (id(model), B * K, Ts, max_len, K, sup)
+                    cache = getattr(self, "_graph_cache", None)
+                    if cache is not None and cache[0] == key:
+                        graphed = cache[1]
+                        graphed.load_batch(states[0]["enc"],
+                                           states[0]["mask"])
+                    else:
+                        graphed = GraphedDecodeStep(
+                            model, states[0]["enc"], states[0]["mask"],
+                            B * K, max_len, K, sup)
+                        graphed.capture()
+                        self._graph_cache = (key, graphed)
                     if not getattr(BeamSearch, "_graph_logged", False):
                         from ..common.logging_utils import log_info
                         log_info("beam search: HIP-graph decode step active")

@@ -52,11 +52,12 @@ class GraphedDecodeStep:
         d = model.dim
         self.L, self.H, self.dh, self.d = L, H, dh, d
 
-        # ---- static inputs -------------------------------------------------
-        self.enc = enc_out.contiguous()
-        self.src_mask = src_mask
-        self.src_lmask = ((1.0 - src_mask[:, None, None, :]) * NEG_INF).to(dt)
-        self.src_lens = src_mask.sum(dim=1).to(torch.int32).contiguous()
+        # ---- static inputs (load_batch refreshes them; graphs reusable
+        # across batches of the same shape) ---------------------------------
+        Ts = enc_out.shape[1]
+        self.Ts = Ts
+        self.src_lmask = torch.zeros(n_rows, 1, 1, Ts, device=dev, dtype=dt)
+        self.src_lens = torch.zeros(n_rows, dtype=torch.int32, device=dev)
         self.prev_tokens = torch.zeros(n_rows, dtype=torch.long, device=dev)
         self.emb_scale = torch.ones(1, device=dev, dtype=dt)  # 0 at step 0
         self.beam_sel = torch.arange(n_rows, dtype=torch.long, device=dev)
@@ -69,23 +70,37 @@ class GraphedDecodeStep:
         # precomputed positional table
         self.pos_table = sinusoidal_position_embeddings(
             max_len + 1, d, device=dev, dtype=dt)
-        # static cross-attention K/V per layer (computed once, eagerly)
-        self.cross_k = []
-        self.cross_v = []
-        with torch.no_grad():
-            for layer in model.decoder:
-                a = layer.context_attn
-                W = torch.cat([a.Wk.W, a.Wv.W], dim=1)
-                bias = torch.cat([a.Wk.b, a.Wv.b])
-                kv = ops.linear(self.enc, W, bias)
-                kh, vh = kv.split(d, dim=-1)
-                self.cross_k.append(a._split(kh).contiguous())
-                self.cross_v.append(a._split(vh).contiguous())
+        # static cross-attention K/V buffers (filled by load_batch)
+        self.cross_k = [torch.zeros(n_rows, H, Ts, dh, device=dev, dtype=dt)
+                        for _ in range(L)]
+        self.cross_v = [torch.zeros(n_rows, H, Ts, dh, device=dev, dtype=dt)
+                        for _ in range(L)]
+        self.load_batch(enc_out, src_mask)
         # outputs
         V = model.trg_vocab_size
         self.out_scores = torch.zeros(self.B, beam_k, device=dev)
         self.out_idx = torch.zeros(self.B, beam_k, dtype=torch.long, device=dev)
         self.graphs = [None, None]
+        self._which = 0
+
+    # ----------------------------------------------------------------------
+    def load_batch(self, enc_out, src_mask):
+        """Refresh the per-batch static inputs (graphs stay valid)."""
+        d = self.d
+        dt = self.dt
+        self.src_lmask.copy_(
+            ((1.0 - src_mask[:, None, None, :]) * NEG_INF).to(dt))
+        self.src_lens.copy_(src_mask.sum(dim=1).to(torch.int32))
+        enc = enc_out.contiguous()
+        with torch.no_grad():
+            for i, layer in enumerate(self.model.decoder):
+                a = layer.context_attn
+                W = torch.cat([a.Wk.W, a.Wv.W], dim=1)
+                bias = torch.cat([a.Wk.b, a.Wv.b])
+                kv = ops.linear(enc, W, bias)
+                kh, vh = kv.split(d, dim=-1)
+                self.cross_k[i].copy_(a._split(kh))
+                self.cross_v[i].copy_(a._split(vh))
         self._which = 0
 
     # ----------------------------------------------------------------------
