@@ -140,7 +140,7 @@ class Train:
         self.scheduler = Scheduler(options, self.state)
 
     # ------------------------------------------------------------------
-    def _batch_generator(self, skip_batches: int = 0):
+    def _batch_generator(self, skip_sentences: int = 0):
         opts = self.opt
         ga = opts.get("guided-alignment", "none")
         corpus = Corpus(opts.list("train-sets"), self.vocabs,
@@ -157,7 +157,7 @@ class Train:
         shuffle = opts.get("shuffle", "data") != "none" \
             and not opts.get("no-shuffle", False)
         gen = BatchGenerator(
-            corpus.iter_epoch(shuffle=shuffle),
+            corpus.iter_epoch(shuffle=shuffle, skip=skip_sentences),
             mini_batch=int(opts.get("mini-batch", 64)),
             mini_batch_words=int(opts.get("mini-batch-words", 0) or 0),
             maxi_batch=int(opts.get("maxi-batch", 100)),
@@ -186,8 +186,17 @@ class Train:
         sched = self.scheduler
         first_metric_validator = self._make_validators()
 
+        restore_skip = 0
+        if self.state.batches > 0 and self.state.samples_epoch > 0 \
+                and not self.opt.get("no-restore-corpus", False):
+            # mid-epoch resume: skip the sentences already consumed
+            # (ref: Corpus restore via TrainingState)
+            restore_skip = self.state.samples_epoch
+            log_info("Restoring corpus position: skipping %d sentences",
+                     restore_skip)
         while sched.keep_going() and not _sigterm_seen:
-            gen = self._batch_generator()
+            gen = self._batch_generator(skip_sentences=restore_skip)
+            restore_skip = 0
             got_batches = False
             for i, batch in enumerate(gen):
                 if self.world > 1 and i % self.world != self.rank:

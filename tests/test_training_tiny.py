@@ -215,3 +215,28 @@ def test_sentencepiece_vocab_training_and_use(tmp_path):
                   "--beam-size", "2", "--cpu-threads", "1",
                   "--max-length", "20", "--quiet"])
     assert out.exists()
+
+
+def test_mid_epoch_corpus_restore(workdir):
+    """Resume mid-epoch skips already-consumed sentences."""
+    tmp_path, src, trg = workdir
+    opts = _opts(tmp_path, src, src, extra=["--after-batches", "7",
+                                            "--shuffle", "none"])
+    tr = Train(opts)
+    tr.run()
+    consumed = tr.state.samples_epoch
+    assert 0 < consumed < 300  # mid-epoch
+
+    opts2 = _opts(tmp_path, src, src, extra=["--after-batches", "9",
+                                             "--shuffle", "none"])
+    tr2 = Train(opts2)
+    seen = []
+    orig = tr2._batch_generator
+
+    def spy(skip_sentences=0):
+        seen.append(skip_sentences)
+        return orig(skip_sentences=skip_sentences)
+    tr2._batch_generator = spy
+    tr2.run()
+    assert seen[0] == consumed  # first epoch resumed with skip
+    assert all(s == 0 for s in seen[1:])
