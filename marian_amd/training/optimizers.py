@@ -255,6 +255,10 @@ class ShardedOptimizer:
         self.smoothing = float(options.get("exponential-smoothing", 0.0) or 0.0)
         self.smooth = (self.master.clone() if self.smoothing > 0 else None)
         self._swapped = None
+        # --quantize-bits: QAT re-quantization of the weight shard after
+        # every update (ref: optimizers/quantizer.{h,cpp} ModelQuantizer)
+        self.quantize_bits = int(options.get("quantize-bits", 0) or 0)
+        self.quantize_log = bool(options.get("quantize-log-based", False))
 
     # ------------------------------------------------------------------
     def grad_shard_for_bucket(self, bi: int) -> torch.Tensor:
@@ -281,6 +285,8 @@ class ShardedOptimizer:
             if mult != 1.0:
                 g = g * mult
             self.core.step(self.master, g, lr, self.t)
+        if self.quantize_bits:
+            self._requantize(self.master)
         if self.smooth is not None:
             # marian caps the effective decay early in training
             decay = min(self.smoothing, (self.t + 1.0) / (self.t + 10.0))
@@ -292,6 +298,23 @@ class ShardedOptimizer:
                     self.master[o:o + s].to(self.store.dtype))
         from ..ops.fp8 import bump_weight_version
         bump_weight_version()
+
+    # ------------------------------------------------------------------
+    def _requantize(self, w: torch.Tensor) -> None:
+        """Snap weights to 2^bits levels (fixed-point, or log-scale with
+        --quantize-log-based), keeping the max-abs scale."""
+        bits = self.quantize_bits
+        amax = w.abs().max().clamp_min(1e-8)
+        if self.quantize_log:
+            sign = w.sign()
+            mag = (w.abs() / amax).clamp_min(1e-12)
+            levels = 2 ** (bits - 1) - 1
+            exp = torch.round(torch.log2(mag)).clamp_(min=-float(levels))
+            w.copy_(sign * amax * torch.pow(2.0, exp) * (mag > 1e-9))
+        else:
+            levels = 2 ** (bits - 1) - 1
+            step = amax / levels
+            w.copy_(torch.round(w / step) * step)
 
     # ------------------------------------------------------------------
     # --exponential-smoothing swap (ref: communicator swapParams)

@@ -33,6 +33,13 @@ class Scheduler:
         self.warmup_start_rate = float(options.get("lr-warmup-start-rate", 0.0))
         inv = options.list("lr-decay-inv-sqrt") or ["0"]
         self.decay_inv_sqrt = SchedulingParameter.parse(inv[0])
+        self.lr_decay = float(options.get("lr-decay", 0.0) or 0.0)
+        self.lr_decay_strategy = options.get("lr-decay-strategy",
+                                             "epoch+stalled")
+        start = options.list("lr-decay-start") or [10, 1]
+        self.lr_decay_start_epoch = int(start[0])
+        self.lr_decay_start_stalled = int(start[1]) if len(start) > 1 else 1
+        self.lr_decay_freq = int(options.get("lr-decay-freq", 50000) or 0)
         self.timer = Timer()
         self._start_batches = state.batches
 
@@ -141,6 +148,26 @@ class Scheduler:
         return actions
 
     # ------------------------------------------------------------------
+    def _maybe_decay(self, why: str):
+        """Apply --lr-decay when the strategy matches (ref: scheduler.cpp
+        updateLearningRate)."""
+        if self.lr_decay <= 0:
+            return
+        strat = self.lr_decay_strategy
+        st = self.state
+        if why == "epoch" and "epoch" in strat \
+                and st.epochs >= self.lr_decay_start_epoch:
+            st.factor *= self.lr_decay
+            log_info("Decaying learning rate to factor %.6f (epoch)", st.factor)
+        if why == "stalled" and "stalled" in strat \
+                and st.stalled >= self.lr_decay_start_stalled:
+            st.factor *= self.lr_decay
+            log_info("Decaying learning rate to factor %.6f (stalled)",
+                     st.factor)
+        if why == "batches" and strat == "batches" and self.lr_decay_freq \
+                and st.batches % self.lr_decay_freq == 0:
+            st.factor *= self.lr_decay
+
     def epoch_actions(self) -> dict:
         """'e'-unit save/valid triggers, checked at epoch boundaries."""
         st = self.state
@@ -152,6 +179,7 @@ class Scheduler:
         return out
 
     def increase_epoch(self):
+        self._maybe_decay("epoch")
         st = self.state
         log_info("Seen %d samples in epoch %d", st.samples_epoch, st.epochs)
         st.epochs += 1
@@ -185,6 +213,7 @@ class Scheduler:
             if metric == self._first_metric():
                 st.stalled += 1
                 st.max_stalled = max(st.max_stalled, st.stalled)
+                self._maybe_decay("stalled")
         return improved
 
     def _first_metric(self):

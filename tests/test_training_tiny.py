@@ -240,3 +240,34 @@ def test_mid_epoch_corpus_restore(workdir):
     tr2.run()
     assert seen[0] == consumed  # first epoch resumed with skip
     assert all(s == 0 for s in seen[1:])
+
+
+def test_quantize_bits(workdir):
+    """--quantize-bits snaps weights to 2^bits levels after each update."""
+    tmp_path, src, trg = workdir
+    opts = _opts(tmp_path, src, src, extra=["--quantize-bits", "4",
+                                            "--after-batches", "3"])
+    tr = Train(opts)
+    tr.run()
+    import numpy as np
+    w = tr.graph_group.optimizer.master.numpy()
+    amax = np.abs(w).max()
+    step = amax / (2 ** 3 - 1)
+    frac = np.abs(w / step - np.round(w / step))
+    assert frac.max() < 1e-4  # all weights on the quantization grid
+
+
+def test_lr_decay_on_stalled():
+    from marian_amd.common.options import Options
+    from marian_amd.training.scheduler import Scheduler
+    from marian_amd.training.training_state import TrainingState
+    opts = Options({"learn-rate": 0.1, "lr-decay": 0.5,
+                    "lr-decay-strategy": "stalled",
+                    "lr-decay-start": [10, 1], "valid-metrics": ["cross-entropy"]})
+    st = TrainingState()
+    sched = Scheduler(opts, st)
+    assert sched.learning_rate() == pytest.approx(0.1)
+    sched.validated("cross-entropy", 5.0)          # best
+    sched.validated("cross-entropy", 6.0)          # stalled -> decay
+    assert st.factor == pytest.approx(0.5)
+    assert sched.learning_rate() == pytest.approx(0.05)
