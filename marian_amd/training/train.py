@@ -143,7 +143,20 @@ class Train:
     def _batch_generator(self, skip_sentences: int = 0):
         opts = self.opt
         ga = opts.get("guided-alignment", "none")
-        corpus = Corpus(opts.list("train-sets"), self.vocabs,
+        sqlite_mode = opts.get("sqlite", "")
+        if sqlite_mode:
+            from ..data.corpus_sqlite import CorpusSQLite
+            corpus = CorpusSQLite(
+                opts.list("train-sets"), self.vocabs,
+                db_path="" if sqlite_mode in ("temporary", "true") else sqlite_mode,
+                drop=bool(opts.get("sqlite-drop", False)),
+                max_length=int(opts.get("max-length", 50)),
+                max_length_crop=bool(opts.get("max-length-crop", False)),
+                tsv=bool(opts.get("tsv", False)),
+                tsv_fields=int(opts.get("tsv-fields", 0) or 0),
+                seed=int(opts.get("seed", 0) or 1234))
+        else:
+            corpus = Corpus(opts.list("train-sets"), self.vocabs,
                         max_length=int(opts.get("max-length", 50)),
                         max_length_crop=bool(opts.get("max-length-crop", False)),
                         tsv=bool(opts.get("tsv", False)),

@@ -271,3 +271,27 @@ def test_lr_decay_on_stalled():
     sched.validated("cross-entropy", 6.0)          # stalled -> decay
     assert st.factor == pytest.approx(0.5)
     assert sched.learning_rate() == pytest.approx(0.05)
+
+
+def test_sqlite_corpus(workdir, tmp_path):
+    tmp_path2, src, trg = workdir
+    from marian_amd.data.corpus_sqlite import CorpusSQLite
+    from marian_amd.data.vocab import Vocab
+    v = Vocab.create_from_corpus([src])
+    db = str(tmp_path2 / "c.sqlite")
+    c = CorpusSQLite([src, src], [v, v], db_path=db, max_length=30)
+    all1 = list(c.iter_epoch(shuffle=False))
+    assert len(all1) == 300
+    assert os.path.exists(db)
+    order1 = [t[0] for t in c.iter_epoch(shuffle=True)]
+    c2 = CorpusSQLite([src, src], [v, v], db_path=db, max_length=30, seed=1234)
+    c2.epoch = 1
+    order2 = [t[0] for t in c2.iter_epoch(shuffle=True)]
+    assert order1 == order2          # seeded epoch shuffle reproducible
+    assert order1 != sorted(order1)  # and actually shuffled
+    # end-to-end training over sqlite
+    opts = _opts(tmp_path2, src, src, extra=["--sqlite", db,
+                                             "--after-batches", "5"])
+    tr = Train(opts)
+    tr.run()
+    assert tr.state.batches == 5
