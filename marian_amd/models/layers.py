@@ -149,7 +149,11 @@ class PrePostProcessor(nn.Module):
 class MultiHeadAttention(nn.Module):
     """Multi-head attention with marian parameter names Wq/bq..Wo/bo
     (ref: transformer.h LayerAttention/MultiHead). The core SDPA runs
-    through ops.attention -> fused CDNA4 kernel on device."""
+    through ops.attention -> fused CDNA4 kernel on device.
+
+    Q/K/V projections are stored as ONE fused [d, 3d] parameter (one GEMM,
+    no per-forward weight concat); marian's per-matrix names are exported
+    as views (see _name_map in transformer.py)."""
 
     def __init__(self, dim: int, heads: int, dropout: float = 0.0):
         super().__init__()
@@ -157,11 +161,19 @@ class MultiHeadAttention(nn.Module):
         self.dim = dim
         self.heads = heads
         self.dh = dim // heads
-        self.Wq = Dense(dim, dim)
-        self.Wk = Dense(dim, dim)
-        self.Wv = Dense(dim, dim)
+        self.W_qkv = nn.Parameter(torch.empty(dim, 3 * dim))
+        self.b_qkv = nn.Parameter(torch.zeros(3 * dim))
+        for i in range(3):
+            nn.init.xavier_uniform_(self.W_qkv[:, i * dim:(i + 1) * dim])
         self.Wo = Dense(dim, dim)
         self.dropout = dropout
+
+    # marian-named views of the fused parameter
+    def w_slice(self, i):
+        return self.W_qkv[:, i * self.dim:(i + 1) * self.dim]
+
+    def b_slice(self, i):
+        return self.b_qkv[i * self.dim:(i + 1) * self.dim]
 
     def _split(self, x):
         B, T, _ = x.shape
@@ -181,30 +193,30 @@ class MultiHeadAttention(nn.Module):
         dt = query.dtype
         fuse_qkv = keys is query and values is query and cache is None
         if fuse_qkv:
-            # one [d, 3d] GEMM for Q/K/V (the cat is 6 MB vs three GEMM
-            # launches); split()s are views the fused kernel takes strided
-            W = torch.cat([self.Wq.W, self.Wk.W, self.Wv.W], dim=1)
-            bias = torch.cat([self.Wq.b, self.Wk.b, self.Wv.b])
-            qkv = ops.linear(query, W, bias)
+            # one [d, 3d] GEMM for Q/K/V on the persistent fused parameter;
+            # split()s are views the fused attention kernel takes strided
+            qkv = ops.linear(query, self.W_qkv, self.b_qkv)
             qh, kh, vh = qkv.split(self.dim, dim=-1)
             q = self._split(qh)
             k = self._split(kh)
             v = self._split(vh)
         else:
-            q = self._split(self.Wq(query))
+            q = self._split(ops.linear(query, self.w_slice(0),
+                                       self.b_slice(0)))
         if cache is not None and cache_key + "_static" in cache:
             k, v = cache[cache_key + "_k"], cache[cache_key + "_v"]
         elif not fuse_qkv:
             if keys is values:
-                W = torch.cat([self.Wk.W, self.Wv.W], dim=1)
-                bias = torch.cat([self.Wk.b, self.Wv.b])
-                kv = ops.linear(keys, W, bias)
+                kv = ops.linear(keys, self.W_qkv[:, self.dim:],
+                                self.b_qkv[self.dim:])
                 kh, vh = kv.split(self.dim, dim=-1)
                 k = self._split(kh)
                 v = self._split(vh)
             else:
-                k = self._split(self.Wk(keys))
-                v = self._split(self.Wv(values))
+                k = self._split(ops.linear(keys, self.w_slice(1),
+                                           self.b_slice(1)))
+                v = self._split(ops.linear(values, self.w_slice(2),
+                                           self.b_slice(2)))
             if cache is not None:
                 if cache_key.endswith("context"):   # cross-attn: static
                     cache[cache_key + "_k"] = k

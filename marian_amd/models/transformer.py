@@ -335,12 +335,13 @@ class TransformerModel(nn.Module):
             m["decoder_ff_logit_out_b"] = self.out_b
 
         def attn(prefix: str, a: MultiHeadAttention, post: PrePostProcessor):
-            m[prefix + "_Wq"] = a.Wq.W
-            m[prefix + "_bq"] = a.Wq.b
-            m[prefix + "_Wk"] = a.Wk.W
-            m[prefix + "_bk"] = a.Wk.b
-            m[prefix + "_Wv"] = a.Wv.W
-            m[prefix + "_bv"] = a.Wv.b
+            # marian names are VIEWS into the fused [d,3d] QKV parameter
+            m[prefix + "_Wq"] = a.w_slice(0)
+            m[prefix + "_bq"] = a.b_slice(0)
+            m[prefix + "_Wk"] = a.w_slice(1)
+            m[prefix + "_bk"] = a.b_slice(1)
+            m[prefix + "_Wv"] = a.w_slice(2)
+            m[prefix + "_bv"] = a.b_slice(2)
             m[prefix + "_Wo"] = a.Wo.W
             m[prefix + "_bo"] = a.Wo.b
             if hasattr(post, "ln"):

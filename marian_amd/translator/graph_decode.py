@@ -95,9 +95,7 @@ class GraphedDecodeStep:
         with torch.no_grad():
             for i, layer in enumerate(self.model.decoder):
                 a = layer.context_attn
-                W = torch.cat([a.Wk.W, a.Wv.W], dim=1)
-                bias = torch.cat([a.Wk.b, a.Wv.b])
-                kv = ops.linear(enc, W, bias)
+                kv = ops.linear(enc, a.W_qkv[:, a.dim:], a.b_qkv[a.dim:])
                 kh, vh = kv.split(d, dim=-1)
                 self.cross_k[i].copy_(a._split(kh))
                 self.cross_v[i].copy_(a._split(vh))
@@ -121,9 +119,7 @@ class GraphedDecodeStep:
         for i, layer in enumerate(m.decoder):
             h = layer.pre_self(x)
             a = layer.self_attn
-            W = torch.cat([a.Wq.W, a.Wk.W, a.Wv.W], dim=1)
-            bias = torch.cat([a.Wq.b, a.Wk.b, a.Wv.b])
-            qkv = ops.linear(h, W, bias)
+            qkv = ops.linear(h, a.W_qkv, a.b_qkv)
             qh, kh, vh = qkv.split(d, dim=-1)
             q = a._split(qh)           # [N, H, 1, dh]
             # write this step's K/V at position `step` (dim 2 of [N,H,T,dh])
@@ -135,7 +131,7 @@ class GraphedDecodeStep:
             x = layer.post_self(h, residual=x)
             h = layer.pre_ctx(x)
             ca = layer.context_attn
-            q2 = ca._split(ca.Wq(h))
+            q2 = ca._split(ops.linear(h, ca.w_slice(0), ca.b_slice(0)))
             att2 = ops.attention(q2, self.cross_k[i], self.cross_v[i],
                                  mask=self.src_lmask, lens=self.src_lens)
             h = ca.Wo(ca._join(att2))

@@ -177,4 +177,4 @@ def test_tied_layers():
     logits = m(b)
     logits.float().pow(2).mean().backward()
     # shared params got gradients once
-    assert m.encoder[0].self_attn.Wq.W.grad is not None
+    assert m.encoder[0].self_attn.W_qkv.grad is not None
