@@ -27,6 +27,7 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                                          c10::optional<torch::Tensor> kv_lens,
                                          bool causal, double scale,
                                          bool need_lse);
+torch::Tensor col_sum(torch::Tensor x);
 std::vector<torch::Tensor> beam_topk(torch::Tensor logprobs,
                                      torch::Tensor scores, int64_t n_beams,
                                      int64_t k, int64_t suppress);
@@ -52,6 +53,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused flash-style attention forward (bf16, Dh=64)");
   m.def("attention_bwd", &attention_bwd,
         "fused flash-style attention backward (bf16, Dh=64)");
+  m.def("col_sum", &col_sum, "column sum (bias gradients)");
   m.def("beam_topk", &beam_topk,
         "fused beam-search top-k (path-score add + UNK suppression)");
 }

@@ -28,6 +28,9 @@ if os.path.exists(attn):
 topk = os.path.join(HIP_DIR, "topk.hip")
 if os.path.exists(topk):
     sources.append(topk)
+colsum = os.path.join(HIP_DIR, "colsum.hip")
+if os.path.exists(colsum):
+    sources.append(colsum)
 
 ext = CUDAExtension(
     name="marian_amd.ops._hip_ext",

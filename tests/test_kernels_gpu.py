@@ -584,6 +584,35 @@ def test_transformer_base_T128_fused_grads_match_composite(ext):
         assert diff < 0.08 * ref + 1e-3, (n, diff, ref)
 
 
+def test_col_sum_numerics():
+    from marian_amd.ops import hip_ext
+    ext = hip_ext(required=True)
+    for rows, cols in [(65536, 4096), (4096, 1024), (333, 1000), (7, 5)]:
+        x = torch.randn(rows, cols, device="cuda", dtype=torch.bfloat16)
+        got = ext.col_sum(x)
+        ref = x.float().sum(dim=0)
+        assert torch.allclose(got, ref, rtol=1e-2, atol=1e-1), (rows, cols)
+
+
+def test_linear_bias_grad():
+    from marian_amd import ops
+    x = torch.randn(512, 256, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    W = torch.randn(256, 128, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.randn(128, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ops.linear(x, W, b)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    xr = x.detach().float().requires_grad_(True)
+    Wr = W.detach().float().requires_grad_(True)
+    br = b.detach().float().requires_grad_(True)
+    (xr @ Wr + br).backward(dy.float())
+    for got, ref in [(x.grad, xr.grad), (W.grad, Wr.grad), (b.grad, br.grad)]:
+        assert torch.allclose(got.float(), ref, rtol=5e-2, atol=5e-1)
+
+
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
