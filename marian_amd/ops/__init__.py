@@ -94,7 +94,7 @@ def linear(x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor] = None
         return fp8_linear(x, W.to(x.dtype), b.to(x.dtype) if b is not None else None)
     W = W.to(x.dtype)
     shp = x.shape[:-1] + (W.shape[1],)
-    if _on_gpu(x):
+    if _on_gpu(x) and "MAMD_NO_CUSTOM_LINEAR" not in os.environ:
         x2 = x.reshape(-1, W.shape[0])
         return _LinearFn.apply(x2, W.contiguous() if not W.is_contiguous()
                                and W.stride(0) != 1 else W,
