@@ -237,6 +237,16 @@ def topk(scores: torch.Tensor, k: int):
 # Fused Adam step (ref: the Element(...) Adam functor + castOptimizerType)
 # ---------------------------------------------------------------------------
 
+def grad_sumsq_finite(g: torch.Tensor) -> torch.Tensor:
+    """Returns [2] fp32: (sum of squares, count of non-finite values) in
+    ONE pass over the flat gradient (no fp32 temp, no g*g temp)."""
+    if _on_gpu(g):
+        return hip_ext(required=True).sumsq_finite(g.contiguous())
+    gf = g.float()
+    return torch.stack([(gf * gf).sum(),
+                        (~torch.isfinite(gf)).sum().float()])
+
+
 def adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd, step,
               grad_scale: float = 1.0):
     """One fused Adam update; grad is multiplied by grad_scale first

@@ -28,6 +28,7 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                                          bool causal, double scale,
                                          bool need_lse);
 torch::Tensor col_sum(torch::Tensor x);
+torch::Tensor sumsq_finite(torch::Tensor g);
 std::vector<torch::Tensor> beam_topk(torch::Tensor logprobs,
                                      torch::Tensor scores, int64_t n_beams,
                                      int64_t k, int64_t suppress);
@@ -54,6 +55,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_bwd", &attention_bwd,
         "fused flash-style attention backward (bf16, Dh=64)");
   m.def("col_sum", &col_sum, "column sum (bias gradients)");
+  m.def("sumsq_finite", &sumsq_finite, "fused grad sumsq + nonfinite count");
   m.def("beam_topk", &beam_topk,
         "fused beam-search top-k (path-score add + UNK suppression)");
 }

@@ -1,8 +1,16 @@
-// Column sum for bias gradients (db = sum over rows of dY) on CDNA4.
-// torch's bf16 column reduce runs ~340us on [65536, 4096]; this stripe-
-// parallel kernel with vectorized loads + fp32 stripe atomics targets HBM
-// bandwidth. (ref semantics: the reference's bias gradients come from
-// gAggregate row reductions, SURVEY §3.3 #2.)
+// Column sum for bias gradients (db = sum over rows of dY) and a fused
+// single-pass gradient sum-of-squares + finiteness check, on CDNA4.
+//
+// col_sum: torch's bf16 column reduce runs ~340us on [65536, 4096]; the
+// first version here used per-stripe fp32 atomics and collapsed to
+// 0.6 TB/s on narrow (cols=1024) outputs from 2048-way address
+// contention. This version writes per-stripe partials to scratch
+// (atomics-free) and reduces them in a second tiny kernel.
+//
+// sumsq_finite: replaces `isfinite(g.float().sum())` + `(g*g).sum()`
+// (which cost two 840 MB fp32 temps + multi-pass reduces per step) with
+// ONE read of the flat gradient buffer.
+// (ref semantics: gAggregate row reductions / gClip)
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -13,17 +21,16 @@ using namespace mamd;
 namespace {
 
 template <typename T>
-__global__ void col_sum_kernel(const T* __restrict__ x,
-                               float* __restrict__ out,
-                               long rows, long cols, int stripes) {
+__global__ void col_sum_part_kernel(const T* __restrict__ x,
+                                    float* __restrict__ part,
+                                    long rows, long cols, int stripes) {
   const long c0 = (long)blockIdx.x * blockDim.x * 4 + threadIdx.x * 4;
   if (c0 >= cols) return;
   const long per = (rows + stripes - 1) / stripes;
   const long r0 = (long)blockIdx.y * per;
   const long r1 = min(r0 + per, rows);
   float acc[4] = {0.f, 0.f, 0.f, 0.f};
-  const bool vec = (cols % 4 == 0) && (c0 + 4 <= cols);
-  if (vec) {
+  if ((cols % 4 == 0) && (c0 + 4 <= cols)) {
     for (long r = r0; r < r1; ++r) {
       float v4[4];
       load4_f32(x + r * cols + c0, v4);
@@ -36,14 +43,69 @@ __global__ void col_sum_kernel(const T* __restrict__ x,
       for (int j = 0; j < 4; ++j)
         if (c0 + j < cols) acc[j] += to_f32(x[r * cols + c0 + j]);
   }
+  float* p = part + (long)blockIdx.y * cols + c0;
 #pragma unroll
   for (int j = 0; j < 4; ++j)
-    if (c0 + j < cols) {
-      if (gridDim.y == 1)
-        out[c0 + j] = acc[j];
-      else
-        atomicAdd(&out[c0 + j], acc[j]);
+    if (c0 + j < cols) p[j] = acc[j];
+}
+
+// part [stripes, cols] -> out [cols]; gridDim.y stripe-chunks, few-way atomics
+__global__ void col_sum_fin_kernel(const float* __restrict__ part,
+                                   float* __restrict__ out,
+                                   long cols, int stripes) {
+  const long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= cols) return;
+  const int chunks = gridDim.y;
+  const int per = (stripes + chunks - 1) / chunks;
+  const int s0 = blockIdx.y * per;
+  const int s1 = min(s0 + per, stripes);
+  float acc = 0.f;
+  for (int s = s0; s < s1; ++s) acc += part[(long)s * cols + c];
+  if (chunks == 1) out[c] = acc;
+  else atomicAdd(&out[c], acc);
+}
+
+template <typename T>
+__global__ void sumsq_finite_kernel(const T* __restrict__ g, long n,
+                                    float* __restrict__ out) {
+  // out[0] += sum(g*g), out[1] += count of non-finite values
+  float ss = 0.f, bad = 0.f;
+  const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  const long stride = (long)gridDim.x * blockDim.x * 4;
+  for (long i = i0; i < n; i += stride) {
+    if (i + 4 <= n && (n % 4 == 0 || i + 4 <= (n & ~3L))) {
+      float v4[4];
+      load4_f32(g + i, v4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        ss += v4[j] * v4[j];
+        if (!isfinite(v4[j])) bad += 1.f;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        if (i + j < n) {
+          float v = to_f32(g[i + j]);
+          ss += v * v;
+          if (!isfinite(v)) bad += 1.f;
+        }
     }
+  }
+  __shared__ float red[2][8];
+  ss = wave_sum(ss);
+  bad = wave_sum(bad);
+  const int wid = threadIdx.x / 64, lid = threadIdx.x % 64;
+  if (lid == 0) { red[0][wid] = ss; red[1][wid] = bad; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t0 = 0.f, t1 = 0.f;
+    for (int w = 0; w < (int)(blockDim.x / 64); ++w) {
+      t0 += red[0][w];
+      t1 += red[1][w];
+    }
+    atomicAdd(&out[0], t0);
+    atomicAdd(&out[1], t1);
+  }
 }
 
 }  // namespace
@@ -52,22 +114,48 @@ torch::Tensor col_sum(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2);
   auto xc = x.contiguous();
   const long rows = xc.size(0), cols = xc.size(1);
-  auto out = torch::zeros({cols}, xc.options().dtype(torch::kFloat32));
+  auto out = torch::empty({cols}, xc.options().dtype(torch::kFloat32));
   const int block = 256;
   const long col_blocks = (cols + block * 4 - 1) / (block * 4);
-  // enough stripes to fill the chip, few enough that atomics stay cheap
   int stripes = (int)std::min<long>(rows,
       std::max<long>(1, 2048 / std::max<long>(col_blocks, 1)));
+  auto part = torch::empty({stripes, cols},
+                           xc.options().dtype(torch::kFloat32));
   auto stream = at::cuda::getCurrentHIPStream();
-#define RUN(T)                                                             \
-  hipLaunchKernelGGL((col_sum_kernel<T>), dim3(col_blocks, stripes),       \
-                     dim3(block), 0, stream,                               \
-                     reinterpret_cast<const T*>(xc.data_ptr()),            \
-                     out.data_ptr<float>(), rows, cols, stripes)
+#define RUN(T)                                                              \
+  hipLaunchKernelGGL((col_sum_part_kernel<T>), dim3(col_blocks, stripes),   \
+                     dim3(block), 0, stream,                                \
+                     reinterpret_cast<const T*>(xc.data_ptr()),             \
+                     part.data_ptr<float>(), rows, cols, stripes)
   if (xc.scalar_type() == torch::kBFloat16) RUN(__hip_bfloat16);
   else if (xc.scalar_type() == torch::kHalf) RUN(__half);
   else if (xc.scalar_type() == torch::kFloat32) RUN(float);
   else TORCH_CHECK(false, "col_sum: unsupported dtype");
+#undef RUN
+  const int chunks = stripes >= 64 ? 8 : 1;
+  if (chunks > 1) out.zero_();
+  hipLaunchKernelGGL(col_sum_fin_kernel,
+                     dim3((cols + block - 1) / block, chunks), dim3(block),
+                     0, stream, part.data_ptr<float>(),
+                     out.data_ptr<float>(), cols, stripes);
+  return out;
+}
+
+torch::Tensor sumsq_finite(torch::Tensor g) {
+  TORCH_CHECK(g.is_cuda() && g.is_contiguous());
+  const long n = g.numel();
+  auto out = torch::zeros({2}, g.options().dtype(torch::kFloat32));
+  const int block = 256;
+  const int grid = (int)std::min<long>(2048, (n + block * 4 - 1) / (block * 4));
+  auto stream = at::cuda::getCurrentHIPStream();
+#define RUN(T)                                                              \
+  hipLaunchKernelGGL((sumsq_finite_kernel<T>), dim3(grid), dim3(block), 0,  \
+                     stream, reinterpret_cast<const T*>(g.data_ptr()), n,   \
+                     out.data_ptr<float>())
+  if (g.scalar_type() == torch::kBFloat16) RUN(__hip_bfloat16);
+  else if (g.scalar_type() == torch::kHalf) RUN(__half);
+  else if (g.scalar_type() == torch::kFloat32) RUN(float);
+  else TORCH_CHECK(false, "sumsq_finite: unsupported dtype");
 #undef RUN
   return out;
 }

@@ -220,20 +220,18 @@ class SyncGraphGroup:
         grad_scale *= self.scaler.scale
 
         skip = False
-        if self.check_grad_nan:
-            finite = torch.isfinite(opt.shard_grad.float().sum())
+        stats = None
+        if self.check_grad_nan or self.clip_norm > 0:
+            stats = opt.grad_stats().clone()  # [sumsq, nonfinite] one pass
             if self.world > 1:
-                f = finite.float()
-                self.comm.all_reduce_scalar(f)
-                finite = f >= self.world
-            if not bool(finite.item()):
+                self.comm.all_reduce_scalar(stats)
+        if self.check_grad_nan and stats is not None:
+            if float(stats[1].item()) > 0:
                 skip = True
 
         grad_norm = None
         if not skip and self.clip_norm > 0:
-            sq = opt.local_grad_sq_norm() / (grad_scale * grad_scale)
-            if self.world > 1:
-                self.comm.all_reduce_scalar(sq)
+            sq = stats[0] / (grad_scale * grad_scale)
             grad_norm = float(sq.sqrt().item())
             self._last_gnorm = grad_norm
 

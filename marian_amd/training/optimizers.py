@@ -115,9 +115,10 @@ class ParameterStore:
 
     def grad_finite(self) -> torch.Tensor:
         """1.0 if every gradient is finite (for dynamic cost scaling)."""
+        from .. import ops
         ok = torch.ones((), device=self.device, dtype=torch.float32)
         for b in self.buckets:
-            ok = ok * torch.isfinite(b.flat_grad.float().sum()).float()
+            ok = ok * (ops.grad_sumsq_finite(b.flat_grad)[1] == 0).float()
         return ok
 
     def param_buffers(self) -> List[torch.Tensor]:
@@ -257,6 +258,7 @@ class ShardedOptimizer:
         self._swapped = None
         # --quantize-bits: QAT re-quantization of the weight shard after
         # every update (ref: optimizers/quantizer.{h,cpp} ModelQuantizer)
+        self._grad_stats = None
         self.quantize_bits = int(options.get("quantize-bits", 0) or 0)
         self.quantize_log = bool(options.get("quantize-log-based", False))
 
@@ -266,11 +268,19 @@ class ShardedOptimizer:
         return self.shard_grad[o:o + s]
 
     def local_grad_sq_norm(self) -> torch.Tensor:
-        g = self.shard_grad.float()
-        return (g * g).sum()
+        return self.grad_stats()[0]
+
+    def grad_stats(self) -> torch.Tensor:
+        """[sumsq, nonfinite-count] of the local grad shard — ONE fused
+        pass, shared by the nan check and clip-norm (cleared each step)."""
+        if self._grad_stats is None:
+            from .. import ops
+            self._grad_stats = ops.grad_sumsq_finite(self.shard_grad)
+        return self._grad_stats
 
     def step(self, lr: float, grad_scale: float = 1.0,
              grad_norm: Optional[float] = None) -> None:
+        self._grad_stats = None
         self.t += 1
         mult = 1.0 / grad_scale
         if self.clip_norm > 0 and grad_norm is not None \

@@ -613,6 +613,23 @@ def test_linear_bias_grad():
         assert torch.allclose(got.float(), ref, rtol=5e-2, atol=5e-1)
 
 
+def test_sumsq_finite():
+    from marian_amd import ops
+    g = torch.randn(1_000_003, device="cuda", dtype=torch.bfloat16)
+    out = ops.grad_sumsq_finite(g)
+    ref = (g.float() ** 2).sum()
+    assert torch.allclose(out[0], ref, rtol=1e-2)
+    assert out[1].item() == 0
+    g[1234] = float("inf")
+    g[999_999] = float("nan")
+    out = ops.grad_sumsq_finite(g)
+    assert out[1].item() == 2
+    # fp32 path
+    g32 = torch.randn(4097, device="cuda")
+    out = ops.grad_sumsq_finite(g32)
+    assert torch.allclose(out[0], (g32 ** 2).sum(), rtol=1e-4)
+
+
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
