@@ -193,9 +193,16 @@ class MultiHeadAttention(nn.Module):
         dt = query.dtype
         fuse_qkv = keys is query and values is query and cache is None
         if fuse_qkv:
-            # one [d, 3d] GEMM for Q/K/V on the persistent fused parameter;
-            # split()s are views the fused attention kernel takes strided
+            # one [d, 3d] GEMM for Q/K/V on the persistent fused parameter
             qkv = ops.linear(query, self.W_qkv, self.b_qkv)
+            drop0 = self.dropout if self.training else 0.0
+            if collect is None:
+                out = ops.attention_qkv(qkv, self.heads, mask=mask,
+                                        causal=causal, dropout_p=drop0,
+                                        lens=lens)
+                if out is not None:
+                    return self.Wo(self._join(out))
+            # composite fallback: split()s are strided views
             qh, kh, vh = qkv.split(self.dim, dim=-1)
             q = self._split(qh)
             k = self._split(kh)

@@ -157,6 +157,18 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return attention_torch(q, k, v, mask, causal, dropout_p)
 
 
+def attention_qkv(qkv: torch.Tensor, heads: int, mask=None, causal=False,
+                  dropout_p: float = 0.0,
+                  lens: Optional[torch.Tensor] = None):
+    """Self-attention on the fused [B,T,3d] QKV projection (one buffer in,
+    one buffer out in both directions). Returns [B,H,T,dh] or None for
+    the composite fallback."""
+    if dropout_p != 0.0 or not _on_gpu(qkv):
+        return None
+    from ._attention import fused_qkv_attention
+    return fused_qkv_attention(qkv, heads, mask, causal, lens=lens)
+
+
 def attention_torch(q, k, v, mask=None, causal=False, dropout_p=0.0):
     scale = q.shape[-1] ** -0.5
     scores = torch.matmul(q, k.transpose(-1, -2)) * scale  # [B,H,Tq,Tk]

@@ -87,3 +87,76 @@ def fused_attention(q, k, v, mask, causal,
     scale = D ** -0.5
     o, _lse = ext.attention_fwd(q, k, v, lens, causal, scale, False)
     return o
+
+
+def _qkv_views(qkv: torch.Tensor, H: int):
+    """[B, T, 3d] fused projection -> q/k/v [B,H,T,dh] strided views
+    (zero-copy; the kernels read/write through TStride)."""
+    B, T, d3 = qkv.shape
+    d = d3 // 3
+    dh = d // H
+    def view(off):
+        return qkv.as_strided((B, H, T, dh), (T * d3, dh, d3, 1), off)
+    return view(0), view(d), view(2 * d)
+
+
+class FusedQKVAttentionFn(torch.autograd.Function):
+    """Self-attention on the fused [B,T,3d] QKV projection. The backward
+    writes dq/dk/dv directly into ONE [B,T,3d] buffer through strided
+    views — no CatArrayBatchedCopy from split()'s backward (was ~4.7
+    ms/step at B1024)."""
+
+    @staticmethod
+    def forward(ctx, qkv, H, lens, causal):
+        ext = hip_ext(required=True)
+        q, k, v = _qkv_views(qkv, H)
+        scale = q.shape[-1] ** -0.5
+        o, lse = ext.attention_fwd(q, k, v, lens, causal, scale, True)
+        ctx.save_for_backward(qkv, o, lse,
+                              lens if lens is not None else torch.Tensor())
+        ctx.H = H
+        ctx.causal = causal
+        ctx.scale = scale
+        ctx.has_lens = lens is not None
+        return o
+
+    @staticmethod
+    def backward(ctx, d_o):
+        ext = hip_ext(required=True)
+        qkv, o, lse, lens = ctx.saved_tensors
+        q, k, v = _qkv_views(qkv, ctx.H)
+        dqkv = torch.empty_like(qkv)
+        dq, dk, dv = _qkv_views(dqkv, ctx.H)
+        ext.attention_bwd_into(q, k, v, o, d_o, lse,
+                               lens if ctx.has_lens else None,
+                               ctx.causal, ctx.scale, dq, dk, dv)
+        return dqkv, None, None, None
+
+
+def fused_qkv_attention(qkv, H: int, mask, causal,
+                        lens: Optional[torch.Tensor] = None):
+    """Fused self-attention straight off the [B,T,3d] QKV projection.
+    Returns [B,H,T,dh] (BTHD storage, so _join is a free view), or None
+    when the fused kernel does not cover the case."""
+    if _DISABLED or not qkv.is_cuda or qkv.dtype != torch.bfloat16:
+        return None
+    ext = hip_ext(required=True)
+    if not hasattr(ext, "attention_bwd_into"):
+        return None
+    B, T, d3 = qkv.shape
+    d = d3 // 3
+    if d % H or (d // H) != 64:
+        return None
+    if lens is None:
+        lens = mask_to_lens(mask, B, T)
+        if lens is None:
+            return None
+        lens = lens if lens.numel() else None
+    elif lens.numel() == 0:
+        lens = None
+    if torch.is_grad_enabled() and qkv.requires_grad:
+        return FusedQKVAttentionFn.apply(qkv, H, lens, causal)
+    q, k, v = _qkv_views(qkv, H)
+    o, _ = hip_ext(required=True).attention_fwd(q, k, v, lens, causal,
+                                                (d // H) ** -0.5, False)
+    return o

@@ -757,11 +757,11 @@ __global__ void attn_dsum_kernel(const __hip_bfloat16* __restrict__ dO,
 
 // returns (dq, dk, dv) - all bf16, allocated in [B,T,H,D] storage and
 // returned as [B,H,T,D] views (caller reshapes for free)
-std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
-                                         torch::Tensor v, torch::Tensor o,
-                                         torch::Tensor d_o, torch::Tensor lse,
-                                         c10::optional<torch::Tensor> kv_lens,
-                                         bool causal, double scale) {
+std::vector<torch::Tensor> attention_bwd_into(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor o,
+    torch::Tensor d_o, torch::Tensor lse, c10::optional<torch::Tensor> kv_lens,
+    bool causal, double scale,
+    torch::Tensor dq, torch::Tensor dk, torch::Tensor dv) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   auto qc = view_ok(q) ? q : q.contiguous();
   auto kc = view_ok(k) ? k : k.contiguous();
@@ -775,9 +775,9 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
   const int B = qc.size(0), H = qc.size(1), Tq = qc.size(2), D = qc.size(3);
   const int Tk = kc.size(2);
   TORCH_CHECK(D == 64, "attention_bwd: only Dh=64 supported");
-  auto dq = torch::empty({B, Tq, H, (int)D}, qc.options()).permute({0, 2, 1, 3});
-  auto dk = torch::empty({B, Tk, H, (int)D}, qc.options()).permute({0, 2, 1, 3});
-  auto dv = torch::empty({B, Tk, H, (int)D}, qc.options()).permute({0, 2, 1, 3});
+  TORCH_CHECK(view_ok(dq) && view_ok(dk) && view_ok(dv) &&
+              dk.strides() == dv.strides(),
+              "attention_bwd: grad outputs must be 8-aligned strided views");
   // D = rowsum(dO o O) per query row, fp32 — single fused pass
   auto dsum = torch::empty({B, H, Tq},
                            qc.options().dtype(torch::kFloat32));
@@ -853,4 +853,19 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
 #undef RUNB2
 #undef RUNB1
   return {dq, dk, dv};
+}
+
+// returns (dq, dk, dv) allocated in [B,T,H,D] storage, [B,H,T,D] views
+std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
+                                         torch::Tensor v, torch::Tensor o,
+                                         torch::Tensor d_o, torch::Tensor lse,
+                                         c10::optional<torch::Tensor> kv_lens,
+                                         bool causal, double scale) {
+  const int B = q.size(0), H = q.size(1), Tq = q.size(2), D = q.size(3);
+  const int Tk = k.size(2);
+  auto dq = torch::empty({B, Tq, H, D}, q.options()).permute({0, 2, 1, 3});
+  auto dk = torch::empty({B, Tk, H, D}, q.options()).permute({0, 2, 1, 3});
+  auto dv = torch::empty({B, Tk, H, D}, q.options()).permute({0, 2, 1, 3});
+  return attention_bwd_into(q, k, v, o, d_o, lse, kv_lens, causal, scale,
+                            dq, dk, dv);
 }

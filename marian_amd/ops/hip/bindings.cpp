@@ -32,6 +32,11 @@ torch::Tensor sumsq_finite(torch::Tensor g);
 std::vector<torch::Tensor> beam_topk(torch::Tensor logprobs,
                                      torch::Tensor scores, int64_t n_beams,
                                      int64_t k, int64_t suppress);
+std::vector<torch::Tensor> attention_bwd_into(
+    torch::Tensor q, torch::Tensor k, torch::Tensor v, torch::Tensor o,
+    torch::Tensor d_o, torch::Tensor lse, c10::optional<torch::Tensor> kv_lens,
+    bool causal, double scale,
+    torch::Tensor dq, torch::Tensor dk, torch::Tensor dv);
 std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v, torch::Tensor o,
                                          torch::Tensor d_o, torch::Tensor lse,
@@ -52,6 +57,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step", &adam_step, "fused Adam step");
   m.def("attention_fwd", &attention_fwd,
         "fused flash-style attention forward (bf16, Dh=64)");
+  m.def("attention_bwd_into", &attention_bwd_into,
+        "fused attention backward writing into given strided views");
   m.def("attention_bwd", &attention_bwd,
         "fused flash-style attention backward (bf16, Dh=64)");
   m.def("col_sum", &col_sum, "column sum (bias gradients)");

@@ -630,6 +630,39 @@ def test_sumsq_finite():
     assert torch.allclose(out[0], (g32 ** 2).sum(), rtol=1e-4)
 
 
+def test_fused_qkv_attention():
+    """FusedQKVAttentionFn (strided in/out on one buffer) vs fp32 SDPA."""
+    from marian_amd.ops._attention import fused_qkv_attention
+    torch.manual_seed(5)
+    B, T, H, dh = 8, 48, 16, 64
+    d = H * dh
+    lens = torch.randint(8, T + 1, (B,), device="cuda", dtype=torch.int32)
+    lens[0] = T
+    for causal in (False, True):
+        qkv = torch.randn(B, T, 3 * d, device="cuda", dtype=torch.bfloat16,
+                          requires_grad=True)
+        out = fused_qkv_attention(qkv, H, None, causal, lens=lens)
+        assert out is not None
+        dy = torch.randn_like(out)
+        out.backward(dy)
+        # fp32 reference
+        qkv32 = qkv.detach().float().requires_grad_(True)
+        q, k, v = [t.reshape(B, T, H, dh).transpose(1, 2)
+                   for t in qkv32.split(d, dim=-1)]
+        am = torch.zeros(B, 1, T, T, device="cuda")
+        km = (torch.arange(T, device="cuda")[None, :] >= lens[:, None])
+        am.masked_fill_(km[:, None, None, :], float("-inf"))
+        if causal:
+            cm = torch.ones(T, T, device="cuda", dtype=torch.bool).tril()
+            am.masked_fill_(~cm, float("-inf"))
+        p = torch.softmax(q @ k.transpose(-1, -2) * dh ** -0.5 + am, dim=-1)
+        ref = p @ v
+        ref.backward(dy.float())
+        assert torch.allclose(out.float(), ref, rtol=5e-2, atol=5e-2), causal
+        assert torch.allclose(qkv.grad.float(), qkv32.grad, rtol=8e-2,
+                              atol=2e-1), causal
+
+
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
