@@ -208,14 +208,20 @@ class MultiHeadAttention(nn.Module):
             k = self._split(kh)
             v = self._split(vh)
         else:
-            q = self._split(ops.linear(query, self.w_slice(0),
-                                       self.b_slice(0)))
+            q2d = ops.linear(query, self.w_slice(0), self.b_slice(0))
+            q = self._split(q2d)
         if cache is not None and cache_key + "_static" in cache:
             k, v = cache[cache_key + "_k"], cache[cache_key + "_v"]
         elif not fuse_qkv:
             if keys is values:
                 kv = ops.linear(keys, self.W_qkv[:, self.dim:],
                                 self.b_qkv[self.dim:])
+                if (cache is None and collect is None
+                        and not (self.dropout if self.training else 0.0)):
+                    out = ops.attention_kv(q2d, kv, self.heads, mask=mask,
+                                           lens=lens)
+                    if out is not None:
+                        return self.Wo(self._join(out))
                 kh, vh = kv.split(self.dim, dim=-1)
                 k = self._split(kh)
                 v = self._split(vh)

@@ -169,6 +169,17 @@ def attention_qkv(qkv: torch.Tensor, heads: int, mask=None, causal=False,
     return fused_qkv_attention(qkv, heads, mask, causal, lens=lens)
 
 
+def attention_kv(q2: torch.Tensor, kv: torch.Tensor, heads: int, mask=None,
+                 dropout_p: float = 0.0,
+                 lens: Optional[torch.Tensor] = None):
+    """Cross-attention on fused projections q2 [B,Tq,d] / kv [B,Tk,2d];
+    returns [B,H,Tq,dh] or None for the composite fallback."""
+    if dropout_p != 0.0 or not _on_gpu(q2):
+        return None
+    from ._attention import fused_kv_attention
+    return fused_kv_attention(q2, kv, heads, mask, lens=lens)
+
+
 def attention_torch(q, k, v, mask=None, causal=False, dropout_p=0.0):
     scale = q.shape[-1] ** -0.5
     scores = torch.matmul(q, k.transpose(-1, -2)) * scale  # [B,H,Tq,Tk]

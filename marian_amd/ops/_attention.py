@@ -160,3 +160,78 @@ def fused_qkv_attention(qkv, H: int, mask, causal,
     o, _ = hip_ext(required=True).attention_fwd(q, k, v, lens, causal,
                                                 (d // H) ** -0.5, False)
     return o
+
+
+class FusedKVAttentionFn(torch.autograd.Function):
+    """Cross-attention with the fused [B,Tk,2d] KV projection and a
+    [B,Tq,d] Q projection; both gradients land in single buffers through
+    strided views (no split-backward cat)."""
+
+    @staticmethod
+    def forward(ctx, q2, kv, H, lens):
+        ext = hip_ext(required=True)
+        B, Tq, d = q2.shape
+        dh = d // H
+        q = q2.as_strided((B, H, Tq, dh), (Tq * d, dh, d, 1))
+        Tk = kv.shape[1]
+        k = kv.as_strided((B, H, Tk, dh), (Tk * 2 * d, dh, 2 * d, 1), 0)
+        v = kv.as_strided((B, H, Tk, dh), (Tk * 2 * d, dh, 2 * d, 1), d)
+        scale = dh ** -0.5
+        o, lse = ext.attention_fwd(q, k, v, lens, False, scale, True)
+        ctx.save_for_backward(q2, kv, o, lse,
+                              lens if lens is not None else torch.Tensor())
+        ctx.H = H
+        ctx.scale = scale
+        ctx.has_lens = lens is not None
+        return o
+
+    @staticmethod
+    def backward(ctx, d_o):
+        ext = hip_ext(required=True)
+        q2, kv, o, lse, lens = ctx.saved_tensors
+        H = ctx.H
+        B, Tq, d = q2.shape
+        dh = d // H
+        Tk = kv.shape[1]
+        q = q2.as_strided((B, H, Tq, dh), (Tq * d, dh, d, 1))
+        k = kv.as_strided((B, H, Tk, dh), (Tk * 2 * d, dh, 2 * d, 1), 0)
+        v = kv.as_strided((B, H, Tk, dh), (Tk * 2 * d, dh, 2 * d, 1), d)
+        dq2 = torch.empty_like(q2)
+        dkv = torch.empty_like(kv)
+        dq = dq2.as_strided((B, H, Tq, dh), (Tq * d, dh, d, 1))
+        dk = dkv.as_strided((B, H, Tk, dh), (Tk * 2 * d, dh, 2 * d, 1), 0)
+        dv = dkv.as_strided((B, H, Tk, dh), (Tk * 2 * d, dh, 2 * d, 1), d)
+        ext.attention_bwd_into(q, k, v, o, d_o, lse,
+                               lens if ctx.has_lens else None,
+                               False, ctx.scale, dq, dk, dv)
+        return dq2, dkv, None, None
+
+
+def fused_kv_attention(q2, kv, H: int, mask,
+                       lens: Optional[torch.Tensor] = None):
+    """Cross-attention off fused projections. q2 [B,Tq,d], kv [B,Tk,2d].
+    Returns [B,H,Tq,dh] or None for the composite fallback."""
+    if _DISABLED or not q2.is_cuda or q2.dtype != torch.bfloat16:
+        return None
+    ext = hip_ext(required=True)
+    if not hasattr(ext, "attention_bwd_into"):
+        return None
+    B, Tq, d = q2.shape
+    Tk = kv.shape[1]
+    if d % H or (d // H) != 64:
+        return None
+    if lens is None:
+        lens = mask_to_lens(mask, B, Tk)
+        if lens is None:
+            return None
+        lens = lens if lens.numel() else None
+    elif lens.numel() == 0:
+        lens = None
+    if torch.is_grad_enabled() and (q2.requires_grad or kv.requires_grad):
+        return FusedKVAttentionFn.apply(q2, kv, H, lens)
+    dh = d // H
+    q = q2.as_strided((B, H, Tq, dh), (Tq * d, dh, d, 1))
+    k = kv.as_strided((B, H, Tk, dh), (Tk * 2 * d, dh, 2 * d, 1), 0)
+    v = kv.as_strided((B, H, Tk, dh), (Tk * 2 * d, dh, 2 * d, 1), d)
+    o, _ = ext.attention_fwd(q, k, v, lens, False, dh ** -0.5, False)
+    return o

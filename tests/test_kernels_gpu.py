@@ -663,6 +663,36 @@ def test_fused_qkv_attention():
                               atol=2e-1), causal
 
 
+def test_fused_kv_cross_attention():
+    from marian_amd.ops._attention import fused_kv_attention
+    torch.manual_seed(7)
+    B, Tq, Tk, H, dh = 6, 32, 48, 16, 64
+    d = H * dh
+    lens = torch.randint(8, Tk + 1, (B,), device="cuda", dtype=torch.int32)
+    q2 = torch.randn(B, Tq, d, device="cuda", dtype=torch.bfloat16,
+                     requires_grad=True)
+    kv = torch.randn(B, Tk, 2 * d, device="cuda", dtype=torch.bfloat16,
+                     requires_grad=True)
+    out = fused_kv_attention(q2, kv, H, None, lens=lens)
+    assert out is not None
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    q32 = q2.detach().float().requires_grad_(True)
+    kv32 = kv.detach().float().requires_grad_(True)
+    q = q32.reshape(B, Tq, H, dh).transpose(1, 2)
+    k, v = [t.reshape(B, Tk, H, dh).transpose(1, 2)
+            for t in kv32.split(d, dim=-1)]
+    am = torch.zeros(B, 1, 1, Tk, device="cuda")
+    km = (torch.arange(Tk, device="cuda")[None, :] >= lens[:, None])
+    am.masked_fill_(km[:, None, None, :], float("-inf"))
+    p = torch.softmax(q @ k.transpose(-1, -2) * dh ** -0.5 + am, dim=-1)
+    ref = p @ v
+    ref.backward(dy.float())
+    assert torch.allclose(out.float(), ref, rtol=5e-2, atol=5e-2)
+    assert torch.allclose(q2.grad.float(), q32.grad, rtol=8e-2, atol=2e-1)
+    assert torch.allclose(kv.grad.float(), kv32.grad, rtol=8e-2, atol=2e-1)
+
+
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
