@@ -302,13 +302,15 @@ class TransformerModel(nn.Module):
             W = self.out_W.to(dt)
         if shortlist is not None:
             W = W.index_select(1, shortlist)
-        logits = torch.matmul(x, W)
         if self.out_b is not None:
             b = self.out_b.to(dt)
             if shortlist is not None:
                 b = b.index_select(0, shortlist)
-            logits = logits + b
-        return logits
+            # bias folded into the GEMM epilogue (a separate broadcast add
+            # costs a full extra read+write of the [N, V] logits tensor)
+            shp = x.shape[:-1] + (W.shape[1],)
+            return torch.addmm(b, x.reshape(-1, W.shape[0]), W).view(shp)
+        return torch.matmul(x, W)
 
     # ------------------------------------------------------------------
     def forward(self, batch):
