@@ -295,11 +295,12 @@ class FFN(nn.Module):
         super().__init__()
         self.W1 = Dense(dim, dim_ffn)
         self.W2 = Dense(dim_ffn, dim)
+        self.act_name = activation
         self.act = _ACT[activation]
         self.dropout = dropout
 
     def forward(self, x):
-        h = self.act(self.W1(x))
+        h = ops.linear_act(x, self.W1.W, self.W1.b, self.act_name)
         if self.dropout > 0 and self.training:
             h = torch.nn.functional.dropout(h, self.dropout, True)
         return self.W2(h)

@@ -60,6 +60,60 @@ def _on_gpu(t: torch.Tensor) -> bool:
 # Linear (GEMM + bias): addmm -> hipBLASLt epilogue; fp8 path for config #5
 # ---------------------------------------------------------------------------
 
+class _LtLinearFn(torch.autograd.Function):
+    """Linear through direct hipBLASLt (ops/hip/ltgemm.cpp): forward GEMM
+    with fused bias(+ReLU) epilogue; backward dW GEMM emits the bias
+    gradient via the BGRADA epilogue (no separate dY column reduce).
+    ReLU backward uses the post-activation output (relu'(x) == [y > 0])."""
+
+    @staticmethod
+    def forward(ctx, x2, W, b, relu):
+        ext = hip_ext(required=True)
+        y = ext.lt_linear_act(x2, W, b, relu)
+        if relu:
+            ctx.save_for_backward(x2, W, y)
+        else:
+            ctx.save_for_backward(x2, W)
+        ctx.relu = relu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = hip_ext(required=True)
+        if ctx.relu:
+            x2, W, y = ctx.saved_tensors
+            dyc = torch.ops.aten.threshold_backward(dy.contiguous(), y, 0)
+        else:
+            x2, W = ctx.saved_tensors
+            dyc = dy.contiguous()
+        dW, db = ext.lt_dw_bgrad(x2, dyc, True)
+        dx = torch.mm(dyc, W.t() if W.stride(1) == 1 else W.t().contiguous())
+        return dx, dW, db, None
+
+
+def _lt_ok(x, W, b) -> bool:
+    return (b is not None and "MAMD_NO_LT" not in os.environ
+            and x.dtype == torch.bfloat16 and W.dtype == torch.bfloat16
+            and W.stride(1) == 1 and x.is_cuda)
+
+
+def linear_act(x: torch.Tensor, W: torch.Tensor,
+               b: Optional[torch.Tensor], act: str) -> torch.Tensor:
+    """act(x @ W + b) with the activation fused into the GEMM epilogue
+    where hipBLASLt supports it (relu)."""
+    if act == "relu" and _lt_ok(x, W, b) and torch.is_grad_enabled()             and (x.requires_grad or W.requires_grad):
+        x2 = x.reshape(-1, W.shape[0])
+        shp = x.shape[:-1] + (W.shape[1],)
+        return _LtLinearFn.apply(x2.contiguous(), W, b, True).view(shp)
+    if act == "relu" and _lt_ok(x, W, b) and not torch.is_grad_enabled():
+        ext = hip_ext(required=True)
+        x2 = x.reshape(-1, W.shape[0]).contiguous()
+        shp = x.shape[:-1] + (W.shape[1],)
+        return ext.lt_linear_act(x2, W, b, True).view(shp)
+    from ..models.layers import _ACT
+    return _ACT[act](linear(x, W, b))
+
+
 class _LinearFn(torch.autograd.Function):
     """addmm forward; backward with the HIP column-sum kernel for the bias
     gradient. Measured on MI355X: torch's bf16 column reduce runs at
@@ -102,6 +156,9 @@ def linear(x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor] = None
                                and W.stride(0) != 1 else W,
                                b.to(x.dtype) if b is not None else None
                                ).view(shp)
+    if _lt_ok(x, W, b) and torch.is_grad_enabled()             and (x.requires_grad or W.requires_grad):
+        x2 = x.reshape(-1, W.shape[0]).contiguous()
+        return _LtLinearFn.apply(x2, W, b, False).view(shp)
     if b is None:
         return torch.matmul(x, W)
     return torch.addmm(b.to(x.dtype), x.reshape(-1, W.shape[0]), W).view(shp)

@@ -31,10 +31,14 @@ if os.path.exists(topk):
 colsum = os.path.join(HIP_DIR, "colsum.hip")
 if os.path.exists(colsum):
     sources.append(colsum)
+ltgemm = os.path.join(HIP_DIR, "ltgemm.cpp")
+if os.path.exists(ltgemm):
+    sources.append(ltgemm)
 
 ext = CUDAExtension(
     name="marian_amd.ops._hip_ext",
     sources=sources,
+    libraries=["hipblaslt"],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
         "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],

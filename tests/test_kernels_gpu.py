@@ -693,6 +693,68 @@ def test_fused_kv_cross_attention():
     assert torch.allclose(kv.grad.float(), kv32.grad, rtol=8e-2, atol=2e-1)
 
 
+def test_lt_linear_act_numerics():
+    """hipBLASLt fused bias(+relu) epilogue vs fp32 reference."""
+    from marian_amd.ops import hip_ext
+    ext = hip_ext(required=True)
+    torch.manual_seed(11)
+    x = torch.randn(512, 256, device="cuda", dtype=torch.bfloat16)
+    W = torch.randn(256, 384, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(384, device="cuda", dtype=torch.bfloat16)
+    for relu in (False, True):
+        y = ext.lt_linear_act(x, W, b, relu)
+        ref = x.float() @ W.float() + b.float()
+        if relu:
+            ref = torch.relu(ref)
+        assert torch.allclose(y.float(), ref, rtol=5e-2, atol=5e-1), relu
+    # column-slice view weight (row stride > ncols)
+    Wbig = torch.randn(256, 768, device="cuda", dtype=torch.bfloat16)
+    Wv = Wbig[:, 128:512]
+    y = ext.lt_linear_act(x, Wv, b, False)
+    ref = x.float() @ Wv.float() + b.float()
+    assert torch.allclose(y.float(), ref, rtol=5e-2, atol=5e-1)
+
+
+def test_lt_dw_bgrad_numerics():
+    from marian_amd.ops import hip_ext
+    ext = hip_ext(required=True)
+    x = torch.randn(2048, 256, device="cuda", dtype=torch.bfloat16)
+    dy = torch.randn(2048, 192, device="cuda", dtype=torch.bfloat16)
+    dW, db = ext.lt_dw_bgrad(x, dy, True)
+    refW = x.float().t() @ dy.float()
+    refb = dy.float().sum(0)
+    assert torch.allclose(dW.float(), refW, rtol=5e-2, atol=2.0)
+    assert torch.allclose(db.float(), refb, rtol=2e-2, atol=2.0)
+
+
+def test_lt_linear_autograd():
+    """ops.linear / linear_act on the Lt path vs fp32 autograd."""
+    from marian_amd import ops
+    torch.manual_seed(13)
+    for act in (None, "relu"):
+        x = torch.randn(400, 128, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        W = torch.randn(128, 96, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        b = torch.randn(96, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        y = (ops.linear_act(x, W, b, "relu") if act
+             else ops.linear(x, W, b))
+        dy = torch.randn_like(y)
+        y.backward(dy)
+        xr = x.detach().float().requires_grad_(True)
+        Wr = W.detach().float().requires_grad_(True)
+        br = b.detach().float().requires_grad_(True)
+        ref = xr @ Wr + br
+        if act:
+            ref = torch.relu(ref)
+        ref.backward(dy.float())
+        for got, want in [(x.grad, xr.grad), (W.grad, Wr.grad),
+                          (b.grad, br.grad)]:
+            assert torch.allclose(got.float(), want, rtol=8e-2,
+                                  atol=5e-1), act
+
+
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
