@@ -60,6 +60,9 @@ def _on_gpu(t: torch.Tensor) -> bool:
 # Linear (GEMM + bias): addmm -> hipBLASLt epilogue; fp8 path for config #5
 # ---------------------------------------------------------------------------
 
+_LT_BGRAD_OK = True
+
+
 class _LtLinearFn(torch.autograd.Function):
     """Linear through direct hipBLASLt (ops/hip/ltgemm.cpp): forward GEMM
     with fused bias(+ReLU) epilogue; backward dW GEMM emits the bias
@@ -86,9 +89,16 @@ class _LtLinearFn(torch.autograd.Function):
         else:
             x2, W = ctx.saved_tensors
             dyc = dy.contiguous()
-        dW, db = ext.lt_dw_bgrad(x2, dyc, True)
+        global _LT_BGRAD_OK
         dx = torch.mm(dyc, W.t() if W.stride(1) == 1 else W.t().contiguous())
-        return dx, dW, db, None
+        if _LT_BGRAD_OK:
+            try:
+                dWt, db = ext.lt_dwt_bgrad(x2, dyc, True)
+                return dx, dWt.t(), db.to(dyc.dtype), None
+            except RuntimeError:
+                _LT_BGRAD_OK = False  # epilogue unsupported: torch fallback
+        dW = torch.mm(x2.t(), dyc)
+        return dx, dW, dyc.sum(0), None
 
 
 def _lt_ok(x, W, b) -> bool:

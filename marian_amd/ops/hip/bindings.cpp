@@ -29,8 +29,8 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                                          bool need_lse);
 torch::Tensor lt_linear_act(torch::Tensor x, torch::Tensor W,
                             torch::Tensor b, bool relu);
-std::vector<torch::Tensor> lt_dw_bgrad(torch::Tensor x, torch::Tensor dy,
-                                       bool want_db);
+std::vector<torch::Tensor> lt_dwt_bgrad(torch::Tensor x, torch::Tensor dy,
+                                        bool want_db);
 torch::Tensor col_sum(torch::Tensor x);
 torch::Tensor sumsq_finite(torch::Tensor g);
 std::vector<torch::Tensor> beam_topk(torch::Tensor logprobs,
@@ -67,8 +67,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused flash-style attention backward (bf16, Dh=64)");
   m.def("lt_linear_act", &lt_linear_act,
         "hipBLASLt GEMM with fused bias(+ReLU) epilogue");
-  m.def("lt_dw_bgrad", &lt_dw_bgrad,
-        "hipBLASLt dW GEMM with fused bias-gradient (BGRADA) epilogue");
+  m.def("lt_dwt_bgrad", &lt_dwt_bgrad,
+        "hipBLASLt dW^T GEMM with fused bias-gradient (BGRADB) epilogue");
   m.def("col_sum", &col_sum, "column sum (bias gradients)");
   m.def("sumsq_finite", &sumsq_finite, "fused grad sumsq + nonfinite count");
   m.def("beam_topk", &beam_topk,

@@ -56,7 +56,7 @@ struct Desc {
   hipblasLtMatmulAlgo_t algo;
 };
 
-using Key = std::tuple<int, int, long, long, long, long, long, long, int>;
+using Key = std::tuple<int, int, long, long, long, long, long, long, int, int>;
 std::map<Key, Desc> g_cache;
 std::mutex g_mu;
 
@@ -65,9 +65,11 @@ std::mutex g_mu;
 // shape enumerates supported algos and keeps the fastest.
 void lt_run(bool transA, bool transB, long m, long n, long k, const void* A,
             long lda, const void* B, long ldb, void* C, long ldc,
-            const void* bias, hipblasLtEpilogue_t epi) {
+            const void* bias, hipblasLtEpilogue_t epi,
+            hipDataType bias_type = HIP_R_16BF) {
   auto stream = at::cuda::getCurrentHIPStream();
-  Key key{(int)transA, (int)transB, m, n, k, lda, ldb, ldc, (int)epi};
+  Key key{(int)transA, (int)transB, m, n, k, lda, ldb, ldc, (int)epi,
+          (int)bias_type};
   std::unique_lock<std::mutex> lk(g_mu);
   auto it = g_cache.find(key);
   if (it == g_cache.end()) {
@@ -85,6 +87,9 @@ void lt_run(bool transA, bool transB, long m, long n, long k, const void* A,
     if (bias) {
       LT_CHECK(hipblasLtMatmulDescSetAttribute(
           d.op, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias)));
+      int32_t bt = (int32_t)bias_type;
+      LT_CHECK(hipblasLtMatmulDescSetAttribute(
+          d.op, HIPBLASLT_MATMUL_DESC_BIAS_DATA_TYPE, &bt, sizeof(bt)));
     }
     LT_CHECK(hipblasLtMatrixLayoutCreate(&d.a, HIP_R_16BF,
                                          transA ? k : m, transA ? m : k,
@@ -151,7 +156,9 @@ void lt_run(bool transA, bool transB, long m, long n, long k, const void* A,
                                                d.c, d.c, pref, 1, hr,
                                                &found));
       (void)hipblasLtMatmulPreferenceDestroy(pref);
-      TORCH_CHECK(found > 0, "hipBLASLt: no algorithm for this GEMM");
+      TORCH_CHECK(found > 0, "hipBLASLt: no algorithm for GEMM m=", m,
+                  " n=", n, " k=", k, " epi=", (int)epi,
+                  " candidates=", (long)algos.size());
       best_algo = hr[0].algo;
     }
     d.algo = best_algo;
@@ -191,24 +198,28 @@ torch::Tensor lt_linear_act(torch::Tensor x, torch::Tensor W,
   return y;
 }
 
-// dW [K,N] = x^T [K,M] @ dy [M,N]; db [N] = colsum(dy) fused via BGRADA.
-std::vector<torch::Tensor> lt_dw_bgrad(torch::Tensor x, torch::Tensor dy,
-                                       bool want_db) {
+// dW^T [N,K] (row-major) = dy^T @ x, with db [N] = colsum(dy) fused via
+// the BGRADB epilogue (BGRADA is not implemented by this hipBLASLt build
+// for bf16 — measured: 0 of 2205 candidates support it; BGRADB on the
+// swapped formulation is). Caller transposes the returned dW^T view.
+std::vector<torch::Tensor> lt_dwt_bgrad(torch::Tensor x, torch::Tensor dy,
+                                        bool want_db) {
   TORCH_CHECK(x.is_cuda() && rm_ok(x) && rm_ok(dy) &&
               x.size(0) == dy.size(0));
   const long M = x.size(0), K = x.size(1), N = dy.size(1);
-  auto dW = torch::empty({K, N}, x.options());
+  auto dWt = torch::empty({N, K}, x.options());
   torch::Tensor db;
   const void* bias = nullptr;
   if (want_db) {
-    db = torch::empty({N}, x.options());
+    db = torch::empty({N}, x.options().dtype(torch::kFloat32));
     bias = db.data_ptr();
   }
-  // column-major: dW_cm [N,K] = dy_cm[N,M] (opN) * x_cm[K,M]^T (opT);
-  // BGRADA reduces op(A)=dy over its columns (M) -> vector length N.
-  lt_run(false, true, N, K, M, dy.data_ptr(), dy.stride(0), x.data_ptr(),
-         x.stride(0), dW.data_ptr(), N, bias,
-         want_db ? HIPBLASLT_EPILOGUE_BGRADA : HIPBLASLT_EPILOGUE_DEFAULT);
+  // column-major: dWt_cm [K,N] = x_cm[K,M] (opN) * dy_cm[N,M]^T (opT);
+  // BGRADB reduces op(B)=dy over the k dim (M) -> vector length N.
+  lt_run(false, true, K, N, M, x.data_ptr(), x.stride(0), dy.data_ptr(),
+         dy.stride(0), dWt.data_ptr(), K, bias,
+         want_db ? HIPBLASLT_EPILOGUE_BGRADB : HIPBLASLT_EPILOGUE_DEFAULT,
+         HIP_R_32F);
   if (!want_db) db = torch::empty({0}, x.options());
-  return {dW, db};
+  return {dWt, db};
 }

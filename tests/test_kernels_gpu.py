@@ -715,16 +715,16 @@ def test_lt_linear_act_numerics():
     assert torch.allclose(y.float(), ref, rtol=5e-2, atol=5e-1)
 
 
-def test_lt_dw_bgrad_numerics():
+def test_lt_dwt_bgrad_numerics():
     from marian_amd.ops import hip_ext
     ext = hip_ext(required=True)
     x = torch.randn(2048, 256, device="cuda", dtype=torch.bfloat16)
     dy = torch.randn(2048, 192, device="cuda", dtype=torch.bfloat16)
-    dW, db = ext.lt_dw_bgrad(x, dy, True)
+    dWt, db = ext.lt_dwt_bgrad(x, dy, True)
     refW = x.float().t() @ dy.float()
     refb = dy.float().sum(0)
-    assert torch.allclose(dW.float(), refW, rtol=5e-2, atol=2.0)
-    assert torch.allclose(db.float(), refb, rtol=2e-2, atol=2.0)
+    assert torch.allclose(dWt.t().float(), refW, rtol=5e-2, atol=2.0)
+    assert torch.allclose(db, refb, rtol=2e-2, atol=2.0)
 
 
 def test_lt_linear_autograd():
