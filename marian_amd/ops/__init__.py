@@ -60,14 +60,18 @@ def _on_gpu(t: torch.Tensor) -> bool:
 # Linear (GEMM + bias): addmm -> hipBLASLt epilogue; fp8 path for config #5
 # ---------------------------------------------------------------------------
 
-_LT_BGRAD_OK = True
+# BGRADB fused bias-grad: opt-in. Measured on MI355X the BGRADB-capable
+# NT solution pool loses more on the dW GEMM than the fused reduce saves
+# (B1024 transformer-big 585k -> 508k tok/s with it on everywhere).
+_LT_BGRAD_OK = os.environ.get("MAMD_LT_BGRAD", "") == "1"
 
 
 class _LtLinearFn(torch.autograd.Function):
     """Linear through direct hipBLASLt (ops/hip/ltgemm.cpp): forward GEMM
-    with fused bias(+ReLU) epilogue; backward dW GEMM emits the bias
-    gradient via the BGRADA epilogue (no separate dY column reduce).
-    ReLU backward uses the post-activation output (relu'(x) == [y > 0])."""
+    with fused bias+ReLU epilogue (a separate ReLU pass costs a full
+    read+write of the FFN activation). Backward runs TunableOp-tuned mm;
+    ReLU backward uses the post-activation output (relu'(x) == [y > 0]),
+    so the pre-activation is never materialized."""
 
     @staticmethod
     def forward(ctx, x2, W, b, relu):
@@ -89,15 +93,14 @@ class _LtLinearFn(torch.autograd.Function):
         else:
             x2, W = ctx.saved_tensors
             dyc = dy.contiguous()
-        global _LT_BGRAD_OK
         dx = torch.mm(dyc, W.t() if W.stride(1) == 1 else W.t().contiguous())
         if _LT_BGRAD_OK:
             try:
                 dWt, db = ext.lt_dwt_bgrad(x2, dyc, True)
                 return dx, dWt.t(), db.to(dyc.dtype), None
             except RuntimeError:
-                _LT_BGRAD_OK = False  # epilogue unsupported: torch fallback
-        dW = torch.mm(x2.t(), dyc)
+                pass  # epilogue unsupported on this shape: torch fallback
+        dW = torch.mm(x2.t(), dyc)  # TunableOp-tuned TN
         return dx, dW, dyc.sum(0), None
 
 
@@ -166,9 +169,6 @@ def linear(x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor] = None
                                and W.stride(0) != 1 else W,
                                b.to(x.dtype) if b is not None else None
                                ).view(shp)
-    if _lt_ok(x, W, b) and torch.is_grad_enabled()             and (x.requires_grad or W.requires_grad):
-        x2 = x.reshape(-1, W.shape[0]).contiguous()
-        return _LtLinearFn.apply(x2, W, b, False).view(shp)
     if b is None:
         return torch.matmul(x, W)
     return torch.addmm(b.to(x.dtype), x.reshape(-1, W.shape[0]), W).view(shp)
