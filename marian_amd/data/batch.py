@@ -72,6 +72,14 @@ class CorpusBatch:
     def target(self) -> SubBatch:
         return self.sub[-1]
 
+    @target.setter
+    def target(self, sb: SubBatch):
+        """Models may rewrite the target stream (BERT masking transform)."""
+        if len(self.sub) == 1:
+            self.sub.append(sb)
+        else:
+            self.sub[-1] = sb
+
     def size(self) -> int:
         return self.sub[0].batch_size
 

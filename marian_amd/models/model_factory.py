@@ -13,6 +13,12 @@ def create_model(options: Options):
     if mtype == "s2s":
         from .s2s import S2SModel
         return S2SModel(options)
+    if mtype == "bert":
+        from .bert import BertModel
+        return BertModel(options)
+    if mtype == "bert-classifier":
+        from .bert import BertClassifier
+        return BertClassifier(options)
     raise NotImplementedError(
         f"--type {mtype} is not implemented yet in marian_amd "
         "(transformer and s2s are)")

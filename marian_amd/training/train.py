@@ -116,6 +116,8 @@ class Train:
         options.set("dim-vocabs", dims)
 
         self.model = create_model(options)
+        if hasattr(self.model, "resolve_bert_ids"):
+            self.model.resolve_bert_ids(self.vocabs[0], options)
         self.graph_group = SyncGraphGroup(options, self.model, self.device,
                                           self.rank, self.world)
         self.state = TrainingState()
