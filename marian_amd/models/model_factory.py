@@ -10,9 +10,13 @@ def create_model(options: Options):
     mtype = options.get("type", "transformer")
     if mtype in ("transformer", "transformer-encoder"):
         return TransformerModel(options)
-    if mtype == "s2s":
+    if mtype in ("s2s", "amun", "nematus"):
         from .s2s import S2SModel
-        return S2SModel(options)
+        m = S2SModel(options)
+        # amun/nematus = s2s computation + GroundHog-era weight names
+        # (ref: src/models/amun.h, nematus.h)
+        m.legacy_names = mtype in ("amun", "nematus")
+        return m
     if mtype == "bert":
         from .bert import BertModel
         return BertModel(options)

@@ -79,6 +79,39 @@ class MLPAttention(nn.Module):
         return ctx, a
 
 
+
+# Legacy weight-name translation for GroundHog-lineage checkpoints
+# (ref: src/models/amun.h, src/models/nematus.h — name maps only, the
+# computation is the marian s2s cell stack). [M on exact tail names]
+_LEGACY_NAME_MAP = {
+    "encoder_Wemb": "Wemb",
+    "decoder_Wemb": "Wemb_dec",
+    "encoder_bi_W": "encoder_W", "encoder_bi_U": "encoder_U",
+    "encoder_bi_b": "encoder_b", "encoder_bi_Wx": "encoder_Wx",
+    "encoder_bi_Ux": "encoder_Ux", "encoder_bi_bx": "encoder_bx",
+    "encoder_bi_r_W": "encoder_r_W", "encoder_bi_r_U": "encoder_r_U",
+    "encoder_bi_r_b": "encoder_r_b", "encoder_bi_r_Wx": "encoder_r_Wx",
+    "encoder_bi_r_Ux": "encoder_r_Ux", "encoder_bi_r_bx": "encoder_r_bx",
+    "decoder_ff_state_W": "ff_state_W", "decoder_ff_state_b": "ff_state_b",
+    "decoder_cell1_W": "decoder_W", "decoder_cell1_U": "decoder_U",
+    "decoder_cell1_b": "decoder_b", "decoder_cell1_Wx": "decoder_Wx",
+    "decoder_cell1_Ux": "decoder_Ux", "decoder_cell1_bx": "decoder_bx",
+    "decoder_cell2_W": "decoder_Wc", "decoder_cell2_U": "decoder_U_nl",
+    "decoder_cell2_b": "decoder_b_nl", "decoder_cell2_Wx": "decoder_Wcx",
+    "decoder_cell2_Ux": "decoder_Ux_nl", "decoder_cell2_bx": "decoder_bx_nl",
+    "decoder_att_W": "decoder_W_comb_att", "decoder_att_U": "decoder_Wc_att",
+    "decoder_att_b": "decoder_b_att", "decoder_att_v": "decoder_U_att",
+    "decoder_ff_logit_l1_W0": "ff_logit_lstm_W",
+    "decoder_ff_logit_l1_b0": "ff_logit_lstm_b",
+    "decoder_ff_logit_l1_W1": "ff_logit_prev_W",
+    "decoder_ff_logit_l1_b1": "ff_logit_prev_b",
+    "decoder_ff_logit_l1_W2": "ff_logit_ctx_W",
+    "decoder_ff_logit_l1_b2": "ff_logit_ctx_b",
+    "decoder_ff_logit_l2_W": "ff_logit_W",
+    "decoder_ff_logit_l2_b": "ff_logit_b",
+}
+
+
 class S2SModel(nn.Module):
     def __init__(self, options: Options):
         super().__init__()
@@ -269,9 +302,16 @@ class S2SModel(nn.Module):
         m["decoder_ff_logit_l2_b"] = self.out_b
         return m
 
+    def _named(self) -> Dict[str, torch.Tensor]:
+        """Name map, translated for --type amun/nematus checkpoints."""
+        m = self._name_map()
+        if getattr(self, "legacy_names", False):
+            m = {_LEGACY_NAME_MAP.get(k, k): v for k, v in m.items()}
+        return m
+
     def marian_items(self) -> Dict[str, np.ndarray]:
         out = {}
-        for name, t in self._name_map().items():
+        for name, t in self._named().items():
             arr = t.detach().cpu().float().numpy()
             if arr.ndim == 1:
                 arr = arr.reshape(1, -1)
@@ -279,7 +319,11 @@ class S2SModel(nn.Module):
         return out
 
     def load_marian_items(self, items, strict: bool = True) -> None:
-        nm = self._name_map()
+        nm = self._named()
+        if not getattr(self, "legacy_names", False) \
+                and "decoder_W_comb_att" in items:
+            # auto-detect a legacy (amun/nematus) checkpoint
+            nm = {_LEGACY_NAME_MAP.get(k, k): v for k, v in nm.items()}
         with torch.no_grad():
             for name, t in nm.items():
                 if name not in items:

@@ -290,7 +290,28 @@ class TransformerModel(nn.Module):
                 cache.setdefault("alignments", []).append(collect["probs"])
         if self.post_top_dec is not None:
             x = self.post_top_dec(x)
-        return self.output_logits(x[:, 0], shortlist)
+        h = x[:, 0]
+        lsh = getattr(self, "lsh_index", None)
+        if lsh is not None and shortlist is None:
+            # LSH approximate output (--output-approx-knn): hamming top-k
+            # candidates per row, exact re-score on the union, scattered
+            # back to full width so beam/top-k machinery is unchanged
+            ids = lsh.candidates(h)
+            small = self.output_logits(h, ids).float()
+            logits = torch.full((h.shape[0], self.trg_vocab_size), -1e9,
+                                device=h.device)
+            logits.scatter_(1, ids.unsqueeze(0).expand(h.shape[0], -1),
+                            small)
+            return logits
+        return self.output_logits(h, shortlist)
+
+    # ------------------------------------------------------------------
+    def enable_lsh(self, k: int, nbits: int):
+        """Build the LSH output shortlist index (--output-approx-knn K
+        NBITS; ref: layers/lsh.cpp) over the output-projection rows."""
+        from ..data.lsh import LSHIndex
+        W = self.trg_embed.E if self.tied_out else self.out_W.t()
+        self.lsh_index = LSHIndex(k, nbits).build(W.detach())
 
     # ------------------------------------------------------------------
     def output_logits(self, x, shortlist: Optional[torch.Tensor] = None):

@@ -40,6 +40,9 @@ def load_model_for_inference(path: str, options: Options, device):
         model = model.to(torch.bfloat16)
         model.set_compute_dtype(torch.bfloat16)
     model.eval()
+    knn = [int(x) for x in (options.list("output-approx-knn") or [])]
+    if len(knn) >= 2 and hasattr(model, "enable_lsh"):
+        model.enable_lsh(knn[0], knn[1])
     return model
 
 

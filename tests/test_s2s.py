@@ -98,3 +98,31 @@ def test_s2s_trains_on_copy_task():
         first = first if first is not None else ce
         last = ce
     assert last < first - 0.3, (first, last)
+
+
+def test_amun_legacy_names_roundtrip():
+    """--type amun saves/loads GroundHog-era names (ref: models/amun.h)."""
+    import numpy as np
+    from marian_amd.models.model_factory import create_model
+    o = _opts() if "_opts" in dir() else None
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    o = ConfigParser(MODE_TRAINING).parse_options(
+        ["--train-sets", "a", "b", "--vocabs", "a.yml", "b.yml",
+         "--type", "amun", "--dim-emb", "16", "--dim-rnn", "32"],
+        validate=False)
+    o.set("dim-vocabs", [30, 30])
+    m = create_model(o)
+    items = m.marian_items()
+    assert "decoder_W_comb_att" in items          # legacy attention name
+    assert "ff_logit_lstm_W" in items
+    assert "decoder_cell1_W" not in items
+    # loading the legacy items back reproduces parameters
+    m2 = create_model(o)
+    m2.load_marian_items(items, strict=True)
+    for a, b in zip(m.parameters(), m2.parameters()):
+        assert torch.allclose(a, b)
+    # auto-detection: a plain --type s2s model loads amun items too
+    o.set("type", "s2s")
+    m3 = create_model(o)
+    m3.load_marian_items(items, strict=False)
+    assert torch.allclose(next(m.parameters()), next(m3.parameters()))
