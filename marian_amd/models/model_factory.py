@@ -17,6 +17,9 @@ def create_model(options: Options):
         # (ref: src/models/amun.h, nematus.h)
         m.legacy_names = mtype in ("amun", "nematus")
         return m
+    if mtype in ("lm", "lm-transformer"):
+        from .lm import TransformerLM
+        return TransformerLM(options)
     if mtype == "bert":
         from .bert import BertModel
         return BertModel(options)

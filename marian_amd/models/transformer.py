@@ -96,11 +96,13 @@ class DecoderLayer(nn.Module):
             h = self.self_attn(h, h, h, causal=causal, cache=cache,
                                cache_key=f"l{layer_id}_self")
         x = self.post_self(h, residual=x)
-        h = self.pre_ctx(x)
-        h = self.context_attn(h, enc_out, enc_out, mask=src_mask, cache=cache,
-                              cache_key=f"l{layer_id}_context", lens=src_lens,
-                              collect=collect)
-        x = self.post_ctx(h, residual=x)
+        if enc_out is not None:   # LM decoders have no cross-attention
+            h = self.pre_ctx(x)
+            h = self.context_attn(h, enc_out, enc_out, mask=src_mask,
+                                  cache=cache,
+                                  cache_key=f"l{layer_id}_context",
+                                  lens=src_lens, collect=collect)
+            x = self.post_ctx(h, residual=x)
         h = self.pre_ffn(x)
         h = self.ffn(h)
         return self.post_ffn(h, residual=x)
@@ -405,12 +407,14 @@ class TransformerModel(nn.Module):
                     m[f"decoder_l{i}_aan_ln_bias"] = layer.post_self.ln.bias
             else:
                 attn(f"decoder_l{i}_self", layer.self_attn, layer.post_self)
-            attn(f"decoder_l{i}_context", layer.context_attn, layer.post_ctx)
+            if layer.context_attn is not None:   # absent in LMs
+                attn(f"decoder_l{i}_context", layer.context_attn,
+                     layer.post_ctx)
             ffn(f"decoder_l{i}_ffn", layer.ffn, layer.post_ffn)
             if hasattr(layer.pre_self, "ln"):
                 m[f"decoder_l{i}_self_Wq_ln_scale_pre"] = layer.pre_self.ln.scale
                 m[f"decoder_l{i}_self_Wq_ln_bias_pre"] = layer.pre_self.ln.bias
-            if hasattr(layer.pre_ctx, "ln"):
+            if layer.pre_ctx is not None and hasattr(layer.pre_ctx, "ln"):
                 m[f"decoder_l{i}_context_Wq_ln_scale_pre"] = layer.pre_ctx.ln.scale
                 m[f"decoder_l{i}_context_Wq_ln_bias_pre"] = layer.pre_ctx.ln.bias
             if hasattr(layer.pre_ffn, "ln"):
