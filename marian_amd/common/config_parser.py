@@ -90,7 +90,7 @@ class ConfigParser:
         a("pretrained-model", str, None, "Initialize weights from this model", "model", modes={T})
         a("ignore-model-config", "bool", False, "Ignore the config embedded in the model file", "model")
         a("type", str, "amun" if False else "transformer", "Model type", "model",
-          choices=["transformer", "s2s", "amun", "nematus", "lm", "lm-transformer",
+          choices=["transformer", "s2s", "amun", "nematus", "char-s2s", "lm", "lm-transformer",
                    "bert", "bert-classifier", "transformer-encoder"])
         a("dim-vocabs", int, [0, 0], "Max vocabulary sizes", "model", nargs="+")
         a("dim-emb", int, 512, "Size of embedding vector", "model")
@@ -137,6 +137,8 @@ class ConfigParser:
         a("transformer-train-position-embeddings", "bool", False,
           "Train positional embeddings instead of sinusoidal", "model")
         a("transformer-depth-scaling", "bool", False, "Scale init by layer depth", "model")
+        a("char-stride", int, 5, "Width of max-pooling layer after convolution layer in char-s2s", "model")
+        a("char-highway", int, 4, "Number of highway network layers after max-pooling in char-s2s", "model")
         a("bert-mask-symbol", str, "[MASK]", "BERT mask symbol", "model")
         a("bert-sep-symbol", str, "[SEP]", "BERT sep symbol", "model")
         a("bert-class-symbol", str, "[CLS]", "BERT class symbol", "model")

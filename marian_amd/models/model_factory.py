@@ -17,6 +17,9 @@ def create_model(options: Options):
         # (ref: src/models/amun.h, nematus.h)
         m.legacy_names = mtype in ("amun", "nematus")
         return m
+    if mtype == "char-s2s":
+        from .char_s2s import CharS2SModel
+        return CharS2SModel(options)
     if mtype in ("lm", "lm-transformer"):
         from .lm import TransformerLM
         return TransformerLM(options)
