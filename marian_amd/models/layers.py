@@ -88,10 +88,17 @@ class Embedding(nn.Module):
     def reset_parameters(self):
         nn.init.normal_(self.E, std=self.dim ** -0.5)
 
+    def table(self, compute_dtype=torch.float32):
+        """The [V, d] lookup table; factored vocabs override it with a
+        per-forward factor-sum (data/factored_vocab.py)."""
+        if getattr(self, "factored", None) is not None:
+            return self.factored.surface_table().to(compute_dtype)
+        return self.E.to(compute_dtype)
+
     def words(self, indices: torch.Tensor, compute_dtype=torch.float32):
         """Scaled word embeddings only (marian adds positions AFTER the
         decoder's target shift, so the two halves are separable)."""
-        x = torch.nn.functional.embedding(indices, self.E.to(compute_dtype))
+        x = torch.nn.functional.embedding(indices, self.table(compute_dtype))
         return x * math.sqrt(self.dim)
 
     def add_positions(self, x: torch.Tensor, start_pos: int = 0):

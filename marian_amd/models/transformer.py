@@ -308,6 +308,23 @@ class TransformerModel(nn.Module):
         return self.output_logits(h, shortlist)
 
     # ------------------------------------------------------------------
+    def enable_factored_vocab(self, fv, side: str = "trg"):
+        """Switch a side to factored embeddings (+ factored output on trg)
+        (ref: data/factored_vocab.cpp, layers/logits.cpp)."""
+        from ..data.factored_vocab import FactoredEmbedding, FactoredOutput
+        femb = FactoredEmbedding(fv.group_sizes, self.dim, fv.factor_matrix)
+        if side in ("src", "both"):
+            self.src_embed.factored = femb
+            self.src_embed.E = torch.nn.Parameter(torch.zeros(0, 0),
+                                                  requires_grad=False)
+        if side in ("trg", "both"):
+            self.trg_embed.factored = femb if side == "both" else                 FactoredEmbedding(fv.group_sizes, self.dim,
+                                  fv.factor_matrix)
+            self.trg_embed.E = torch.nn.Parameter(torch.zeros(0, 0),
+                                                  requires_grad=False)
+            self.factored_out = FactoredOutput(self.trg_embed.factored)
+
+    # ------------------------------------------------------------------
     def enable_lsh(self, k: int, nbits: int):
         """Build the LSH output shortlist index (--output-approx-knn K
         NBITS; ref: layers/lsh.cpp) over the output-projection rows."""
@@ -318,6 +335,12 @@ class TransformerModel(nn.Module):
     # ------------------------------------------------------------------
     def output_logits(self, x, shortlist: Optional[torch.Tensor] = None):
         """Output projection; tied -> Wemb^T (ref: layers/output.cpp)."""
+        fo = getattr(self, "factored_out", None)
+        if fo is not None:
+            scores = fo(x)              # [N, V] summed per-group log-probs
+            if shortlist is not None:
+                scores = scores.index_select(-1, shortlist)
+            return scores
         dt = x.dtype
         if self.tied_out:
             W = self.trg_embed.E.to(dt).t()  # [d, V]

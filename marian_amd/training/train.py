@@ -74,7 +74,10 @@ def load_or_create_vocabs(options: Options) -> List[Vocab]:
     dim_vocabs = options.list("dim-vocabs") or [0, 0]
     for i, p in enumerate(paths):
         max_size = int(dim_vocabs[i]) if i < len(dim_vocabs) else 0
-        if os.path.exists(p):
+        if p.endswith(".fsv") and os.path.exists(p):
+            from ..data.factored_vocab import FactoredVocab
+            v = FactoredVocab.load(p, max_size=max_size)
+        elif os.path.exists(p):
             v = Vocab.load(p, max_size=max_size)
         elif p.endswith(".spm") or p.endswith(".model"):
             _train_spm(p, train_sets[min(i, len(train_sets) - 1)], options,
@@ -118,6 +121,16 @@ class Train:
         self.model = create_model(options)
         if hasattr(self.model, "resolve_bert_ids"):
             self.model.resolve_bert_ids(self.vocabs[0], options)
+        if hasattr(self.model, "enable_factored_vocab"):
+            src_f = getattr(self.vocabs[0], "is_factored", False)
+            trg_f = getattr(self.vocabs[-1], "is_factored", False)
+            if src_f and trg_f and self.vocabs[0] is self.vocabs[-1]:
+                self.model.enable_factored_vocab(self.vocabs[-1], "both")
+            else:
+                if src_f:
+                    self.model.enable_factored_vocab(self.vocabs[0], "src")
+                if trg_f:
+                    self.model.enable_factored_vocab(self.vocabs[-1], "trg")
         self.graph_group = SyncGraphGroup(options, self.model, self.device,
                                           self.rank, self.world)
         self.state = TrainingState()

@@ -1,0 +1,91 @@
+"""Factored vocabulary (ref: src/data/factored_vocab.{h,cpp},
+src/layers/logits.{h,cpp})."""
+import torch
+
+from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+from marian_amd.data.batch import SubBatch, CorpusBatch
+from marian_amd.data.factored_vocab import (FactoredVocab, FactoredEmbedding,
+                                            FactoredOutput)
+from marian_amd.models.model_factory import create_model
+
+FSV = """\
+# surface<TAB>lemma + trait factors
+the\tthe _c0
+The\tthe _c1
+dog\tdog _c0
+dogs\tdog _c0 _pl1
+Dog\tdog _c1
+run\trun _c0
+"""
+
+
+def _write(tmp_path):
+    p = tmp_path / "v.fsv"
+    p.write_text(FSV)
+    return str(p)
+
+
+def test_fsv_parsing(tmp_path):
+    v = FactoredVocab.load(_write(tmp_path))
+    assert len(v) == 8                 # 6 surfaces + </s> + <unk>
+    assert v.groups[0] == "(lemma)"
+    assert "_c" in v.groups and "_pl" in v.groups
+    F = v.factor_matrix
+    the, The = v.token2id["the"], v.token2id["The"]
+    dog, Dog = v.token2id["dog"], v.token2id["Dog"]
+    assert F[the, 0] == F[The, 0]       # same lemma
+    assert F[dog, 0] == F[Dog, 0]
+    assert F[the, 0] != F[dog, 0]
+    ci = v.groups.index("_c")
+    assert F[the, ci] != F[The, ci]     # different caps factor
+    # ids round-trip through the surface pipeline
+    assert v.encode("the dog run", add_eos=False) == [the, dog,
+                                                      v.token2id["run"]]
+
+
+def test_factored_embedding_shares_lemma(tmp_path):
+    v = FactoredVocab.load(_write(tmp_path))
+    emb = FactoredEmbedding(v.group_sizes, 16, v.factor_matrix)
+    T = emb.surface_table()
+    assert T.shape == (len(v), 16)
+    the, The = v.token2id["the"], v.token2id["The"]
+    dog = v.token2id["dog"]
+    # "the" and "The" share lemma but differ in caps factor
+    assert not torch.allclose(T[the], T[The])
+    lemma_part = emb.tables[0][v.factor_matrix[:, 0]]
+    assert torch.allclose(lemma_part[the], lemma_part[The])
+    assert not torch.allclose(lemma_part[the], lemma_part[dog])
+
+
+def test_factored_output_distribution(tmp_path):
+    v = FactoredVocab.load(_write(tmp_path))
+    emb = FactoredEmbedding(v.group_sizes, 16, v.factor_matrix)
+    out = FactoredOutput(emb)
+    h = torch.randn(3, 16)
+    scores = out(h)
+    assert scores.shape == (3, len(v))
+    assert torch.isfinite(scores).all()
+    # renormalized distribution sums to 1
+    p = torch.softmax(scores.float(), dim=-1)
+    assert torch.allclose(p.sum(-1), torch.ones(3), atol=1e-4)
+
+
+def test_factored_transformer_train_step(tmp_path):
+    v = FactoredVocab.load(_write(tmp_path))
+    o = ConfigParser(MODE_TRAINING).parse_options(
+        ["--train-sets", "a", "b", "--vocabs", "a.fsv", "b.fsv",
+         "--dim-emb", "32", "--enc-depth", "1", "--dec-depth", "1",
+         "--transformer-heads", "4", "--transformer-dim-ffn", "64"],
+        validate=False)
+    o.set("dim-vocabs", [len(v), len(v)])
+    m = create_model(o)
+    m.enable_factored_vocab(v, "both")
+    idx = torch.randint(2, len(v), (2, 5))
+    b = CorpusBatch([SubBatch(idx, torch.ones(2, 5)),
+                     SubBatch(idx.clone(), torch.ones(2, 5))])
+    logits = m(b)
+    assert logits.shape == (2, 5, len(v))
+    loss = -logits.float().log_softmax(-1).gather(
+        -1, b.target.indices.unsqueeze(-1)).sum()
+    loss.backward()
+    assert m.trg_embed.factored.tables[0].grad is not None
