@@ -755,6 +755,34 @@ def test_lt_linear_autograd():
                                   atol=5e-1), act
 
 
+def test_new_model_families_gpu():
+    """bert / lm-transformer / factored output run on the fused GPU path."""
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.data.batch import SubBatch, CorpusBatch
+    from marian_amd.models.model_factory import create_model
+
+    def opts(extra):
+        o = ConfigParser(MODE_TRAINING).parse_options(
+            ["--train-sets", "a", "b", "--vocabs", "a.yml", "b.yml",
+             "--dim-emb", "128", "--transformer-heads", "2",
+             "--transformer-dim-ffn", "256", "--enc-depth", "2",
+             "--dec-depth", "2"] + extra, validate=False)
+        o.set("dim-vocabs", [64, 64])
+        return o
+
+    for mtype in ("bert", "lm-transformer"):
+        m = create_model(opts(["--type", mtype])).cuda().bfloat16()
+        m.set_compute_dtype(torch.bfloat16)
+        idx = torch.randint(3, 64, (4, 12), device="cuda")
+        mask = torch.ones(4, 12, device="cuda")
+        b = CorpusBatch([SubBatch(idx, mask),
+                         SubBatch(idx.clone(), mask.clone())])
+        logits = m(b)
+        assert logits.shape[-1] == 64 and torch.isfinite(
+            logits.float()).all(), mtype
+        logits.float().sum().backward()
+
+
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
