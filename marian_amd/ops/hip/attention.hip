@@ -93,7 +93,10 @@ struct TStride {
 // LDS V^T image: [D][KVBLK] with row pad to dodge bank conflicts.
 constexpr int KVBLK = 32;
 constexpr int QBLK = 32;
-constexpr int VT_PAD = 8;  // elements
+constexpr int VT_PAD = 2;  // elements; row stride = odd dword count
+// (N+2 bf16 elems = odd number of 4B banks: stride 34 elem = 17 dwords,
+// 66 elem = 33 dwords -> consecutive rows walk all 64 LDS banks; the old
+// pad 8 gave 20/36-dword strides, gcd 4 with 64 banks = 4-way conflicts)
 
 // Forward kernel: NW waves per workgroup, each owning one 32-row q-tile of
 // the same (b,h); the KVBLK=64 K/V^T LDS stage is shared by all waves.
