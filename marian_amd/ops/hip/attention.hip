@@ -112,7 +112,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
                      int B, int H, int Tq, int Tk, float scale) {
   constexpr int KVB = 64;                       // kv tile (2 x 32-row mfma)
   __shared__ __hip_bfloat16 vt[D][KVB + VT_PAD];  // V^T image (shared)
-  __shared__ float obounce[NW][QBLK][D + 4];
+  __shared__ float obounce[NW][QBLK][D + 5];  // odd dword row stride (bank-conflict free)
 
   const int lane = threadIdx.x & 63;
   const int tid = threadIdx.x;
@@ -410,7 +410,7 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
                         TStride qs, TStride ks, TStride dos, TStride dqs,
                         int B, int H, int Tq, int Tk, float scale) {
   __shared__ __hip_bfloat16 kt[D][KVBLK + VT_PAD];  // K^T image (shared)
-  __shared__ float dq_b[NW][QBLK][D + 4];
+  __shared__ float dq_b[NW][QBLK][D + 5];  // odd dword row stride (bank-conflict free)
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -564,7 +564,7 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
                          int B, int H, int Tq, int Tk, float scale) {
   __shared__ __hip_bfloat16 qt[D][QBLK + VT_PAD];   // Q^T image (shared)
   __shared__ __hip_bfloat16 dot[D][QBLK + VT_PAD];  // dO^T image (shared)
-  __shared__ float kv_b[NW][KVBLK][D + 4];
+  __shared__ float kv_b[NW][KVBLK][D + 5];  // odd dword row stride (bank-conflict free)
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
