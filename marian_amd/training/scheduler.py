@@ -27,6 +27,8 @@ class Scheduler:
         self.lr_report = bool(options.get("lr-report", False))
         self.after_epochs = int(options.get("after-epochs", 0) or 0)
         self.after_batches = int(options.get("after-batches", 0) or 0)
+        # generic --after "Ne|Nu|Nt" (ref: training.h Train::run stop set)
+        self.after = SchedulingParameter.parse(options.get("after", "0e"))
         self.early_stopping = int(options.get("early-stopping", 10) or 0)
         self.base_lr = float(options.get("learn-rate", 1e-4))
         self.warmup = SchedulingParameter.parse(options.get("lr-warmup", "0"))
@@ -192,6 +194,13 @@ class Scheduler:
             return False
         if self.after_batches and st.batches >= self.after_batches:
             return False
+        if self.after:
+            if self.after.unit == "e" and st.epochs > self.after.n:
+                return False
+            if self.after.unit == "u" and st.batches >= self.after.n:
+                return False
+            if self.after.unit == "t" and st.labels_total >= self.after.n:
+                return False
         if self.early_stopping and st.stalled >= self.early_stopping:
             return False
         return True

@@ -89,3 +89,39 @@ def test_factored_transformer_train_step(tmp_path):
         -1, b.target.indices.unsqueeze(-1)).sum()
     loss.backward()
     assert m.trg_embed.factored.tables[0].grad is not None
+
+
+def test_factored_cli_train_and_decode(tmp_path):
+    """marian-train + marian-decoder end-to-end with an .fsv vocab."""
+    import subprocess, sys, os
+    fsv = tmp_path / "v.fsv"
+    words = ["the", "The", "dog", "dogs", "Dog", "run", "runs", "cat",
+             "cats", "Cat"]
+    lines = []
+    for w in words:
+        lemma = w.lower().rstrip("s")
+        caps = "_c1" if w[0].isupper() else "_c0"
+        pl = " _pl1" if w.endswith("s") and len(w) > 3 else ""
+        lines.append(f"{w}\t{lemma} {caps}{pl}")
+    fsv.write_text("\n".join(lines) + "\n")
+    src = tmp_path / "train.src"
+    trg = tmp_path / "train.trg"
+    import random
+    rng = random.Random(0)
+    pairs = [" ".join(rng.choices(words, k=rng.randint(2, 5)))
+             for _ in range(64)]
+    src.write_text("\n".join(pairs) + "\n")
+    trg.write_text("\n".join(pairs) + "\n")
+    model = tmp_path / "model.npz"
+    env = dict(os.environ, PYTHONPATH=os.getcwd())
+    r = subprocess.run(
+        [sys.executable, "-m", "marian_amd.cli.marian_train",
+         "--train-sets", str(src), str(trg),
+         "--vocabs", str(fsv), str(fsv), "--model", str(model),
+         "--dim-emb", "32", "--enc-depth", "1", "--dec-depth", "1",
+         "--transformer-heads", "4", "--transformer-dim-ffn", "64",
+         "--mini-batch", "16", "--after", "6u", "--disp-freq", "3",
+         "--cpu-threads", "1", "--max-length", "20", "--seed", "7"],
+        capture_output=True, text=True, env=env, timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert model.exists()
