@@ -58,6 +58,16 @@ def main():
         bench("attn_fwd  B512 H16 T64 D64",
               lambda: ext.attention_fwd(q2, k2, v2, None, False, D ** -0.5,
                                         False), iters, flops=f2)
+        o2, lse2 = ext.attention_fwd(q2, k2, v2, None, False, D ** -0.5, True)
+        do2 = torch.randn_like(o2)
+        bench("attn_bwd  B512 H16 T64 D64",
+              lambda: ext.attention_bwd(q2, k2, v2, o2, do2, lse2, None,
+                                        False, D ** -0.5),
+              iters, flops=int(2.5 * f2))
+        bench("attn_bwd  B512 H16 T64 D64 causal",
+              lambda: ext.attention_bwd(q2, k2, v2, o2, do2, lse2, None,
+                                        True, D ** -0.5),
+              iters, flops=int(2.5 * f2))
 
     if which in ("layernorm", "all"):
         R, d = 32768, 1024

@@ -409,7 +409,10 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
                         __hip_bfloat16* __restrict__ dQ,
                         TStride qs, TStride ks, TStride dos, TStride dqs,
                         int B, int H, int Tq, int Tk, float scale) {
-  __shared__ __hip_bfloat16 kt[D][KVBLK + VT_PAD];  // K^T image (shared)
+  // K^T image staged 2 tiles at a time: one cooperative pass + one
+  // barrier pair per 64 kv rows instead of per 32 (T<=64 shapes are
+  // staging/barrier-bound: 313us/call at B1024 T64 vs ~0.5us of MFMA)
+  __shared__ __hip_bfloat16 kt[D][2 * KVBLK + VT_PAD];
   __shared__ float dq_b[NW][QBLK][D + 5];  // odd dword row stride (bank-conflict free)
 
   const int tid = threadIdx.x;
@@ -456,15 +459,16 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
   const int q_hi = min((blockIdx.x * NW + NW) * QBLK, Tq);
   const int kv_end_blk = CAUSAL ? min(kv_len, q_hi) : kv_len;
   const int kv_end = CAUSAL ? min(kv_len, q0 + QBLK) : kv_len;
-  for (int k0 = 0; k0 < kv_end_blk; k0 += KVBLK) {
+  constexpr int STW = 2 * KVBLK;   // staging width (2 inner tiles)
+  for (int k0s = 0; k0s < kv_end_blk; k0s += STW) {
     {
       // cooperative K^T staging with all NW*64 threads
       const int rows_per_pass = (NW * 64) / (D / 8);
-      for (int kk = 0; kk < KVBLK; kk += rows_per_pass) {
+      for (int kk = 0; kk < STW; kk += rows_per_pass) {
         const int kl = kk + tid / (D / 8);
-        const int krow2 = k0 + kl;
+        const int krow2 = k0s + kl;
         const int dcol = (tid % (D / 8)) * 8;
-        if (kl < KVBLK) {
+        if (kl < STW) {
           bf16x8 kv = (krow2 < kv_len)
               ? *reinterpret_cast<const bf16x8*>(Kb + (size_t)krow2 * ks.s + dcol)
               : (bf16x8)(__bf16)0.f;
@@ -476,10 +480,10 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
     }
     __builtin_amdgcn_s_waitcnt(0);
     __builtin_amdgcn_s_barrier();
-    if (!wave_live || k0 >= kv_end) {
-      __builtin_amdgcn_s_barrier();
+    const int k_tile_hi = min(k0s + STW, kv_end_blk);
+    for (int k0 = k0s; k0 < k_tile_hi; k0 += KVBLK) {
+    if (!wave_live || k0 >= kv_end)
       continue;
-    }
     bf16x8 kfrag[D / 16], vfrag[D / 16];
     {
       const int krow = min(k0 + lq, Tk - 1);
@@ -515,11 +519,12 @@ void attn_bwd_dq_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
         bf16x8 ka = *reinterpret_cast<const bf16x8*>(
-            &kt[32 * dt2 + lq][16 * c + 8 * hi]);
+            &kt[32 * dt2 + lq][(k0 - k0s) + 16 * c + 8 * hi]);
         dqt[dt2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, dst_op[c],
                                                            dqt[dt2], 0, 0, 0);
       }
     }
+    }  // inner k0 tiles
     __builtin_amdgcn_s_barrier();
   }
 
@@ -562,8 +567,9 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
                          __hip_bfloat16* __restrict__ dV,
                          TStride qs, TStride ks, TStride dos, TStride dks,
                          int B, int H, int Tq, int Tk, float scale) {
-  __shared__ __hip_bfloat16 qt[D][QBLK + VT_PAD];   // Q^T image (shared)
-  __shared__ __hip_bfloat16 dot[D][QBLK + VT_PAD];  // dO^T image (shared)
+  // Q^T / dO^T images staged 2 tiles at a time (see dq kernel note)
+  __shared__ __hip_bfloat16 qt[D][2 * QBLK + VT_PAD];
+  __shared__ __hip_bfloat16 dot[D][2 * QBLK + VT_PAD];
   __shared__ float kv_b[NW][KVBLK][D + 5];  // odd dword row stride (bank-conflict free)
 
   const int tid = threadIdx.x;
@@ -608,15 +614,17 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
   const int k0_first = (blockIdx.x * NW) * KVBLK;
   const int q_start_blk = CAUSAL ? (k0_first / QBLK) * QBLK : 0;
   const int q_start_own = CAUSAL ? (k0 / QBLK) * QBLK : 0;
-  for (int q0 = q_start_blk; q0 < Tq; q0 += QBLK) {
+  constexpr int STW = 2 * QBLK;    // staging width (2 inner tiles)
+  const int q_start_stage = CAUSAL ? (k0_first / STW) * STW : 0;
+  for (int q0s = q_start_stage; q0s < Tq; q0s += STW) {
     {
       // cooperative Q^T/dO^T staging with all NW*64 threads
       const int rows_per_pass = (NW * 64) / (D / 8);
-      for (int qq = 0; qq < QBLK; qq += rows_per_pass) {
+      for (int qq = 0; qq < STW; qq += rows_per_pass) {
         const int qlocal = qq + tid / (D / 8);
-        const int qrow2 = q0 + qlocal;
+        const int qrow2 = q0s + qlocal;
         const int dcol = (tid % (D / 8)) * 8;
-        if (qlocal < QBLK) {
+        if (qlocal < STW) {
           const bool live = qrow2 < Tq;
           const int qr = live ? qrow2 : (Tq - 1);
           bf16x8 vq = *reinterpret_cast<const bf16x8*>(Qb + (size_t)qr * qs.s + dcol);
@@ -635,10 +643,10 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
     }
     __builtin_amdgcn_s_waitcnt(0);
     __builtin_amdgcn_s_barrier();
-    if (!wave_live || q0 < q_start_own) {
-      __builtin_amdgcn_s_barrier();
+    const int q_tile_hi = min(q0s + STW, Tq);
+    for (int q0 = max(q0s, q_start_blk); q0 < q_tile_hi; q0 += QBLK) {
+    if (!wave_live || q0 < q_start_own)
       continue;
-    }
     // per-q-tile fragments (direct loads)
     bf16x8 qfrag[D / 16], dofrag[D / 16];
     {
@@ -682,15 +690,16 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
         bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-            &dot[32 * dt2 + lq][16 * c + 8 * hi]);
+            &dot[32 * dt2 + lq][(q0 - q0s) + 16 * c + 8 * hi]);
         bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-            &qt[32 * dt2 + lq][16 * c + 8 * hi]);
+            &qt[32 * dt2 + lq][(q0 - q0s) + 16 * c + 8 * hi]);
         dv_acc[dt2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(p_op[c], dob,
                                                               dv_acc[dt2], 0, 0, 0);
         dk_acc[dt2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ds_op[c], qb,
                                                               dk_acc[dt2], 0, 0, 0);
       }
     }
+    }  // inner q0 tiles
     __builtin_amdgcn_s_barrier();
   }
 
