@@ -99,6 +99,8 @@ class Embedding(nn.Module):
         """Scaled word embeddings only (marian adds positions AFTER the
         decoder's target shift, so the two halves are separable)."""
         x = torch.nn.functional.embedding(indices, self.table(compute_dtype))
+        if getattr(self, "ulr", None) is not None:
+            x = self.ulr(indices, x)              # universal-rep mix (--ulr)
         return x * math.sqrt(self.dim)
 
     def add_positions(self, x: torch.Tensor, start_pos: int = 0):

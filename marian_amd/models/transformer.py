@@ -308,6 +308,16 @@ class TransformerModel(nn.Module):
         return self.output_logits(h, shortlist)
 
     # ------------------------------------------------------------------
+    def enable_ulr(self, options):
+        """Universal Language Representation on the source embedding
+        (--ulr; ref: layers/embedding.cpp ULREmbedding)."""
+        from .ulr import ULREmbedding
+        self.src_embed.ulr = ULREmbedding(self.src_vocab_size, self.dim,
+                                          options)
+        # register as a submodule so parameters train/save
+        self.src_embed.add_module("ulr_mod", self.src_embed.ulr)
+
+    # ------------------------------------------------------------------
     def enable_factored_vocab(self, fv, side: str = "trg"):
         """Switch a side to factored embeddings (+ factored output on trg)
         (ref: data/factored_vocab.cpp, layers/logits.cpp)."""

@@ -121,6 +121,8 @@ class Train:
         self.model = create_model(options)
         if hasattr(self.model, "resolve_bert_ids"):
             self.model.resolve_bert_ids(self.vocabs[0], options)
+        if options.get("ulr", False) and hasattr(self.model, "enable_ulr"):
+            self.model.enable_ulr(options)
         if hasattr(self.model, "enable_factored_vocab"):
             src_f = getattr(self.vocabs[0], "is_factored", False)
             trg_f = getattr(self.vocabs[-1], "is_factored", False)
