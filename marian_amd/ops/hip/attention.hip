@@ -737,6 +737,283 @@ void attn_bwd_dkv_kernel(const __hip_bfloat16* __restrict__ Q,
   }
 }
 
+// Merged small-T backward: for Tq,Tk <= 64 the split dQ/dKV kernels are
+// launch/staging-bound (91 TF vs 201 TF at T512). One 4-wave block per
+// (b,h): waves 0/1 own the two kv-tiles (dK/dV), waves 2/3 the two
+// q-tiles (dQ); Q^T, dO^T and K^T images are staged once and shared, the
+// dO*O rowsum is computed into LDS by the q-waves (no dsum kernel, no
+// global dsum tensor), and the fp32 epilogue bounce aliases the image
+// memory — two barriers total.
+template <int D, bool CAUSAL, bool HAS_LEN>
+__global__ __launch_bounds__(256)
+void attn_bwd_fused64_kernel(const __hip_bfloat16* __restrict__ Q,
+                             const __hip_bfloat16* __restrict__ K,
+                             const __hip_bfloat16* __restrict__ V,
+                             const __hip_bfloat16* __restrict__ O,
+                             const __hip_bfloat16* __restrict__ dO,
+                             const float* __restrict__ LSE,
+                             const int* __restrict__ kv_lens,
+                             __hip_bfloat16* __restrict__ dQ,
+                             __hip_bfloat16* __restrict__ dK,
+                             __hip_bfloat16* __restrict__ dV,
+                             TStride qs, TStride ks, TStride dos,
+                             TStride oss, TStride dqs, TStride dks,
+                             int B, int H, int Tq, int Tk, float scale) {
+  constexpr int TW = 64;                  // staged width
+  constexpr int IMG = D * (TW + VT_PAD);  // one bf16 image
+  constexpr int BNC = 32 * (D + 5);       // one wave's fp32 bounce slice
+  __shared__ __align__(16) char smem[4 * BNC * 4 > 3 * IMG * 2
+                                     ? 4 * BNC * 4 : 3 * IMG * 2];
+  __shared__ float dsum_sh[TW];
+  __shared__ float lse_sh[TW];
+  auto qt = [&](int r, int c) -> __hip_bfloat16& {
+    return reinterpret_cast<__hip_bfloat16*>(smem)[r * (TW + VT_PAD) + c];
+  };
+  auto dot_ = [&](int r, int c) -> __hip_bfloat16& {
+    return reinterpret_cast<__hip_bfloat16*>(smem)[IMG + r * (TW + VT_PAD) + c];
+  };
+  auto kt = [&](int r, int c) -> __hip_bfloat16& {
+    return reinterpret_cast<__hip_bfloat16*>(smem)[2 * IMG + r * (TW + VT_PAD) + c];
+  };
+  auto bounce = [&](int w, int r, int c) -> float& {
+    return reinterpret_cast<float*>(smem)[w * BNC + r * (D + 5) + c];
+  };
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;               // 0,1: kv role; 2,3: q role
+  const int hi = lane >> 5;
+  const int lq = lane & 31;
+  const int bh = blockIdx.x;
+  const int b = bh / H;
+  const int h = bh % H;
+  const int kv_len = HAS_LEN ? kv_lens[b] : Tk;
+  const __hip_bfloat16* Qb = Q + (size_t)b * qs.b + (size_t)h * qs.h;
+  const __hip_bfloat16* Kb = K + (size_t)b * ks.b + (size_t)h * ks.h;
+  const __hip_bfloat16* Vb = V + (size_t)b * ks.b + (size_t)h * ks.h;
+  const __hip_bfloat16* Ob = O + (size_t)b * oss.b + (size_t)h * oss.h;
+  const __hip_bfloat16* dOb = dO + (size_t)b * dos.b + (size_t)h * dos.h;
+
+  // ---- stage Q^T, dO^T, K^T cooperatively (256 threads, 8 lanes/row) --
+  {
+    const int rows_per_pass = 256 / (D / 8);          // 32
+    const int rl = tid / (D / 8);
+    const int dcol = (tid % (D / 8)) * 8;
+    for (int rr = 0; rr < TW; rr += rows_per_pass) {
+      const int r = rr + rl;
+      const bool ql = r < Tq;
+      const bool kl = r < kv_len && r < Tk;
+      bf16x8 vq = ql ? *reinterpret_cast<const bf16x8*>(
+                           Qb + (size_t)r * qs.s + dcol)
+                     : (bf16x8)(__bf16)0.f;
+      bf16x8 vd = ql ? *reinterpret_cast<const bf16x8*>(
+                           dOb + (size_t)r * dos.s + dcol)
+                     : (bf16x8)(__bf16)0.f;
+      bf16x8 vk = kl ? *reinterpret_cast<const bf16x8*>(
+                           Kb + (size_t)r * ks.s + dcol)
+                     : (bf16x8)(__bf16)0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        qt(dcol + j, r) = (__hip_bfloat16)(float)vq[j];
+        dot_(dcol + j, r) = (__hip_bfloat16)(float)vd[j];
+        kt(dcol + j, r) = (__hip_bfloat16)(float)vk[j];
+      }
+    }
+  }
+
+  // role tile base: kv waves own kv rows, q waves own q rows
+  const int t0 = (wid & 1) * 32;
+  bf16x8 rfrag[D / 16], dofrag[D / 16];   // role-row fragments
+  const bool q_role = wid >= 2;
+  {
+    const int row = t0 + lq;
+    const int lim = q_role ? Tq : min(kv_len, Tk);
+    const int rr = min(row, (q_role ? Tq : Tk) - 1);
+    const __hip_bfloat16* p0 = q_role ? (Qb + (size_t)rr * qs.s)
+                                      : (Kb + (size_t)rr * ks.s);
+    const __hip_bfloat16* p1 = q_role ? (dOb + (size_t)rr * dos.s)
+                                      : (Vb + (size_t)rr * ks.s);
+    const bool live = row < lim;
+#pragma unroll
+    for (int c = 0; c < D / 16; ++c) {
+      bf16x8 v0 = *reinterpret_cast<const bf16x8*>(p0 + 16 * c + 8 * hi);
+      bf16x8 v1 = *reinterpret_cast<const bf16x8*>(p1 + 16 * c + 8 * hi);
+      if (!live) {
+        v0 = (bf16x8)(__bf16)0.f;
+        v1 = (bf16x8)(__bf16)0.f;
+      }
+      rfrag[c] = v0;     // q-role: Q row; kv-role: K row
+      dofrag[c] = v1;    // q-role: dO row; kv-role: V row
+    }
+    if (q_role) {
+      // dsum + lse for my q row into LDS (shared with the kv waves)
+      float ds = 0.f;
+      const __hip_bfloat16* op = Ob + (size_t)rr * oss.s;
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        const bf16x8 ov = *reinterpret_cast<const bf16x8*>(op + 16 * c + 8 * hi);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) ds += (float)dofrag[c][j] * (float)ov[j];
+      }
+      ds += __shfl_xor(ds, 32, 64);
+      if (live && hi == 0) {
+        dsum_sh[row] = ds;
+        lse_sh[row] = LSE[(size_t)bh * Tq + row];
+      }
+    }
+  }
+  __builtin_amdgcn_s_waitcnt(0);
+  __builtin_amdgcn_s_barrier();
+
+  // ---- compute ------------------------------------------------------
+  f32x16 acc0[D / 32], acc1[D / 32];      // q: dQ,- ; kv: dV,dK
+#pragma unroll
+  for (int d2 = 0; d2 < D / 32; ++d2) {
+    acc0[d2] = (f32x16)0.f;
+    acc1[d2] = (f32x16)0.f;
+  }
+
+  if (q_role) {
+    const int q0 = t0;
+    const int q_abs = q0 + lq;
+    const float lse_l = (q_abs < Tq) ? lse_sh[q_abs] : 0.f;
+    const float ds_l = (q_abs < Tq) ? dsum_sh[q_abs] : 0.f;
+    const int kv_end = CAUSAL ? min(kv_len, q0 + 32) : kv_len;
+    for (int k0 = 0; k0 < kv_end; k0 += 32) {
+      bf16x8 kfrag[D / 16], vfrag[D / 16];
+      const int krow = min(k0 + lq, Tk - 1);
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        kfrag[c] = *reinterpret_cast<const bf16x8*>(
+            Kb + (size_t)krow * ks.s + 16 * c + 8 * hi);
+        vfrag[c] = *reinterpret_cast<const bf16x8*>(
+            Vb + (size_t)krow * ks.s + 16 * c + 8 * hi);
+      }
+      f32x16 st = (f32x16)0.f, dpt = (f32x16)0.f;
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag[c], rfrag[c], st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag[c], dofrag[c], dpt, 0, 0, 0);
+      }
+      float ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int k_abs = k0 + crow(r, hi);
+        const bool dead = (k_abs >= kv_len) || (CAUSAL && k_abs > q_abs)
+                          || (q_abs >= Tq);
+        const float pval = dead ? 0.f : __expf(st[r] * scale - lse_l);
+        ds[r] = pval * (dpt[r] - ds_l) * scale;
+      }
+      bf16x8 dst_op[2];
+      pack_p_to_b(ds, dst_op);
+#pragma unroll
+      for (int d2 = 0; d2 < D / 32; ++d2)
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+              &kt(32 * d2 + lq, k0 + 16 * c + 8 * hi));
+          acc0[d2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              ka, dst_op[c], acc0[d2], 0, 0, 0);
+        }
+    }
+  } else {
+    const int k0 = t0;
+    const int k_lane = k0 + lq;
+    const int q_start = CAUSAL ? (k0 / 32) * 32 : 0;
+    for (int q0 = q_start; q0 < Tq; q0 += 32) {
+      bf16x8 qfr[D / 16], dofr[D / 16];
+      const int qrow = min(q0 + lq, Tq - 1);
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        qfr[c] = *reinterpret_cast<const bf16x8*>(
+            Qb + (size_t)qrow * qs.s + 16 * c + 8 * hi);
+        bf16x8 dv = *reinterpret_cast<const bf16x8*>(
+            dOb + (size_t)qrow * dos.s + 16 * c + 8 * hi);
+        if (q0 + lq >= Tq) dv = (bf16x8)(__bf16)0.f;
+        dofr[c] = dv;
+      }
+      f32x16 st = (f32x16)0.f, dpt = (f32x16)0.f;
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr[c], rfrag[c], st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofr[c], dofrag[c], dpt, 0, 0, 0);
+      }
+      float pv[16], ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int q_r = q0 + crow(r, hi);
+        const bool qlive = q_r < Tq;
+        const float lse_r = qlive ? lse_sh[q_r] : 0.f;
+        const float dsum_r = qlive ? dsum_sh[q_r] : 0.f;
+        const bool dead = (k_lane >= kv_len) || (CAUSAL && k_lane > q_r)
+                          || !qlive;
+        const float pval = dead ? 0.f : __expf(st[r] * scale - lse_r);
+        pv[r] = pval;
+        ds[r] = pval * (dpt[r] - dsum_r) * scale;
+      }
+      bf16x8 p_op[2], ds_op[2];
+      pack_p_to_b(pv, p_op);
+      pack_p_to_b(ds, ds_op);
+#pragma unroll
+      for (int d2 = 0; d2 < D / 32; ++d2)
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          bf16x8 dob = *reinterpret_cast<const bf16x8*>(
+              &dot_(32 * d2 + lq, q0 + 16 * c + 8 * hi));
+          bf16x8 qb = *reinterpret_cast<const bf16x8*>(
+              &qt(32 * d2 + lq, q0 + 16 * c + 8 * hi));
+          acc0[d2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              p_op[c], dob, acc0[d2], 0, 0, 0);
+          acc1[d2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              ds_op[c], qb, acc1[d2], 0, 0, 0);
+        }
+    }
+  }
+  __builtin_amdgcn_s_barrier();   // images dead; bounce aliases them
+
+  // ---- epilogue: per-wave fp32 bounce -> bf16 rows -------------------
+  __hip_bfloat16* dKb = dK + (size_t)b * dks.b + (size_t)h * dks.h;
+  __hip_bfloat16* dVb = dV + (size_t)b * dks.b + (size_t)h * dks.h;
+  __hip_bfloat16* dQb = dQ + (size_t)b * dqs.b + (size_t)h * dqs.h;
+  const int npass = q_role ? 1 : 2;       // kv waves: dV then dK
+  for (int which = 0; which < npass; ++which) {
+    f32x16* acc = (which == 0) ? acc0 : acc1;
+    if (q_role) {
+      // dQ accumulator layout: lane lq = q row, crow = d rows
+#pragma unroll
+      for (int d2 = 0; d2 < D / 32; ++d2)
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          bounce(wid, lq, 32 * d2 + crow(r, hi)) = acc[d2][r];
+    } else {
+#pragma unroll
+      for (int d2 = 0; d2 < D / 32; ++d2)
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          bounce(wid, crow(r, hi), 32 * d2 + lq) = acc[d2][r];
+    }
+    const int lanes_per_row = D / 8;
+    const int rows_per_pass = 64 / lanes_per_row;
+    for (int rr = 0; rr < 32; rr += rows_per_pass) {
+      const int local = rr + lane / lanes_per_row;
+      const int row = t0 + local;
+      const int dcol = (lane % lanes_per_row) * 8;
+      __hip_bfloat16* out_base =
+          q_role ? dQb : ((which == 0) ? dVb : dKb);
+      const TStride& os_ = q_role ? dqs : dks;
+      const int lim = q_role ? Tq : Tk;
+      const bool live = row < lim && (q_role || row < kv_len);
+      if (row < lim) {
+        bf16x8 out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          out[j] = live ? (__bf16)bounce(wid, local, dcol + j) : (__bf16)0.f;
+        *reinterpret_cast<bf16x8*>(out_base + (size_t)row * os_.s + dcol) = out;
+      }
+    }
+  }
+}
+
 // Fused delta = rowsum(dO * O) in one pass (bf16 reads, fp32 out).
 // The torch composite (two fp32 upcasts + mul + a dim=-1 reduce) cost
 // ~680us per call at B1024; this is a single ~45us streaming kernel.
@@ -790,6 +1067,46 @@ std::vector<torch::Tensor> attention_bwd_into(
   TORCH_CHECK(view_ok(dq) && view_ok(dk) && view_ok(dv) &&
               dk.strides() == dv.strides(),
               "attention_bwd: grad outputs must be 8-aligned strided views");
+  const int* lens_ptr = nullptr;
+  torch::Tensor lensc;
+  if (kv_lens.has_value() && kv_lens->defined()) {
+    lensc = kv_lens->to(torch::kInt32).contiguous();
+    lens_ptr = lensc.data_ptr<int>();
+  }
+  const int* lens_ptr_early = lens_ptr;
+  auto stream_f = at::cuda::getCurrentHIPStream();
+  // Merged-kernel path: measured NEUTRAL non-causal (237us vs 235) and
+  // SLOWER causal (385 vs 229) at B512 T64 — the dual-role accumulators
+  // push past the register budget. Kept opt-in for round-2 iteration.
+  static const bool use_fused64 = getenv("MAMD_FUSED_BWD64") != nullptr;
+  if (use_fused64 && Tq <= 64 && Tk <= 64) {
+    // merged one-launch backward for the NMT training shapes
+    dim3 gf(B * H);
+    const bool has_len = lens_ptr_early != nullptr;
+#define RUNF(CAUSAL, HASLEN)                                                  \
+    hipLaunchKernelGGL((attn_bwd_fused64_kernel<64, CAUSAL, HASLEN>), gf,     \
+                       dim3(256), 0, stream_f,                                \
+                       reinterpret_cast<const __hip_bfloat16*>(qc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(kc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(vc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(oc.data_ptr()),\
+                       reinterpret_cast<const __hip_bfloat16*>(doc.data_ptr()),\
+                       lse.data_ptr<float>(), lens_ptr_early,                 \
+                       reinterpret_cast<__hip_bfloat16*>(dq.data_ptr()),      \
+                       reinterpret_cast<__hip_bfloat16*>(dk.data_ptr()),      \
+                       reinterpret_cast<__hip_bfloat16*>(dv.data_ptr()),      \
+                       tstride(qc), tstride(kc), tstride(doc), tstride(oc),   \
+                       tstride(dq), tstride(dk), B, H, Tq, Tk, (float)scale)
+    if (causal) {
+      if (has_len) RUNF(true, true);
+      else RUNF(true, false);
+    } else {
+      if (has_len) RUNF(false, true);
+      else RUNF(false, false);
+    }
+#undef RUNF
+    return {dq, dk, dv};
+  }
   // D = rowsum(dO o O) per query row, fp32 — single fused pass
   auto dsum = torch::empty({B, H, Tq},
                            qc.options().dtype(torch::kFloat32));
@@ -806,17 +1123,16 @@ std::vector<torch::Tensor> attention_bwd_into(
                        dsum.data_ptr<float>(), tstride(doc), tstride(oc),
                        H, Tq, nrows);
   }
-  const int* lens_ptr = nullptr;
-  torch::Tensor lensc;
-  if (kv_lens.has_value() && kv_lens->defined()) {
-    lensc = kv_lens->to(torch::kInt32).contiguous();
-    lens_ptr = lensc.data_ptr<int>();
-  }
   auto stream = at::cuda::getCurrentHIPStream();
   const int nqt = (Tq + 31) / 32;
   const int nkt = (Tk + 31) / 32;
-  const int NWQ = (nqt % 4 == 0) ? 4 : (nqt % 2 == 0 ? 2 : 1);
-  const int NWK = (nkt % 4 == 0) ? 4 : (nkt % 2 == 0 ? 2 : 1);
+  int NWQ = (nqt % 4 == 0) ? 4 : (nqt % 2 == 0 ? 2 : 1);
+  int NWK = (nkt % 4 == 0) ? 4 : (nkt % 2 == 0 ? 2 : 1);
+  if (const char* e = getenv("MAMD_BWD_NW")) {   // A/B experiments
+    NWQ = NWK = atoi(e);
+    if (nqt % NWQ) NWQ = 1;
+    if (nkt % NWK) NWK = 1;
+  }
   dim3 gq(nqt / NWQ, B * H);
   dim3 gk(nkt / NWK, B * H);
 #define RUNB1(NQ, NK, CAUSAL, HASLEN)                                         \
