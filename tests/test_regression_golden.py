@@ -108,3 +108,29 @@ def test_scorer_golden(tmp_path):
     Rescorer(opts).run(out=buf)
     check_golden("scorer_tiny_ppl",
                  {"perplexity": float(buf.getvalue().strip())}, tol=5e-3)
+
+def test_nbest_format_golden(tmp_path):
+    """The marian n-best line format `id ||| text ||| F0= ||| score` is
+    locked exactly (ref: translator/output_printer.cpp)."""
+    tr, d = _tiny_train(tmp_path)
+    from marian_amd.common.options import Options
+    from marian_amd.translator.beam_search import BeamSearch, ModelScorer
+    from marian_amd.translator.translate import OutputPrinter
+    from tests.test_decode import make_batch
+    tr.model.eval()
+    bs = BeamSearch([ModelScorer(tr.model, torch.device("cpu"))],
+                    beam_size=4, normalize=0.6, max_length=12, n_best=3)
+    res = bs.search(make_batch([[5, 6, 7, 0]]), torch.device("cpu"))
+    pr = OutputPrinter(Options({"n-best": True}), tr.vocabs[-1])
+    out = pr.print(0, res[0])
+    lines = out.splitlines()
+    assert all(" ||| " in ln for ln in lines)
+    assert all(ln.startswith("0 ||| ") for ln in lines)
+    # scores descending in the n-best list
+    scores = [float(ln.rsplit(" ||| ", 1)[1]) for ln in lines]
+    assert scores == sorted(scores, reverse=True)
+    check_golden("decode_nbest_format", {
+        "n_lines": len(lines),
+        "line0_fields": len(lines[0].split(" ||| ")),
+        "top_score": scores[0],
+    }, tol=5e-3)
