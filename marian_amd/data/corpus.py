@@ -31,8 +31,17 @@ class Corpus:
                  shuffle_in_ram: bool = False, seed: int = 1234,
                  align_path: Optional[str] = None,
                  weight_path: Optional[str] = None,
-                 right_left: bool = False):
+                 right_left: bool = False,
+                 all_caps_every: int = 0, title_case_every: int = 0,
+                 caps_chunk: int = 64):
         self.right_left = right_left  # --right-left: reverse target side
+        # --all-caps-every / --english-title-case-every robustness
+        # augmentation (ref: data/corpus_base.cpp maybe-uppercase). The
+        # reference applies it per mini-batch; here it is applied per
+        # caps_chunk consecutive sentences (batches are formed later).
+        self.all_caps_every = int(all_caps_every or 0)
+        self.title_case_every = int(title_case_every or 0)
+        self.caps_chunk = max(1, caps_chunk)
         assert len(paths) >= 1
         self.align_path = align_path     # --guided-alignment (pharaoh file)
         self.weight_path = weight_path   # --data-weighting
@@ -91,8 +100,18 @@ class Corpus:
         for pos, rows in source:
             tup = []
             ok = True
+            chunk = pos // self.caps_chunk
+            upper = self.all_caps_every and \
+                (chunk % self.all_caps_every) == self.all_caps_every - 1
+            title = (not upper) and self.title_case_every and \
+                (chunk % self.title_case_every) == self.title_case_every - 1
             for vi, text in enumerate(rows):
                 vocab = self.vocabs[min(vi, len(self.vocabs) - 1)]
+                if upper:
+                    text = text.upper()
+                elif title:
+                    text = " ".join(w[:1].upper() + w[1:]
+                                    for w in text.split(" "))
                 ids = vocab.encode(text, add_eos=True)
                 if len(ids) > self.max_length + 1:
                     if self.max_length_crop:

@@ -182,7 +182,11 @@ class Train:
                         seed=int(opts.get("seed", 0) or 1234),
                         align_path=None if ga in (None, "none", "") else ga,
                         weight_path=opts.get("data-weighting"),
-                        right_left=bool(opts.get("right-left", False)))
+                        right_left=bool(opts.get("right-left", False)),
+                        all_caps_every=int(opts.get("all-caps-every", 0) or 0),
+                        title_case_every=int(
+                            opts.get("english-title-case-every", 0) or 0),
+                        caps_chunk=int(opts.get("mini-batch", 64) or 64))
         corpus.epoch = self.state.epochs - 1
         shuffle = opts.get("shuffle", "data") != "none" \
             and not opts.get("no-shuffle", False)

@@ -105,3 +105,29 @@ def test_shuffle_deterministic(tiny_corpus):
     l2 = [t[0] for t in c2.iter_epoch()]
     assert l1 == l2
     assert l1 != sorted(l1)
+
+
+def test_all_caps_augmentation(tmp_path):
+    """--all-caps-every / --english-title-case-every (ref:
+    data/corpus_base.cpp maybe-uppercase robustness augmentation)."""
+    from marian_amd.data.corpus import Corpus
+    from marian_amd.data.vocab import Vocab
+    src = tmp_path / "a.src"
+    trg = tmp_path / "a.trg"
+    lines = [f"tok{i} word" for i in range(8)]
+    src.write_text("\n".join(lines) + "\n")
+    trg.write_text("\n".join(lines) + "\n")
+    v = Vocab.create_from_corpus(
+        [str(src)], max_size=0)
+    # add uppercase forms so encoding distinguishes
+    c = Corpus([str(src), str(trg)], [v, v], max_length=20,
+               all_caps_every=2, caps_chunk=2)
+    tuples = list(c.iter_epoch(shuffle=False))
+    assert len(tuples) == 8
+    # chunks of 2: chunk idx 1 and 3 (sentences 2,3 and 6,7) upper-cased
+    # -> unknown tokens (vocab only has lowercase) => encoded as UNK
+    unk = v.unk_id
+    s0 = tuples[0][1][0]
+    s2 = tuples[2][1][0]
+    assert unk not in s0[:-1]
+    assert all(t == unk for t in s2[:-1])
