@@ -101,6 +101,13 @@ class Embedding(nn.Module):
         x = torch.nn.functional.embedding(indices, self.table(compute_dtype))
         if getattr(self, "ulr", None) is not None:
             x = self.ulr(indices, x)              # universal-rep mix (--ulr)
+        wd = getattr(self, "word_dropout", 0.0)
+        if wd and self.training:
+            # --dropout-src/--dropout-trg: drop WHOLE word vectors
+            # (ref: layers/embedding.cpp dropout on the word dimension)
+            keep = (torch.rand(indices.shape, device=x.device)
+                    >= wd).to(x.dtype) / (1.0 - wd)
+            x = x * keep.unsqueeze(-1)
         return x * math.sqrt(self.dim)
 
     def add_positions(self, x: torch.Tensor, start_pos: int = 0):

@@ -140,6 +140,16 @@ class TransformerModel(nn.Module):
                 self.trg_vocab_size, d, drop,
                 options.get("transformer-train-position-embeddings", False))
 
+        self.src_embed.word_dropout = float(options.get("dropout-src", 0)
+                                            or 0.0)
+        self.trg_embed.word_dropout = float(options.get("dropout-trg", 0)
+                                            or 0.0)
+        # --embedding-fix-src/trg: freeze embedding tables
+        if options.get("embedding-fix-src", False):
+            self.src_embed.E.requires_grad_(False)
+        if options.get("embedding-fix-trg", False):
+            self.trg_embed.E.requires_grad_(False)
+
         self.post_emb_enc = PrePostProcessor(
             options.get("transformer-postprocess-emb", "d"), d, drop)
         self.post_emb_dec = PrePostProcessor(

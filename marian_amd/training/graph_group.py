@@ -101,8 +101,20 @@ class SyncGraphGroup:
         self.scaler = CostScaler(options.list("cost-scaling"),
                                  self.compute_dtype == torch.float16)
         self.check_grad_nan = bool(options.get("check-gradient-nan", False)) \
-            or self.scaler.enabled
+            or bool(options.get("check-nan", False)) or self.scaler.enabled
         self.clip_norm = float(options.get("clip-norm", 0.0) or 0.0)
+        # --dynamic-gradient-scaling factor: rescale outlier gradients to
+        # factor x the running-average norm (ref: graph_group.cpp
+        # dynamicGradientScaling, --gradient-norm-average-window)
+        dgs = [x for x in (options.list("dynamic-gradient-scaling") or [])
+               if str(x) not in ("fp16", "log")]
+        self.dgs_factor = float(dgs[0]) if dgs else 0.0
+        self.gnorm_window = int(
+            options.get("gradient-norm-average-window", 100) or 100)
+        self._gnorm_avg = 0.0
+        self._gnorm_count = 0
+        self.throw_on_divergence = bool(
+            options.get("throw-on-divergence", False))
         self._accum = 0
         self._accum_labels = 0
         self._accum_sents = 0
@@ -221,7 +233,8 @@ class SyncGraphGroup:
 
         skip = False
         stats = None
-        if self.check_grad_nan or self.clip_norm > 0:
+        if self.check_grad_nan or self.clip_norm > 0 \
+                or self.dgs_factor > 0:
             stats = opt.grad_stats().clone()  # [sumsq, nonfinite] one pass
             if self.world > 1:
                 self.comm.all_reduce_scalar(stats)
@@ -230,11 +243,26 @@ class SyncGraphGroup:
                 skip = True
 
         grad_norm = None
-        if not skip and self.clip_norm > 0:
+        if not skip and (self.clip_norm > 0 or self.dgs_factor > 0):
             sq = stats[0] / (grad_scale * grad_scale)
             grad_norm = float(sq.sqrt().item())
             self._last_gnorm = grad_norm
+            if self.dgs_factor > 0 and grad_norm > 0:
+                cap = self.dgs_factor * self._gnorm_avg
+                if self._gnorm_count >= 8 and grad_norm > cap > 0:
+                    # outlier: rescale the gradient down to the cap
+                    grad_scale *= grad_norm / cap
+                    used = cap
+                else:
+                    used = grad_norm
+                n = min(self._gnorm_count, self.gnorm_window - 1)
+                self._gnorm_avg = (self._gnorm_avg * n + used) / (n + 1)
+                self._gnorm_count += 1
 
+        if skip and self.throw_on_divergence and not self.scaler.enabled:
+            raise RuntimeError(
+                "Training diverged: non-finite gradient "
+                "(--throw-on-divergence)")
         if skip:
             self.scaler.on_overflow()
             log_info("Non-finite gradient, skipping update "

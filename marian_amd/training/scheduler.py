@@ -30,6 +30,11 @@ class Scheduler:
         # generic --after "Ne|Nu|Nt" (ref: training.h Train::run stop set)
         self.after = SchedulingParameter.parse(options.get("after", "0e"))
         self.early_stopping = int(options.get("early-stopping", 10) or 0)
+        self.disp_label_counts = bool(options.get("disp-label-counts", False))
+        le = (options.list("logical-epoch") or ["1e"])[0]
+        self.logical_epoch = SchedulingParameter.parse(le)
+        if self.logical_epoch.unit == "e" and self.logical_epoch.n <= 1:
+            self.logical_epoch = None   # default: real epochs
         self.base_lr = float(options.get("learn-rate", 1e-4))
         self.warmup = SchedulingParameter.parse(options.get("lr-warmup", "0"))
         self.warmup_start_rate = float(options.get("lr-warmup-start-rate", 0.0))
@@ -126,8 +131,17 @@ class Scheduler:
             elapsed = self.timer.elapsed()
             wps = st.words_disp / max(1e-9, elapsed)
             avg_cost = st.cost_sum / max(1, st.cost_count)
-            line = (f"Ep. {st.epochs} : Up. {st.batches} : Sen. {st.samples:,} : "
+            ep_disp = st.epochs
+            if self.logical_epoch:
+                le = self.logical_epoch
+                if le.unit == "u":
+                    ep_disp = f"{st.batches / le.n:.2f}"
+                elif le.unit == "t":
+                    ep_disp = f"{st.labels_total / le.n:.2f}"
+            line = (f"Ep. {ep_disp} : Up. {st.batches} : Sen. {st.samples:,} : "
                     f"Cost {avg_cost:.8f} : Time {elapsed:.2f}s : {wps:.2f} words/s")
+            if self.disp_label_counts:
+                line += f" : Avg. labels {st.words_disp}"
             if grad_norm is not None:
                 line += f" : gNorm {grad_norm:.4f}"
             if self.lr_report:
@@ -207,22 +221,38 @@ class Scheduler:
 
     # ------------------------------------------------------------------
     def validated(self, metric: str, value: float, lower_is_better: bool = True):
-        """Track best/stalled per metric (ref: validator.h trackBest)."""
+        """Track best/stalled per metric; improvements smaller than
+        --early-stopping-epsilon do not reset the stall counter
+        (ref: validator.h trackBest + --early-stopping-on)."""
         st = self.state
         v = st.validators.setdefault(metric, {"best": None, "stalled": 0})
         best = v["best"]
-        improved = best is None or (value < best if lower_is_better else value > best)
+        eps_list = [float(x) for x in
+                    (self.opt.list("early-stopping-epsilon") or [0.0])]
+        metrics = self.opt.list("valid-metrics") or ["cross-entropy"]
+        mi = metrics.index(metric) if metric in metrics else 0
+        eps = eps_list[min(mi, len(eps_list) - 1)]
+        improved = best is None or             (value < best - eps if lower_is_better else value > best + eps)
         if improved:
             v["best"] = value
             v["stalled"] = 0
-            if metric == self._first_metric():
-                st.stalled = 0
         else:
             v["stalled"] += 1
-            if metric == self._first_metric():
-                st.stalled += 1
-                st.max_stalled = max(st.max_stalled, st.stalled)
-                self._maybe_decay("stalled")
+        # combine per-metric stall counters per --early-stopping-on
+        mode = self.opt.get("early-stopping-on", "first")
+        tracked = [m for m in (metrics if mode != "first"
+                               else metrics[:1])]
+        stalls = [st.validators.get(m, {"stalled": 0})["stalled"]
+                  for m in tracked if m in st.validators]
+        stalls = stalls or [0]
+        # "all": stop only when ALL stall (combined = min);
+        # "any": stop when ANY stalls (combined = max); "first": first metric
+        combined = min(stalls) if mode == "all" else max(stalls)
+        prev = st.stalled
+        st.stalled = combined
+        if combined > prev:
+            st.max_stalled = max(st.max_stalled, st.stalled)
+            self._maybe_decay("stalled")
         return improved
 
     def _first_metric(self):

@@ -123,6 +123,17 @@ class Train:
             self.model.resolve_bert_ids(self.vocabs[0], options)
         if options.get("ulr", False) and hasattr(self.model, "enable_ulr"):
             self.model.enable_ulr(options)
+        pre = options.get("pretrained-model", "")
+        if pre and os.path.exists(pre):
+            # --pretrained-model: initialize matching parameters from an
+            # existing checkpoint (ref: models/model_factory.cpp)
+            from ..common import io_npz
+            items, _ = io_npz.load_items(pre)
+            self.model.load_marian_items(items, strict=False)
+            log_info("Initialized parameters from %s", pre)
+        embv = options.list("embedding-vectors") or []
+        if embv:
+            self._load_embedding_vectors(embv, options)
         if hasattr(self.model, "enable_factored_vocab"):
             src_f = getattr(self.vocabs[0], "is_factored", False)
             trg_f = getattr(self.vocabs[-1], "is_factored", False)
@@ -213,6 +224,39 @@ class Train:
             return None
 
     # ------------------------------------------------------------------
+    def _load_embedding_vectors(self, paths, options):
+        """--embedding-vectors: word2vec-style text vectors ("V d" header,
+        then "word v1 v2 ..." lines) into the embedding tables, optionally
+        L2-normalized (--embedding-normalization)
+        (ref: layers/embedding.cpp + node_initializers fromWord2vec)."""
+        import numpy as np
+        import torch
+        targets = [(self.vocabs[0], self.model.src_embed),
+                   (self.vocabs[-1], self.model.trg_embed)]
+        for i, path in enumerate(paths[:2]):
+            if not os.path.exists(path):
+                continue
+            vocab, emb = targets[min(i, 1)]
+            t2i = getattr(vocab, "token2id", {})
+            with torch.no_grad():
+                n = 0
+                with open(path, encoding="utf-8") as f:
+                    first = f.readline().split()
+                    if len(first) > 2:  # no header: it is a data line
+                        f.seek(0)
+                    for line in f:
+                        parts = line.rstrip("\n").split(" ")
+                        if len(parts) < 3 or parts[0] not in t2i:
+                            continue
+                        vec = torch.tensor([float(x) for x in parts[1:]])
+                        if vec.numel() != emb.E.shape[1]:
+                            continue
+                        if options.get("embedding-normalization", False):
+                            vec = vec / vec.norm().clamp_min(1e-8)
+                        emb.E[t2i[parts[0]]] = vec.to(emb.E.dtype)
+                        n += 1
+            log_info("Loaded %d embedding vectors from %s", n, path)
+
     def run(self):
         signal.signal(signal.SIGTERM, _on_sigterm)
         self.scheduler.started()

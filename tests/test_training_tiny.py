@@ -295,3 +295,75 @@ def test_sqlite_corpus(workdir, tmp_path):
     tr = Train(opts)
     tr.run()
     assert tr.state.batches == 5
+
+
+def test_embedding_vectors_and_pretrained(tmp_path):
+    """--embedding-vectors (word2vec text) + --pretrained-model init."""
+    import numpy as np
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    from marian_amd.training.train import Train
+
+    src, trg = str(tmp_path / "c.src"), str(tmp_path / "c.trg")
+    write_synthetic_corpus(src, trg, 40, 30, 3, 7, seed=5)
+    # build a vocab first by running a tiny config once
+    argv = ["--train-sets", src, trg,
+            "--vocabs", str(tmp_path / "v.src.yml"), str(tmp_path / "v.trg.yml"),
+            "--model", str(tmp_path / "m.npz"),
+            "--dim-emb", "16", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "32",
+            "--mini-batch", "8", "--after", "2u", "--cpu-threads", "1",
+            "--seed", "3"]
+    o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    t = Train(o)
+    t.run()
+    # word2vec file for two known tokens
+    from marian_amd.data.vocab import Vocab
+    v = Vocab.load(str(tmp_path / "v.src.yml"))
+    toks = [v.id_to_token(3), v.id_to_token(4)]
+    vecs = tmp_path / "w2v.txt"
+    with open(vecs, "w") as f:
+        f.write("2 16\n")
+        for i, tok in enumerate(toks):
+            f.write(tok + " " + " ".join(str(float(i + 1))
+                                         for _ in range(16)) + "\n")
+    o2 = ConfigParser(MODE_TRAINING).parse_options(
+        argv[:-2] + ["--seed", "4", "--no-reload",
+                     "--model", str(tmp_path / "m2.npz"),
+                     "--embedding-vectors", str(vecs), str(vecs),
+                     "--pretrained-model", str(tmp_path / "m.npz"),
+                     "--embedding-normalization"], validate=False)
+    t2 = Train(o2)
+    row = t2.model.src_embed.E[v.token2id[toks[0]]]
+    assert torch.allclose(row, row.mean().expand_as(row), atol=1e-5)
+    assert abs(float(row.norm()) - 1.0) < 1e-4   # normalized
+
+
+def test_dynamic_gradient_scaling_caps_outliers():
+    """--dynamic-gradient-scaling rescales outlier gradient norms to
+    factor x running average (ref: graph_group.cpp)."""
+    from marian_amd.common.options import Options
+    from marian_amd.data.synthetic import synthetic_batch
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.training.graph_group import SyncGraphGroup
+    o = Options({
+        "type": "transformer", "dim-emb": 16, "enc-depth": 1, "dec-depth": 1,
+        "transformer-heads": 2, "transformer-dim-ffn": 32,
+        "transformer-preprocess": "", "transformer-postprocess": "dan",
+        "transformer-postprocess-emb": "d", "transformer-postprocess-top": "",
+        "transformer-ffn-activation": "relu", "transformer-dropout": 0.0,
+        "dim-vocabs": [30, 30], "tied-embeddings-all": True,
+        "precision": ["float32", "float32"],
+        "cost-type": "ce-sum", "label-smoothing": 0.0,
+        "optimizer": "sgd", "learn-rate": 0.01, "clip-norm": 0.0,
+        "dynamic-gradient-scaling": [2.0],
+        "gradient-norm-average-window": 16,
+    })
+    torch.manual_seed(0)
+    gg = SyncGraphGroup(o, create_model(o), torch.device("cpu"))
+    gg.set_lr(0.01)
+    assert gg.dgs_factor == 2.0
+    for step in range(12):
+        gg.update(synthetic_batch(4, 6, 6, 30, seed=step))
+    assert gg._gnorm_count == 12
+    assert gg._gnorm_avg > 0
