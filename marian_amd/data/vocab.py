@@ -123,7 +123,15 @@ class Vocab:
 
     def encode(self, line: str, add_eos: bool = True) -> List[int]:
         if self._spm is not None:
-            ids = self._spm.encode(line, out_type=int)
+            alpha = getattr(self, "sample_alpha", 0.0)
+            if alpha:
+                # --sentencepiece-alphas: subword-regularization sampling
+                # (ref: data/sentencepiece_vocab.cpp)
+                ids = self._spm.encode(line, out_type=int,
+                                       enable_sampling=True, alpha=alpha,
+                                       nbest_size=-1)
+            else:
+                ids = self._spm.encode(line, out_type=int)
         else:
             ids = [self.token2id.get(t, self.unk_id) for t in line.split()]
         if add_eos:
@@ -135,6 +143,9 @@ class Vocab:
         if remove_eos:
             ids = [i for i in ids if i != self.eos_id]
         if self._spm is not None:
+            if not getattr(self, "spm_decode", True):
+                # --no-spm-decode: emit raw subword pieces
+                return " ".join(self._spm.id_to_piece(int(i)) for i in ids)
             return self._spm.decode(ids)
         return " ".join(self.id2token.get(i, UNK_STR) for i in ids)
 

@@ -261,6 +261,9 @@ class ShardedOptimizer:
         self._grad_stats = None
         self.quantize_bits = int(options.get("quantize-bits", 0) or 0)
         self.quantize_log = bool(options.get("quantize-log-based", False))
+        self.quantize_opt_steps = int(
+            options.get("quantize-optimization-steps", 0) or 0)
+        self.options = options
 
     # ------------------------------------------------------------------
     def grad_shard_for_bucket(self, bi: int) -> torch.Tensor:
@@ -310,21 +313,56 @@ class ShardedOptimizer:
         bump_weight_version()
 
     # ------------------------------------------------------------------
+    def _quant_mask(self, w: torch.Tensor):
+        """Shard-local mask of entries to quantize: matrices always,
+        biases only with --quantize-biases (ref: optimizers/quantizer.cpp
+        isQuantizable)."""
+        if getattr(self, "_qmask", None) is not None \
+                and self._qmask.numel() == w.numel():
+            return self._qmask
+        mask = torch.zeros(w.numel(), dtype=torch.bool, device=w.device)
+        qbias = bool(self.options.get("quantize-biases", False)) \
+            if hasattr(self, "options") else False
+        shard_off = 0
+        for b, (o, sz) in zip(self.store.buckets, self.seg):
+            S = b.padded // self.world
+            lo = self.rank * S
+            # walk this bucket's params, intersect with my shard slice
+            for name, p in b.params:
+                boff = self.store.offsets[name][1]
+                a0, a1 = boff, boff + p.numel()
+                i0, i1 = max(a0, lo), min(a1, lo + S)
+                if i1 > i0 and (p.dim() > 1 or qbias):
+                    mask[shard_off + (i0 - lo):shard_off + (i1 - lo)] = True
+            shard_off += sz
+        self._qmask = mask
+        return mask
+
     def _requantize(self, w: torch.Tensor) -> None:
         """Snap weights to 2^bits levels (fixed-point, or log-scale with
-        --quantize-log-based), keeping the max-abs scale."""
+        --quantize-log-based), keeping the max-abs scale; the scale is
+        refined with --quantize-optimization-steps least-squares passes."""
         bits = self.quantize_bits
-        amax = w.abs().max().clamp_min(1e-8)
+        mask = self._quant_mask(w)
+        wq = w[mask]
+        amax = wq.abs().max().clamp_min(1e-8)
         if self.quantize_log:
-            sign = w.sign()
-            mag = (w.abs() / amax).clamp_min(1e-12)
+            sign = wq.sign()
+            mag = (wq.abs() / amax).clamp_min(1e-12)
             levels = 2 ** (bits - 1) - 1
             exp = torch.round(torch.log2(mag)).clamp_(min=-float(levels))
-            w.copy_(sign * amax * torch.pow(2.0, exp) * (mag > 1e-9))
+            w[mask] = sign * amax * torch.pow(2.0, exp) * (mag > 1e-9)
         else:
             levels = 2 ** (bits - 1) - 1
             step = amax / levels
-            w.copy_(torch.round(w / step) * step)
+            q = torch.round(wq / step)
+            # --quantize-optimization-steps: least-squares refine the step
+            # (scale) against the fixed level assignment, re-assign, repeat
+            for _ in range(int(self.quantize_opt_steps)):
+                denom = float((q * q).sum().clamp_min(1e-8))
+                step = float((wq * q).sum()) / denom
+                q = torch.round(wq / max(step, 1e-12)).clamp_(-levels, levels)
+            w[mask] = q * step
 
     # ------------------------------------------------------------------
     # --exponential-smoothing swap (ref: communicator swapParams)

@@ -115,6 +115,11 @@ class Train:
             init_process_group(device=self.device)
 
         self.vocabs = load_or_create_vocabs(options)
+        alphas = [float(a) for a in
+                  (options.list("sentencepiece-alphas") or [])]
+        for i, v in enumerate(self.vocabs):
+            if i < len(alphas) and alphas[i] > 0:
+                v.sample_alpha = alphas[i]
         dims = [len(v) for v in self.vocabs]
         options.set("dim-vocabs", dims)
 

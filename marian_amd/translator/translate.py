@@ -117,6 +117,8 @@ class Translator:
             raise ValueError("marian-decoder needs --vocabs")
         self.src_vocab = Vocab.load(vocab_paths[0])
         self.trg_vocab = Vocab.load(vocab_paths[-1])
+        if options.get("no-spm-decode", False):
+            self.trg_vocab.spm_decode = False
         self.models = [load_model_for_inference(m, options, self.device)
                        for m in options.list("models")]
         scorers = [ModelScorer(m, self.device) for m in self.models]

@@ -243,18 +243,35 @@ def test_mid_epoch_corpus_restore(workdir):
 
 
 def test_quantize_bits(workdir):
-    """--quantize-bits snaps weights to 2^bits levels after each update."""
+    """--quantize-bits snaps MATRIX weights to 2^bits levels after each
+    update; biases stay full precision unless --quantize-biases."""
     tmp_path, src, trg = workdir
     opts = _opts(tmp_path, src, src, extra=["--quantize-bits", "4",
+                                            "--quantize-biases",
                                             "--after-batches", "3"])
     tr = Train(opts)
     tr.run()
     import numpy as np
-    w = tr.graph_group.optimizer.master.numpy()
+    opt = tr.graph_group.optimizer
+    w = opt.master.numpy()
     amax = np.abs(w).max()
     step = amax / (2 ** 3 - 1)
     frac = np.abs(w / step - np.round(w / step))
     assert frac.max() < 1e-4  # all weights on the quantization grid
+
+    # without --quantize-biases, bias entries stay off-grid (full precision)
+    opts2 = _opts(tmp_path, src, src, extra=["--quantize-bits", "4",
+                                             "--after-batches", "3",
+                                             "--no-reload", "--seed", "99"])
+    tr2 = Train(opts2)
+    tr2.run()
+    opt2 = tr2.graph_group.optimizer
+    mask = opt2._quant_mask(opt2.master).numpy()
+    assert mask.any() and not mask.all()
+    wq = opt2.master.numpy()[mask]
+    step2 = np.abs(wq).max() / (2 ** 3 - 1)
+    frac2 = np.abs(wq / step2 - np.round(wq / step2))
+    assert frac2.max() < 1e-3
 
 
 def test_lr_decay_on_stalled():
