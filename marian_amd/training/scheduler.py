@@ -53,18 +53,23 @@ class Scheduler:
     # ------------------------------------------------------------------
     def learning_rate(self) -> float:
         """ref: scheduler.h getScheduledLRate: warmup then inv-sqrt decay,
-        both in units of updates (or labels with 't')."""
-        n = self.state.batches + 1
+        both in units of updates (or labels with 't'). --lr-warmup-at-reload
+        restarts warmup relative to the resume point; --lr-warmup-cycle
+        repeats the warmup ramp every `warm` updates."""
+        n = self.state.batches + 1 - getattr(self.state, "warmup_start", 0)
         lr = self.base_lr
         warm = self.warmup.n
+        if warm > 0 and bool(self.opt.get("lr-warmup-cycle", False)):
+            n = (n - 1) % warm + 1
         if warm > 0 and n <= warm:
             s = self.warmup_start_rate
             progress = n / warm
             lr = lr * ((1.0 - s) * progress + s) if s else lr * progress
         if self.decay_inv_sqrt.n > 0:
             start = self.decay_inv_sqrt.n
-            if n > start:
-                lr = lr * math.sqrt(start / n)
+            nd = self.state.batches + 1
+            if nd > start:
+                lr = lr * math.sqrt(start / nd)
         lr *= self.state.factor
         self.state.eta = lr
         return lr

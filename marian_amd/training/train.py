@@ -160,6 +160,8 @@ class Train:
                 prog = self.model_path + ".progress.yml"
                 if os.path.exists(prog):
                     self.state = TrainingState.load(prog)
+                    if options.get("lr-warmup-at-reload", False):
+                        self.state.warmup_start = self.state.batches
                 if options.get("valid-reset-all", False):
                     self.state.validators = {}
                     self.state.stalled = 0

@@ -49,6 +49,7 @@ class TrainingState:
     labels_total: int = 0        # total target labels
     stalled: int = 0
     max_stalled: int = 0
+    warmup_start: int = 0        # --lr-warmup-at-reload base
     validators: Dict[str, dict] = field(default_factory=dict)
     eta: float = 0.0             # current learning rate
     factor: float = 1.0          # lr decay factor

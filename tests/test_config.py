@@ -89,3 +89,26 @@ def test_validation_requires_train_sets():
     p = ConfigParser(MODE_TRAINING)
     with pytest.raises(ValueError):
         p.parse_options([])
+
+
+def test_lr_warmup_cycle_and_reload_base():
+    """--lr-warmup-cycle repeats the ramp; warmup_start shifts it."""
+    from marian_amd.common.options import Options
+    from marian_amd.training.scheduler import Scheduler
+    from marian_amd.training.training_state import TrainingState
+    st = TrainingState()
+    o = Options({"learn-rate": 1.0, "lr-warmup": "4", "lr-warmup-cycle": True,
+                 "valid-metrics": []})
+    sch = Scheduler(o, st)
+    lrs = []
+    for b in range(8):
+        st.batches = b
+        lrs.append(sch.learning_rate())
+    assert lrs[0] == lrs[4] and lrs[3] == lrs[7]  # cyclic ramp
+    # warmup-at-reload base: restart ramp at batch 100
+    o2 = Options({"learn-rate": 1.0, "lr-warmup": "4", "valid-metrics": []})
+    st2 = TrainingState()
+    st2.batches = 100
+    st2.warmup_start = 100
+    sch2 = Scheduler(o2, st2)
+    assert sch2.learning_rate() == 0.25  # first warmup step again
