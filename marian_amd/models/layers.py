@@ -303,23 +303,38 @@ _ACT = {"relu": torch.nn.functional.relu,
 
 
 class FFN(nn.Module):
-    """Transformer FFN block, marian names W1/b1/W2/b2
-    (ref: transformer.h LayerFFN)."""
+    """Transformer FFN block, marian names W1/b1..Wn/bn
+    (ref: transformer.h LayerFFN, --transformer-ffn-depth). Depth-2 is
+    the hot path (fused bias+ReLU GEMM epilogue on the first layer);
+    deeper stacks keep the hidden width for all inner layers."""
 
     def __init__(self, dim: int, dim_ffn: int, activation: str = "relu",
-                 dropout: float = 0.0):
+                 dropout: float = 0.0, depth: int = 2):
         super().__init__()
-        self.W1 = Dense(dim, dim_ffn)
-        self.W2 = Dense(dim_ffn, dim)
+        self.depth = max(1, int(depth or 2))
+        dims = [dim] + [dim_ffn] * (self.depth - 1) + [dim]
+        self.Ws = nn.ModuleList([Dense(dims[i], dims[i + 1])
+                                 for i in range(self.depth)])
         self.act_name = activation
         self.act = _ACT[activation]
         self.dropout = dropout
 
+    @property
+    def W1(self):
+        return self.Ws[0]
+
+    @property
+    def W2(self):
+        return self.Ws[-1]
+
     def forward(self, x):
-        h = ops.linear_act(x, self.W1.W, self.W1.b, self.act_name)
-        if self.dropout > 0 and self.training:
-            h = torch.nn.functional.dropout(h, self.dropout, True)
-        return self.W2(h)
+        h = x
+        for li in range(self.depth - 1):
+            h = ops.linear_act(h, self.Ws[li].W, self.Ws[li].b,
+                               self.act_name)
+            if self.dropout > 0 and self.training:
+                h = torch.nn.functional.dropout(h, self.dropout, True)
+        return self.Ws[-1](h)
 
 
 class AverageAttention(nn.Module):

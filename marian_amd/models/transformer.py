@@ -48,7 +48,8 @@ class EncoderLayer(nn.Module):
             d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
         self.ffn = FFN(d, opt["transformer-dim-ffn"],
                        opt.get("transformer-ffn-activation", "relu"),
-                       opt.get("transformer-dropout-ffn", 0.0))
+                       opt.get("transformer-dropout-ffn", 0.0),
+                       opt.get("transformer-ffn-depth", 2))
 
     def forward(self, x, mask, lens=None):
         h = self.pre_self(x)
@@ -83,8 +84,11 @@ class DecoderLayer(nn.Module):
                 d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
         self.context_attn = MultiHeadAttention(
             d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
+        dec_depth_ffn = opt.get("transformer-decoder-ffn-depth", 0) \
+            or opt.get("transformer-ffn-depth", 2)
         self.ffn = FFN(d, dim_ffn, opt.get("transformer-ffn-activation", "relu"),
-                       opt.get("transformer-dropout-ffn", 0.0))
+                       opt.get("transformer-dropout-ffn", 0.0),
+                       dec_depth_ffn)
 
     def forward(self, x, enc_out, src_mask, causal: bool,
                 cache: Optional[dict] = None, layer_id: int = 0,
@@ -417,10 +421,9 @@ class TransformerModel(nn.Module):
                 m[prefix + "_Wo_ln_bias"] = post.ln.bias
 
         def ffn(prefix: str, f: FFN, post: PrePostProcessor):
-            m[prefix + "_W1"] = f.W1.W
-            m[prefix + "_b1"] = f.W1.b
-            m[prefix + "_W2"] = f.W2.W
-            m[prefix + "_b2"] = f.W2.b
+            for li, d in enumerate(f.Ws, start=1):
+                m[f"{prefix}_W{li}"] = d.W
+                m[f"{prefix}_b{li}"] = d.b
             if hasattr(post, "ln"):
                 m[prefix + "_ffn_ln_scale"] = post.ln.scale
                 m[prefix + "_ffn_ln_bias"] = post.ln.bias

@@ -178,3 +178,34 @@ def test_tied_layers():
     logits.float().pow(2).mean().backward()
     # shared params got gradients once
     assert m.encoder[0].self_attn.W_qkv.grad is not None
+
+
+def test_ffn_depth_3():
+    """--transformer-ffn-depth builds deeper FFN stacks with marian
+    W1..Wn naming (ref: transformer.h LayerFFN)."""
+    import torch
+    from marian_amd.common.options import Options
+    from marian_amd.models.model_factory import create_model
+    o = Options({
+        "type": "transformer", "dim-emb": 16, "enc-depth": 1, "dec-depth": 1,
+        "transformer-heads": 2, "transformer-dim-ffn": 32,
+        "transformer-ffn-depth": 3, "transformer-decoder-ffn-depth": 2,
+        "transformer-preprocess": "", "transformer-postprocess": "dan",
+        "transformer-postprocess-emb": "d", "transformer-postprocess-top": "",
+        "transformer-ffn-activation": "relu", "transformer-dropout": 0.0,
+        "dim-vocabs": [20, 20], "tied-embeddings-all": True,
+        "transformer-decoder-dim-ffn": 0,
+    })
+    m = create_model(o)
+    assert len(m.encoder[0].ffn.Ws) == 3
+    assert len(m.decoder[0].ffn.Ws) == 2
+    names = m._name_map()
+    assert "encoder_l1_ffn_W3" in names
+    assert "decoder_l1_ffn_W2" in names and "decoder_l1_ffn_W3" not in names
+    # hidden width preserved in the middle layer
+    assert m.encoder[0].ffn.Ws[1].W.shape == (32, 32)
+    from marian_amd.data.batch import SubBatch, CorpusBatch
+    idx = torch.randint(2, 20, (2, 5))
+    b = CorpusBatch([SubBatch(idx, torch.ones(2, 5)),
+                     SubBatch(idx.clone(), torch.ones(2, 5))])
+    m(b).sum().backward()
