@@ -171,7 +171,8 @@ class MultiHeadAttention(nn.Module):
     no per-forward weight concat); marian's per-matrix names are exported
     as views (see _name_map in transformer.py)."""
 
-    def __init__(self, dim: int, heads: int, dropout: float = 0.0):
+    def __init__(self, dim: int, heads: int, dropout: float = 0.0,
+                 no_projection: bool = False):
         super().__init__()
         assert dim % heads == 0
         self.dim = dim
@@ -181,7 +182,9 @@ class MultiHeadAttention(nn.Module):
         self.b_qkv = nn.Parameter(torch.zeros(3 * dim))
         for i in range(3):
             nn.init.xavier_uniform_(self.W_qkv[:, i * dim:(i + 1) * dim])
-        self.Wo = Dense(dim, dim)
+        # --transformer-no-projection: heads re-joined without Wo
+        self.Wo = (lambda x: x) if no_projection else Dense(dim, dim)
+        self.no_projection = no_projection
         self.dropout = dropout
 
     # marian-named views of the fused parameter

@@ -45,7 +45,9 @@ class EncoderLayer(nn.Module):
         self.pre_ffn = PrePostProcessor(opt["transformer-preprocess"], d, drop)
         self.post_ffn = PrePostProcessor(opt["transformer-postprocess"], d, drop)
         self.self_attn = MultiHeadAttention(
-            d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
+            d, opt["transformer-heads"],
+            opt.get("transformer-dropout-attention", 0.0),
+            opt.get("transformer-no-projection", False))
         self.ffn = FFN(d, opt["transformer-dim-ffn"],
                        opt.get("transformer-ffn-activation", "relu"),
                        opt.get("transformer-dropout-ffn", 0.0),
@@ -81,9 +83,13 @@ class DecoderLayer(nn.Module):
                 not opt.get("transformer-aan-nogate", False), drop)
         else:
             self.self_attn = MultiHeadAttention(
-                d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
+                d, opt["transformer-heads"],
+                opt.get("transformer-dropout-attention", 0.0),
+                opt.get("transformer-no-projection", False))
         self.context_attn = MultiHeadAttention(
-            d, opt["transformer-heads"], opt.get("transformer-dropout-attention", 0.0))
+            d, opt["transformer-heads"],
+            opt.get("transformer-dropout-attention", 0.0),
+            opt.get("transformer-no-projection", False))
         dec_depth_ffn = opt.get("transformer-decoder-ffn-depth", 0) \
             or opt.get("transformer-ffn-depth", 2)
         self.ffn = FFN(d, dim_ffn, opt.get("transformer-ffn-activation", "relu"),
@@ -414,8 +420,9 @@ class TransformerModel(nn.Module):
             m[prefix + "_bk"] = a.b_slice(1)
             m[prefix + "_Wv"] = a.w_slice(2)
             m[prefix + "_bv"] = a.b_slice(2)
-            m[prefix + "_Wo"] = a.Wo.W
-            m[prefix + "_bo"] = a.Wo.b
+            if not getattr(a, "no_projection", False):
+                m[prefix + "_Wo"] = a.Wo.W
+                m[prefix + "_bo"] = a.Wo.b
             if hasattr(post, "ln"):
                 m[prefix + "_Wo_ln_scale"] = post.ln.scale
                 m[prefix + "_Wo_ln_bias"] = post.ln.bias
