@@ -21,12 +21,16 @@ from .corpus import SentenceTuple
 class BatchGenerator:
     def __init__(self, source: Iterable[SentenceTuple],
                  mini_batch: int = 64, mini_batch_words: int = 0,
+                 batch_scale_fn=None,
                  maxi_batch: int = 100, maxi_batch_sort: str = "trg",
                  mini_batch_fit_table=None, prefetch: bool = True,
                  pad_ids: Optional[List[int]] = None):
         self.source = source
         self.mini_batch = mini_batch
         self.mini_batch_words = mini_batch_words
+        # --mini-batch-warmup: callable -> multiplier in (0, 1] applied to
+        # the sentence/token budget (ref: graph_group.cpp ramp-up logic)
+        self.batch_scale_fn = batch_scale_fn
         self.maxi_batch = maxi_batch
         self.maxi_batch_sort = maxi_batch_sort
         self.fit_table = mini_batch_fit_table  # BatchStats when --mini-batch-fit
@@ -82,13 +86,15 @@ class BatchGenerator:
             nm_src = max(max_src, s_len)
             if cur:
                 full = False
+                scale = self.batch_scale_fn() if self.batch_scale_fn else 1.0
                 if self.fit_table is not None:
                     cap = self.fit_table.find(nm_src, nm_trg)
-                    full = len(cur) + 1 > cap
+                    full = len(cur) + 1 > max(1, int(cap * scale))
                 elif self.mini_batch_words > 0:
-                    full = (len(cur) + 1) * nm_trg > self.mini_batch_words
+                    full = (len(cur) + 1) * nm_trg > \
+                        max(1, int(self.mini_batch_words * scale))
                 else:
-                    full = len(cur) + 1 > self.mini_batch
+                    full = len(cur) + 1 > max(1, int(self.mini_batch * scale))
                 if full:
                     yield self._make_batch(cur)
                     cur = []

@@ -28,7 +28,7 @@ from ..models.model_factory import create_model
 from ..parallel.communicator import dist_env, init_process_group
 from .graph_group import SyncGraphGroup
 from .scheduler import Scheduler
-from .training_state import TrainingState
+from .training_state import SchedulingParameter,  TrainingState
 
 _sigterm_seen = False
 
@@ -208,8 +208,20 @@ class Train:
         corpus.epoch = self.state.epochs - 1
         shuffle = opts.get("shuffle", "data") != "none" \
             and not opts.get("no-shuffle", False)
+        warm = SchedulingParameter.parse(
+            opts.get("mini-batch-warmup", "0") or "0")
+
+        def batch_scale():
+            # --mini-batch-warmup: linear batch-size ramp over N updates
+            if not warm.n:
+                return 1.0
+            n = self.state.labels_total if warm.unit == "t" \
+                else self.state.batches
+            return min(1.0, max(0.05, n / warm.n))
+
         gen = BatchGenerator(
             corpus.iter_epoch(shuffle=shuffle, skip=skip_sentences),
+            batch_scale_fn=batch_scale if warm.n else None,
             mini_batch=int(opts.get("mini-batch", 64)),
             mini_batch_words=int(opts.get("mini-batch-words", 0) or 0),
             maxi_batch=int(opts.get("maxi-batch", 100)),

@@ -131,3 +131,18 @@ def test_all_caps_augmentation(tmp_path):
     s2 = tuples[2][1][0]
     assert unk not in s0[:-1]
     assert all(t == unk for t in s2[:-1])
+
+
+def test_mini_batch_warmup_scaling():
+    """--mini-batch-warmup ramps the batch size (batch_scale_fn)."""
+    from marian_amd.data.batch_generator import BatchGenerator
+    tuples = [(i, [[1, 2, 0], [1, 2, 0]], {}) for i in range(40)]
+    state = {"n": 0}
+    gen = BatchGenerator(iter(tuples), mini_batch=8, maxi_batch=5,
+                         maxi_batch_sort="none", prefetch=False,
+                         batch_scale_fn=lambda: 0.5)
+    sizes = [b.size() for b in gen]
+    assert max(sizes) == 4          # half of mini-batch 8
+    gen2 = BatchGenerator(iter(tuples), mini_batch=8, maxi_batch=5,
+                          maxi_batch_sort="none", prefetch=False)
+    assert max(b.size() for b in gen2) == 8
