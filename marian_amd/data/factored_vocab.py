@@ -133,9 +133,10 @@ class FactoredOutput(torch.nn.Module):
     The result is a coherent product-of-experts distribution; downstream
     log-softmax only renormalises it."""
 
-    def __init__(self, emb: FactoredEmbedding):
+    def __init__(self, emb: FactoredEmbedding, factor_weight: float = 1.0):
         super().__init__()
         self.emb = emb
+        self.factor_weight = factor_weight   # --factor-weight
         self.biases = torch.nn.ParameterList(
             [torch.nn.Parameter(torch.zeros(t.shape[0]))
              for t in emb.tables])
@@ -147,5 +148,7 @@ class FactoredOutput(torch.nn.Module):
             lg = torch.log_softmax(
                 (h.float() @ W.t().to(h.dtype).float()) + b, dim=-1)
             picked = lg.index_select(-1, F[:, g].to(h.device))
+            if g > 0 and self.factor_weight != 1.0:
+                picked = picked * self.factor_weight
             out = picked if out is None else out + picked
         return out.to(h.dtype)                       # [..., V]

@@ -352,7 +352,9 @@ class TransformerModel(nn.Module):
                                   fv.factor_matrix)
             self.trg_embed.E = torch.nn.Parameter(torch.zeros(0, 0),
                                                   requires_grad=False)
-            self.factored_out = FactoredOutput(self.trg_embed.factored)
+            self.factored_out = FactoredOutput(
+                self.trg_embed.factored,
+                float(self.opt.get("factor-weight", 1.0) or 1.0))
 
     # ------------------------------------------------------------------
     def enable_lsh(self, k: int, nbits: int):

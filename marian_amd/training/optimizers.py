@@ -266,6 +266,13 @@ class ShardedOptimizer:
         self.options = options
 
     # ------------------------------------------------------------------
+    def reset_state(self):
+        """Zero the optimizer moments (--lr-decay-reset-optimizer)."""
+        if isinstance(self.core, Adam):
+            self.core.m.zero_()
+            self.core.v.zero_()
+        self.t = 0
+
     def grad_shard_for_bucket(self, bi: int) -> torch.Tensor:
         o, s = self.seg[bi]
         return self.shard_grad[o:o + s]

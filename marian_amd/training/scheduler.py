@@ -30,6 +30,7 @@ class Scheduler:
         # generic --after "Ne|Nu|Nt" (ref: training.h Train::run stop set)
         self.after = SchedulingParameter.parse(options.get("after", "0e"))
         self.early_stopping = int(options.get("early-stopping", 10) or 0)
+        self.reset_optimizer_requested = False
         self.disp_label_counts = bool(options.get("disp-label-counts", False))
         le = (options.list("logical-epoch") or ["1e"])[0]
         self.logical_epoch = SchedulingParameter.parse(le)
@@ -188,6 +189,12 @@ class Scheduler:
         if why == "batches" and strat == "batches" and self.lr_decay_freq \
                 and st.batches % self.lr_decay_freq == 0:
             st.factor *= self.lr_decay
+        if st.factor != getattr(self, "_prev_factor", 1.0):
+            self._prev_factor = st.factor
+            if self.opt.get("lr-decay-repeat-warmup", False):
+                st.warmup_start = st.batches   # re-run the warmup ramp
+            if self.opt.get("lr-decay-reset-optimizer", False):
+                self.reset_optimizer_requested = True
 
     def epoch_actions(self) -> dict:
         """'e'-unit save/valid triggers, checked at epoch boundaries."""

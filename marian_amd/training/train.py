@@ -307,6 +307,10 @@ class Train:
                 actions = sched.update(
                     float(lv.ce_sum), n_sents, lv.label_count,
                     grad_norm=getattr(lv, "grad_norm", None))
+                if self.scheduler.reset_optimizer_requested:
+                    self.scheduler.reset_optimizer_requested = False
+                    self.graph_group.optimizer.reset_state()
+                    log_info("Optimizer state reset (--lr-decay-reset-optimizer)")
                 if actions["save"]:
                     self.save()
                 if actions["validate"]:
