@@ -20,9 +20,11 @@ class LossValue:
 
 
 class CrossEntropyCost:
-    def __init__(self, label_smoothing: float = 0.0, cost_type: str = "ce-sum"):
+    def __init__(self, label_smoothing: float = 0.0, cost_type: str = "ce-sum",
+                 unlikelihood: bool = False):
         self.eps = label_smoothing
         self.cost_type = cost_type
+        self.unlikelihood = unlikelihood
 
     def __call__(self, logits: torch.Tensor, batch) -> LossValue:
         """logits: [B, T, V]; labels/mask from batch.target."""
@@ -38,7 +40,16 @@ class CrossEntropyCost:
                 w = w[:, None].expand(B, T).reshape(B * T)
             else:
                 w = w.reshape(B * T)
-            losses = losses * w.to(losses.dtype)
+            w = w.to(losses.dtype)
+            if self.unlikelihood:
+                # --unlikelihood-loss: negative weights mark sequences to
+                # SUPPRESS: -|w|*log(1 - p(label)) (ref: layers/loss.cpp
+                # data-weighting + unlikelihood objective)
+                p = torch.exp(-losses.clamp_min(1e-6))
+                unlik = -torch.log1p(-p.clamp(max=1 - 1e-6))
+                losses = torch.where(w < 0, unlik * w.abs(), losses * w)
+            else:
+                losses = losses * w
         ce_sum = losses.sum()
         n_labels = int(mask.sum().item())
         if self.cost_type in ("ce-mean", "perplexity"):

@@ -151,6 +151,8 @@ class S2SModel(nn.Module):
         self.cell2 = GRUCellM(ctx_dim, d)
         self.dec_deep = nn.ModuleList(
             [GRUCellM(d, d) for _ in range(max(0, self.dec_depth - 1))])
+        self.dropout_rnn = float(options.get("dropout-rnn", 0.0) or 0.0)
+        self.skip_connections = bool(options.get("skip", False))
 
         # readout (ref: s2s.h logit layer W1*state + W2*ctx + W3*emb)
         self.ff_logit_l1_W0 = Dense(d, e)
@@ -184,12 +186,19 @@ class S2SModel(nn.Module):
         out = torch.stack([torch.cat([f, b], dim=-1)
                            for f, b in zip(fwd_states, bwd_states)], dim=1)
         for cell in self.enc_deep:
+            if self.dropout_rnn and self.training:   # --dropout-rnn
+                out = torch.nn.functional.dropout(out, self.dropout_rnn,
+                                                  True)
             h = torch.zeros(B, self.dim_rnn, device=emb.device, dtype=dt)
             states = []
             for t in range(T):
                 h = cell(out[:, t], h)
                 states.append(h)
-            out = torch.stack(states, dim=1)
+            new_out = torch.stack(states, dim=1)
+            # --skip: residual connections between deep RNN layers
+            out = new_out + out if (self.skip_connections
+                                    and new_out.shape == out.shape) \
+                else new_out
         return out
 
     # ------------------------------------------------------------------

@@ -90,7 +90,8 @@ class SyncGraphGroup:
 
         self.criterion = CrossEntropyCost(
             float(options.get("label-smoothing", 0.0) or 0.0),
-            options.get("cost-type", "ce-sum"))
+            options.get("cost-type", "ce-sum"),
+            unlikelihood=bool(options.get("unlikelihood-loss", False)))
         ga = options.get("guided-alignment", "none")
         self.guided_alignment = (ga not in (None, "none", ""))
         self.guided_weight = float(options.get("guided-alignment-weight", 0.1))

@@ -126,3 +126,19 @@ def test_amun_legacy_names_roundtrip():
     m3 = create_model(o)
     m3.load_marian_items(items, strict=False)
     assert torch.allclose(next(m.parameters()), next(m3.parameters()))
+
+
+def test_s2s_skip_and_dropout_rnn():
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.models.model_factory import create_model
+    o = ConfigParser(MODE_TRAINING).parse_options(
+        ["--train-sets", "a", "b", "--vocabs", "a.yml", "b.yml",
+         "--type", "s2s", "--dim-emb", "16", "--dim-rnn", "24",
+         "--enc-depth", "3", "--dropout-rnn", "0.1", "--skip"],
+        validate=False)
+    o.set("dim-vocabs", [30, 30])
+    m = create_model(o)
+    assert m.skip_connections and m.dropout_rnn == 0.1
+    src = torch.randint(3, 30, (2, 6))
+    out = m.encode(src, torch.ones(2, 6))
+    assert out.shape == (2, 6, 24)
