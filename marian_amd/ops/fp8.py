@@ -36,6 +36,7 @@ def enable_fp8(flag: bool = True):
     _FP8_ENABLED = flag
     if not flag:
         _wcache.clear()
+        _acache.clear()
 
 
 def fp8_enabled() -> bool:
@@ -56,9 +57,27 @@ def _scale_of(t: torch.Tensor) -> torch.Tensor:
     return (amax / E4M3_MAX).float()
 
 
-def _quantize_x(t: torch.Tensor):
-    scale = _scale_of(t)
-    return (t * scale.reciprocal()).to(torch.float8_e4m3fn), scale
+# --- delayed scaling for activations (per call-site = per weight id) ----
+# The fused HIP kernel quantizes with LAST step's scale (saturating cast,
+# so stale scales are safe) and returns this step's amax in the same
+# pass; the history update is two tiny [1]-tensor device ops. No host
+# sync anywhere.
+_acache: Dict[int, torch.Tensor] = {}
+
+
+def _quantize_x(t: torch.Tensor, site: int):
+    from . import hip_ext
+    ext = hip_ext(required=True)
+    s = _acache.get(site)
+    if s is None:
+        s = _scale_of(t)          # first call: honest one-off amax pass
+        _acache[site] = s
+    x8, amax = ext.fp8_quantize(t, s)
+    scale_used = s.clone()
+    # history: decay-max with 5% headroom, ready for the NEXT step
+    s.copy_(torch.maximum(amax[0] * (1.05 / E4M3_MAX),
+                          s * 0.75).clamp_min(1e-6))
+    return x8, scale_used
 
 
 def _quantize_w(W: torch.Tensor):
@@ -77,7 +96,7 @@ def _quantize_w(W: torch.Tensor):
 class Fp8LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, W, b):
-        x8, sx = _quantize_x(x)
+        x8, sx = _quantize_x(x, id(W))
         w8, sw = _quantize_w(W)
         y = torch._scaled_mm(x8, w8, scale_a=sx, scale_b=sw,
                              bias=b.to(torch.bfloat16) if b is not None else None,

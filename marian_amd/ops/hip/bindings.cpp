@@ -31,6 +31,8 @@ torch::Tensor lt_linear_act(torch::Tensor x, torch::Tensor W,
                             torch::Tensor b, bool relu);
 std::vector<torch::Tensor> lt_dwt_bgrad(torch::Tensor x, torch::Tensor dy,
                                         bool want_db);
+std::vector<torch::Tensor> fp8_quantize(torch::Tensor x,
+                                        torch::Tensor scale);
 torch::Tensor col_sum(torch::Tensor x);
 torch::Tensor sumsq_finite(torch::Tensor g);
 std::vector<torch::Tensor> beam_topk(torch::Tensor logprobs,
@@ -69,6 +71,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "hipBLASLt GEMM with fused bias(+ReLU) epilogue");
   m.def("lt_dwt_bgrad", &lt_dwt_bgrad,
         "hipBLASLt dW^T GEMM with fused bias-gradient (BGRADB) epilogue");
+  m.def("fp8_quantize", &fp8_quantize,
+        "fused e4m3fn quantize + amax (delayed scaling)");
   m.def("col_sum", &col_sum, "column sum (bias gradients)");
   m.def("sumsq_finite", &sumsq_finite, "fused grad sumsq + nonfinite count");
   m.def("beam_topk", &beam_topk,

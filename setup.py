@@ -31,6 +31,9 @@ if os.path.exists(topk):
 colsum = os.path.join(HIP_DIR, "colsum.hip")
 if os.path.exists(colsum):
     sources.append(colsum)
+fp8q = os.path.join(HIP_DIR, "fp8q.hip")
+if os.path.exists(fp8q):
+    sources.append(fp8q)
 ltgemm = os.path.join(HIP_DIR, "ltgemm.cpp")
 if os.path.exists(ltgemm):
     sources.append(ltgemm)

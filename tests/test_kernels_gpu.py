@@ -783,6 +783,48 @@ def test_new_model_families_gpu():
         logits.float().sum().backward()
 
 
+def test_fp8_quantize_kernel():
+    """Fused e4m3fn quantize+amax vs torch reference (saturating cast)."""
+    from marian_amd.ops import hip_ext
+    ext = hip_ext(required=True)
+    torch.manual_seed(21)
+    x = torch.randn(4096, 512, device="cuda", dtype=torch.bfloat16) * 3
+    x[0, 0] = 1000.0   # saturation case
+    amax_true = x.float().abs().max()
+    scale = (amax_true / 448.0).reshape(1)
+    x8, amax = ext.fp8_quantize(x, scale)
+    assert x8.dtype == torch.float8_e4m3fn
+    assert torch.allclose(amax[0], amax_true, rtol=1e-3)
+    ref = (x.float() / scale).clamp(-448, 448).to(torch.float8_e4m3fn)
+    assert (x8.view(torch.uint8) == ref.view(torch.uint8)).float().mean() \
+        > 0.999
+    # stale (too-small) scale must SATURATE, not produce NaN
+    x8b, _ = ext.fp8_quantize(x, scale * 0.01)
+    assert not torch.isnan(x8b.float()).any()
+
+
+def test_fp8_delayed_scaling_linear():
+    """fp8_linear trains a step without host syncs and matches bf16
+    within fp8 tolerance."""
+    from marian_amd.ops import fp8
+    fp8.enable_fp8(True)
+    try:
+        x = torch.randn(2048, 512, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        W = torch.randn(512, 2048, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        b = torch.randn(2048, device="cuda", dtype=torch.bfloat16)
+        for _ in range(3):   # exercise the delayed-scale history
+            y = fp8.fp8_linear(x, W, b)
+        ref = x.float() @ W.float() + b.float()
+        err = (y.float() - ref).abs().mean() / ref.abs().mean()
+        assert float(err) < 0.05
+        y.sum().backward()
+        assert x.grad is not None and W.grad is not None
+    finally:
+        fp8.enable_fp8(False)
+
+
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
