@@ -24,6 +24,8 @@ __global__ void fp8_quantize_kernel(const __hip_bfloat16* __restrict__ x,
                                     float* __restrict__ amax_out,
                                     long n) {
   const float inv = 1.0f / fmaxf(*scale, 1e-12f);
+  // v_cvt_pk_fp8_f32 does NOT saturate on overflow (NaN): clamp first
+  auto q = [&](float v) { return fminf(fmaxf(v * inv, -448.f), 448.f); };
   float amax = 0.f;
   const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   const long stride = (long)gridDim.x * blockDim.x * 8;
@@ -40,9 +42,9 @@ __global__ void fp8_quantize_kernel(const __hip_bfloat16* __restrict__ x,
         amax = fmaxf(amax, fmaxf(fmaxf(fabsf(a0), fabsf(a1)),
                                  fmaxf(fabsf(a2), fabsf(a3))));
         // cvt_pk writes 2 saturated e4m3 bytes into the selected half
-        uint32_t w = __builtin_amdgcn_cvt_pk_fp8_f32(a0 * inv, a1 * inv,
+        uint32_t w = __builtin_amdgcn_cvt_pk_fp8_f32(q(a0), q(a1),
                                                      0u, false);
-        w = __builtin_amdgcn_cvt_pk_fp8_f32(a2 * inv, a3 * inv, w, true);
+        w = __builtin_amdgcn_cvt_pk_fp8_f32(q(a2), q(a3), w, true);
         packed[h] = w;
       }
       *reinterpret_cast<uint2*>(out + i) = make_uint2(packed[0], packed[1]);
@@ -50,7 +52,7 @@ __global__ void fp8_quantize_kernel(const __hip_bfloat16* __restrict__ x,
       for (long k = i; k < n; ++k) {
         const float a = to_f32(x[k]);
         amax = fmaxf(amax, fabsf(a));
-        uint32_t w = __builtin_amdgcn_cvt_pk_fp8_f32(a * inv, 0.f, 0, false);
+        uint32_t w = __builtin_amdgcn_cvt_pk_fp8_f32(q(a), 0.f, 0, false);
         out[k] = (uint8_t)(w & 0xff);
       }
     }
