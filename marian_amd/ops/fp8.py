@@ -24,8 +24,9 @@ import torch
 
 _FP8_ENABLED = False
 E4M3_MAX = 448.0
-MIN_N = 2048          # columns below this: bf16 wins
-MIN_ROWS = 1024
+import os as _os
+MIN_N = int(_os.environ.get("MAMD_FP8_MIN_N", "2048"))
+MIN_ROWS = int(_os.environ.get("MAMD_FP8_MIN_ROWS", "1024"))
 
 _weight_version = 0
 _wcache: Dict[int, Tuple[int, torch.Tensor, torch.Tensor]] = {}
