@@ -18,6 +18,8 @@ NOTE: this is a capability the CUDA reference does not have.
 """
 from __future__ import annotations
 
+import os
+
 from typing import Dict, Tuple
 
 import torch
@@ -82,36 +84,71 @@ def _quantize_x(t: torch.Tensor, site: int):
 
 
 def _quantize_w(W: torch.Tensor):
-    """Version-cached, column-major fp8 weight."""
+    """Version-cached fp8 weight in BOTH layouts: column-major [K,N] for
+    the forward (B operand) and row-major [K,N] whose transposed view is
+    the column-major [N,K] B operand of the dgrad GEMM."""
     key = id(W)
     hit = _wcache.get(key)
     if hit is not None and hit[0] == _weight_version:
-        return hit[1], hit[2]
+        return hit[1], hit[2], hit[3]
     scale = _scale_of(W)
-    w8 = (W * scale.reciprocal()).to(torch.float8_e4m3fn)
-    w8 = w8.t().contiguous().t()  # column-major for hipBLASLt
-    _wcache[key] = (_weight_version, w8, scale)
-    return w8, scale
+    w8_rm = (W * scale.reciprocal()).clamp(-E4M3_MAX, E4M3_MAX) \
+        .to(torch.float8_e4m3fn)
+    w8_cm = w8_rm.t().contiguous().t()  # column-major for hipBLASLt
+    _wcache[key] = (_weight_version, w8_cm, scale, w8_rm)
+    return w8_cm, scale, w8_rm
+
+
+
+_FP8_BWD = os.environ.get("MAMD_FP8_BWD", "1") == "1"
 
 
 class Fp8LinearFn(torch.autograd.Function):
+    """fp8 forward AND backward GEMMs (delayed-scaled e4m3):
+      y  = x8 @ w8_cm                       (fwd)
+      dx = dy8_rm @ w8_rm^T                 (dgrad; B = rm weight .t())
+      dW = (x8_cm)^T @ dy8_cm               (wgrad; byte-transposed copies)
+    Only x8 (+scale) is saved — LESS activation memory than the bf16
+    path. db stays a bf16 column reduce."""
+
     @staticmethod
     def forward(ctx, x, W, b):
         x8, sx = _quantize_x(x, id(W))
-        w8, sw = _quantize_w(W)
+        w8, sw, w8_rm = _quantize_w(W)
         y = torch._scaled_mm(x8, w8, scale_a=sx, scale_b=sw,
                              bias=b.to(torch.bfloat16) if b is not None else None,
                              out_dtype=torch.bfloat16)
-        ctx.save_for_backward(x, W)
+        if _FP8_BWD:
+            ctx.save_for_backward(x8, sx, W)
+        else:
+            ctx.save_for_backward(x, W)
         ctx.has_bias = b is not None
+        ctx.wkey = id(W)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, W = ctx.saved_tensors
         dyc = dy.contiguous()
-        dx = dyc @ W.t()
-        dW = x.t() @ dyc
+        if not _FP8_BWD:
+            x, W = ctx.saved_tensors
+            dx = dyc @ W.t()
+            dW = x.t() @ dyc
+            db = dyc.sum(0) if ctx.has_bias else None
+            return dx, dW, db
+        x8, sx, W = ctx.saved_tensors
+        w8_cm, sw, w8_rm = _quantize_w(W)
+        dy8, sdy = _quantize_x(dyc, ctx.wkey ^ 0x5F5F)
+        # dgrad [M,K]: A = dy8 (rm), B = w8_rm.t() ([N,K] column-major)
+        dx = torch._scaled_mm(dy8, w8_rm.t(), scale_a=sdy, scale_b=sw,
+                              out_dtype=torch.bfloat16)
+        # wgrad [K,N]: A = x^T ([K,M] rm byte-transpose), B = dy8 as
+        # [M,N] column-major (byte-transpose-copy)
+        x8_t = x8.view(torch.uint8).t().contiguous().view(
+            torch.float8_e4m3fn)              # [K,M] row-major
+        dy8_cm = dy8.view(torch.uint8).t().contiguous().view(
+            torch.float8_e4m3fn).t()          # [M,N] column-major
+        dW = torch._scaled_mm(x8_t, dy8_cm, scale_a=sx, scale_b=sdy,
+                              out_dtype=torch.bfloat16)
         db = dyc.sum(0) if ctx.has_bias else None
         return dx, dW, db
 

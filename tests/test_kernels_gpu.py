@@ -825,6 +825,32 @@ def test_fp8_delayed_scaling_linear():
         fp8.enable_fp8(False)
 
 
+def test_fp8_full_backward():
+    """Full-fp8 backward (dgrad/wgrad via byte-transposed e4m3 operands)
+    vs fp32 reference within fp8 tolerance."""
+    from marian_amd.ops import fp8
+    if not fp8._FP8_BWD:
+        pytest.skip("fp8 backward disabled")
+    fp8.enable_fp8(True)
+    try:
+        torch.manual_seed(31)
+        x = torch.randn(2048, 512, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        W = torch.randn(512, 2048, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        b = torch.randn(2048, device="cuda", dtype=torch.bfloat16)
+        y = fp8.fp8_linear(x, W, b)
+        dy = torch.randn_like(y)
+        y.backward(dy)
+        ref_dx = dy.float() @ W.float().t()
+        ref_dW = x.float().t() @ dy.float()
+        ex = (x.grad.float() - ref_dx).abs().mean() / ref_dx.abs().mean()
+        ew = (W.grad.float() - ref_dW).abs().mean() / ref_dW.abs().mean()
+        assert float(ex) < 0.08 and float(ew) < 0.08, (float(ex), float(ew))
+    finally:
+        fp8.enable_fp8(False)
+
+
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
