@@ -100,7 +100,12 @@ def _quantize_w(W: torch.Tensor):
 
 
 
-_FP8_BWD = os.environ.get("MAMD_FP8_BWD", "1") == "1"
+# Full-fp8 backward is OPT-IN: numerically validated (grad error <8% vs
+# fp32) but measured -14% end-to-end on config #5 — the wgrad
+# ([K,M] rm x [M,N] cm) fp8 layout hits hipBLASLt fallback kernels and
+# the byte-transposes add passes. Revisit when hipBLASLt gains fast fp8
+# TN solutions.
+_FP8_BWD = os.environ.get("MAMD_FP8_BWD", "0") == "1"
 
 
 class Fp8LinearFn(torch.autograd.Function):

@@ -829,9 +829,8 @@ def test_fp8_full_backward():
     """Full-fp8 backward (dgrad/wgrad via byte-transposed e4m3 operands)
     vs fp32 reference within fp8 tolerance."""
     from marian_amd.ops import fp8
-    if not fp8._FP8_BWD:
-        pytest.skip("fp8 backward disabled")
     fp8.enable_fp8(True)
+    fp8._FP8_BWD, _saved = True, fp8._FP8_BWD
     try:
         torch.manual_seed(31)
         x = torch.randn(2048, 512, device="cuda", dtype=torch.bfloat16,
@@ -848,6 +847,7 @@ def test_fp8_full_backward():
         ew = (W.grad.float() - ref_dW).abs().mean() / ref_dW.abs().mean()
         assert float(ex) < 0.08 and float(ew) < 0.08, (float(ex), float(ew))
     finally:
+        fp8._FP8_BWD = _saved
         fp8.enable_fp8(False)
 
 
