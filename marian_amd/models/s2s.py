@@ -31,7 +31,8 @@ class GRUCellM(nn.Module):
     Gate layout: W/U produce [r, z] stacked; Wx/Ux the candidate.
     """
 
-    def __init__(self, dim_in: int, dim: int):
+    def __init__(self, dim_in: int, dim: int, layer_norm: bool = False,
+                 transition_depth: int = 1):
         super().__init__()
         self.W = nn.Parameter(torch.empty(dim_in, 2 * dim))   # input -> r,z
         self.U = nn.Parameter(torch.empty(dim, 2 * dim))      # state -> r,z
@@ -42,15 +43,41 @@ class GRUCellM(nn.Module):
         self.dim = dim
         for w in (self.W, self.U, self.Wx, self.Ux):
             nn.init.xavier_uniform_(w)
+        # --layer-normalization: LN on the gate pre-activations
+        # (ref: rnn/cells.h GRU with layerNorm=true)
+        self.ln = layer_norm
+        if layer_norm:
+            self.ln_g = nn.Parameter(torch.ones(2 * dim))
+            self.ln_gb = nn.Parameter(torch.zeros(2 * dim))
+            self.ln_x = nn.Parameter(torch.ones(dim))
+            self.ln_xb = nn.Parameter(torch.zeros(dim))
+        # deep-transition cell (--enc-cell-depth / --dec-cell-base-depth /
+        # --dec-cell-high-depth): extra U-only GRU transitions per step
+        # (ref: rnn/rnn.h StackedCell deep transitions)
+        self.transitions = nn.ModuleList(
+            [GRUCellM(dim, dim, layer_norm) for _ in
+             range(max(0, transition_depth - 1))])
+
+    def _norm(self, t, g, b):
+        mu = t.mean(-1, keepdim=True)
+        sd = t.var(-1, unbiased=False, keepdim=True).add(1e-9).sqrt()
+        return (t - mu) / sd * g.to(t.dtype) + b.to(t.dtype)
 
     def forward(self, x, h):
         dt = x.dtype
         gates = x @ self.W.to(dt) + h @ self.U.to(dt) + self.b.to(dt)
+        if self.ln:
+            gates = self._norm(gates, self.ln_g, self.ln_gb)
         r, z = gates.chunk(2, dim=-1)
         r, z = torch.sigmoid(r), torch.sigmoid(z)
-        hbar = torch.tanh(x @ self.Wx.to(dt) + (r * h) @ self.Ux.to(dt)
-                          + self.bx.to(dt))
-        return (1 - z) * hbar + z * h
+        pre = x @ self.Wx.to(dt) + (r * h) @ self.Ux.to(dt) + self.bx.to(dt)
+        if self.ln:
+            pre = self._norm(pre, self.ln_x, self.ln_xb)
+        hbar = torch.tanh(pre)
+        h = (1 - z) * hbar + z * h
+        for cell in self.transitions:    # deep transitions: state-only GRU
+            h = cell(torch.zeros_like(h), h)
+        return h
 
 
 class MLPAttention(nn.Module):
@@ -137,20 +164,26 @@ class S2SModel(nn.Module):
             nn.init.normal_(self.trg_embed, std=self.dim_emb ** -0.5)
 
         d, e = self.dim_rnn, self.dim_emb
-        self.enc_fwd = GRUCellM(e, d)
-        self.enc_bwd = GRUCellM(e, d)
+        ln = bool(options.get("layer-normalization", False))
+        ecd = int(options.get("enc-cell-depth", 1) or 1)
+        dbd = int(options.get("dec-cell-base-depth", 2) or 2)
+        dhd = int(options.get("dec-cell-high-depth", 1) or 1)
+        self.enc_fwd = GRUCellM(e, d, ln, ecd)
+        self.enc_bwd = GRUCellM(e, d, ln, ecd)
         self.enc_deep = nn.ModuleList(
-            [GRUCellM(2 * d if i == 0 else d, d)
+            [GRUCellM(2 * d if i == 0 else d, d, ln, ecd)
              for i in range(max(0, self.enc_depth - 1))])
         ctx_dim = 2 * d if not self.enc_deep else d
         self.ctx_dim = ctx_dim
 
         self.dec_init = Dense(ctx_dim, d)        # mean-pooled init state
-        self.cell1 = GRUCellM(e, d)
+        # marian's conditional-GRU base depth counts cell1+cell2; extra
+        # base transitions attach to cell2 (ref: s2s.h decoder cell)
+        self.cell1 = GRUCellM(e, d, ln)
         self.att = MLPAttention(d, ctx_dim, d)
-        self.cell2 = GRUCellM(ctx_dim, d)
+        self.cell2 = GRUCellM(ctx_dim, d, ln, max(1, dbd - 1))
         self.dec_deep = nn.ModuleList(
-            [GRUCellM(d, d) for _ in range(max(0, self.dec_depth - 1))])
+            [GRUCellM(d, d, ln, dhd) for _ in range(max(0, self.dec_depth - 1))])
         self.dropout_rnn = float(options.get("dropout-rnn", 0.0) or 0.0)
         self.skip_connections = bool(options.get("skip", False))
 

@@ -142,3 +142,29 @@ def test_s2s_skip_and_dropout_rnn():
     src = torch.randint(3, 30, (2, 6))
     out = m.encode(src, torch.ones(2, 6))
     assert out.shape == (2, 6, 24)
+
+
+def test_s2s_cell_depths_and_layer_norm():
+    """--enc-cell-depth / --dec-cell-base-depth deep transitions +
+    --layer-normalization in GRU cells (ref: rnn/cells.h, rnn/rnn.h)."""
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.data.batch import SubBatch, CorpusBatch
+    from marian_amd.models.model_factory import create_model
+    o = ConfigParser(MODE_TRAINING).parse_options(
+        ["--train-sets", "a", "b", "--vocabs", "a.yml", "b.yml",
+         "--type", "s2s", "--dim-emb", "16", "--dim-rnn", "24",
+         "--enc-cell-depth", "2", "--dec-cell-base-depth", "3",
+         "--layer-normalization"], validate=False)
+    o.set("dim-vocabs", [30, 30])
+    m = create_model(o)
+    assert len(m.enc_fwd.transitions) == 1
+    assert len(m.cell2.transitions) == 1
+    assert m.enc_fwd.ln and m.enc_fwd.ln_g.shape == (48,)
+    src = torch.randint(3, 30, (2, 5))
+    trg = torch.randint(3, 30, (2, 4))
+    b = CorpusBatch([SubBatch(src, torch.ones(2, 5)),
+                     SubBatch(trg, torch.ones(2, 4))])
+    logits = m(b)
+    assert logits.shape == (2, 4, 30)
+    logits.sum().backward()
+    assert m.enc_fwd.transitions[0].Ux.grad is not None
