@@ -318,6 +318,13 @@ class S2SModel(nn.Module):
             m[prefix + "_Wx"] = cell.Wx
             m[prefix + "_Ux"] = cell.Ux
             m[prefix + "_bx"] = cell.bx
+            if cell.ln:   # --layer-normalization scales/biases [M names]
+                m[prefix + "_gamma1"] = cell.ln_g
+                m[prefix + "_beta1"] = cell.ln_gb
+                m[prefix + "_gamma2"] = cell.ln_x
+                m[prefix + "_beta2"] = cell.ln_xb
+            for j, tcell in enumerate(cell.transitions, start=2):
+                gru(f"{prefix}_t{j}", tcell)   # deep transitions [M names]
 
         gru("encoder_bi", self.enc_fwd)
         gru("encoder_bi_r", self.enc_bwd)
