@@ -179,7 +179,15 @@ class SyncGraphGroup:
                 self.model.last_cross_attn,
                 batch.guided_alignment.to(self.device),
                 batch.target.mask, self.guided_cost)
-            bloss = bloss + self.guided_weight * ga_loss * max(1, lv.label_count)
+            # --multi-loss-type: how the partial losses combine
+            # (ref: layers/loss.h MultiRationalLoss sum/scaled/mean)
+            mlt = self.opt.get("multi-loss-type", "sum") \
+                if hasattr(self, "opt") else "sum"
+            if mlt == "mean":
+                bloss = bloss + self.guided_weight * ga_loss
+            else:  # "sum"/"scaled": GA loss scaled to label count
+                bloss = bloss + self.guided_weight * ga_loss \
+                    * max(1, lv.label_count)
         if self.scaler.enabled and self.scaler.scale != 1.0:
             bloss = bloss * self.scaler.scale
         bloss.backward()
