@@ -54,28 +54,6 @@ __device__ __forceinline__ void store4_f32(T* __restrict__ p,
   *reinterpret_cast<Pack4<T>*>(p) = t;
 }
 
-template <typename T>
-struct __align__(16) Pack8 {
-  T v[8];
-};
-
-template <typename T>
-__device__ __forceinline__ void load8_f32(const T* __restrict__ p,
-                                          float o[8]) {
-  Pack8<T> t = *reinterpret_cast<const Pack8<T>*>(p);
-#pragma unroll
-  for (int j = 0; j < 8; ++j) o[j] = to_f32(t.v[j]);
-}
-
-template <typename T>
-__device__ __forceinline__ void store8_f32(T* __restrict__ p,
-                                           const float o[8]) {
-  Pack8<T> t;
-#pragma unroll
-  for (int j = 0; j < 8; ++j) t.v[j] = from_f32<T>(o[j]);
-  *reinterpret_cast<Pack8<T>*>(p) = t;
-}
-
 // ---- wave reductions (64-wide) -------------------------------------------
 __device__ __forceinline__ float wave_sum(float v) {
 #pragma unroll

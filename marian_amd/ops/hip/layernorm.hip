@@ -107,42 +107,39 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
     if constexpr (MAXC <= 16) {
       // register-cache xhat and dy*gamma across the two passes (d <= 1024
       // per 64-lane wave) - the re-read variant ran at 1.3 TB/s
-      // 16B (8 x bf16) loads per lane: half the load instructions of the
-      // VEC=4 version (measured 3.37 -> target ~5 TB/s)
-      constexpr int V8 = 8;
       float xh[MAXC], gg[MAXC];
 #pragma unroll
-      for (int i = 0; i < MAXC / V8; ++i) {
-        const int c = lane * V8 + i * WAVE * V8;
+      for (int i = 0; i < MAXC / VEC; ++i) {
+        const int c = lane * VEC + i * WAVE * VEC;
         if (c < d) {
-          float x8[V8], g8[V8], gm8[V8];
-          load8_f32(xr + c, x8);
-          load8_f32(dyr + c, g8);
-          load8_f32(gamma + c, gm8);
+          float x4[VEC], g4[VEC], gm4[VEC];
+          load4_f32(xr + c, x4);
+          load4_f32(dyr + c, g4);
+          load4_f32(gamma + c, gm4);
 #pragma unroll
-          for (int j = 0; j < V8; ++j) {
-            const float xhv = (x8[j] - mu) * rs;
-            const float ggv = g8[j] * gm8[j];
-            xh[i * V8 + j] = xhv;
-            gg[i * V8 + j] = ggv;
+          for (int j = 0; j < VEC; ++j) {
+            const float xhv = (x4[j] - mu) * rs;
+            const float ggv = g4[j] * gm4[j];
+            xh[i * VEC + j] = xhv;
+            gg[i * VEC + j] = ggv;
             h1 += ggv;
             h2 += ggv * xhv;
-            dg[i * V8 + j] += g8[j] * xhv;
-            db[i * V8 + j] += g8[j];
+            dg[i * VEC + j] += g4[j] * xhv;
+            db[i * VEC + j] += g4[j];
           }
         }
       }
       h1 = wave_sum(h1) / d;
       h2 = wave_sum(h2) / d;
 #pragma unroll
-      for (int i = 0; i < MAXC / V8; ++i) {
-        const int c = lane * V8 + i * WAVE * V8;
+      for (int i = 0; i < MAXC / VEC; ++i) {
+        const int c = lane * VEC + i * WAVE * VEC;
         if (c < d) {
-          float o8[V8];
+          float o4[VEC];
 #pragma unroll
-          for (int j = 0; j < V8; ++j)
-            o8[j] = rs * (gg[i * V8 + j] - h1 - xh[i * V8 + j] * h2);
-          store8_f32(dxr + c, o8);
+          for (int j = 0; j < VEC; ++j)
+            o4[j] = rs * (gg[i * VEC + j] - h1 - xh[i * VEC + j] * h2);
+          store4_f32(dxr + c, o4);
         }
       }
     } else {
@@ -274,10 +271,9 @@ void ln_bwd_launch(const torch::Tensor& x, const torch::Tensor& g,
                      dg_scratch.data_ptr<float>(),                           \
                      db_scratch.data_ptr<float>(), rows, d)
   const int per_lane = (d + WAVE - 1) / WAVE;  // columns per lane
-  // the register-cached (MAXC<=16) fast path uses 16B packs: d%8 required
-  if (per_lane <= 8 && d % 8 == 0) {
+  if (per_lane <= 8) {
     LAUNCH_BWD(8);
-  } else if (per_lane <= 16 && d % 8 == 0) {
+  } else if (per_lane <= 16) {
     LAUNCH_BWD(16);
   } else if (per_lane <= 32) {
     LAUNCH_BWD(32);
