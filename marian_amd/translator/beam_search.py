@@ -145,7 +145,8 @@ class BeamSearch:
                     from .graph_decode import GraphedDecodeStep
                     sup = -1 if self.allow_unk else self.unk_id
                     Ts = states[0]["enc"].shape[1]
-                    key = (id(model), B * K, Ts, max_len, K, sup)
+                    key = (id(model), B * K, Ts, max_len, K, sup,
+                           self.eos_id, self.n_best)
                     cache = getattr(self, "_graph_cache", None)
                     if cache is not None and cache[0] == key:
                         graphed = cache[1]
@@ -154,7 +155,8 @@ class BeamSearch:
                     else:
                         graphed = GraphedDecodeStep(
                             model, states[0]["enc"], states[0]["mask"],
-                            B * K, max_len, K, sup)
+                            B * K, max_len, K, sup,
+                            eos_id=self.eos_id, n_best=self.n_best)
                         graphed.capture()
                         self._graph_cache = (key, graphed)
                     if not getattr(BeamSearch, "_graph_logged", False):
@@ -172,51 +174,30 @@ class BeamSearch:
                     + torch.arange(K, device=device)[None, :]).reshape(-1)
         align_steps: List[torch.Tensor] = []
         V = None
-        for step in range(max_len):
-            if graphed is not None:
-                V = self.scorers[0].model.trg_vocab_size
-                top_scores, top_idx = graphed.run_step(
-                    step, prev_tokens, flat_sel, scores.view(B * K))
-                beam_idx = top_idx // V
-                word_idx = top_idx % V
-                real_words = word_idx
-                is_eos = (real_words == self.eos_id) \
-                    & (top_scores > NEG_INF / 2)
-                cap = fin_scores.shape[1]
-                eos_rank = torch.cumsum(is_eos.long(), dim=1) - 1
-                slot = fin_count[:, None] + eos_rank
-                valid = is_eos & (slot < cap) & (slot >= 0)
-                slot_c = slot.clamp(0, cap - 1)
-                kidx = torch.arange(K, device=device)[None, :].expand(B, K)
-                fin_scores.scatter_(1, slot_c,
-                                    torch.where(valid, top_scores,
-                                                fin_scores.gather(1, slot_c)))
-                fin_len.scatter_(1, slot_c,
-                                 torch.where(valid,
-                                             torch.full_like(slot_c, step + 1),
-                                             fin_len.gather(1, slot_c)))
-                fin_step.scatter_(1, slot_c,
-                                  torch.where(valid,
-                                              torch.full_like(slot_c, step),
-                                              fin_step.gather(1, slot_c)))
-                fin_beam.scatter_(1, slot_c,
-                                  torch.where(valid, kidx,
-                                              fin_beam.gather(1, slot_c)))
-                fin_count = fin_count + valid.sum(dim=1)
-                scores = torch.where(is_eos,
-                                     torch.full_like(top_scores, NEG_INF),
-                                     top_scores)
-                bp_steps.append(beam_idx)
-                tok_steps.append(real_words)
-                alive = (scores > NEG_INF / 2).reshape(B * K)
+        if graphed is not None:
+            # closed-loop replay: the graph feeds itself (beam update is
+            # IN-graph) — one launch per step, host syncs only the stop
+            # check every 4 steps
+            graphed.reset_state()
+            n_steps = 0
+            for step in range(max_len):
+                graphed.replay_once()
+                n_steps = step + 1
                 if (step & 3) == 3 or step == max_len - 1:
-                    done = (fin_count >= self.n_best).all() | ~alive.any()
-                    if bool(done):
+                    alive_t = graphed.scores > NEG_INF / 2
+                    done = (graphed.fin_count >= self.n_best).all() \
+                        | ~alive_t.any()
+                    if bool(done.item()):
                         break
-                flat_sel = (torch.arange(B, device=device)[:, None] * K
-                            + beam_idx).reshape(-1)
-                prev_tokens = real_words.reshape(B * K)
-                continue
+            bp_steps = [graphed.bp_store[t] for t in range(n_steps)]
+            tok_steps = [graphed.tok_store[t] for t in range(n_steps)]
+            fin_scores = graphed.fin_scores
+            fin_len = graphed.fin_len
+            fin_step = graphed.fin_step
+            fin_beam = graphed.fin_beam
+            fin_count = graphed.fin_count
+            scores = graphed.scores.view(B, K)
+        for step in range(max_len) if graphed is None else []:
             # ensemble step: weighted sum of log-probs
             logprobs = None
             for w, sc, st in zip(self.weights, self.scorers, states):

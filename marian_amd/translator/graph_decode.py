@@ -33,7 +33,8 @@ from ..models.layers import sinusoidal_position_embeddings
 
 class GraphedDecodeStep:
     def __init__(self, model: TransformerModel, enc_out, src_mask,
-                 n_rows: int, max_len: int, beam_k: int, suppress: int):
+                 n_rows: int, max_len: int, beam_k: int, suppress: int,
+                 eos_id: int = 0, n_best: int = 1):
         from ..ops import hip_ext
         self.ext = hip_ext(required=True)
         self.model = model
@@ -78,8 +79,27 @@ class GraphedDecodeStep:
         self.load_batch(enc_out, src_mask)
         # outputs
         V = model.trg_vocab_size
+        self.V = V
+        self.eos_id = eos_id
+        self.n_best = n_best
         self.out_scores = torch.zeros(self.B, beam_k, device=dev)
         self.out_idx = torch.zeros(self.B, beam_k, dtype=torch.long, device=dev)
+        # ---- in-graph beam state: the WHOLE beam update happens inside
+        # the replay, so steady-state decoding is one graph launch per
+        # step with zero per-step host work ---------------------------------
+        B, K = self.B, beam_k
+        cap = n_best + K
+        self.fin_scores = torch.full((B, cap), NEG_INF, device=dev)
+        self.fin_len = torch.zeros((B, cap), dtype=torch.long, device=dev)
+        self.fin_step = torch.full((B, cap), -1, dtype=torch.long, device=dev)
+        self.fin_beam = torch.zeros((B, cap), dtype=torch.long, device=dev)
+        self.fin_count = torch.zeros(B, dtype=torch.long, device=dev)
+        self.bp_store = torch.zeros(max_len, B, K, dtype=torch.long,
+                                    device=dev)
+        self.tok_store = torch.zeros(max_len, B, K, dtype=torch.long,
+                                     device=dev)
+        self._kidx = torch.arange(K, device=dev)[None, :].expand(B, K)
+        self._bbase = (torch.arange(B, device=dev)[:, None] * K)
         self.graphs = [None, None]
         self._which = 0
 
@@ -147,6 +167,37 @@ class GraphedDecodeStep:
                                   self.suppress)
         self.out_scores.copy_(v)
         self.out_idx.copy_(i)
+        # ---- in-graph beam update (mirrors beam_search's tensor ops) ------
+        beam_idx = i // self.V
+        word = i % self.V
+        is_eos = (word == self.eos_id) & (v > NEG_INF / 2)
+        cap = self.fin_scores.shape[1]
+        eos_rank = torch.cumsum(is_eos.long(), dim=1) - 1
+        slot = self.fin_count[:, None] + eos_rank
+        valid = is_eos & (slot < cap) & (slot >= 0)
+        slot_c = slot.clamp(0, cap - 1)
+        self.fin_scores.scatter_(
+            1, slot_c, torch.where(valid, v, self.fin_scores.gather(1, slot_c)))
+        self.fin_len.scatter_(
+            1, slot_c, torch.where(valid, (self.step_t + 1).expand_as(slot_c),
+                                   self.fin_len.gather(1, slot_c)))
+        self.fin_step.scatter_(
+            1, slot_c, torch.where(valid, self.step_t.expand_as(slot_c),
+                                   self.fin_step.gather(1, slot_c)))
+        self.fin_beam.scatter_(
+            1, slot_c, torch.where(valid, self._kidx,
+                                   self.fin_beam.gather(1, slot_c)))
+        self.fin_count.add_(valid.sum(dim=1))
+        new_scores = torch.where(is_eos, torch.full_like(v, NEG_INF), v)
+        self.bp_store.index_copy_(0, self.step_t, beam_idx.unsqueeze(0))
+        self.tok_store.index_copy_(0, self.step_t, word.unsqueeze(0))
+        # feed the NEXT replay
+        self.scores.copy_(new_scores.reshape(-1))
+        self.prev_tokens.copy_(word.reshape(-1))
+        self.beam_sel.copy_((self._bbase + beam_idx).reshape(-1))
+        self.step_t.add_(1)
+        self.kv_lens.add_(1)
+        self.emb_scale.fill_(1.0)
 
     # ----------------------------------------------------------------------
     def capture(self):
@@ -162,18 +213,28 @@ class GraphedDecodeStep:
             self.graphs[w] = g
 
     # ----------------------------------------------------------------------
-    def run_step(self, step: int, prev_tokens, beam_sel, scores):
-        """Replay one captured step. Inputs are [N] device tensors; the
-        fresh KV state lives in the DESTINATION buffer afterwards."""
-        self.prev_tokens.copy_(prev_tokens)
-        self.beam_sel.copy_(beam_sel)
-        self.scores.copy_(scores)
-        self.step_t.fill_(step)
-        self.kv_lens.fill_(step + 1)
-        self.emb_scale.fill_(0.0 if step == 0 else 1.0)
+    def reset_state(self):
+        """Prepare for a fresh decode: the first replay runs step 0 with a
+        zero embedding (emb_scale=0) and only beam 0 live."""
+        self.step_t.zero_()
+        self.kv_lens.fill_(1)
+        self.emb_scale.fill_(0.0)
+        self.prev_tokens.zero_()
+        self.beam_sel.copy_(torch.arange(self.N, device=self.device))
+        sc = torch.full((self.B, self.K), NEG_INF, device=self.device)
+        sc[:, 0] = 0.0
+        self.scores.copy_(sc.reshape(-1))
+        self.fin_scores.fill_(NEG_INF)
+        self.fin_len.zero_()
+        self.fin_step.fill_(-1)
+        self.fin_beam.zero_()
+        self.fin_count.zero_()
+        self._which = 0
+
+    def replay_once(self):
+        """One full decode step: model + beam update, one graph launch."""
         self.graphs[self._which].replay()
         self._which ^= 1
-        return self.out_scores, self.out_idx
 
     def active_kv(self):
         """Buffer holding the CURRENT state (destination of the last step)."""
