@@ -851,6 +851,47 @@ def test_fp8_full_backward():
         fp8.enable_fp8(False)
 
 
+def test_graphed_beam_equals_eager():
+    """The closed-loop hipGraph beam path produces the same tokens as the
+    eager path on the same bf16 model."""
+    import os
+    from marian_amd.common.options import Options
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.translator.beam_search import BeamSearch, ModelScorer
+    from marian_amd.data.batch import SubBatch, CorpusBatch
+    torch.manual_seed(3)
+    o = Options({
+        "type": "transformer", "dim-emb": 128, "enc-depth": 2,
+        "dec-depth": 2, "transformer-heads": 2, "transformer-dim-ffn": 256,
+        "transformer-preprocess": "", "transformer-postprocess": "dan",
+        "transformer-postprocess-emb": "d", "transformer-postprocess-top": "",
+        "transformer-ffn-activation": "relu", "transformer-dropout": 0.0,
+        "dim-vocabs": [64, 64], "tied-embeddings-all": True,
+        "transformer-decoder-dim-ffn": 0,
+    })
+    m = create_model(o).cuda().bfloat16()
+    m.set_compute_dtype(torch.bfloat16)
+    m.eval()
+    idx = torch.randint(3, 64, (5, 7), device="cuda")
+    mask = torch.ones(5, 7, device="cuda")
+
+    bs_g = BeamSearch([ModelScorer(m, torch.device("cuda"))],
+                      beam_size=4, normalize=0.6, max_length=16, n_best=2)
+    b1 = CorpusBatch([SubBatch(idx, mask), SubBatch(idx.clone(), mask)])
+    res_g = bs_g.search(b1, torch.device("cuda"))
+    assert getattr(bs_g, "_graph_cache", None) is not None, \
+        "graph path did not activate"
+    # eager reference: temporarily disable by asking for alignment
+    bs_e = BeamSearch([ModelScorer(m, torch.device("cuda"))],
+                      beam_size=4, normalize=0.6, max_length=16, n_best=2,
+                      collect_alignment=True)
+    b2 = CorpusBatch([SubBatch(idx, mask), SubBatch(idx.clone(), mask)])
+    res_e = bs_e.search(b2, torch.device("cuda"))
+    for rg, re in zip(res_g, res_e):
+        assert rg.best.tokens == re.best.tokens
+        assert abs(rg.best.score - re.best.score) < 0.05
+
+
 def test_beam_topk_kernel(ext):
     """Fused beam top-k vs torch reference incl. dead beams + suppression."""
     torch.manual_seed(51)
