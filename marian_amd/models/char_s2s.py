@@ -73,34 +73,4 @@ class CharS2SModel(S2SModel):
                                             self.src_embed.to(dt))
         pooled, pmask = self.char_encoder(emb, src_mask)
         self._pooled_mask = pmask.to(src_mask.dtype)
-        return self._char_rnn_encode(pooled, self._pooled_mask)
-
-    def _char_rnn_encode(self, emb, mask):
-        """The standard bi-GRU stack over the pooled char segments."""
-        # reuse S2SModel.encode's RNN body by faking the embedding lookup:
-        # inline here to keep the loop identical to the word-level path
-        dt = emb.dtype
-        B, T, _ = emb.shape
-        hf = torch.zeros(B, self.dim_rnn, device=emb.device, dtype=dt)
-        hb = torch.zeros_like(hf)
-        m = mask.to(dt)
-        fwd_states, bwd_states = [], [None] * T
-        for t in range(T):
-            hf_new = self.enc_fwd(emb[:, t], hf)
-            hf = m[:, t:t + 1] * hf_new + (1 - m[:, t:t + 1]) * hf
-            fwd_states.append(hf)
-        for t in range(T - 1, -1, -1):
-            hb_new = self.enc_bwd(emb[:, t], hb)
-            hb = m[:, t:t + 1] * hb_new + (1 - m[:, t:t + 1]) * hb
-            bwd_states[t] = hb
-        out = torch.stack([torch.cat([f, b], dim=-1)
-                           for f, b in zip(fwd_states, bwd_states)], dim=1)
-        for cell in self.enc_deep:
-            h = torch.zeros(B, self.dim_rnn, device=emb.device, dtype=dt)
-            states = []
-            for t in range(T):
-                h_new = cell(out[:, t], h)
-                h = m[:, t:t + 1] * h_new + (1 - m[:, t:t + 1]) * h
-                states.append(h)
-            out = torch.stack(states, dim=1)
-        return out
+        return self.rnn_encode(pooled, self._pooled_mask)

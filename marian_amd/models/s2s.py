@@ -202,8 +202,14 @@ class S2SModel(nn.Module):
     # ------------------------------------------------------------------
     def encode(self, src_indices, src_mask):
         dt = self.compute_dtype
-        B, T = src_indices.shape
         emb = torch.nn.functional.embedding(src_indices, self.src_embed.to(dt))
+        return self.rnn_encode(emb, src_mask)
+
+    def rnn_encode(self, emb, src_mask):
+        """Bi-GRU encoder stack over already-embedded inputs (shared with
+        the char-s2s front-end)."""
+        dt = emb.dtype
+        B, T = emb.shape[:2]
         mask = src_mask.to(dt)
         hf = torch.zeros(B, self.dim_rnn, device=emb.device, dtype=dt)
         hb = torch.zeros_like(hf)

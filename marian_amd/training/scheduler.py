@@ -32,6 +32,8 @@ class Scheduler:
         self.early_stopping = int(options.get("early-stopping", 10) or 0)
         self.reset_optimizer_requested = False
         self.disp_label_counts = bool(options.get("disp-label-counts", False))
+        self.stat_freq = SchedulingParameter.parse(
+            options.get("stat-freq", "0") or "0")
         le = (options.list("logical-epoch") or ["1e"])[0]
         self.logical_epoch = SchedulingParameter.parse(le)
         if self.logical_epoch.unit == "e" and self.logical_epoch.n <= 1:
@@ -98,6 +100,7 @@ class Scheduler:
     _disp_labels = 0
     _save_labels = 0
     _valid_labels = 0
+    _stat_labels = 0
 
     # ------------------------------------------------------------------
     def update(self, cost: float, n_sentences: int, n_labels: int,
@@ -158,6 +161,19 @@ class Scheduler:
             self.timer.reset()
             actions["display"] = True
 
+        if due(self.stat_freq, "_stat_labels"):
+            # --stat-freq: memory/throughput stats line
+            import torch
+            if torch.cuda.is_available():
+                mem = torch.cuda.memory_allocated() / 2**30
+                peak = torch.cuda.max_memory_allocated() / 2**30
+                log_info(f"[stats] Up. {st.batches} : GPU mem {mem:.2f} GiB "
+                         f"(peak {peak:.2f} GiB)")
+            else:
+                import resource
+                rss = resource.getrusage(
+                    resource.RUSAGE_SELF).ru_maxrss / 2**20
+                log_info(f"[stats] Up. {st.batches} : RSS {rss:.2f} GiB")
         if due(self.save_freq, "_save_labels"):
             actions["save"] = True
         if due(self.valid_freq, "_valid_labels"):
