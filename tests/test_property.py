@@ -52,3 +52,67 @@ def test_batch_split_preserves_content(B, T, n):
     assert sum(p.size() for p in parts) == B
     recon = torch.cat([p.source.indices for p in parts])
     assert torch.equal(recon, b.source.indices)
+
+
+@given(st.lists(st.tuples(
+    st.text(alphabet="abcdef", min_size=1, max_size=6),
+    st.integers(0, 1), st.integers(0, 2)), min_size=1, max_size=30))
+@settings(max_examples=30, deadline=None)
+def test_factored_vocab_parse_properties(entries):
+    """Factored vocab invariants: ids dense, factor matrix complete, every
+    surface encodes to its own id."""
+    import tempfile, os
+    from marian_amd.data.factored_vocab import FactoredVocab
+    lines = []
+    seen = set()
+    for w, c, pl in entries:
+        if w in seen:
+            continue
+        seen.add(w)
+        lines.append(f"{w}\t{w.rstrip('s') or w} _c{c}" +
+                     (f" _pl{pl}" if pl else ""))
+    fd, path = tempfile.mkstemp(suffix=".fsv")
+    os.close(fd)
+    try:
+        with open(path, "w") as f:
+            f.write("\n".join(lines) + "\n")
+        v = FactoredVocab.load(path)
+        assert len(v) == len(seen) + 2          # + eos/unk
+        F = v.factor_matrix
+        assert F.shape[0] == len(v)
+        assert (F >= 0).all()
+        for g, size in enumerate(v.group_sizes):
+            assert int(F[:, g].max()) < size
+        for w in seen:
+            assert v.encode(w, add_eos=False) == [v.token2id[w]]
+    finally:
+        os.unlink(path)
+
+
+@given(st.integers(1, 10**7), st.sampled_from(["", "k", "K", "M"]),
+       st.sampled_from(["u", "t", "e", ""]))
+@settings(max_examples=50, deadline=None)
+def test_scheduling_parameter_parse(n, mult, unit):
+    from marian_amd.training.training_state import SchedulingParameter
+    sp = SchedulingParameter.parse(f"{n}{mult}{unit}")
+    factor = {"": 1, "k": 1000, "K": 1000, "M": 10**6}[mult]
+    assert sp.n == n * factor
+    assert sp.unit == (unit or "u").lower()
+
+
+@given(st.integers(2, 6), st.integers(2, 40), st.integers(1, 5))
+@settings(max_examples=20, deadline=None)
+def test_beam_backtrace_roundtrip(K, V, steps):
+    """Backtracing a hand-built beam lattice returns the planted path."""
+    import torch
+    from marian_amd.translator.beam_search import BeamSearch
+    bs = BeamSearch.__new__(BeamSearch)
+    bs.eos_id = 0
+    # plant a path through beam 0 with known tokens
+    toks = [(i % (V - 1)) + 1 for i in range(steps)]
+    bp = torch.zeros(steps, 1, K, dtype=torch.long)
+    tk = torch.zeros(steps, 1, K, dtype=torch.long)
+    for t, w in enumerate(toks):
+        tk[t, 0, 0] = w
+    got, _ = bs._backtrace(bp, tk, 0, steps - 1, 0, [], K)
+    assert got == toks
