@@ -19,7 +19,6 @@ training computes exactly the global cost gradient.
 """
 from __future__ import annotations
 
-import math
 from typing import Dict, List, Optional
 
 import numpy as np

@@ -13,7 +13,6 @@ shard (ref: SyncGraphGroup sharded optimizer + castOptimizerType).
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Tuple
 

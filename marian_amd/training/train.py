@@ -23,7 +23,6 @@ from ..common.options import Options
 from ..data.batch_generator import BatchGenerator, BatchStats
 from ..data.corpus import Corpus
 from ..data.vocab import Vocab
-from ..models.losses import CrossEntropyCost
 from ..models.model_factory import create_model
 from ..parallel.communicator import dist_env, init_process_group
 from .graph_group import SyncGraphGroup

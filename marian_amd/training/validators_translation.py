@@ -10,7 +10,6 @@ import math
 from collections import Counter
 from typing import List
 
-import torch
 
 from ..common.logging_utils import log_info
 from ..common.options import Options
