@@ -17,8 +17,7 @@ from __future__ import annotations
 
 import argparse
 import os
-import sys
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, Dict, List, Optional
 
 import yaml

@@ -9,7 +9,6 @@ NUL-terminated YAML text of the model config. We reproduce that exactly with
 """
 from __future__ import annotations
 
-import io as _io
 import os
 from typing import Dict, Tuple
 

@@ -10,9 +10,9 @@ from __future__ import annotations
 import os
 import random
 import sqlite3
-from typing import Iterator, List, Optional, Sequence
+from typing import Iterator, Sequence
 
-from .corpus import Corpus, SentenceTuple, _open
+from .corpus import Corpus, SentenceTuple
 from .vocab import Vocab
 
 

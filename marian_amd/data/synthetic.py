@@ -6,7 +6,7 @@ shape a real corpus would (random token ids, EOS-terminated, full masks).
 """
 from __future__ import annotations
 
-from typing import Iterator, List
+from typing import Iterator
 
 import torch
 

@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import os
 from collections import Counter
-from typing import Dict, Iterable, List, Optional
+from typing import Dict, Iterable, List
 
 import yaml
 

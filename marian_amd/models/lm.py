@@ -12,7 +12,7 @@ from __future__ import annotations
 import torch
 
 from ..common.options import Options
-from .transformer import TransformerModel, _log_mask
+from .transformer import TransformerModel
 
 
 def _lm_options(options: Options) -> Options:

@@ -14,8 +14,7 @@ see SURVEY.md §0.
 """
 from __future__ import annotations
 
-import math
-from typing import Dict, Optional
+from typing import Dict
 
 import numpy as np
 import torch

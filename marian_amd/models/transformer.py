@@ -16,16 +16,14 @@ Marian semantics reproduced:
 """
 from __future__ import annotations
 
-import math
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import numpy as np
 import torch
 import torch.nn as nn
 
-from .. import ops
 from ..common.options import Options
-from .layers import (Dense, Embedding, FFN, LayerNorm, MultiHeadAttention,
+from .layers import (Dense, Embedding, FFN, MultiHeadAttention, 
                      PrePostProcessor)
 
 NEG_INF = -1e9  # marian uses additive log-masks with large negative value

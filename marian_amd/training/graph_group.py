@@ -21,7 +21,6 @@ from __future__ import annotations
 
 from typing import Dict, List, Optional
 
-import numpy as np
 import torch
 
 from ..common.logging_utils import log_info

@@ -8,7 +8,6 @@ Display line format follows the reference::
 from __future__ import annotations
 
 import math
-import time
 from typing import Optional
 
 from ..common.logging_utils import log_info, Timer

@@ -247,7 +247,6 @@ class Train:
         then "word v1 v2 ..." lines) into the embedding tables, optionally
         L2-normalized (--embedding-normalization)
         (ref: layers/embedding.cpp + node_initializers fromWord2vec)."""
-        import numpy as np
         import torch
         targets = [(self.vocabs[0], self.model.src_embed),
                    (self.vocabs[-1], self.model.trg_embed)]

@@ -7,7 +7,7 @@ from __future__ import annotations
 import os
 import re
 from dataclasses import dataclass, field
-from typing import Dict, List
+from typing import Dict
 
 import yaml
 

@@ -11,7 +11,6 @@ from collections import Counter
 from typing import List
 
 
-from ..common.logging_utils import log_info
 from ..common.options import Options
 
 

@@ -22,7 +22,6 @@ case); capture happens at step 2.
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 

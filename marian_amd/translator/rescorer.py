@@ -6,12 +6,11 @@ from __future__ import annotations
 
 import math
 import sys
-from typing import List, Optional
 
 import torch
 
 from .. import ops
-from ..common.logging_utils import create_loggers, log_info
+from ..common.logging_utils import create_loggers
 from ..common.options import Options
 from ..data.batch_generator import BatchGenerator
 from ..data.corpus import Corpus

@@ -8,7 +8,6 @@ protocol, plus an HTTP POST /translate convenience endpoint.
 import sys
 
 from ..common.logging_utils import create_loggers, log_info
-from ..common.options import Options
 from .translate import TranslateService
 
 

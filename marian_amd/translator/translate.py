@@ -4,14 +4,13 @@ output_printer.{h,cpp}, command/marian_decoder.cpp).
 """
 from __future__ import annotations
 
-import os
 import sys
-from typing import Iterator, List, Optional
+from typing import List
 
 import torch
 
 from ..common import io_npz
-from ..common.logging_utils import create_loggers, log_info
+from ..common.logging_utils import create_loggers
 from ..common.options import Options
 from ..data.batch_generator import BatchGenerator
 from ..data.corpus import TextInput
