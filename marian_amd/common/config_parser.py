@@ -299,6 +299,14 @@ class ConfigParser:
             a("tsv", "bool", False, "Tab-separated input", "translation")
             a("tsv-fields", int, 0, "Number of TSV fields", "translation")
             a("data-threads", int, 8, "Data reading threads", "translation")
+        if self.mode == T:
+            # validation-time decoding options (marian-train accepts the
+            # decoder flags for its translation validators)
+            a("beam-size", int, 12, "Beam size (validation decoding)", "validation")
+            a("normalize", float, 0.0, "Divide score by length^alpha (validation decoding)", "validation")
+            a("word-penalty", float, 0.0, "Word penalty (validation decoding)", "validation")
+            a("max-length-factor", float, 3.0, "Max target length factor (validation decoding)", "validation")
+            a("allow-unk", "bool", False, "Allow <unk> in validation output", "validation")
         if self.mode in (D, MODE_SERVER):
             a("beam-size", int, 12, "Beam size", "translation")
             a("normalize", float, 0.0, "Divide score by length^alpha", "translation")

@@ -1,0 +1,74 @@
+"""Validator coverage (ref: src/training/validator.{h,cpp})."""
+import math
+
+import torch
+
+from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+from marian_amd.training.validators import create_validators
+from marian_amd.training.validators_translation import corpus_bleu, corpus_chrf
+from marian_amd.data.synthetic import write_synthetic_corpus
+from marian_amd.training.train import Train
+
+
+def test_corpus_bleu_hand_values():
+    # perfect match -> 100
+    assert abs(corpus_bleu(["a b c d e"], ["a b c d e"]) - 100.0) < 1e-6
+    # no overlap -> 0
+    assert corpus_bleu(["x y z w v"], ["a b c d e"]) < 1e-6
+    # known partial case: hyp 4 tokens, 3/4 unigrams, 2/3 bigrams...
+    b = corpus_bleu(["a b c x"], ["a b c d"])
+    assert 0 < b < 100
+
+
+def test_chrf_hand_values():
+    assert corpus_chrf(["hello world"], ["hello world"]) > 99.0
+    assert corpus_chrf(["aaaaaa"], ["zzzzzz"]) < 1.0
+    mid = corpus_chrf(["hello there"], ["hello world"])
+    assert 10 < mid < 90
+
+
+def test_validators_run_in_training(tmp_path):
+    """--valid-sets + --valid-metrics drive validators; best/stalled are
+    tracked in the progress file; --valid-translation-output written."""
+    src, trg = str(tmp_path / "c.src"), str(tmp_path / "c.trg")
+    write_synthetic_corpus(src, trg, 60, 25, 3, 7, seed=1)
+    vout = str(tmp_path / "dev.out")
+    argv = ["--train-sets", src, trg,
+            "--vocabs", str(tmp_path / "v1.yml"), str(tmp_path / "v2.yml"),
+            "--model", str(tmp_path / "m.npz"),
+            "--dim-emb", "16", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "32",
+            "--mini-batch", "8", "--after", "4u", "--cpu-threads", "1",
+            "--valid-sets", src, trg, "--valid-freq", "2u",
+            "--valid-metrics", "cross-entropy", "perplexity", "bleu",
+            "--valid-mini-batch", "8", "--beam-size", "2",
+            "--valid-translation-output", vout,
+            "--seed", "5"]
+    o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    t = Train(o)
+    vals = t._validators() if hasattr(t, "_validators") else None
+    t.run()
+    st = t.state
+    assert "cross-entropy" in st.validators
+    assert st.validators["cross-entropy"]["best"] is not None
+    assert "perplexity" in st.validators
+    ppl = st.validators["perplexity"]["best"]
+    ce = st.validators["cross-entropy"]["best"]
+    assert abs(math.log(ppl) - ce) < 0.2   # ppl = exp(ce-mean-words)
+    assert "bleu" in st.validators
+    import os
+    assert os.path.exists(vout)
+
+
+def test_accuracy_validator_metric(tmp_path):
+    src, trg = str(tmp_path / "c.src"), str(tmp_path / "c.trg")
+    write_synthetic_corpus(src, trg, 30, 20, 3, 6, seed=2)
+    argv = ["--train-sets", src, trg,
+            "--vocabs", str(tmp_path / "v1.yml"), str(tmp_path / "v2.yml"),
+            "--valid-sets", src, trg, "--valid-metrics", "accuracy"]
+    o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    from marian_amd.data.vocab import Vocab
+    v = Vocab.create_from_corpus([src])
+    vals = create_validators(o, [v, v])
+    assert len(vals) == 1 and vals[0].name == "accuracy"
+    assert vals[0].lower_is_better is False
