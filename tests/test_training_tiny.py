@@ -384,3 +384,44 @@ def test_dynamic_gradient_scaling_caps_outliers():
         gg.update(synthetic_batch(4, 6, 6, 30, seed=step))
     assert gg._gnorm_count == 12
     assert gg._gnorm_avg > 0
+
+
+def test_sigterm_graceful_save(tmp_path):
+    """SIGTERM triggers a checkpoint + progress save before exit
+    (ref: common/signal_handling.cpp + training loop save)."""
+    import os
+    import signal
+    import subprocess
+    import sys
+    import time
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    src, trg = str(tmp_path / "c.src"), str(tmp_path / "c.trg")
+    write_synthetic_corpus(src, trg, 400, 30, 3, 8, seed=9)
+    model = tmp_path / "m.npz"
+    env = dict(os.environ, PYTHONPATH=os.getcwd())
+    p = subprocess.Popen(
+        [sys.executable, "-m", "marian_amd.cli.marian_train",
+         "--train-sets", src, trg,
+         "--vocabs", str(tmp_path / "v1.yml"), str(tmp_path / "v2.yml"),
+         "--model", str(model), "--dim-emb", "16", "--enc-depth", "1",
+         "--dec-depth", "1", "--transformer-heads", "2",
+         "--transformer-dim-ffn", "32", "--mini-batch", "4",
+         "--after", "100000u", "--disp-freq", "5", "--cpu-threads", "1",
+         "--seed", "3"],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    try:
+        deadline = time.time() + 60
+        # wait until training is actually stepping (first disp line)
+        seen = False
+        while time.time() < deadline and not seen:
+            line = p.stdout.readline()
+            if "Up. " in line:
+                seen = True
+        assert seen, "training never started stepping"
+        p.send_signal(signal.SIGTERM)
+        p.wait(timeout=60)
+    finally:
+        if p.poll() is None:
+            p.kill()
+    assert model.exists(), "no checkpoint written on SIGTERM"
+    assert (tmp_path / "m.npz.progress.yml").exists()
