@@ -425,3 +425,40 @@ def test_sigterm_graceful_save(tmp_path):
             p.kill()
     assert model.exists(), "no checkpoint written on SIGTERM"
     assert (tmp_path / "m.npz.progress.yml").exists()
+
+
+def test_cost_scaler_dynamics():
+    """CostScaler: overflow halves (floored), N clean steps double
+    (ref: graph_group.cpp DynamicLossScaling)."""
+    from marian_amd.training.graph_group import CostScaler
+    cs = CostScaler(["8", "4", "2", "2"], enabled=True)
+    assert cs.scale == 256.0 and cs.freq == 4 and cs.min_scale == 4.0
+    cs.on_overflow()
+    assert cs.scale == 128.0 and cs.clean == 0
+    for _ in range(4):
+        cs.on_clean()
+    assert cs.scale == 256.0 and cs.clean == 0
+    for _ in range(20):
+        cs.on_overflow()
+    assert cs.scale == 4.0          # min floor
+    assert not CostScaler([], enabled=True).enabled
+    assert not CostScaler(["8"], enabled=False).enabled
+
+
+def test_adagrad_optimizer(workdir):
+    """--optimizer adagrad trains and its state round-trips."""
+    tmp_path, src, trg = workdir
+    opts = _opts(tmp_path, src, src, extra=["--optimizer", "adagrad",
+                                            "--after-batches", "3"])
+    tr = Train(opts)
+    tr.run()
+    opt = tr.graph_group.optimizer
+    assert opt.core.name == "adagrad"
+    assert float(opt.core.gt.abs().sum()) > 0
+    # resume restores accumulator state
+    opts2 = _opts(tmp_path, src, src, extra=["--optimizer", "adagrad",
+                                             "--after-batches", "5"])
+    tr2 = Train(opts2)
+    assert float((tr2.graph_group.optimizer.core.gt
+                  - opt.core.gt).abs().max()) < 1e-6
+    tr2.run()
