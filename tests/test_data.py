@@ -146,3 +146,23 @@ def test_mini_batch_warmup_scaling():
     gen2 = BatchGenerator(iter(tuples), mini_batch=8, maxi_batch=5,
                           maxi_batch_sort="none", prefetch=False)
     assert max(b.size() for b in gen2) == 8
+
+
+def test_tsv_corpus(tmp_path):
+    """--tsv: one tab-separated file carries both streams
+    (ref: corpus_base.cpp TSV handling)."""
+    from marian_amd.data.corpus import Corpus
+    from marian_amd.data.vocab import Vocab
+    tsv = tmp_path / "data.tsv"
+    rows = [("a b c", "x y"), ("d e", "z w v"), ("f", "u")]
+    tsv.write_text("\n".join(f"{s}\t{t}" for s, t in rows) + "\n")
+    plain_src = tmp_path / "s.txt"
+    plain_src.write_text("\n".join(s for s, _ in rows) + "\n")
+    plain_trg = tmp_path / "t.txt"
+    plain_trg.write_text("\n".join(t for _, t in rows) + "\n")
+    v = Vocab.create_from_corpus([str(plain_src), str(plain_trg)])
+    c_tsv = Corpus([str(tsv)], [v, v], max_length=20, tsv=True, tsv_fields=2)
+    c_sep = Corpus([str(plain_src), str(plain_trg)], [v, v], max_length=20)
+    got = [t[1] for t in c_tsv.iter_epoch(shuffle=False)]
+    want = [t[1] for t in c_sep.iter_epoch(shuffle=False)]
+    assert got == want
