@@ -160,3 +160,33 @@ def test_distributed_checkpoint_shards(tmp_path):
     items, _ = io_npz.load_items(str(tmp_path / "model.npz.optimizer.npz"))
     assert any(k.startswith("rank0_") for k in items)
     assert any(k.startswith("rank1_") for k in items)
+
+
+@pytest.mark.timeout(300)
+def test_multinode_style_rendezvous(tmp_path):
+    """Two separate torchrun launchers (--nnodes=2) rendezvous and train
+    (the reference's MPI multi-node ≙ torchrun multi-launcher)."""
+    import subprocess
+    import sys
+    env = dict(os.environ, PYTHONPATH=os.getcwd())
+    args = [sys.executable, "-m", "torch.distributed.run", "--nnodes=2",
+            "--nproc-per-node", "1", "--master-addr", "127.0.0.1",
+            "--master-port", "29743", os.path.join(os.getcwd(), "bench.py"),
+            "--gpus", "2", "--steps", "2", "--warmup", "1",
+            "--model", "transformer-tiny"]
+    a0 = list(args)
+    a0.insert(3, "--node-rank=0")
+    p0 = subprocess.Popen(a0 + ["--seq-len", "16"],
+                          env=dict(env), stdout=subprocess.PIPE, text=True,
+                          stderr=subprocess.STDOUT)
+    # patch node rank for the second launcher
+    a1 = list(args)
+    a1.insert(3, "--node-rank=1")
+    a0_extra = ["--seq-len", "16"]
+    p1 = subprocess.Popen(a1 + a0_extra, env=dict(env),
+                          stdout=subprocess.PIPE, text=True,
+                          stderr=subprocess.STDOUT)
+    out0, _ = p0.communicate(timeout=240)
+    p1.wait(timeout=240)
+    assert p0.returncode == 0 and p1.returncode == 0, out0[-800:]
+    assert '"n_gpus": 2' in out0 and '"parallelism": "dp2"' in out0
