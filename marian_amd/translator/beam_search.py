@@ -148,8 +148,11 @@ class BeamSearch:
                     key = (id(model), B * K, Ts, max_len, K, sup,
                            self.eos_id, self.n_best)
                     cache = getattr(self, "_graph_cache", None)
-                    if cache is not None and cache[0] == key:
-                        graphed = cache[1]
+                    if cache is None or not isinstance(cache, dict):
+                        cache = {}
+                        self._graph_cache = cache
+                    if key in cache:
+                        graphed = cache[key]
                         graphed.load_batch(states[0]["enc"],
                                            states[0]["mask"])
                     else:
@@ -158,7 +161,9 @@ class BeamSearch:
                             B * K, max_len, K, sup,
                             eos_id=self.eos_id, n_best=self.n_best)
                         graphed.capture()
-                        self._graph_cache = (key, graphed)
+                        if len(cache) >= 4:   # bound captured-graph memory
+                            cache.pop(next(iter(cache)))
+                        cache[key] = graphed
                     if not getattr(BeamSearch, "_graph_logged", False):
                         from ..common.logging_utils import log_info
                         log_info("beam search: HIP-graph decode step active")
