@@ -462,3 +462,20 @@ def test_adagrad_optimizer(workdir):
     assert float((tr2.graph_group.optimizer.core.gt
                   - opt.core.gt).abs().max()) < 1e-6
     tr2.run()
+
+
+def test_keep_best_checkpoint(workdir):
+    """--keep-best writes model.npz.best-<metric>.npz on improvement."""
+    import os
+    tmp_path, src, trg = workdir
+    opts = _opts(tmp_path, src, src,
+                 extra=["--after", "4u", "--keep-best",
+                        "--valid-sets", src, src, "--valid-freq", "2u",
+                        "--valid-metrics", "cross-entropy",
+                        "--valid-mini-batch", "8"])
+    tr = Train(opts)
+    tr.run()
+    model = opts.get("model")
+    cands = [f for f in os.listdir(os.path.dirname(model))
+             if "best" in f]
+    assert cands, "no best-checkpoint written"

@@ -72,3 +72,30 @@ def test_accuracy_validator_metric(tmp_path):
     vals = create_validators(o, [v, v])
     assert len(vals) == 1 and vals[0].name == "accuracy"
     assert vals[0].lower_is_better is False
+
+
+def test_valid_script_path(tmp_path):
+    """--valid-script-path: external script scores the validation output
+    (ref: validator.h ScriptValidator)."""
+    import os
+    import stat
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    src, trg = str(tmp_path / "c.src"), str(tmp_path / "c.trg")
+    write_synthetic_corpus(src, trg, 40, 20, 3, 6, seed=4)
+    script = tmp_path / "score.sh"
+    script.write_text("#!/bin/sh\necho 41.5\n")
+    script.chmod(script.stat().st_mode | stat.S_IEXEC)
+    argv = ["--train-sets", src, trg,
+            "--vocabs", str(tmp_path / "v1.yml"), str(tmp_path / "v2.yml"),
+            "--model", str(tmp_path / "m.npz"),
+            "--dim-emb", "16", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "32",
+            "--mini-batch", "8", "--after", "2u", "--cpu-threads", "1",
+            "--valid-sets", src, trg, "--valid-freq", "2u",
+            "--valid-metrics", "translation", "--beam-size", "1",
+            "--valid-script-path", str(script), "--seed", "6"]
+    o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    t = Train(o)
+    t.run()
+    assert "translation" in t.state.validators
+    assert abs(t.state.validators["translation"]["best"] - 41.5) < 1e-6
