@@ -108,9 +108,11 @@ class Translator:
 
     def __init__(self, options: Options):
         self.opt = options
+        nthreads = int(options.get("cpu-threads", 0) or 0)
         self.device = (torch.device("cuda", 0) if torch.cuda.is_available()
-                       and int(options.get("cpu-threads", 0) or 0) == 0
-                       else torch.device("cpu"))
+                       and nthreads == 0 else torch.device("cpu"))
+        if self.device.type == "cpu" and nthreads > 0:
+            torch.set_num_threads(nthreads)   # --cpu-threads for decoding
         vocab_paths = options.list("vocabs")
         if not vocab_paths:
             raise ValueError("marian-decoder needs --vocabs")
