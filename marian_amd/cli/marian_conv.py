@@ -78,7 +78,39 @@ def main(argv=None):
     p.add_argument("--to", dest="dst", required=True)
     p.add_argument("--gemm-type", default="float32",
                    choices=list(_DT.keys()))
+    p.add_argument("--shortlist", nargs="+", default=[],
+                   help="binarize a lexical shortlist: "
+                        "--shortlist DIC SRC_VOCAB TRG_VOCAB "
+                        "(--from/--to are the dic in/out paths)")
     args = p.parse_args(argv)
+
+    if args.shortlist:
+        # text .dic -> id-resolved .blex.npz (ref: marian-conv --shortlist)
+        from ..data.vocab import Vocab
+        dic = args.shortlist[0] if args.shortlist else args.src
+        sv = Vocab.load(args.shortlist[1])
+        tv = Vocab.load(args.shortlist[2])
+        src_ids, trg_ids, probs = [], [], []
+        with open(dic, encoding="utf-8") as f:
+            for line in f:
+                parts = line.rstrip("\n").split("\t")
+                if len(parts) < 2:
+                    parts = line.split()
+                if len(parts) < 2:
+                    continue
+                sid = sv.token2id.get(parts[0])
+                tid = tv.token2id.get(parts[1])
+                if sid is None or tid is None:
+                    continue
+                src_ids.append(sid)
+                trg_ids.append(tid)
+                probs.append(float(parts[2]) if len(parts) > 2 else 1.0)
+        np.savez(args.dst, src=np.asarray(src_ids, dtype=np.int64),
+                 trg=np.asarray(trg_ids, dtype=np.int64),
+                 prob=np.asarray(probs, dtype=np.float32))
+        print(f"Binarized shortlist {dic} -> {args.dst} "
+              f"({len(src_ids)} pairs)")
+        return
 
     if args.src.endswith(".npz"):
         items, cfg = io_npz.load_items(args.src)

@@ -8,6 +8,11 @@ A shortlist restricts the output vocabulary per batch to:
 Dictionary format (the reference's text ``.dic``/tsv form):
     src_word<TAB>trg_word<TAB>prob
 One line per pair; probabilities sort candidates per source word.
+
+Binary form (``.blex.npz``, written by ``marian-conv --shortlist``,
+ref: the reference's binary shortlist in marian-conv): three aligned
+id-resolved arrays ``src``, ``trg``, ``prob`` — loads without vocab
+lookups or text parsing.
 """
 from __future__ import annotations
 
@@ -24,6 +29,15 @@ class LexicalShortlistGenerator:
         self.best_num = best_num
         self.trg_vocab_size = len(trg_vocab)
         table: Dict[int, List[tuple]] = defaultdict(list)
+        if path.endswith(".npz"):
+            import numpy as np
+            z = np.load(path)
+            for sid, tid, p in zip(z["src"], z["trg"], z["prob"]):
+                table[int(sid)].append((float(p), int(tid)))
+            self.table = {
+                sid: [tid for _, tid in sorted(c, reverse=True)[:best_num]]
+                for sid, c in table.items()}
+            return
         with open(path, "r", encoding="utf-8") as f:
             for line in f:
                 parts = line.rstrip("\n").split("\t")

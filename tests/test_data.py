@@ -166,3 +166,28 @@ def test_tsv_corpus(tmp_path):
     got = [t[1] for t in c_tsv.iter_epoch(shuffle=False)]
     want = [t[1] for t in c_sep.iter_epoch(shuffle=False)]
     assert got == want
+
+
+def test_binary_shortlist_roundtrip(tmp_path):
+    """marian-conv --shortlist binarizes a .dic; the generator loads the
+    binary form identically (ref: marian-conv shortlist binarization)."""
+    from marian_amd.cli.marian_conv import main as conv_main
+    from marian_amd.data.shortlist import LexicalShortlistGenerator
+    from marian_amd.data.vocab import Vocab
+    import torch
+    corpus = tmp_path / "c.txt"
+    corpus.write_text("aa bb cc dd\n")
+    v = Vocab.create_from_corpus([str(corpus)])
+    v.save(str(tmp_path / "v.yml"))
+    dic = tmp_path / "lex.dic"
+    dic.write_text("aa\tbb\t0.9\naa\tcc\t0.5\nbb\tdd\t0.7\n")
+    out = tmp_path / "lex.blex.npz"
+    conv_main(["--from", str(dic), "--to", str(out),
+               "--shortlist", str(dic), str(tmp_path / "v.yml"),
+               str(tmp_path / "v.yml")])
+    g_txt = LexicalShortlistGenerator(str(dic), v, v, first_num=2,
+                                      best_num=5)
+    g_bin = LexicalShortlistGenerator(str(out), v, v, first_num=2,
+                                      best_num=5)
+    src = torch.tensor([[v.token2id["aa"], v.token2id["bb"]]])
+    assert g_txt.generate(src).tolist() == g_bin.generate(src).tolist()
