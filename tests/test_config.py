@@ -112,3 +112,26 @@ def test_lr_warmup_cycle_and_reload_base():
     st2.warmup_start = 100
     sch2 = Scheduler(o2, st2)
     assert sch2.learning_rate() == 0.25  # first warmup step again
+
+
+def test_devices_flag_spawns_torchrun(monkeypatch):
+    """`marian-train --devices 0 1` self-launches torchrun (one process
+    per GPU), matching the reference's single-command multi-GPU UX."""
+    import os
+    from marian_amd.cli.marian_train import maybe_spawn_workers
+    monkeypatch.delenv("RANK", raising=False)
+    monkeypatch.delenv("LOCAL_RANK", raising=False)
+    got = maybe_spawn_workers(["--devices", "0", "1", "--model", "m.npz"])
+    assert got is not None
+    cmd, env = got
+    assert "--nproc-per-node=2" in cmd
+    assert env["CUDA_VISIBLE_DEVICES"] == "0,1"
+    assert cmd[-2:] == ["--model", "m.npz"]
+    # under a launcher: no respawn
+    monkeypatch.setenv("RANK", "0")
+    assert maybe_spawn_workers(["--devices", "0", "1"]) is None
+    monkeypatch.delenv("RANK")
+    # single device: no respawn
+    assert maybe_spawn_workers(["--devices", "3"]) is None
+    got = maybe_spawn_workers(["--num-devices", "4"])
+    assert got and "--nproc-per-node=4" in got[0]
