@@ -12,6 +12,14 @@ from ..translator.translate import main_decoder
 
 
 def main():
+    from .marian_train import maybe_spawn_workers
+    spawn = maybe_spawn_workers(sys.argv[1:])
+    if spawn is not None:
+        import subprocess
+        cmd, env = spawn
+        cmd[cmd.index("marian_amd.cli.marian_train")] = \
+            "marian_amd.cli.marian_decoder"
+        sys.exit(subprocess.call(cmd, env=env))
     sys.exit(main_decoder())
 
 

@@ -4,6 +4,7 @@ output_printer.{h,cpp}, command/marian_decoder.cpp).
 """
 from __future__ import annotations
 
+import os
 import sys
 from typing import List
 
@@ -109,8 +110,10 @@ class Translator:
     def __init__(self, options: Options):
         self.opt = options
         nthreads = int(options.get("cpu-threads", 0) or 0)
-        self.device = (torch.device("cuda", 0) if torch.cuda.is_available()
-                       and nthreads == 0 else torch.device("cpu"))
+        local_rank = int(os.environ.get("LOCAL_RANK", 0) or 0)
+        self.device = (torch.device("cuda", local_rank)
+                       if torch.cuda.is_available() and nthreads == 0
+                       else torch.device("cpu"))
         if self.device.type == "cpu" and nthreads > 0:
             torch.set_num_threads(nthreads)   # --cpu-threads for decoding
         vocab_paths = options.list("vocabs")
@@ -218,14 +221,50 @@ def main_decoder(argv=None):
     tr = Translator(opts)
     printer = OutputPrinter(opts, tr.trg_vocab)
 
+    from ..parallel.communicator import dist_env, init_process_group
+    rank0, _l0, world0 = dist_env()
     inputs = opts.list("input") or ["stdin"]
     if inputs[0] == "stdin":
-        lines = [l.rstrip("\n") for l in sys.stdin]
+        if world0 > 1:
+            # only rank 0 owns stdin; broadcast the lines
+            import torch.distributed as dist
+            init_process_group()
+            obj = [[l.rstrip("\n") for l in sys.stdin]] if rank0 == 0 \
+                else [None]
+            dist.broadcast_object_list(obj, src=0)
+            lines = obj[0]
+        else:
+            lines = [l.rstrip("\n") for l in sys.stdin]
     else:
         with open(inputs[0], "r", encoding="utf-8") as f:
             lines = [l.rstrip("\n") for l in f]
 
     out_path = opts.get("output", "stdout")
+
+    # multi-process decode (ref: Translate<BeamSearch> ThreadPool across
+    # devices + ordered OutputCollector): under torchrun each rank
+    # translates a strided shard; rank 0 gathers and writes in order.
+    rank, _local, world = dist_env()
+    if world > 1:
+        import torch.distributed as dist
+        init_process_group()
+        shard = lines[rank::world]
+        results = tr.translate_lines(shard)
+        mine = [(rank + i * world, printer.print(rank + i * world, r))
+                for i, r in enumerate(results)]
+        gathered = [None] * world if rank == 0 else None
+        dist.gather_object(mine, gathered, dst=0)
+        if rank == 0:
+            out = sys.stdout if out_path == "stdout" else open(out_path, "w")
+            collector = OutputCollector(out)
+            for part in gathered:
+                for i, text in part:
+                    collector.write(i, text)
+            if out is not sys.stdout:
+                out.close()
+        dist.barrier()
+        return 0
+
     out = sys.stdout if out_path == "stdout" else open(out_path, "w")
     collector = OutputCollector(out)
     results = tr.translate_lines(lines)

@@ -190,3 +190,40 @@ def test_multinode_style_rendezvous(tmp_path):
     p1.wait(timeout=240)
     assert p0.returncode == 0 and p1.returncode == 0, out0[-800:]
     assert '"n_gpus": 2' in out0 and '"parallelism": "dp2"' in out0
+
+
+@pytest.mark.timeout(300)
+def test_multiprocess_decode_ordered(tmp_path):
+    """marian-decoder --devices N shards lines across ranks; rank 0
+    merges in order — output identical to single-process decoding
+    (ref: Translate<BeamSearch> ThreadPool + ordered OutputCollector)."""
+    import subprocess
+    import sys
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    src, trg = str(tmp_path / "c.src"), str(tmp_path / "c.trg")
+    write_synthetic_corpus(src, trg, 24, 20, 3, 6, seed=8)
+    env = dict(os.environ, PYTHONPATH=os.getcwd())
+    model = str(tmp_path / "m.npz")
+    base = [sys.executable, "-m", "marian_amd.cli.marian_train",
+            "--train-sets", src, trg,
+            "--vocabs", str(tmp_path / "v1.yml"), str(tmp_path / "v2.yml"),
+            "--model", model, "--dim-emb", "16", "--enc-depth", "1",
+            "--dec-depth", "1", "--transformer-heads", "2",
+            "--transformer-dim-ffn", "32", "--mini-batch", "8",
+            "--after", "2u", "--cpu-threads", "1", "--seed", "3"]
+    assert subprocess.run(base, env=env, capture_output=True,
+                          timeout=240).returncode == 0
+    dec = [sys.executable, "-m", "marian_amd.cli.marian_decoder",
+           "--models", model,
+           "--vocabs", str(tmp_path / "v1.yml"), str(tmp_path / "v2.yml"),
+           "--beam-size", "2", "--cpu-threads", "1", "--input", src,
+           "--quiet"]
+    r1 = subprocess.run(dec + ["--output", str(tmp_path / "o1.txt")],
+                        env=env, capture_output=True, timeout=240)
+    assert r1.returncode == 0, r1.stderr[-500:]
+    r2 = subprocess.run(dec + ["--devices", "0", "1",
+                               "--output", str(tmp_path / "o2.txt")],
+                        env=env, capture_output=True, timeout=240)
+    assert r2.returncode == 0, r2.stderr[-500:]
+    assert (tmp_path / "o1.txt").read_text() == \
+        (tmp_path / "o2.txt").read_text()
