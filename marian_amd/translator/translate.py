@@ -52,6 +52,7 @@ class OutputPrinter:
     def __init__(self, options: Options, vocab: Vocab):
         self.vocab = vocab
         self.n_best = bool(options.get("n-best", False))
+        self.feature = options.get("n-best-feature", "F0") or "F0"
         self.word_scores = bool(options.get("word-scores", False))
         self.alignment = options.get("alignment")
 
@@ -79,7 +80,8 @@ class OutputPrinter:
                 line = f"{sentence_id} ||| {self.text(h)}"
                 if self.alignment:
                     line += f" ||| {self._align_str(h)}"
-                line += f" ||| F0= {h.score:.6f} ||| {h.normalized:.6f}"
+                line += (f" ||| {self.feature}= {h.score:.6f}"
+                         f" ||| {h.normalized:.6f}")
                 lines.append(line)
             return "\n".join(lines)
         out = self.text(result.best)
