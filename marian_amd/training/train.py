@@ -293,9 +293,20 @@ class Train:
             gen = self._batch_generator(skip_sentences=restore_skip)
             restore_skip = 0
             got_batches = False
+            if self.world > 1:
+                # round-robin in COMPLETE rounds of `world` batches: every
+                # rank must run the same number of updates (collectives
+                # match); an incomplete tail round is dropped on all ranks
+                # alike (the generator is identical everywhere).
+                def rounds(g):
+                    buf = []
+                    for b in g:
+                        buf.append(b)
+                        if len(buf) == self.world:
+                            yield buf[self.rank]
+                            buf = []
+                gen = rounds(gen)
             for i, batch in enumerate(gen):
-                if self.world > 1 and i % self.world != self.rank:
-                    continue
                 got_batches = True
                 gg.set_lr(sched.learning_rate())
                 lv = gg.update(batch)

@@ -227,3 +227,43 @@ def test_multiprocess_decode_ordered(tmp_path):
     assert r2.returncode == 0, r2.stderr[-500:]
     assert (tmp_path / "o1.txt").read_text() == \
         (tmp_path / "o2.txt").read_text()
+
+
+def _train_worker(rank, world, port, workdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    argv = ["--train-sets", f"{workdir}/c.src", f"{workdir}/c.trg",
+            "--vocabs", f"{workdir}/v1.yml", f"{workdir}/v2.yml",
+            "--model", f"{workdir}/m{world}.npz",
+            "--dim-emb", "16", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "32",
+            "--mini-batch", "4", "--after-epochs", "2",
+            "--cpu-threads", "1", "--seed", "3", "--overwrite", "--quiet"]
+    o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    Train(o).run()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_full_train_loop_world2_odd_batches(tmp_path):
+    """Train.run with world=2 and an ODD number of batches per epoch must
+    not deadlock (incomplete rounds are dropped on all ranks alike)."""
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    # 18 sentences / mini-batch 4 -> 5 batches per epoch (odd)
+    write_synthetic_corpus(str(tmp_path / "c.src"), str(tmp_path / "c.trg"),
+                           18, 25, 3, 7, seed=11)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_train_worker,
+                         args=(r, 2, 29547, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    assert (tmp_path / "m2.npz").exists()
