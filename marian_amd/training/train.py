@@ -371,8 +371,19 @@ class Train:
         if validators is None:
             return
         for v in validators:
-            value = v.run(self.model, self.device,
-                          self.graph_group.compute_dtype)
+            # rank 0 validates (avoids duplicate beam search, N copies of
+            # --valid-script runs and races on the output file); the value
+            # is broadcast so every rank's scheduler state stays in step
+            if self.world > 1:
+                import torch.distributed as dist
+                obj = [v.run(self.model, self.device,
+                             self.graph_group.compute_dtype)] \
+                    if self.rank == 0 else [None]
+                dist.broadcast_object_list(obj, src=0)
+                value = obj[0]
+            else:
+                value = v.run(self.model, self.device,
+                              self.graph_group.compute_dtype)
             improved = self.scheduler.validated(v.name, value,
                                                 v.lower_is_better)
             st = self.state.validators.get(v.name, {})

@@ -243,6 +243,9 @@ def _train_worker(rank, world, port, workdir):
             "--dim-emb", "16", "--enc-depth", "1", "--dec-depth", "1",
             "--transformer-heads", "2", "--transformer-dim-ffn", "32",
             "--mini-batch", "4", "--after-epochs", "2",
+            "--valid-sets", f"{workdir}/c.src", f"{workdir}/c.trg",
+            "--valid-freq", "2u", "--valid-metrics", "cross-entropy",
+            "--valid-mini-batch", "4",
             "--cpu-threads", "1", "--seed", "3", "--overwrite", "--quiet"]
     o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
     Train(o).run()
