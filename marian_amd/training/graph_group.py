@@ -304,12 +304,26 @@ class SyncGraphGroup:
 
     # ------------------------------------------------------------------
     # checkpointing (4-file contract, ref: SURVEY.md §6 checkpoint row)
+    def swap_smoothed(self, into: bool):
+        """All-rank smoothing swap (collective all-gather inside):
+        marian validates and SAVES the Polyak-averaged parameters
+        (ref: SyncGraphGroup swapParams + exponential_smoothing)."""
+        opt = self.optimizer
+        if opt.smooth is None:
+            return
+        if into:
+            opt.swap_in_smoothed(self.comm if self.world > 1 else None)
+        else:
+            opt.swap_out_smoothed(self.comm if self.world > 1 else None)
+
     def save_checkpoint(self, prefix: str, model_config: dict,
                         training_state=None) -> None:
         from ..common import io_npz
+        self.swap_smoothed(True)   # marian saves the smoothed model
         if self.rank == 0:
             items = self.model.marian_items()
             io_npz.save_items(prefix, items, config=model_config)
+        self.swap_smoothed(False)
         # every rank contributes its optimizer shard; single-process path
         # gathers locally. (multi-proc: each rank writes its shard file,
         # rank 0 merges on load; simple + correct for single node)

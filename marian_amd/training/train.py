@@ -370,6 +370,14 @@ class Train:
     def _validate(self, validators):
         if validators is None:
             return
+        # validate the smoothed parameters (all ranks swap: collective)
+        self.graph_group.swap_smoothed(True)
+        try:
+            self._run_validators(validators)
+        finally:
+            self.graph_group.swap_smoothed(False)
+
+    def _run_validators(self, validators):
         for v in validators:
             # rank 0 validates (avoids duplicate beam search, N copies of
             # --valid-script runs and races on the output file); the value

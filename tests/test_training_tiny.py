@@ -176,6 +176,17 @@ def test_exponential_smoothing(workdir):
     opt.swap_out_smoothed()
     for a, b in zip(before, tr.graph_group.store.buckets):
         assert torch.equal(a, b.flat_param)
+    # the SAVED checkpoint holds the smoothed parameters (marian saves
+    # the Polyak average; raw params stay live for training)
+    import numpy as np
+    from marian_amd.common import io_npz
+    items, _ = io_npz.load_items(opts.get("model"))
+    opt.swap_in_smoothed()
+    name, t = next(iter(tr.model._name_map().items()))
+    want = t.detach().float().numpy().copy()   # copy: numpy() aliases
+    opt.swap_out_smoothed()
+    got = items[name].reshape(want.shape)
+    np.testing.assert_allclose(got, want, atol=1e-5)
 
 
 def test_sentencepiece_vocab_training_and_use(tmp_path):
