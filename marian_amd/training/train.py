@@ -71,6 +71,27 @@ def load_or_create_vocabs(options: Options) -> List[Vocab]:
         options.set("vocabs", paths)
     vocabs = []
     dim_vocabs = options.list("dim-vocabs") or [0, 0]
+    # multi-rank: only rank 0 CREATES missing vocab files (concurrent
+    # writers raced and crashed peers); the rest wait at the barrier
+    from ..parallel.communicator import dist_env
+    rank, _lr, world = dist_env()
+    missing = [p for p in paths if not os.path.exists(p)]
+    if world > 1 and missing:
+        import torch.distributed as dist
+        if rank == 0:
+            for i, p in enumerate(paths):
+                if os.path.exists(p):
+                    continue
+                max_size = int(dim_vocabs[i]) if i < len(dim_vocabs) else 0
+                if p.endswith(".spm") or p.endswith(".model"):
+                    _train_spm(p, train_sets[min(i, len(train_sets) - 1)],
+                               options, max_size)
+                else:
+                    log_info("Building vocabulary from %s", train_sets[i])
+                    v0 = Vocab.create_from_corpus([train_sets[i]],
+                                                  max_size=max_size)
+                    v0.save(p)
+        dist.barrier()
     for i, p in enumerate(paths):
         max_size = int(dim_vocabs[i]) if i < len(dim_vocabs) else 0
         if p.endswith(".fsv") and os.path.exists(p):
