@@ -245,7 +245,7 @@ def _train_worker(rank, world, port, workdir):
             "--mini-batch", "4", "--after-epochs", "2",
             "--valid-sets", f"{workdir}/c.src", f"{workdir}/c.trg",
             "--valid-freq", "2u", "--valid-metrics", "cross-entropy",
-            "--valid-mini-batch", "4",
+            "--valid-mini-batch", "4", "--exponential-smoothing", "0.999",
             "--cpu-threads", "1", "--seed", "3", "--overwrite", "--quiet"]
     o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
     Train(o).run()
