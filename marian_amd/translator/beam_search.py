@@ -82,51 +82,11 @@ class BeamSearch:
 
     # ------------------------------------------------------------------
     @torch.no_grad()
-    def search(self, batch, device, prefix=None) -> List[SentenceResult]:
-        """prefix: optional (ids [B, Tp], mask [B, Tp]) forcing the first
-        target tokens (--force-decode)."""
-        src = batch.source.to(device)
-        B = src.indices.shape[0]
-        K = self.K
-        src_lens = src.mask.sum(dim=1)
-
-        # shortlist: restrict output vocab for this batch
-        shortlist = None
-        if self.shortlist_generator is not None:
-            shortlist = self.shortlist_generator.generate(src.indices).to(device)
-
-        # encoder pass per scorer; expand to beam immediately
-        states = []
-        for si, sc in enumerate(self.scorers):
-            enc_out, src_mask = sc.start(src)
-            enc_out = enc_out.repeat_interleave(K, dim=0)
-            src_mask = src_mask.repeat_interleave(K, dim=0)
-            cache = {}
-            if self.collect_alignment and si == 0:
-                cache["collect_alignment"] = True
-            states.append({"enc": enc_out, "mask": src_mask, "cache": cache})
-
-        max_len = min(self.max_length,
-                      int(self.max_length_factor * src.indices.shape[1]) + 1)
-        for st in states:
-            st["cache"]["max_len"] = max_len
-
-        scores = torch.full((B, K), NEG_INF, device=device)
-        scores[:, 0] = 0.0  # only beam 0 live at step 0
-        alive = torch.ones(B * K, dtype=torch.bool, device=device)
-        prev_tokens = torch.zeros(B * K, dtype=torch.long, device=device)
-        bp_steps: List[torch.Tensor] = []   # backpointers per step [B,K]
-        tok_steps: List[torch.Tensor] = []  # chosen tokens per step [B,K]
-        # finished hypothesis bookkeeping (device tensors, harvested at end)
-        fin_scores = torch.full((B, self.n_best + K), NEG_INF, device=device)
-        fin_len = torch.zeros((B, self.n_best + K), dtype=torch.long,
-                              device=device)
-        fin_step = torch.full((B, self.n_best + K), -1, dtype=torch.long,
-                              device=device)
-        fin_beam = torch.zeros((B, self.n_best + K), dtype=torch.long,
-                               device=device)
-        fin_count = torch.zeros(B, dtype=torch.long, device=device)
-
+    # ------------------------------------------------------------------
+    def _try_graphed(self, device, states, B, K, max_len, shortlist,
+                     prefix):
+        """Set up (or reuse) the closed-loop hipGraph decode when the
+        batch qualifies; returns the GraphedDecodeStep or None."""
         # HIP-graph fast path: whole decode step captured once, replayed
         # per step (single transformer scorer, bf16/Dh=64, no alignment/
         # sampling/shortlist/prefix)
@@ -175,33 +135,123 @@ class BeamSearch:
                         BeamSearch._graph_warned = True
                     graphed = None
 
+        return graphed
+
+    # ------------------------------------------------------------------
+    def _graphed_loop(self, graphed, max_len, B, K):
+        """Closed-loop replay: the graph feeds itself (beam update is
+        IN-graph) — one launch per step, host syncs only the stop check
+        every 4 steps. Returns the harvest state."""
+        graphed.reset_state()
+        n_steps = 0
+        for step in range(max_len):
+            graphed.replay_once()
+            n_steps = step + 1
+            if (step & 3) == 3 or step == max_len - 1:
+                alive_t = graphed.scores > NEG_INF / 2
+                done = (graphed.fin_count >= self.n_best).all() \
+                    | ~alive_t.any()
+                if bool(done.item()):
+                    break
+        return ([graphed.bp_store[t] for t in range(n_steps)],
+                [graphed.tok_store[t] for t in range(n_steps)],
+                graphed.fin_scores, graphed.fin_len, graphed.fin_step,
+                graphed.fin_beam, graphed.fin_count,
+                graphed.scores.view(B, K))
+
+    # ------------------------------------------------------------------
+    def _harvest(self, B, K, bp_steps, tok_steps, fin_scores, fin_len,
+                 fin_step, fin_beam, fin_count, scores, align_steps):
+        # ---- harvest: reconstruct token sequences from backpointers -------
+        bp = torch.stack(bp_steps).cpu() if bp_steps else torch.zeros(0, B, K).long()
+        tk = torch.stack(tok_steps).cpu() if tok_steps else torch.zeros(0, B, K).long()
+        fin_scores_h = fin_scores.cpu()
+        fin_len_h = fin_len.cpu()
+        fin_step_h = fin_step.cpu()
+        fin_beam_h = fin_beam.cpu()
+        fin_count_h = fin_count.cpu()
+        scores_h = scores.cpu()
+
+        results: List[SentenceResult] = []
+        for b in range(B):
+            hyps: List[Hypothesis] = []
+            for c in range(int(fin_count_h[b])):
+                toks, al = self._backtrace(bp, tk, b, int(fin_step_h[b, c]),
+                                           int(fin_beam_h[b, c]),
+                                           align_steps, K)
+                score = float(fin_scores_h[b, c])
+                hyps.append(Hypothesis(
+                    tokens=toks, score=score, alignment=al,
+                    normalized=self._normalized(score, int(fin_len_h[b, c]))))
+            if not hyps:
+                # no EOS reached: take the best alive beam at the last step
+                kk = int(scores_h[b].argmax())
+                toks, al = self._backtrace(bp, tk, b, len(tok_steps) - 1, kk,
+                                           align_steps, K)
+                score = float(scores_h[b, kk])
+                hyps.append(Hypothesis(tokens=toks, score=score, alignment=al,
+                                       normalized=self._normalized(
+                                           score, max(len(toks), 1))))
+            hyps.sort(key=lambda h: h.normalized, reverse=True)
+            results.append(SentenceResult(nbest=hyps[: self.n_best]))
+        return results
+
+    def search(self, batch, device, prefix=None) -> List[SentenceResult]:
+        """prefix: optional (ids [B, Tp], mask [B, Tp]) forcing the first
+        target tokens (--force-decode)."""
+        src = batch.source.to(device)
+        B = src.indices.shape[0]
+        K = self.K
+        src_lens = src.mask.sum(dim=1)
+
+        # shortlist: restrict output vocab for this batch
+        shortlist = None
+        if self.shortlist_generator is not None:
+            shortlist = self.shortlist_generator.generate(src.indices).to(device)
+
+        # encoder pass per scorer; expand to beam immediately
+        states = []
+        for si, sc in enumerate(self.scorers):
+            enc_out, src_mask = sc.start(src)
+            enc_out = enc_out.repeat_interleave(K, dim=0)
+            src_mask = src_mask.repeat_interleave(K, dim=0)
+            cache = {}
+            if self.collect_alignment and si == 0:
+                cache["collect_alignment"] = True
+            states.append({"enc": enc_out, "mask": src_mask, "cache": cache})
+
+        max_len = min(self.max_length,
+                      int(self.max_length_factor * src.indices.shape[1]) + 1)
+        for st in states:
+            st["cache"]["max_len"] = max_len
+
+        scores = torch.full((B, K), NEG_INF, device=device)
+        scores[:, 0] = 0.0  # only beam 0 live at step 0
+        alive = torch.ones(B * K, dtype=torch.bool, device=device)
+        prev_tokens = torch.zeros(B * K, dtype=torch.long, device=device)
+        bp_steps: List[torch.Tensor] = []   # backpointers per step [B,K]
+        tok_steps: List[torch.Tensor] = []  # chosen tokens per step [B,K]
+        # finished hypothesis bookkeeping (device tensors, harvested at end)
+        fin_scores = torch.full((B, self.n_best + K), NEG_INF, device=device)
+        fin_len = torch.zeros((B, self.n_best + K), dtype=torch.long,
+                              device=device)
+        fin_step = torch.full((B, self.n_best + K), -1, dtype=torch.long,
+                              device=device)
+        fin_beam = torch.zeros((B, self.n_best + K), dtype=torch.long,
+                               device=device)
+        fin_count = torch.zeros(B, dtype=torch.long, device=device)
+
+        graphed = self._try_graphed(device, states, B, K, max_len,
+                                    shortlist, prefix)
+
         flat_sel = (torch.arange(B, device=device)[:, None] * K
                     + torch.arange(K, device=device)[None, :]).reshape(-1)
         align_steps: List[torch.Tensor] = []
         V = None
         if graphed is not None:
-            # closed-loop replay: the graph feeds itself (beam update is
-            # IN-graph) — one launch per step, host syncs only the stop
-            # check every 4 steps
-            graphed.reset_state()
-            n_steps = 0
-            for step in range(max_len):
-                graphed.replay_once()
-                n_steps = step + 1
-                if (step & 3) == 3 or step == max_len - 1:
-                    alive_t = graphed.scores > NEG_INF / 2
-                    done = (graphed.fin_count >= self.n_best).all() \
-                        | ~alive_t.any()
-                    if bool(done.item()):
-                        break
-            bp_steps = [graphed.bp_store[t] for t in range(n_steps)]
-            tok_steps = [graphed.tok_store[t] for t in range(n_steps)]
-            fin_scores = graphed.fin_scores
-            fin_len = graphed.fin_len
-            fin_step = graphed.fin_step
-            fin_beam = graphed.fin_beam
-            fin_count = graphed.fin_count
-            scores = graphed.scores.view(B, K)
+            (bp_steps, tok_steps, fin_scores, fin_len, fin_step,
+             fin_beam, fin_count, scores) = self._graphed_loop(
+                graphed, max_len, B, K)
         for step in range(max_len) if graphed is None else []:
             # ensemble step: weighted sum of log-probs
             logprobs = None
@@ -347,39 +397,9 @@ class BeamSearch:
                     cache[key] = t.index_select(0, flat_sel)
             prev_tokens = real_words.reshape(B * K)
 
-        # ---- harvest: reconstruct token sequences from backpointers -------
-        bp = torch.stack(bp_steps).cpu() if bp_steps else torch.zeros(0, B, K).long()
-        tk = torch.stack(tok_steps).cpu() if tok_steps else torch.zeros(0, B, K).long()
-        fin_scores_h = fin_scores.cpu()
-        fin_len_h = fin_len.cpu()
-        fin_step_h = fin_step.cpu()
-        fin_beam_h = fin_beam.cpu()
-        fin_count_h = fin_count.cpu()
-        scores_h = scores.cpu()
-
-        results: List[SentenceResult] = []
-        for b in range(B):
-            hyps: List[Hypothesis] = []
-            for c in range(int(fin_count_h[b])):
-                toks, al = self._backtrace(bp, tk, b, int(fin_step_h[b, c]),
-                                           int(fin_beam_h[b, c]),
-                                           align_steps, K)
-                score = float(fin_scores_h[b, c])
-                hyps.append(Hypothesis(
-                    tokens=toks, score=score, alignment=al,
-                    normalized=self._normalized(score, int(fin_len_h[b, c]))))
-            if not hyps:
-                # no EOS reached: take the best alive beam at the last step
-                kk = int(scores_h[b].argmax())
-                toks, al = self._backtrace(bp, tk, b, len(tok_steps) - 1, kk,
-                                           align_steps, K)
-                score = float(scores_h[b, kk])
-                hyps.append(Hypothesis(tokens=toks, score=score, alignment=al,
-                                       normalized=self._normalized(
-                                           score, max(len(toks), 1))))
-            hyps.sort(key=lambda h: h.normalized, reverse=True)
-            results.append(SentenceResult(nbest=hyps[: self.n_best]))
-        return results
+        return self._harvest(B, K, bp_steps, tok_steps, fin_scores,
+                             fin_len, fin_step, fin_beam, fin_count,
+                             scores, align_steps)
 
     # ------------------------------------------------------------------
     @staticmethod
