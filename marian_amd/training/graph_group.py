@@ -361,8 +361,61 @@ class SyncGraphGroup:
         opt_path = prefix + ".optimizer.npz"
         if load_optimizer and os.path.exists(opt_path):
             oitems, _ = io_npz.load_items(opt_path)
-            mine = {k[len(f"rank{self.rank}_"):]: v for k, v in oitems.items()
-                    if k.startswith(f"rank{self.rank}_")}
+            saved_ranks = sorted({int(k.split("_", 1)[0][4:])
+                                  for k in oitems if k.startswith("rank")})
+            w_saved = len(saved_ranks)
+            if w_saved and w_saved != self.world:
+                from ..common.logging_utils import log_info
+                log_info("Resharding optimizer state: checkpoint has %d "
+                         "shards, running with %d ranks", w_saved, self.world)
+                mine = self._reshard_optimizer_items(oitems, w_saved)
+            else:
+                mine = {k[len(f"rank{self.rank}_"):]: v
+                        for k, v in oitems.items()
+                        if k.startswith(f"rank{self.rank}_")}
             if mine:
                 self.optimizer.load_state_items(mine)
         return True
+
+    # ------------------------------------------------------------------
+    def _reshard_optimizer_items(self, oitems, w_saved: int) -> dict:
+        """Resume at a DIFFERENT world size: rebuild each flat state
+        vector's per-bucket global view from the saved shards and re-slice
+        for this run's sharding (bucket padding depends on world size, so
+        shard boundaries move). (ref: the reference's scatter/gatherState
+        in training/communicator.h serve the same purpose.)"""
+        import numpy as np
+        used = [b.used for b in self.store.buckets]
+
+        def layout(world):
+            pad = [max(u, 256 * world) for u in used]
+            pad = [((p + 256 * world - 1) // (256 * world)) * (256 * world)
+                   for p in pad]
+            shard = [p // world for p in pad]
+            return pad, shard
+
+        pad1, shard1 = layout(w_saved)
+        pad2, shard2 = layout(self.world)
+        vec_keys = [k.split("_", 1)[1] for k in oitems
+                    if k.startswith("rank0_")
+                    and np.asarray(oitems[k]).size > 1]
+        out = {}
+        for key in vec_keys:
+            parts = [np.asarray(oitems[f"rank{r}_{key}"]).reshape(-1)
+                     for r in range(w_saved)]
+            mine_segments = []
+            off1 = [0] * w_saved
+            for bi in range(len(used)):
+                # global padded1 vector for this bucket
+                g = np.concatenate(
+                    [parts[r][sum(shard1[:bi]):sum(shard1[:bi]) + shard1[bi]]
+                     for r in range(w_saved)])
+                g = g[:used[bi]]
+                g2 = np.zeros(pad2[bi], dtype=g.dtype)
+                g2[:used[bi]] = g
+                mine_segments.append(
+                    g2[self.rank * shard2[bi]:(self.rank + 1) * shard2[bi]])
+            out[key] = np.concatenate(mine_segments)
+        if "rank0_optimizer_step" in oitems:
+            out["optimizer_step"] = oitems["rank0_optimizer_step"]
+        return out

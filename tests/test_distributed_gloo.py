@@ -270,3 +270,81 @@ def test_full_train_loop_world2_odd_batches(tmp_path):
         p.join(240)
         assert p.exitcode == 0
     assert (tmp_path / "m2.npz").exists()
+
+
+def _resume_worker(rank, world, port, workdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import numpy as np
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    argv = ["--train-sets", f"{workdir}/c.src", f"{workdir}/c.trg",
+            "--vocabs", f"{workdir}/v1.yml", f"{workdir}/v2.yml",
+            "--model", f"{workdir}/m.npz",
+            "--dim-emb", "16", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "32",
+            "--mini-batch", "4", "--after-batches", "4",
+            "--cpu-threads", "1", "--seed", "3", "--overwrite", "--quiet"]
+    o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    t = Train(o)   # resumes from m.npz written at world=1
+    np.save(f"{workdir}/m_rank{rank}.npy",
+            t.graph_group.optimizer.core.m.numpy())
+    t.run()        # continues training without shape errors
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_resume_at_different_world_size(tmp_path):
+    """Optimizer shards written at world=1 reshard correctly when
+    resuming at world=2 (bucket padding moves with world size)."""
+    import numpy as np
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    write_synthetic_corpus(str(tmp_path / "c.src"), str(tmp_path / "c.trg"),
+                           24, 25, 3, 7, seed=12)
+    argv = ["--train-sets", str(tmp_path / "c.src"), str(tmp_path / "c.trg"),
+            "--vocabs", str(tmp_path / "v1.yml"), str(tmp_path / "v2.yml"),
+            "--model", str(tmp_path / "m.npz"),
+            "--dim-emb", "16", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "32",
+            "--mini-batch", "4", "--after-batches", "3",
+            "--cpu-threads", "1", "--seed", "3", "--overwrite", "--quiet"]
+    o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    t1 = Train(o)
+    t1.run()
+    m_full = t1.graph_group.optimizer.core.m.numpy().copy()
+    buckets = t1.graph_group.store.buckets
+    used = [b.used for b in buckets]
+
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_resume_worker,
+                         args=(r, 2, 29553, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    m0 = np.load(tmp_path / "m_rank0.npy")
+    m1 = np.load(tmp_path / "m_rank1.npy")
+
+    def layout(world):
+        pad = [((max(u, 256 * world) + 256 * world - 1)
+                // (256 * world)) * (256 * world) for u in used]
+        return pad, [p // world for p in pad]
+
+    pad1, sh1 = layout(1)
+    pad2, sh2 = layout(2)
+    o1 = o2 = 0
+    for bi, u in enumerate(used):
+        g1 = m_full[o1:o1 + sh1[bi]][:u]           # world-1 bucket state
+        rank0 = m0[o2:o2 + sh2[bi]]
+        rank1 = m1[o2:o2 + sh2[bi]]
+        g2 = np.concatenate([rank0, rank1])[:u]
+        np.testing.assert_allclose(g2, g1, atol=1e-7)
+        o1 += sh1[bi]
+        o2 += sh2[bi]
