@@ -401,12 +401,26 @@ class TransformerModel(nn.Module):
     def _name_map(self) -> Dict[str, torch.Tensor]:
         """marian name -> tensor view (shared storage)."""
         m: Dict[str, torch.Tensor] = {}
+
+        def embed(prefix: str, e):
+            fe = getattr(e, "factored", None)
+            if fe is not None:
+                # factored vocab: per-group factor tables [M names]
+                for g, t in enumerate(fe.tables):
+                    m[f"{prefix}Wemb_factor{g}"] = t
+            else:
+                m[f"{prefix}Wemb"] = e.E
+
         if self.tied_all:
-            m["Wemb"] = self.src_embed.E
+            embed("", self.src_embed)
         else:
-            m["encoder_Wemb"] = self.src_embed.E
+            embed("encoder_", self.src_embed)
             if not self.tied_src:
-                m["decoder_Wemb"] = self.trg_embed.E
+                embed("decoder_", self.trg_embed)
+        fo = getattr(self, "factored_out", None)
+        if fo is not None:
+            for g, b in enumerate(fo.biases):
+                m[f"decoder_ff_logit_factor{g}_b"] = b
         if not self.tied_out:
             m["decoder_ff_logit_out_W"] = self.out_W
         if self.out_b is not None:

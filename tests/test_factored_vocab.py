@@ -125,3 +125,33 @@ def test_factored_cli_train_and_decode(tmp_path):
         capture_output=True, text=True, env=env, timeout=300)
     assert r.returncode == 0, r.stderr[-1500:]
     assert model.exists()
+
+
+def test_factored_checkpoint_roundtrip(tmp_path):
+    """Factored models save/load their factor tables + output biases."""
+    v = FactoredVocab.load(_write(tmp_path))
+    o = ConfigParser(MODE_TRAINING).parse_options(
+        ["--train-sets", "a", "b", "--vocabs", "a.fsv", "b.fsv",
+         "--dim-emb", "32", "--enc-depth", "1", "--dec-depth", "1",
+         "--transformer-heads", "4", "--transformer-dim-ffn", "64",
+         "--tied-embeddings-all"], validate=False)
+    o.set("dim-vocabs", [len(v), len(v)])
+    m1 = create_model(o)
+    m1.enable_factored_vocab(v, "both")
+    names = m1._name_map()
+    assert any("Wemb_factor" in k for k in names)
+    assert any("logit_factor" in k for k in names)
+    assert "Wemb" not in names          # empty placeholder not exported
+    items = {k: t.detach().float().numpy() for k, t in names.items()}
+    m2 = create_model(o)
+    m2.enable_factored_vocab(v, "both")
+    with torch.no_grad():
+        for t in m2.trg_embed.factored.tables:
+            t.add_(1.0)
+    import numpy as np
+    nm2 = m2._name_map()
+    for k, t in nm2.items():
+        t.data.copy_(torch.from_numpy(items[k]).reshape(t.shape))
+    for a, b in zip(m1.trg_embed.factored.tables,
+                    m2.trg_embed.factored.tables):
+        assert torch.allclose(a, b)
