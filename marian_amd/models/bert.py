@@ -94,6 +94,16 @@ class BertModel(TransformerModel):
             x = self.post_top_enc(x)
         return x
 
+    def _name_map(self):
+        m = super()._name_map()
+        m["masked-lm_ff_logit_l1_W"] = self.mlm_ff.W
+        m["masked-lm_ff_logit_l1_b"] = self.mlm_ff.b
+        m["masked-lm_ff_logit_l1_ln_scale"] = self.mlm_ln.scale
+        m["masked-lm_ff_logit_l1_ln_bias"] = self.mlm_ln.bias
+        if self.type_vocab:
+            m["Wtype"] = self.Wtype
+        return m
+
     def forward(self, batch):
         from ..data.batch import SubBatch
         src = batch.source.indices
@@ -121,6 +131,14 @@ class BertClassifier(TransformerModel):
         self.num_classes = self.trg_vocab_size
         self.cls_ff = Dense(d, d)
         self.cls_out = Dense(d, self.num_classes)
+
+    def _name_map(self):
+        m = super()._name_map()
+        m["classifier_ff_logit_l1_W"] = self.cls_ff.W
+        m["classifier_ff_logit_l1_b"] = self.cls_ff.b
+        m["classifier_ff_logit_l2_W"] = self.cls_out.W
+        m["classifier_ff_logit_l2_b"] = self.cls_out.b
+        return m
 
     def forward(self, batch):
         src = batch.source.indices

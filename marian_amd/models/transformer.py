@@ -421,6 +421,11 @@ class TransformerModel(nn.Module):
         if fo is not None:
             for g, b in enumerate(fo.biases):
                 m[f"decoder_ff_logit_factor{g}_b"] = b
+        ulr = getattr(self.src_embed, "ulr", None)
+        if ulr is not None:   # --ulr universal embeddings (+transform)
+            m["ulr_Wemb"] = ulr.U
+            if ulr.A is not None:
+                m["ulr_transform"] = ulr.A
         if not self.tied_out:
             m["decoder_ff_logit_out_W"] = self.out_W
         if self.out_b is not None:
