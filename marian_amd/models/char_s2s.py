@@ -74,3 +74,16 @@ class CharS2SModel(S2SModel):
         pooled, pmask = self.char_encoder(emb, src_mask)
         self._pooled_mask = pmask.to(src_mask.dtype)
         return self.rnn_encode(pooled, self._pooled_mask)
+
+    def _name_map(self):
+        m = super()._name_map()
+        ce = self.char_encoder
+        for i, c in enumerate(ce.convs, start=1):   # conv stack [M names]
+            m[f"char_conv{i}_W"] = c.weight
+            m[f"char_conv{i}_b"] = c.bias
+        for i, hw in enumerate(ce.highway, start=1):
+            m[f"char_highway{i}_W"] = hw.weight
+            m[f"char_highway{i}_b"] = hw.bias
+        m["char_proj_W"] = ce.proj.weight
+        m["char_proj_b"] = ce.proj.bias
+        return m

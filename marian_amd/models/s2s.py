@@ -191,7 +191,8 @@ class S2SModel(nn.Module):
         self.ff_logit_l1_W1 = Dense(ctx_dim, e)
         self.ff_logit_l1_W2 = Dense(e, e)
         if not tied_all:
-            self.ff_logit_l2 = Dense(e, self.trg_vocab_size)
+            # bias lives in out_b (marian ff_logit_b); no Dense bias
+            self.ff_logit_l2 = Dense(e, self.trg_vocab_size, bias=False)
         self.out_b = nn.Parameter(torch.zeros(self.trg_vocab_size))
         self.compute_dtype = torch.float32
 

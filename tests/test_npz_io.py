@@ -31,3 +31,30 @@ def test_npz_readable_by_plain_numpy(tmp_path):
         assert raw.dtype == np.int8
         text = raw.tobytes().rstrip(b"\0").decode()
         assert "a: 1" in text
+
+
+def test_every_parameter_is_checkpointed():
+    """Invariant: every trainable parameter of every model family shares
+    storage with some entry of the marian name map (nothing silently
+    dropped from checkpoints)."""
+    import torch
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.models.model_factory import create_model
+
+    def missing(mtype):
+        argv = ["--train-sets", "a", "b", "--vocabs", "a.yml", "b.yml",
+                "--type", mtype, "--dim-emb", "16", "--dim-rnn", "24",
+                "--enc-depth", "2", "--dec-depth", "2",
+                "--transformer-heads", "2", "--transformer-dim-ffn", "32"]
+        o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+        o.set("dim-vocabs", [30, 30])
+        m = create_model(o)
+        mapped = {t.untyped_storage().data_ptr()
+                  for t in m._name_map().values()}
+        return [n for n, p in m.named_parameters()
+                if p.requires_grad and p.numel()
+                and p.untyped_storage().data_ptr() not in mapped]
+
+    for mt in ("transformer", "s2s", "char-s2s", "bert", "bert-classifier",
+               "lm-transformer", "amun"):
+        assert not missing(mt), (mt, missing(mt))
