@@ -425,6 +425,9 @@ class ShardedOptimizer:
                         self.master[o:o + s].to(self.store.dtype))
         if "optimizer_step" in items:
             self.t = int(np.asarray(items["optimizer_step"]).reshape(-1)[0])
+        if "exp_smoothing" in items and self.smooth is not None:
+            self.smooth.copy_(torch.from_numpy(np.ascontiguousarray(
+                items["exp_smoothing"])).to(self.smooth.device))
         self.core.load_state_items(
             {k: torch.from_numpy(np.ascontiguousarray(v))
              for k, v in items.items()

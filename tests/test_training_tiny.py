@@ -490,3 +490,18 @@ def test_keep_best_checkpoint(workdir):
     cands = [f for f in os.listdir(os.path.dirname(model))
              if "best" in f]
     assert cands, "no best-checkpoint written"
+
+
+def test_smoothing_state_resumes(workdir):
+    """exp_smoothing survives checkpoint/resume (Polyak history kept)."""
+    tmp_path, src, trg = workdir
+    opts = _opts(tmp_path, src, src, extra=["--exponential-smoothing",
+                                            "0.999", "--after-batches", "4"])
+    tr = Train(opts)
+    tr.run()
+    want = tr.graph_group.optimizer.smooth.clone()
+    opts2 = _opts(tmp_path, src, src, extra=["--exponential-smoothing",
+                                             "0.999", "--after-batches", "6"])
+    tr2 = Train(opts2)
+    got = tr2.graph_group.optimizer.smooth
+    assert torch.allclose(got, want, atol=1e-6)
