@@ -182,6 +182,12 @@ class Train:
                     self.state = TrainingState.load(prog)
                     if options.get("lr-warmup-at-reload", False):
                         self.state.warmup_start = self.state.batches
+                    # restore fp16 loss scale + grad-norm window
+                    # (ref: TrainingState costScaleFactor / gradientNormAvg)
+                    if self.state.cost_scale > 0:
+                        self.graph_group.scaler.scale = self.state.cost_scale
+                    self.graph_group._gnorm_avg = self.state.gnorm_avg
+                    self.graph_group._gnorm_count = self.state.gnorm_count
                 if options.get("valid-reset-all", False):
                     self.state.validators = {}
                     self.state.stalled = 0
@@ -368,6 +374,11 @@ class Train:
             base, ext = os.path.splitext(path)
             path = f"{base}{suffix}{ext}"
         cfg = self.model.model_config()
+        # snapshot scaler/grad-norm state into the progress file
+        self.state.cost_scale = float(self.graph_group.scaler.scale) \
+            if self.graph_group.scaler.enabled else 0.0
+        self.state.gnorm_avg = float(self.graph_group._gnorm_avg)
+        self.state.gnorm_count = int(self.graph_group._gnorm_count)
         self.graph_group.save_checkpoint(path, cfg, self.state)
         if self.rank == 0:
             # external config (model.npz.yml), ref: io row in SURVEY §2

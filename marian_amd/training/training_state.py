@@ -50,6 +50,9 @@ class TrainingState:
     stalled: int = 0
     max_stalled: int = 0
     warmup_start: int = 0        # --lr-warmup-at-reload base
+    cost_scale: float = 0.0      # fp16 dynamic loss scale (0 = unset)
+    gnorm_avg: float = 0.0       # --dynamic-gradient-scaling window avg
+    gnorm_count: int = 0
     validators: Dict[str, dict] = field(default_factory=dict)
     eta: float = 0.0             # current learning rate
     factor: float = 1.0          # lr decay factor
@@ -81,6 +84,10 @@ class TrainingState:
             "disp-sentences": self.sentences_disp,
             "seed-batch": self.seed_batch,
             "seed-corpus": self.seed_corpus,
+            "warmup-start": self.warmup_start,
+            "cost-scale-factor": self.cost_scale,
+            "gradient-norm-avg": self.gnorm_avg,
+            "gradient-norm-count": self.gnorm_count,
         }
 
     def save(self, path: str) -> None:
@@ -105,6 +112,10 @@ class TrainingState:
         st.validators = d.get("validators", {}) or {}
         st.eta = d.get("eta", 0.0)
         st.factor = d.get("factor", 1.0)
+        st.warmup_start = d.get("warmup-start", 0)
+        st.cost_scale = d.get("cost-scale-factor", 0.0)
+        st.gnorm_avg = d.get("gradient-norm-avg", 0.0)
+        st.gnorm_count = d.get("gradient-norm-count", 0)
         st.warmed_up = d.get("warmed-up", False)
         st.cost_sum = d.get("cost-sum", 0.0)
         st.cost_count = d.get("cost-count", 0)
