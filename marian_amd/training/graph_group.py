@@ -317,18 +317,30 @@ class SyncGraphGroup:
             opt.swap_out_smoothed(self.comm if self.world > 1 else None)
 
     def save_checkpoint(self, prefix: str, model_config: dict,
-                        training_state=None) -> None:
+                        training_state=None, extra_model_paths=()) -> None:
         from ..common import io_npz
         self.swap_smoothed(True)   # marian saves the smoothed model
         if self.rank == 0:
             items = self.model.marian_items()
             io_npz.save_items(prefix, items, config=model_config)
+            # iteration copies (model.iterN.npz) get the same smoothed
+            # params as the main file (upstream writes them in one pass)
+            for p in extra_model_paths:
+                io_npz.save_items(p, items, config=model_config)
         self.swap_smoothed(False)
         # every rank contributes its optimizer shard; single-process path
         # gathers locally. (multi-proc: each rank writes its shard file,
         # rank 0 merges on load; simple + correct for single node)
         opt_items = {f"rank{self.rank}_{k}": v
                      for k, v in self.optimizer.state_items().items()}
+        if self.rank == 0:
+            # persist the bucket layout so a resume under a different
+            # world size OR bucket grouping (bucketing is disabled at
+            # world=1) can reshard against the layout that WROTE the
+            # file, not the one reading it.
+            import numpy as np
+            opt_items["bucket_used"] = np.array(
+                [b.used for b in self.store.buckets], dtype=np.int64)
         if self.world > 1:
             import torch.distributed as dist
             gathered: List[Optional[dict]] = [None] * self.world
@@ -361,14 +373,22 @@ class SyncGraphGroup:
         opt_path = prefix + ".optimizer.npz"
         if load_optimizer and os.path.exists(opt_path):
             oitems, _ = io_npz.load_items(opt_path)
+            import numpy as np
             saved_ranks = sorted({int(k.split("_", 1)[0][4:])
                                   for k in oitems if k.startswith("rank")})
             w_saved = len(saved_ranks)
-            if w_saved and w_saved != self.world:
+            used_now = [b.used for b in self.store.buckets]
+            used_saved = ([int(x) for x in
+                           np.asarray(oitems["bucket_used"]).reshape(-1)]
+                          if "bucket_used" in oitems else None)
+            layout_differs = (used_saved is not None
+                              and used_saved != used_now)
+            if w_saved and (w_saved != self.world or layout_differs):
                 from ..common.logging_utils import log_info
                 log_info("Resharding optimizer state: checkpoint has %d "
                          "shards, running with %d ranks", w_saved, self.world)
-                mine = self._reshard_optimizer_items(oitems, w_saved)
+                mine = self._reshard_optimizer_items(oitems, w_saved,
+                                                     used_saved)
             else:
                 mine = {k[len(f"rank{self.rank}_"):]: v
                         for k, v in oitems.items()
@@ -378,41 +398,57 @@ class SyncGraphGroup:
         return True
 
     # ------------------------------------------------------------------
-    def _reshard_optimizer_items(self, oitems, w_saved: int) -> dict:
-        """Resume at a DIFFERENT world size: rebuild each flat state
-        vector's per-bucket global view from the saved shards and re-slice
-        for this run's sharding (bucket padding depends on world size, so
-        shard boundaries move). (ref: the reference's scatter/gatherState
+    def _reshard_optimizer_items(self, oitems, w_saved: int,
+                                 used_saved=None) -> dict:
+        """Resume at a DIFFERENT world size or bucket grouping: rebuild
+        each flat state vector's global param-order view from the saved
+        shards (using the layout that wrote the file: per-bucket `used`
+        sizes + saved world, since padding depends on both) and re-slice
+        for this run's layout. (ref: the reference's scatter/gatherState
         in training/communicator.h serve the same purpose.)"""
         import numpy as np
-        used = [b.used for b in self.store.buckets]
+        used_now = [b.used for b in self.store.buckets]
+        if used_saved is None:
+            # pre-layout checkpoint: assume the grouping matched ours
+            used_saved = used_now
+        if sum(used_saved) != sum(used_now):
+            raise ValueError(
+                "optimizer checkpoint parameter count %d != model %d"
+                % (sum(used_saved), sum(used_now)))
 
-        def layout(world):
+        def layout(used, world):
             pad = [max(u, 256 * world) for u in used]
             pad = [((p + 256 * world - 1) // (256 * world)) * (256 * world)
                    for p in pad]
             shard = [p // world for p in pad]
             return pad, shard
 
-        pad1, shard1 = layout(w_saved)
-        pad2, shard2 = layout(self.world)
+        pad1, shard1 = layout(used_saved, w_saved)
+        pad2, shard2 = layout(used_now, self.world)
         vec_keys = [k.split("_", 1)[1] for k in oitems
                     if k.startswith("rank0_")
-                    and np.asarray(oitems[k]).size > 1]
+                    and np.asarray(oitems[k]).size > 1
+                    and k.split("_", 1)[1] != "bucket_used"]
         out = {}
         for key in vec_keys:
             parts = [np.asarray(oitems[f"rank{r}_{key}"]).reshape(-1)
                      for r in range(w_saved)]
-            mine_segments = []
-            off1 = [0] * w_saved
-            for bi in range(len(used)):
-                # global padded1 vector for this bucket
+            # 1) reconstruct the global param-order vector per SAVED layout
+            glob = []
+            for bi in range(len(used_saved)):
                 g = np.concatenate(
                     [parts[r][sum(shard1[:bi]):sum(shard1[:bi]) + shard1[bi]]
                      for r in range(w_saved)])
-                g = g[:used[bi]]
-                g2 = np.zeros(pad2[bi], dtype=g.dtype)
-                g2[:used[bi]] = g
+                glob.append(g[:used_saved[bi]])
+            glob = np.concatenate(glob)
+            # 2) re-slice per CURRENT layout
+            mine_segments = []
+            off = 0
+            for bi in range(len(used_now)):
+                seg = glob[off:off + used_now[bi]]
+                off += used_now[bi]
+                g2 = np.zeros(pad2[bi], dtype=seg.dtype)
+                g2[:used_now[bi]] = seg
                 mine_segments.append(
                     g2[self.rank * shard2[bi]:(self.rank + 1) * shard2[bi]])
             out[key] = np.concatenate(mine_segments)

@@ -307,8 +307,11 @@ class ShardedOptimizer:
         if self.quantize_bits:
             self._requantize(self.master)
         if self.smooth is not None:
-            # marian caps the effective decay early in training
-            decay = min(self.smoothing, (self.t + 1.0) / (self.t + 10.0))
+            # marian's flag value is the decay-BY epsilon (keep factor
+            # 1-eps, typically 1e-4), capped early in training
+            # (ref: optimizers/exponential_smoothing updateAvgParams)
+            decay = min(1.0 - self.smoothing,
+                        (self.t + 1.0) / (self.t + 10.0))
             self.smooth.mul_(decay).add_(self.master, alpha=1.0 - decay)
         # write updated shard back into each bucket's param slice
         with torch.no_grad():

@@ -184,7 +184,11 @@ class Train:
                         self.state.warmup_start = self.state.batches
                     # restore fp16 loss scale + grad-norm window
                     # (ref: TrainingState costScaleFactor / gradientNormAvg)
-                    if self.state.cost_scale > 0:
+                    # only when cost-scaling is active THIS run: resuming
+                    # an fp16 checkpoint with fp32 precision must not
+                    # divide gradients by a stale scale
+                    if self.graph_group.scaler.enabled \
+                            and self.state.cost_scale > 0:
                         self.graph_group.scaler.scale = self.state.cost_scale
                     self.graph_group._gnorm_avg = self.state.gnorm_avg
                     self.graph_group._gnorm_count = self.state.gnorm_count
@@ -379,17 +383,17 @@ class Train:
             if self.graph_group.scaler.enabled else 0.0
         self.state.gnorm_avg = float(self.graph_group._gnorm_avg)
         self.state.gnorm_count = int(self.graph_group._gnorm_count)
-        self.graph_group.save_checkpoint(path, cfg, self.state)
+        extra = []
+        if not self.opt.get("overwrite", False):
+            itr = f".iter{self.state.batches}"
+            base, ext = os.path.splitext(self.model_path)
+            extra.append(f"{base}{itr}{ext}")
+        self.graph_group.save_checkpoint(path, cfg, self.state,
+                                         extra_model_paths=extra)
         if self.rank == 0:
             # external config (model.npz.yml), ref: io row in SURVEY §2
             with open(path + ".yml", "w") as f:
                 f.write(self.opt.to_yaml())
-            if not self.opt.get("overwrite", False):
-                itr = f".iter{self.state.batches}"
-                base, ext = os.path.splitext(self.model_path)
-                iter_path = f"{base}{itr}{ext}"
-                from ..common import io_npz
-                io_npz.save_items(iter_path, self.model.marian_items(), cfg)
             log_info("Saved model to %s", path)
 
     # ------------------------------------------------------------------

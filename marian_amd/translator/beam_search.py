@@ -81,8 +81,6 @@ class BeamSearch:
         return s
 
     # ------------------------------------------------------------------
-    @torch.no_grad()
-    # ------------------------------------------------------------------
     def _try_graphed(self, device, states, B, K, max_len, shortlist,
                      prefix):
         """Set up (or reuse) the closed-loop hipGraph decode when the
@@ -196,6 +194,7 @@ class BeamSearch:
             results.append(SentenceResult(nbest=hyps[: self.n_best]))
         return results
 
+    @torch.no_grad()
     def search(self, batch, device, prefix=None) -> List[SentenceResult]:
         """prefix: optional (ids [B, Tp], mask [B, Tp]) forcing the first
         target tokens (--force-decode)."""

@@ -245,7 +245,7 @@ def _train_worker(rank, world, port, workdir):
             "--mini-batch", "4", "--after-epochs", "2",
             "--valid-sets", f"{workdir}/c.src", f"{workdir}/c.trg",
             "--valid-freq", "2u", "--valid-metrics", "cross-entropy",
-            "--valid-mini-batch", "4", "--exponential-smoothing", "0.999",
+            "--valid-mini-batch", "4", "--exponential-smoothing", "0.0001",
             "--cpu-threads", "1", "--seed", "3", "--overwrite", "--quiet"]
     o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
     Train(o).run()
@@ -348,3 +348,59 @@ def test_resume_at_different_world_size(tmp_path):
         np.testing.assert_allclose(g2, g1, atol=1e-7)
         o1 += sh1[bi]
         o2 += sh2[bi]
+
+
+def test_reshard_across_bucket_groupings():
+    """world=1 (bucketing disabled → ONE bucket) ↔ world=2 (multi-bucket)
+    resharding: the saved `bucket_used` layout must drive the rebuild,
+    not the current run's grouping (advisor finding r1)."""
+    import numpy as np
+    import types
+    from marian_amd.training.optimizers import ParameterStore
+    from marian_amd.training.graph_group import SyncGraphGroup
+
+    torch.manual_seed(7)
+    def make_model():
+        torch.manual_seed(7)
+        return torch.nn.Sequential(
+            torch.nn.Linear(700, 700), torch.nn.Linear(700, 500),
+            torch.nn.Linear(500, 300))
+
+    # saved side: world=1, single bucket (bucket_mb=0 as SyncGraphGroup does)
+    s1 = ParameterStore(make_model(), world_size=1, bucket_mb=0)
+    assert len(s1.buckets) == 1
+    total = sum(b.used for b in s1.buckets)
+    # the canonical param-order vector 0..total-1, padded per world-1 layout
+    master1 = np.full(s1.total_padded, -1.0, dtype=np.float32)
+    master1[:total] = np.arange(total, dtype=np.float32)
+    oitems = {"rank0_master_parameters": master1,
+              "rank0_optimizer_step": np.array([5], dtype=np.int64),
+              "bucket_used": np.array([b.used for b in s1.buckets],
+                                      dtype=np.int64)}
+
+    # resuming side: world=2, 1MB buckets → several buckets
+    s2 = ParameterStore(make_model(), world_size=2, bucket_mb=1)
+    assert len(s2.buckets) > 1
+    shards = []
+    for rank in (0, 1):
+        fake = types.SimpleNamespace(store=s2, rank=rank, world=2)
+        out = SyncGraphGroup._reshard_optimizer_items(
+            fake, oitems, 1, [int(x) for x in oitems["bucket_used"]])
+        shards.append(out["master_parameters"])
+        assert int(out["optimizer_step"][0]) == 5
+    # stitch world-2 shards back to the global vector and compare
+    def layout(used, world):
+        pad = [((max(u, 256 * world) + 256 * world - 1)
+                // (256 * world)) * (256 * world) for u in used]
+        return [p // world for p in pad]
+    used2 = [b.used for b in s2.buckets]
+    sh2 = layout(used2, 2)
+    got = []
+    off = 0
+    for bi, u in enumerate(used2):
+        g = np.concatenate([shards[0][off:off + sh2[bi]],
+                            shards[1][off:off + sh2[bi]]])[:u]
+        got.append(g)
+        off += sh2[bi]
+    got = np.concatenate(got)
+    np.testing.assert_array_equal(got, np.arange(total, dtype=np.float32))

@@ -156,7 +156,7 @@ def test_right_left(workdir):
 
 def test_exponential_smoothing(workdir):
     tmp_path, src, trg = workdir
-    opts = _opts(tmp_path, src, src, extra=["--exponential-smoothing", "0.999",
+    opts = _opts(tmp_path, src, src, extra=["--exponential-smoothing", "0.0001",
                                             "--after-batches", "5"])
     tr = Train(opts)
     opt = tr.graph_group.optimizer
@@ -496,12 +496,12 @@ def test_smoothing_state_resumes(workdir):
     """exp_smoothing survives checkpoint/resume (Polyak history kept)."""
     tmp_path, src, trg = workdir
     opts = _opts(tmp_path, src, src, extra=["--exponential-smoothing",
-                                            "0.999", "--after-batches", "4"])
+                                            "0.0001", "--after-batches", "4"])
     tr = Train(opts)
     tr.run()
     want = tr.graph_group.optimizer.smooth.clone()
     opts2 = _opts(tmp_path, src, src, extra=["--exponential-smoothing",
-                                             "0.999", "--after-batches", "6"])
+                                             "0.0001", "--after-batches", "6"])
     tr2 = Train(opts2)
     got = tr2.graph_group.optimizer.smooth
     assert torch.allclose(got, want, atol=1e-6)
