@@ -14,11 +14,17 @@ import torch
 
 
 class SubBatch:
-    def __init__(self, indices: torch.Tensor, mask: torch.Tensor):
+    def __init__(self, indices: torch.Tensor, mask: torch.Tensor,
+                 n_tokens: Optional[int] = None):
         # indices: [B, T] long; mask: [B, T] float (1 for real token incl. EOS)
         assert indices.shape == mask.shape
         self.indices = indices
         self.mask = mask
+        # token count, computed host-side while the mask is still on CPU so
+        # the training step never has to sync the device for it
+        if n_tokens is None and not mask.is_cuda:
+            n_tokens = int(mask.sum().item())
+        self.n_tokens = n_tokens
 
     @classmethod
     def from_sentences(cls, sents: Sequence[Sequence[int]], pad_id: int = 0) -> "SubBatch":
@@ -45,11 +51,14 @@ class SubBatch:
         return self.indices.shape[1]
 
     def batch_words(self) -> int:
+        if self.n_tokens is not None:
+            return self.n_tokens
         return int(self.mask.sum().item())
 
     def to(self, device, non_blocking: bool = True) -> "SubBatch":
         return SubBatch(self.indices.to(device, non_blocking=non_blocking),
-                        self.mask.to(device, non_blocking=non_blocking))
+                        self.mask.to(device, non_blocking=non_blocking),
+                        n_tokens=self.n_tokens)
 
     def narrow(self, start: int, size: int) -> "SubBatch":
         return SubBatch(self.indices[start:start + size].clone(),

@@ -51,7 +51,9 @@ class CrossEntropyCost:
             else:
                 losses = losses * w
         ce_sum = losses.sum()
-        n_labels = int(mask.sum().item())
+        # n_tokens was counted host-side at batch build: no device sync here
+        n_labels = trg.n_tokens if getattr(trg, "n_tokens", None) is not None \
+            else int(mask.sum().item())
         if self.cost_type in ("ce-mean", "perplexity"):
             loss = ce_sum / max(1, B)
         elif self.cost_type == "ce-mean-words":

@@ -340,17 +340,23 @@ def grad_sumsq_finite(g: torch.Tensor) -> torch.Tensor:
 
 
 def adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd, step,
-              grad_scale: float = 1.0):
+              grad_scale=1.0):
     """One fused Adam update; grad is multiplied by grad_scale first
-    (1/loss-scale x clip factor, folded in to avoid an extra pass)."""
+    (1/loss-scale x clip factor, folded in to avoid an extra pass).
+    grad_scale may be a float OR a 0-dim fp32 device tensor (the async
+    update path computes scaling/clipping on device, no host sync)."""
     if _on_gpu(param32):
         ext = hip_ext(required=True)
-        ext.adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd, step,
-                      grad_scale)
+        if isinstance(grad_scale, torch.Tensor):
+            ext.adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd,
+                          step, 1.0, grad_scale.float().reshape(1))
+        else:
+            ext.adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd,
+                          step, grad_scale)
         return
     # torch reference
     g = grad.float()
-    if grad_scale != 1.0:
+    if isinstance(grad_scale, torch.Tensor) or grad_scale != 1.0:
         g = g * grad_scale
     m.mul_(beta1).add_(g, alpha=1 - beta1)
     v.mul_(beta2).addcmul_(g, g, value=1 - beta2)

@@ -17,7 +17,11 @@ __global__ void adam_kernel(float* __restrict__ p,
                             float* __restrict__ m,
                             float* __restrict__ v,
                             long n, float lr, float b1, float b2, float eps,
-                            float wd, float bc1, float bc2, float gscale) {
+                            float wd, float bc1, float bc2, float gscale,
+                            const float* __restrict__ gscale_dev) {
+  // device-resident scale factor (cost-type/clip computed on GPU so the
+  // host never syncs mid-step); uniform scalar load, hits L2 once per CU
+  if (gscale_dev) gscale *= gscale_dev[0];
   const long i0 = (long)(blockIdx.x) * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = i0; i < n; i += stride) {
@@ -38,9 +42,17 @@ __global__ void adam_kernel(float* __restrict__ p,
 
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, double lr, double b1, double b2, double eps,
-               double wd, int64_t t, double gscale) {
+               double wd, int64_t t, double gscale,
+               c10::optional<torch::Tensor> gscale_t) {
   TORCH_CHECK(p.is_cuda() && p.scalar_type() == torch::kFloat32);
   TORCH_CHECK(p.is_contiguous() && m.is_contiguous() && v.is_contiguous());
+  const float* gsp = nullptr;
+  if (gscale_t.has_value()) {
+    TORCH_CHECK(gscale_t->is_cuda() &&
+                gscale_t->scalar_type() == torch::kFloat32 &&
+                gscale_t->is_contiguous());
+    gsp = gscale_t->data_ptr<float>();
+  }
   auto gc = g.contiguous();
   const long n = p.numel();
   const int block = 256;
@@ -54,7 +66,7 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      reinterpret_cast<const GT*>(gc.data_ptr()),             \
                      m.data_ptr<float>(), v.data_ptr<float>(), n, (float)lr, \
                      (float)b1, (float)b2, (float)eps, (float)wd, bc1, bc2,  \
-                     (float)gscale)
+                     (float)gscale, gsp)
   if (gc.scalar_type() == torch::kBFloat16) RUN(__hip_bfloat16);
   else if (gc.scalar_type() == torch::kHalf) RUN(__half);
   else if (gc.scalar_type() == torch::kFloat32) RUN(float);

@@ -21,7 +21,8 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
                      c10::optional<torch::Tensor> mask);
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, double lr, double b1, double b2, double eps,
-               double wd, int64_t t, double gscale);
+               double wd, int64_t t, double gscale,
+               c10::optional<torch::Tensor> gscale_t);
 std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v,
                                          c10::optional<torch::Tensor> kv_lens,
@@ -60,7 +61,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("log_softmax_bwd", &log_softmax_bwd, "row log-softmax backward");
   m.def("ce_fwd", &ce_fwd, "fused label-smoothed CE forward");
   m.def("ce_bwd", &ce_bwd, "fused label-smoothed CE backward");
-  m.def("adam_step", &adam_step, "fused Adam step");
+  m.def("adam_step", &adam_step, "fused Adam step",
+        py::arg("p"), py::arg("g"), py::arg("m"), py::arg("v"),
+        py::arg("lr"), py::arg("b1"), py::arg("b2"), py::arg("eps"),
+        py::arg("wd"), py::arg("t"), py::arg("gscale"),
+        py::arg("gscale_t") = c10::nullopt);
   m.def("attention_fwd", &attention_fwd,
         "fused flash-style attention forward (bf16, Dh=64)");
   m.def("attention_bwd_into", &attention_bwd_into,

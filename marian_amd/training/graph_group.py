@@ -117,7 +117,7 @@ class SyncGraphGroup:
         self._accum = 0
         self._accum_labels = 0
         self._accum_sents = 0
-        self._accum_ce = 0.0
+        self._accum_ce_t: Optional[torch.Tensor] = None  # device CE sum
         self._overlap = world > 1 and device.type == "cuda"
         if self._overlap:
             self._install_hooks()
@@ -193,7 +193,13 @@ class SyncGraphGroup:
 
         self._accum_labels += lv.label_count
         self._accum_sents += batch.size()
-        self._accum_ce += float(lv.ce_sum.detach().item())
+        # accumulate the CE on DEVICE: a per-step .item() here would
+        # synchronize the stream every update and stall rank progress
+        if self._accum_ce_t is None or \
+                self._accum_ce_t.device != lv.ce_sum.device:
+            self._accum_ce_t = torch.zeros(
+                (), device=lv.ce_sum.device, dtype=torch.float32)
+        self._accum_ce_t += lv.ce_sum.detach().float()
 
         if not is_final:
             return None
@@ -219,9 +225,85 @@ class SyncGraphGroup:
                 self.comm.reduce_scatter_bucket(
                     b.flat_grad, opt.grad_shard_for_bucket(bi))
 
+        # fast path: everything (scaling, clipping, Adam, all-gather) is
+        # computed ON DEVICE and fully enqueued before the single late
+        # host sync for logging — a mid-step .item() would stall rank
+        # progress and cap comm/compute overlap at 8 GPUs.
+        # fp16 cost-scaling / --check-nan / dynamic-gradient-scaling need
+        # a host-side skip decision, so they take the host path below.
+        if (self.device.type == "cuda" or self._force_async) \
+                and not self.check_grad_nan and self.dgs_factor <= 0:
+            return self._apply_update_async(opt)
+        return self._apply_update_host(opt)
+
+    _force_async = False  # test hook: exercise the async path on CPU
+
+    def _apply_update_async(self, opt) -> LossValue:
+        red = torch.empty(3, device=self.device, dtype=torch.float64)
+        red[0] = float(self._accum_labels)
+        red[1] = float(self._accum_sents)
+        red[2] = self._accum_ce_t
+        if self.world > 1:
+            self.comm.all_reduce_scalar(red)
+        ct = self.criterion.cost_type
+        if ct == "ce-mean-words":
+            gs_dev = red[0].clamp(min=1.0)
+        elif ct in ("ce-mean", "perplexity"):
+            gs_dev = red[1].clamp(min=1.0)
+        else:
+            gs_dev = None
+
+        # the arithmetic below mirrors the host path's float32/float64
+        # mix exactly (fp32 sumsq/norm, fp64 scale combine, one final fp32
+        # rounding) so both paths are bit-identical
+        norm_dev = None
+        clipf = None
+        if self.clip_norm > 0:
+            stats = opt.grad_stats()
+            if self.world > 1:
+                stats = stats.clone()
+                self.comm.all_reduce_scalar(stats)
+            sq = stats[0]
+            if gs_dev is not None:
+                sq = sq / (gs_dev * gs_dev).float()
+            norm_dev = sq.sqrt()           # fp32, as the host path computes
+            self._last_gnorm = norm_dev
+            clipf = (self.clip_norm /
+                     norm_dev.double().clamp(min=1e-300)).clamp(max=1.0)
+
+        mult_t = None
+        if gs_dev is not None:
+            mult_t = torch.reciprocal(gs_dev)
+        if clipf is not None:
+            mult_t = clipf if mult_t is None else mult_t * clipf
+        mult = mult_t.float() if mult_t is not None else 1.0
+
+        opt.step(self._current_lr, mult=mult)
+        for bi, b in enumerate(self.store.buckets):
+            self.comm.all_gather_bucket(
+                b.flat_param,
+                b.shard_view(b.flat_param, self.rank, self.world))
+        self.comm.wait_all()
+        self.store.zero_grad()
+        opt.shard_grad.zero_()
+
+        # single late sync: every kernel/collective above is enqueued
+        g_labels, g_sents, g_ce = red.tolist()
+        lv = LossValue(loss=torch.tensor(g_ce), ce_sum=torch.tensor(g_ce),
+                       label_count=int(g_labels))
+        lv.sentences = int(g_sents)   # type: ignore[attr-defined]
+        lv.grad_norm = norm_dev       # type: ignore[attr-defined]
+        self._accum_labels = 0
+        self._accum_sents = 0
+        if self._accum_ce_t is not None:
+            self._accum_ce_t.zero_()
+        return lv
+
+    def _apply_update_host(self, opt) -> LossValue:
         # aggregate label/sentence counts across ranks for exact scaling
-        stats = torch.tensor([self._accum_labels, self._accum_sents,
-                              self._accum_ce],
+        ce = float(self._accum_ce_t.item()) \
+            if self._accum_ce_t is not None else 0.0
+        stats = torch.tensor([self._accum_labels, self._accum_sents, ce],
                              device=self.device, dtype=torch.float64)
         if self.world > 1:
             self.comm.all_reduce_scalar(stats)
@@ -293,7 +375,8 @@ class SyncGraphGroup:
         lv.grad_norm = grad_norm     # type: ignore[attr-defined]
         self._accum_labels = 0
         self._accum_sents = 0
-        self._accum_ce = 0.0
+        if self._accum_ce_t is not None:
+            self._accum_ce_t.zero_()
         return lv
 
     _current_lr = 1e-4

@@ -288,20 +288,24 @@ class ShardedOptimizer:
         return self._grad_stats
 
     def step(self, lr: float, grad_scale: float = 1.0,
-             grad_norm: Optional[float] = None) -> None:
+             grad_norm: Optional[float] = None, mult=None) -> None:
+        """mult: precomputed gradient multiplier (float OR 0-dim device
+        tensor — the async update path keeps scaling/clipping on device);
+        when None it is derived from grad_scale/grad_norm host-side."""
         self._grad_stats = None
         self.t += 1
-        mult = 1.0 / grad_scale
-        if self.clip_norm > 0 and grad_norm is not None \
-                and grad_norm > self.clip_norm:
-            mult *= self.clip_norm / grad_norm
+        if mult is None:
+            mult = 1.0 / grad_scale
+            if self.clip_norm > 0 and grad_norm is not None \
+                    and grad_norm > self.clip_norm:
+                mult *= self.clip_norm / grad_norm
         if isinstance(self.core, Adam):
             # single fused pass: scaling folded into the Adam kernel
             self.core.step(self.master, self.shard_grad, lr, self.t,
                            grad_scale=mult)
         else:
             g = self.shard_grad.float()
-            if mult != 1.0:
+            if isinstance(mult, torch.Tensor) or mult != 1.0:
                 g = g * mult
             self.core.step(self.master, g, lr, self.t)
         if self.quantize_bits:

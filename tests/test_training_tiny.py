@@ -505,3 +505,50 @@ def test_smoothing_state_resumes(workdir):
     tr2 = Train(opts2)
     got = tr2.graph_group.optimizer.smooth
     assert torch.allclose(got, want, atol=1e-6)
+
+
+def test_async_update_path_matches_host_path(workdir):
+    """The device-side (no host sync) update path must produce the same
+    parameters as the host-side path: scaling, clipping and Adam all
+    computed on-device vs on-host."""
+    import numpy as np
+    from marian_amd.common.options import Options
+    from marian_amd.data.synthetic import synthetic_batch
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.training.graph_group import SyncGraphGroup
+
+    o = {
+        "type": "transformer", "dim-emb": 32, "enc-depth": 1, "dec-depth": 1,
+        "transformer-heads": 2, "transformer-dim-ffn": 64,
+        "transformer-preprocess": "", "transformer-postprocess": "dan",
+        "transformer-postprocess-emb": "d", "transformer-postprocess-top": "",
+        "transformer-ffn-activation": "relu", "transformer-dropout": 0.0,
+        "dim-vocabs": [43, 43], "tied-embeddings-all": True,
+        "precision": ["float32", "float32"],
+        "cost-type": "ce-mean-words", "label-smoothing": 0.1,
+        "optimizer": "adam", "optimizer-params": [0.9, 0.98, 1e-9],
+        "clip-norm": 0.5,   # small: clipping actually engages
+    }
+    results = []
+    for force_async in (False, True):
+        torch.manual_seed(11)
+        opts = Options(dict(o))
+        model = create_model(opts)
+        gg = SyncGraphGroup(opts, model, torch.device("cpu"))
+        gg._force_async = force_async
+        gg.set_lr(0.05)
+        lvs = []
+        for step in range(4):
+            batch = synthetic_batch(5, 8, 8, 43, seed=50 + step)
+            lv = gg.update(batch)
+            lvs.append((float(lv.ce_sum), lv.label_count,
+                        getattr(lv, "sentences", None)))
+        params = torch.cat([b.flat_param[:b.used]
+                            for b in gg.store.buckets]).numpy().copy()
+        gn = getattr(lvs, "grad_norm", None)
+        results.append((params, lvs))
+    np.testing.assert_allclose(results[0][0], results[1][0],
+                               rtol=1e-6, atol=1e-7)
+    for a, b in zip(results[0][1], results[1][1]):
+        assert a[1] == b[1] and a[2] == b[2]
+        np.testing.assert_allclose(a[0], b[0], rtol=1e-6)
