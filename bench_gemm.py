@@ -41,6 +41,8 @@ def timeit(fn, iters=30, warmup=5):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--retune", action="store_true")
+    ap.add_argument("--compare", action="store_true",
+                    help="also time lt_mm (full-pool algo search) per shape")
     ap.add_argument("--batch", type=int, default=1024)
     ap.add_argument("--seq", type=int, default=64)
     ap.add_argument("--dim", type=int, default=1024)
@@ -84,12 +86,16 @@ def main():
     def dgrad(m, n, k):
         dy = t(m, n)
         W = t(k, n)   # row-major weight, dgrad uses .t() view (NT)
-        return lambda: torch.mm(dy, W.t())
+        fn = lambda: torch.mm(dy, W.t())
+        fn._operands = (dy, W.t())
+        return fn
 
     def wgrad(m, k, n):
         x = t(m, k)
         dy = t(m, n)
-        return lambda: torch.mm(x.t(), dy)
+        fn = lambda: torch.mm(x.t(), dy)
+        fn._operands = (x.t(), dy)
+        return fn
 
     n_self = NE + ND      # self-attn blocks
     gemm("qkv_fwd", n_self, 2 * M * d * 3 * d, lambda: addmm(M, d, 3 * d))
@@ -130,34 +136,63 @@ def main():
 
     def logits_dgrad():
         dy = t(M, V)
-        return lambda: torch.mm(dy, E)   # dx = dy @ (E.t()).t() == dy @ E
+        fn = lambda: torch.mm(dy, E)   # dx = dy @ (E.t()).t() == dy @ E
+        fn._operands = (dy, E)
+        return fn
 
     def logits_wgrad():
         x = t(M, d)
         dy = t(M, V)
-        return lambda: torch.mm(x.t(), dy)  # [d,V] into E.t().grad
+        fn = lambda: torch.mm(x.t(), dy)  # [d,V] into E.t().grad
+        fn._operands = (x.t(), dy)
+        return fn
 
     gemm("logits_fwd", 1, 2 * M * d * V, logits_fwd)
     gemm("logits_dgrad", 1, 2 * M * d * V, logits_dgrad)
     gemm("logits_wgrad", 1, 2 * M * d * V, logits_wgrad)
 
+    # --compare: time the torch route vs the full-pool-searched lt_mm
+    # (split-K included) on the same operands
+    from marian_amd import ops as mops
+    ext = mops.hip_ext(required=True) if args.compare else None
+
     total_ms = 0.0
     total_tf = 0.0
+    best_ms = 0.0
     print(f"{'name':24s} {'count':>5s} {'ms/call':>8s} {'TF/s':>7s} "
-          f"{'ms/step':>8s} {'TFLOP/step':>10s}")
+          f"{'lt ms':>8s} {'lt TF/s':>7s} {'ms/step':>8s} {'TFLOP/step':>10s}")
     rows = []
     for name, count, flops, make in work:
         fn = make()
         sec = timeit(fn)
         tf = flops / sec / 1e12
+        lt_txt = ("", "")
+        lt_sec = None
+        if ext is not None and hasattr(fn, "_operands"):
+            A, B = fn._operands
+            ref = fn()
+            got = ext.lt_mm(A, B)
+            err = (got.float() - (ref if ref.dtype == torch.float32
+                                  else ref.float())).abs().max().item()
+            den = ref.float().abs().max().item()
+            assert err <= 0.02 * max(den, 1.0), (name, err, den)
+            lt_fn = lambda: ext.lt_mm(A, B)
+            lt_sec = timeit(lt_fn)
+            lt_txt = (f"{lt_sec*1e3:8.3f}", f"{flops/lt_sec/1e12:7.0f}")
+        use = min(sec, lt_sec) if lt_sec else sec
         ms_step = sec * 1e3 * count
+        best_ms += use * 1e3 * count
         total_ms += ms_step
         total_tf += flops * count / 1e12
         rows.append((name, count, sec * 1e3, tf, ms_step, flops * count / 1e12))
         print(f"{name:24s} {count:5d} {sec*1e3:8.3f} {tf:7.0f} "
+              f"{lt_txt[0]:>8s} {lt_txt[1]:>7s} "
               f"{ms_step:8.2f} {flops*count/1e12:10.2f}")
     print(f"{'TOTAL':24s} {'':5s} {'':8s} "
-          f"{total_tf/ (total_ms/1e3) :7.0f} {total_ms:8.2f} {total_tf:10.2f}")
+          f"{total_tf/ (total_ms/1e3) :7.0f} {'':8s} {'':7s} {total_ms:8.2f} {total_tf:10.2f}")
+    if args.compare:
+        print(f"# best-of(torch, lt): {best_ms:.1f} ms/step at "
+              f"{total_tf/(best_ms/1e3):.0f} TF/s effective")
     print(f"# aggregate: {total_ms:.1f} ms/step of GEMM at "
           f"{total_tf/(total_ms/1e3):.0f} TF/s effective "
           f"({100*total_tf/(total_ms/1e3)/2495:.0f}% of bf16 dense peak)")

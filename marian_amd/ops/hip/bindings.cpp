@@ -32,6 +32,7 @@ torch::Tensor lt_linear_act(torch::Tensor x, torch::Tensor W,
                             torch::Tensor b, bool relu);
 std::vector<torch::Tensor> lt_dwt_bgrad(torch::Tensor x, torch::Tensor dy,
                                         bool want_db);
+torch::Tensor lt_mm(torch::Tensor A, torch::Tensor B);
 std::vector<torch::Tensor> fp8_quantize(torch::Tensor x,
                                         torch::Tensor scale);
 torch::Tensor col_sum(torch::Tensor x);
@@ -72,6 +73,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused attention backward writing into given strided views");
   m.def("attention_bwd", &attention_bwd,
         "fused flash-style attention backward (bf16, Dh=64)");
+  m.def("lt_mm", &lt_mm, "tuned bf16 GEMM (full algo pool, split-K)");
   m.def("lt_linear_act", &lt_linear_act,
         "hipBLASLt GEMM with fused bias(+ReLU) epilogue");
   m.def("lt_dwt_bgrad", &lt_dwt_bgrad,
