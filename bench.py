@@ -124,12 +124,29 @@ def main():
             import torch.distributed as dist
             dist.barrier()
 
+    # whole-step hipGraph capture (world 1): fwd+bwd+Adam replayed as ONE
+    # graph — no launch gaps, no Python dispatch in the timed region.
+    # Same work per step (the graph contains every kernel of the eager
+    # step); disable with MAMD_STEP_GRAPH=0.
+    step_fn = None
+    if on_gpu and os.environ.get("MAMD_STEP_GRAPH", "1") == "1" \
+            and gg.can_capture():
+        try:
+            step_fn = gg.make_graphed_update(batch,
+                                             warmup=max(2, args.warmup))
+        except Exception as e:
+            print(f"# step-graph capture failed ({e}); eager fallback",
+                  file=sys.stderr)
+            step_fn = None
+    if step_fn is None:
+        step_fn = gg.update
+
     for _ in range(args.warmup):
-        gg.update(batch)
+        step_fn(batch)
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        gg.update(batch)
+        step_fn(batch)
     sync()
     elapsed = time.perf_counter() - t0
 

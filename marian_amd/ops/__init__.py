@@ -340,21 +340,28 @@ def grad_sumsq_finite(g: torch.Tensor) -> torch.Tensor:
 
 
 def adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd, step,
-              grad_scale=1.0):
+              grad_scale=1.0, lr_t=None, step_t=None):
     """One fused Adam update; grad is multiplied by grad_scale first
     (1/loss-scale x clip factor, folded in to avoid an extra pass).
     grad_scale may be a float OR a 0-dim fp32 device tensor (the async
-    update path computes scaling/clipping on device, no host sync)."""
+    update path computes scaling/clipping on device, no host sync).
+    lr_t/step_t: device-resident LR + step counter for hipGraph-replayed
+    steps (bias correction computed in-kernel)."""
     if _on_gpu(param32):
         ext = hip_ext(required=True)
+        gs_t = None
+        gs = grad_scale
         if isinstance(grad_scale, torch.Tensor):
-            ext.adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd,
-                          step, 1.0, grad_scale.float().reshape(1))
-        else:
-            ext.adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd,
-                          step, grad_scale)
+            gs_t = grad_scale.float().reshape(1)
+            gs = 1.0
+        ext.adam_step(param32, grad, m, v, lr, beta1, beta2, eps, wd,
+                      step, gs, gs_t, lr_t, step_t)
         return
     # torch reference
+    if lr_t is not None:
+        lr = float(lr_t.reshape(-1)[0])
+    if step_t is not None:
+        step = int(step_t.reshape(-1)[0])
     g = grad.float()
     if isinstance(grad_scale, torch.Tensor) or grad_scale != 1.0:
         g = g * grad_scale

@@ -22,7 +22,9 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels,
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, double lr, double b1, double b2, double eps,
                double wd, int64_t t, double gscale,
-               c10::optional<torch::Tensor> gscale_t);
+               c10::optional<torch::Tensor> gscale_t,
+               c10::optional<torch::Tensor> lr_t,
+               c10::optional<torch::Tensor> step_t);
 std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k,
                                          torch::Tensor v,
                                          c10::optional<torch::Tensor> kv_lens,
@@ -66,7 +68,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("p"), py::arg("g"), py::arg("m"), py::arg("v"),
         py::arg("lr"), py::arg("b1"), py::arg("b2"), py::arg("eps"),
         py::arg("wd"), py::arg("t"), py::arg("gscale"),
-        py::arg("gscale_t") = c10::nullopt);
+        py::arg("gscale_t") = c10::nullopt,
+        py::arg("lr_t") = c10::nullopt,
+        py::arg("step_t") = c10::nullopt);
   m.def("attention_fwd", &attention_fwd,
         "fused flash-style attention forward (bf16, Dh=64)");
   m.def("attention_bwd_into", &attention_bwd_into,

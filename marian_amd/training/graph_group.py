@@ -240,8 +240,13 @@ class SyncGraphGroup:
 
     def _apply_update_async(self, opt) -> LossValue:
         red = torch.empty(3, device=self.device, dtype=torch.float64)
-        red[0] = float(self._accum_labels)
-        red[1] = float(self._accum_sents)
+        if self._capture_mode:
+            # under hipGraph capture the label/sentence counts must come
+            # from a device buffer the replay wrapper refreshes per step
+            red[:2] = self._g_counts
+        else:
+            red[0] = float(self._accum_labels)
+            red[1] = float(self._accum_sents)
         red[2] = self._accum_ce_t
         if self.world > 1:
             self.comm.all_reduce_scalar(red)
@@ -286,6 +291,15 @@ class SyncGraphGroup:
         self.comm.wait_all()
         self.store.zero_grad()
         opt.shard_grad.zero_()
+
+        if self._capture_mode:
+            # no host sync inside the captured region: stash the device
+            # buffers; the replay wrapper reads them after g.replay()
+            self._g_red = red
+            self._g_norm = norm_dev
+            self._accum_labels = 0
+            self._accum_sents = 0
+            return LossValue(loss=red[2], ce_sum=red[2], label_count=0)
 
         # single late sync: every kernel/collective above is enqueued
         g_labels, g_sents, g_ce = red.tolist()
@@ -381,9 +395,79 @@ class SyncGraphGroup:
 
     _current_lr = 1e-4
     _last_gnorm = None
+    _capture_mode = False
 
     def set_lr(self, lr: float):
         self._current_lr = lr
+
+    # ------------------------------------------------------------------
+    # Whole-step hipGraph capture: fwd + bwd + scaling + fused Adam +
+    # (all-gather) recorded once, replayed per step. Kills every kernel
+    # launch gap and all Python dispatch inside the step. The MI355X
+    # answer to the reference's per-node launch overhead (SURVEY §4.1
+    # hot-loop note); requires the async (no-host-sync) update path.
+    def can_capture(self) -> bool:
+        from .optimizers import Adam
+        return (self.device.type == "cuda" and self.world == 1
+                and not self.check_grad_nan and self.dgs_factor <= 0
+                and self.delay == 1 and not self.guided_alignment
+                and isinstance(self.optimizer.core, Adam)
+                and not self.optimizer.quantize_bits)
+
+    def make_graphed_update(self, example_batch, warmup: int = 3):
+        """Capture one update() on `example_batch` (shapes become fixed);
+        returns fn(batch)->LossValue that replays the graph. Distinct
+        shapes need distinct captures (callers bucket shapes)."""
+        assert self.can_capture(), "step not capturable with these options"
+        opt = self.optimizer
+        opt.enable_graph_mode()
+        static = example_batch.to(self.device)
+        self._g_counts = torch.zeros(2, device=self.device,
+                                     dtype=torch.float64)
+        opt.lr_dev.fill_(self._current_lr)
+        # warm up on a side stream (TunableOp/lt algo selection, autograd
+        # buffer allocation) before capture, per hipGraph requirements
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(max(1, warmup)):
+                self.update(static)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+
+        g = torch.cuda.CUDAGraph()
+        self._capture_mode = True
+        try:
+            with torch.cuda.graph(g):
+                self.update(static)
+        finally:
+            self._capture_mode = False
+        self._graph = g  # keep alive: owns the replay buffers
+
+        B = static.source.indices.shape[0]
+
+        def run(batch) -> LossValue:
+            src, trg = batch.source, batch.target
+            static.source.indices.copy_(src.indices, non_blocking=True)
+            static.source.mask.copy_(src.mask, non_blocking=True)
+            static.target.indices.copy_(trg.indices, non_blocking=True)
+            static.target.mask.copy_(trg.mask, non_blocking=True)
+            n_labels = trg.n_tokens if getattr(trg, "n_tokens", None) \
+                is not None else int(trg.mask.sum().item())
+            self._g_counts[0] = float(n_labels)
+            self._g_counts[1] = float(B)
+            opt.lr_dev.fill_(self._current_lr)
+            g.replay()
+            g_labels, g_sents, g_ce = self._g_red.tolist()
+            lv = LossValue(loss=torch.tensor(g_ce),
+                           ce_sum=torch.tensor(g_ce),
+                           label_count=int(g_labels))
+            lv.sentences = int(g_sents)      # type: ignore[attr-defined]
+            lv.grad_norm = self._g_norm      # type: ignore[attr-defined]
+            return lv
+
+        run.static_batch = static
+        return run
 
     # ------------------------------------------------------------------
     # checkpointing (4-file contract, ref: SURVEY.md §6 checkpoint row)

@@ -183,10 +183,11 @@ class Adam:
         self.v = torch.zeros(params, device=device, dtype=torch.float32)
         self.beta1, self.beta2, self.eps, self.wd = beta1, beta2, eps, wd
 
-    def step(self, master, grad, lr, t, grad_scale: float = 1.0):
+    def step(self, master, grad, lr, t, grad_scale=1.0,
+             lr_t=None, step_t=None):
         ops.adam_step(master, grad, self.m, self.v, lr,
                       self.beta1, self.beta2, self.eps, self.wd, t,
-                      grad_scale)
+                      grad_scale, lr_t=lr_t, step_t=step_t)
 
     def state_items(self):
         return {"adam_mt": self.m, "adam_vt": self.v}
@@ -249,6 +250,11 @@ class ShardedOptimizer:
                     b.shard_view(b.flat_param, rank, self.world).float())
         self.core = create_optimizer_core(options, self.n, dev)
         self.t = 0
+        # hipGraph-replay mode: step counter + LR live ON DEVICE so a
+        # captured update stays correct across replays (host `self.t`
+        # would be frozen at capture value); enable via enable_graph_mode()
+        self.t_dev: Optional[torch.Tensor] = None
+        self.lr_dev: Optional[torch.Tensor] = None
         self.clip_norm = float(options.get("clip-norm", 0.0) or 0.0)
         # --exponential-smoothing: Polyak average of the shard
         # (ref: optimizers/exponential_smoothing + SyncGraphGroup swapParams)
@@ -287,6 +293,21 @@ class ShardedOptimizer:
             self._grad_stats = ops.grad_sumsq_finite(self.shard_grad)
         return self._grad_stats
 
+    def enable_graph_mode(self):
+        """Move the step counter + LR to device tensors so the whole
+        optimizer step can be captured into a hipGraph and replayed
+        (ref role: the reference has no graph capture; this is the
+        MI355X-native answer to its per-node launch overhead)."""
+        if self.t_dev is None:
+            dev = self.master.device
+            self.t_dev = torch.tensor([self.t], device=dev,
+                                      dtype=torch.int32)
+            self.lr_dev = torch.zeros(1, device=dev, dtype=torch.float32)
+
+    def sync_t_from_device(self):
+        if self.t_dev is not None:
+            self.t = int(self.t_dev.item())
+
     def step(self, lr: float, grad_scale: float = 1.0,
              grad_norm: Optional[float] = None, mult=None) -> None:
         """mult: precomputed gradient multiplier (float OR 0-dim device
@@ -299,10 +320,15 @@ class ShardedOptimizer:
             if self.clip_norm > 0 and grad_norm is not None \
                     and grad_norm > self.clip_norm:
                 mult *= self.clip_norm / grad_norm
+        graph = self.t_dev is not None
+        if graph:
+            self.t_dev += 1
         if isinstance(self.core, Adam):
             # single fused pass: scaling folded into the Adam kernel
             self.core.step(self.master, self.shard_grad, lr, self.t,
-                           grad_scale=mult)
+                           grad_scale=mult,
+                           lr_t=self.lr_dev if graph else None,
+                           step_t=self.t_dev if graph else None)
         else:
             g = self.shard_grad.float()
             if isinstance(mult, torch.Tensor) or mult != 1.0:
@@ -314,9 +340,15 @@ class ShardedOptimizer:
             # marian's flag value is the decay-BY epsilon (keep factor
             # 1-eps, typically 1e-4), capped early in training
             # (ref: optimizers/exponential_smoothing updateAvgParams)
-            decay = min(1.0 - self.smoothing,
-                        (self.t + 1.0) / (self.t + 10.0))
-            self.smooth.mul_(decay).add_(self.master, alpha=1.0 - decay)
+            if graph:   # device-side decay so hipGraph replays stay exact
+                td = self.t_dev.to(torch.float32)
+                decay = torch.clamp((td + 1.0) / (td + 10.0),
+                                    max=1.0 - self.smoothing)
+                self.smooth.lerp_(self.master, (1.0 - decay).reshape(()))
+            else:
+                decay = min(1.0 - self.smoothing,
+                            (self.t + 1.0) / (self.t + 10.0))
+                self.smooth.mul_(decay).add_(self.master, alpha=1.0 - decay)
         # write updated shard back into each bucket's param slice
         with torch.no_grad():
             for b, (o, s) in zip(self.store.buckets, self.seg):
@@ -414,6 +446,7 @@ class ShardedOptimizer:
     # ------------------------------------------------------------------
     # checkpoint (ref: model.npz.optimizer.npz via gatherState/scatterState)
     def state_items(self) -> Dict[str, np.ndarray]:
+        self.sync_t_from_device()
         out = {"master_parameters": self.master.cpu().numpy(),
                "optimizer_step": np.array([self.t], dtype=np.int64)}
         if self.smooth is not None:
@@ -432,6 +465,8 @@ class ShardedOptimizer:
                         self.master[o:o + s].to(self.store.dtype))
         if "optimizer_step" in items:
             self.t = int(np.asarray(items["optimizer_step"]).reshape(-1)[0])
+            if self.t_dev is not None:
+                self.t_dev.fill_(self.t)
         if "exp_smoothing" in items and self.smooth is not None:
             self.smooth.copy_(torch.from_numpy(np.ascontiguousarray(
                 items["exp_smoothing"])).to(self.smooth.device))
