@@ -294,7 +294,10 @@ class SyncGraphGroup:
 
         if self._capture_mode:
             # no host sync inside the captured region: stash the device
-            # buffers; the replay wrapper reads them after g.replay()
+            # buffers; the replay wrapper reads them after g.replay().
+            # The CE accumulator zero MUST be recorded too (after the
+            # `red` copy) or replays accumulate CE across steps.
+            self._accum_ce_t.zero_()
             self._g_red = red
             self._g_norm = norm_dev
             self._accum_labels = 0
