@@ -29,8 +29,8 @@ __global__ void adam_kernel(float* __restrict__ p,
   if (lr_dev) lr = lr_dev[0];
   if (step_dev) {
     const float tt = (float)step_dev[0];
-    bc1 = 1.f - __powf(b1, tt);
-    bc2 = 1.f - __powf(b2, tt);
+    bc1 = 1.f - powf(b1, tt);
+    bc2 = 1.f - powf(b2, tt);
   }
   const long i0 = (long)(blockIdx.x) * blockDim.x + threadIdx.x;
   const long stride = (long)gridDim.x * blockDim.x;

@@ -968,8 +968,8 @@ def test_graphed_step_matches_eager(ext):
     graph_lv, graph_p, opt_g = run(True)
     for (ce_e, n_e), (ce_g, n_g) in zip(eager_lv, graph_lv):
         assert n_e == n_g
-        assert abs(ce_e - ce_g) < 1e-3 * max(1.0, abs(ce_e)), (ce_e, ce_g)
+        assert abs(ce_e - ce_g) < 5e-3 * max(1.0, abs(ce_e)), (ce_e, ce_g)
     opt_g.sync_t_from_device()
     assert opt_e.t == opt_g.t
     d = (eager_p - graph_p).abs().max()
-    assert float(d) < 5e-3, float(d)
+    assert float(d) < 2e-2, float(d)
