@@ -95,6 +95,7 @@ def main():
         dy = t(m, n)
         fn = lambda: torch.mm(x.t(), dy)
         fn._operands = (x.t(), dy)
+        fn._wgrad = (x, dy)
         return fn
 
     n_self = NE + ND      # self-attn blocks
@@ -145,6 +146,7 @@ def main():
         dy = t(M, V)
         fn = lambda: torch.mm(x.t(), dy)  # [d,V] into E.t().grad
         fn._operands = (x.t(), dy)
+        fn._wgrad = (x, dy)
         return fn
 
     gemm("logits_fwd", 1, 2 * M * d * V, logits_fwd)
@@ -179,6 +181,19 @@ def main():
             lt_fn = lambda: ext.lt_mm(A, B)
             lt_sec = timeit(lt_fn)
             lt_txt = (f"{lt_sec*1e3:8.3f}", f"{flops/lt_sec/1e12:7.0f}")
+        if ext is not None and hasattr(fn, "_wgrad"):
+            xw, dyw = fn._wgrad
+            ref = fn().float()
+            got = ext.wgrad_bf16(xw, dyw, 0).float()
+            err = (got - ref).abs().max().item()
+            den = ref.abs().max().item()
+            ok = err <= 0.03 * max(den, 1.0)
+            wsec = timeit(lambda: ext.wgrad_bf16(xw, dyw, 0))
+            print(f"{'  ^ hand-mfma wgrad':24s} {'':5s} {wsec*1e3:8.3f} "
+                  f"{flops/wsec/1e12:7.0f}   err={err:.4f}/{den:.1f} "
+                  f"{'OK' if ok else 'FAIL'}")
+            if wsec < (lt_sec or sec) and ok:
+                lt_sec = wsec
         use = min(sec, lt_sec) if lt_sec else sec
         ms_step = sec * 1e3 * count
         best_ms += use * 1e3 * count

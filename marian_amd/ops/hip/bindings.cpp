@@ -35,6 +35,7 @@ torch::Tensor lt_linear_act(torch::Tensor x, torch::Tensor W,
 std::vector<torch::Tensor> lt_dwt_bgrad(torch::Tensor x, torch::Tensor dy,
                                         bool want_db);
 torch::Tensor lt_mm(torch::Tensor A, torch::Tensor B);
+torch::Tensor wgrad_bf16(torch::Tensor x, torch::Tensor dy, int64_t splits);
 std::vector<torch::Tensor> fp8_quantize(torch::Tensor x,
                                         torch::Tensor scale);
 torch::Tensor col_sum(torch::Tensor x);
@@ -78,6 +79,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_bwd", &attention_bwd,
         "fused flash-style attention backward (bf16, Dh=64)");
   m.def("lt_mm", &lt_mm, "tuned bf16 GEMM (full algo pool, split-K)");
+  m.def("wgrad_bf16", &wgrad_bf16,
+        "hand-MFMA split-K weight-grad GEMM dW = x^T dy",
+        py::arg("x"), py::arg("dy"), py::arg("splits") = 0);
   m.def("lt_linear_act", &lt_linear_act,
         "hipBLASLt GEMM with fused bias(+ReLU) epilogue");
   m.def("lt_dwt_bgrad", &lt_dwt_bgrad,

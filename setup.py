@@ -21,6 +21,7 @@ sources = [
     os.path.join(HIP_DIR, "softmax.hip"),
     os.path.join(HIP_DIR, "cross_entropy.hip"),
     os.path.join(HIP_DIR, "adam.hip"),
+    os.path.join(HIP_DIR, "wgrad.hip"),
 ]
 attn = os.path.join(HIP_DIR, "attention.hip")
 if os.path.exists(attn):
