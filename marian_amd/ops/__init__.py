@@ -127,12 +127,23 @@ def linear_act(x: torch.Tensor, W: torch.Tensor,
     return _ACT[act](linear(x, W, b))
 
 
+def _use_hand_wgrad(K: int, N: int, M: int) -> bool:
+    """The hand-MFMA split-K wgrad (ops/hip/wgrad.hip) wins on the
+    SMALL-output dW shapes where hipBLASLt cannot parallelize the
+    M=B*T reduction (measured r2: [65536,1024]^T@[65536,1024] 551 vs
+    431 TF/s, x2048 573 vs 516); on >=3072-wide outputs the library's
+    bigger tiles win — keep torch there."""
+    tiles = ((K + 127) // 128) * ((N + 127) // 128)
+    return (M >= 4096 and K % 8 == 0 and N % 8 == 0 and tiles <= 128
+            and "MAMD_NO_HAND_WGRAD" not in os.environ)
+
+
 class _LinearFn(torch.autograd.Function):
-    """addmm forward; backward with the HIP column-sum kernel for the bias
-    gradient. Measured on MI355X: torch's bf16 column reduce runs at
-    3.2-4.8 TB/s, so this path is a wash at NMT widths and is OPT-IN via
-    MAMD_CUSTOM_LINEAR=1 (kept for wider-model configs where col_sum's
-    5.7 TB/s at C>=32k wins)."""
+    """addmm forward; backward routes the weight gradient through the
+    hand-MFMA split-K kernel on the shapes where it beats hipBLASLt
+    (see _use_hand_wgrad). With MAMD_CUSTOM_LINEAR=1 the bias gradient
+    additionally uses the HIP column-sum kernel (a wash at NMT widths,
+    kept for wider models)."""
 
     @staticmethod
     def forward(ctx, x2, W, b):
@@ -147,10 +158,17 @@ class _LinearFn(torch.autograd.Function):
         x2, W = ctx.saved_tensors
         dyc = dy.contiguous()
         dx = torch.mm(dyc, W.t())
-        dW = torch.mm(x2.t(), dyc)
+        if x2.dtype == torch.bfloat16 and dyc.dtype == torch.bfloat16 \
+                and _use_hand_wgrad(x2.shape[1], dyc.shape[1], x2.shape[0]):
+            dW = hip_ext(required=True).wgrad_bf16(x2, dyc, 0)
+        else:
+            dW = torch.mm(x2.t(), dyc)
         db = None
         if ctx.has_bias:
-            db = hip_ext(required=True).col_sum(dyc).to(dyc.dtype)
+            if "MAMD_CUSTOM_LINEAR" in os.environ:
+                db = hip_ext(required=True).col_sum(dyc).to(dyc.dtype)
+            else:
+                db = dyc.sum(0)
         return dx, dW, db
 
 
@@ -163,7 +181,11 @@ def linear(x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor] = None
         return fp8_linear(x, W.to(x.dtype), b.to(x.dtype) if b is not None else None)
     W = W.to(x.dtype)
     shp = x.shape[:-1] + (W.shape[1],)
-    if _on_gpu(x) and "MAMD_CUSTOM_LINEAR" in os.environ:
+    if _on_gpu(x) and torch.is_grad_enabled() \
+            and (x.requires_grad or W.requires_grad) \
+            and (("MAMD_CUSTOM_LINEAR" in os.environ)
+                 or (x.dtype == torch.bfloat16 and _use_hand_wgrad(
+                     W.shape[0], W.shape[1], x.numel() // W.shape[0]))):
         x2 = x.reshape(-1, W.shape[0])
         return _LinearFn.apply(x2, W.contiguous() if not W.is_contiguous()
                                and W.stride(0) != 1 else W,

@@ -39,10 +39,12 @@ namespace {
 constexpr int BM = 128;   // kd tile
 constexpr int BN = 128;   // n tile
 constexpr int BK = 64;    // m (reduction) step
-// Row stride 68 elems (136 B = 34 dwords): every row start is 8B-aligned
-// (b64 LDS ops legal) and row*34%32 spreads rows over 2-way bank pairs;
-// the m-block XOR swizzle below breaks the remaining (r, r+16) pairs.
-constexpr int LDS_STRIDE = BK + 4;
+// Row stride 72 elems (144 B): every row start is 16B-aligned, so the
+// fragment reads are single ds_read_b128 (4 LDS cycles vs 8 for the
+// read2_b64 pair the 8B-aligned variant forced). b128 banking is
+// (a/4)%64 over 16-lane groups: 36 dwords/row makes rows within a group
+// distinct; the m-block XOR swizzle spreads the write phase.
+constexpr int LDS_STRIDE = BK + 8;
 
 // Swizzled m-position: 8-element m-blocks XORed by row bits 3..5 so both
 // the transposed write phase and the fragment reads are bank-spread while
@@ -52,10 +54,9 @@ __device__ __forceinline__ int swz(int row, int m) {
 }
 
 __device__ __forceinline__ bf16x8_w lds_read8(const __hip_bfloat16* p) {
-  union { unsigned long long v[2]; bf16x8_w f; } u;
-  u.v[0] = *reinterpret_cast<const unsigned long long*>(p);
-  u.v[1] = *reinterpret_cast<const unsigned long long*>(p + 4);
-  return u.f;
+  // rows are 16B-aligned (stride 144B) and swizzle offsets are 16B
+  // multiples: a true b128 read
+  return *reinterpret_cast<const bf16x8_w*>(p);
 }
 
 // Pack 4 bf16 (one per source register) into a 64-bit value for the
@@ -82,7 +83,8 @@ __global__ __launch_bounds__(256)
 void wgrad_kernel(const __hip_bfloat16* __restrict__ X,
                   const __hip_bfloat16* __restrict__ DY,
                   void* __restrict__ OUT,  // bf16 [KD,N] or fp32 [KD,N]
-                  long M, long KD, long N, long m_per_split) {
+                  long M, long KD, long N, long m_per_split,
+                  int tx, int ty, int S) {
   __shared__ __hip_bfloat16 lx[BM][LDS_STRIDE];   // x^T image [kd][m]
   __shared__ __hip_bfloat16 ly[BN][LDS_STRIDE];   // dy^T image [n][m]
 
@@ -92,9 +94,28 @@ void wgrad_kernel(const __hip_bfloat16* __restrict__ X,
   const int hi = lane >> 5;
   const int lq = lane & 31;
 
-  const long kd0 = (long)blockIdx.x * BM;
-  const long n0 = (long)blockIdx.y * BN;
-  const long m_lo = (long)blockIdx.z * m_per_split;
+  // XCD-pinned kd-tiles (1-D grid; observed: block b runs on XCD b%8,
+  // MI355X_MICROARCH §Workgroup dispatch). All ty*S blocks of one
+  // kd-tile land on ONE XCD, so the x-slab each m-step needs is read
+  // from L3/HBM once and re-served from that XCD's L2 — the 128x128
+  // baseline was L3-bound on x re-reads (bench: 465-600 TF/s).
+  long kdt, nt, z;
+  const long fid = blockIdx.x;
+  if ((tx & 7) == 0) {
+    const long rest = fid >> 3;
+    const long per = (long)ty * S;
+    kdt = (fid & 7) + 8 * (rest / per);
+    const long r2 = rest % per;
+    nt = r2 % ty;
+    z = r2 / ty;
+  } else {
+    kdt = fid % tx;
+    nt = (fid / tx) % ty;
+    z = fid / ((long)tx * ty);
+  }
+  const long kd0 = kdt * BM;
+  const long n0 = nt * BN;
+  const long m_lo = z * m_per_split;
   const long m_hi = min(m_lo + m_per_split, M);
 
   // wave quadrant: wid -> (kd half, n half), each 64x64
@@ -114,13 +135,27 @@ void wgrad_kernel(const __hip_bfloat16* __restrict__ X,
 
   bf16x8_w xr[4], yr[4];  // staged rows (m, m+1, m+2, m+3)
 
+  // interior tiles take the unconditional path: a per-lane
+  // "in-range ? load : zero" select makes hipcc branch around each load and
+  // wait vmcnt(0) per element (guide §5.5 trap 4c) — the loads must be
+  // straight-line so they pipeline under the MFMA work
+  const bool cols_full = (kd0 + BM <= KD) && (n0 + BN <= N);
   auto issue_loads = [&](long m_base) {
+    const long kd = kd0 + s_col;
+    const long n = n0 + s_col;
+    if (cols_full && m_base + BK <= m_hi) {      // wave-uniform branch
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long m = m_base + s_m + r;
+        xr[r] = *reinterpret_cast<const bf16x8_w*>(X + m * KD + kd);
+        yr[r] = *reinterpret_cast<const bf16x8_w*>(DY + m * N + n);
+      }
+      return;
+    }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const long m = m_base + s_m + r;
       const bool mok = m < m_hi;
-      const long kd = kd0 + s_col;
-      const long n = n0 + s_col;
       xr[r] = (mok && kd + 8 <= KD)
           ? *reinterpret_cast<const bf16x8_w*>(X + m * KD + kd)
           : (bf16x8_w)(__bf16)0.f;
@@ -229,19 +264,21 @@ torch::Tensor wgrad_bf16(torch::Tensor x, torch::Tensor dy,
   }
   const long m_per = ((M + S - 1) / S + BK - 1) / BK * BK;
   auto stream = at::cuda::getCurrentHIPStream();
-  dim3 grid((unsigned)tx, (unsigned)ty, (unsigned)S);
+  dim3 grid((unsigned)(tx * ty * S));
   if (S == 1) {
     auto out = torch::empty({KD, N}, xc.options());
     hipLaunchKernelGGL((wgrad_kernel<false>), grid, dim3(256), 0, stream,
                        reinterpret_cast<const __hip_bfloat16*>(xc.data_ptr()),
                        reinterpret_cast<const __hip_bfloat16*>(dyc.data_ptr()),
-                       out.data_ptr(), M, KD, N, m_per);
+                       out.data_ptr(), M, KD, N, m_per,
+                       (int)tx, (int)ty, (int)S);
     return out;
   }
   auto ws = torch::zeros({KD, N}, xc.options().dtype(torch::kFloat32));
   hipLaunchKernelGGL((wgrad_kernel<true>), grid, dim3(256), 0, stream,
                      reinterpret_cast<const __hip_bfloat16*>(xc.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(dyc.data_ptr()),
-                     ws.data_ptr(), M, KD, N, m_per);
+                     ws.data_ptr(), M, KD, N, m_per,
+                     (int)tx, (int)ty, (int)S);
   return ws.to(torch::kBFloat16);
 }
