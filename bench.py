@@ -134,10 +134,14 @@ def main():
         try:
             step_fn = gg.make_graphed_update(batch,
                                              warmup=max(2, args.warmup))
+            print("# step-graph: captured", file=sys.stderr)
         except Exception as e:
             print(f"# step-graph capture failed ({e}); eager fallback",
                   file=sys.stderr)
             step_fn = None
+    elif on_gpu:
+        print(f"# step-graph: not capturable (can={gg.can_capture()})",
+              file=sys.stderr)
     if step_fn is None:
         step_fn = gg.update
 

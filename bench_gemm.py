@@ -81,7 +81,10 @@ def main():
         x = t(m, k)
         W = Wmk if Wmk is not None else t(k, n)
         b = bias[n]
-        return lambda: torch.addmm(b, x, W)
+        fn = lambda: torch.addmm(b, x, W)
+        if W.stride(1) == 1:
+            fn._ltbias = (x, W, b)
+        return fn
 
     def dgrad(m, n, k):
         dy = t(m, n)
@@ -181,6 +184,15 @@ def main():
             lt_fn = lambda: ext.lt_mm(A, B)
             lt_sec = timeit(lt_fn)
             lt_txt = (f"{lt_sec*1e3:8.3f}", f"{flops/lt_sec/1e12:7.0f}")
+        if ext is not None and hasattr(fn, "_ltbias"):
+            xb, Wb, bb = fn._ltbias
+            ref = fn().float()
+            got = ext.lt_linear_act(xb, Wb, bb, False).float()
+            err = (got - ref).abs().max().item()
+            ok = err <= 0.02 * max(ref.abs().max().item(), 1.0)
+            bsec = timeit(lambda: ext.lt_linear_act(xb, Wb, bb, False))
+            print(f"{'  ^ lt-bias fwd':24s} {'':5s} {bsec*1e3:8.3f} "
+                  f"{flops/bsec/1e12:7.0f}   {'OK' if ok else 'FAIL'}")
         if ext is not None and hasattr(fn, "_wgrad"):
             xw, dyw = fn._wgrad
             ref = fn().float()
