@@ -158,13 +158,17 @@ class _LinearFn(torch.autograd.Function):
         x2, W = ctx.saved_tensors
         dyc = dy.contiguous()
         dx = torch.mm(dyc, W.t())
+        db = None
         if x2.dtype == torch.bfloat16 and dyc.dtype == torch.bfloat16 \
                 and _use_hand_wgrad(x2.shape[1], dyc.shape[1], x2.shape[0]):
-            dW = hip_ext(required=True).wgrad_bf16(x2, dyc, 0)
+            # db rides the wgrad kernel's dy reads: zero extra traffic
+            dW, dbf = hip_ext(required=True).wgrad_bias_bf16(
+                x2, dyc, 0, ctx.has_bias)
+            if ctx.has_bias:
+                db = dbf.to(dyc.dtype)
         else:
             dW = torch.mm(x2.t(), dyc)
-        db = None
-        if ctx.has_bias:
+        if ctx.has_bias and db is None:
             if "MAMD_CUSTOM_LINEAR" in os.environ:
                 db = hip_ext(required=True).col_sum(dyc).to(dyc.dtype)
             else:

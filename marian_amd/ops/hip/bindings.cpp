@@ -36,6 +36,8 @@ std::vector<torch::Tensor> lt_dwt_bgrad(torch::Tensor x, torch::Tensor dy,
                                         bool want_db);
 torch::Tensor lt_mm(torch::Tensor A, torch::Tensor B);
 torch::Tensor wgrad_bf16(torch::Tensor x, torch::Tensor dy, int64_t splits);
+std::vector<torch::Tensor> wgrad_bias_bf16(torch::Tensor x, torch::Tensor dy,
+                                           int64_t splits, bool want_db);
 std::vector<torch::Tensor> fp8_quantize(torch::Tensor x,
                                         torch::Tensor scale);
 torch::Tensor col_sum(torch::Tensor x);
@@ -82,6 +84,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_bf16", &wgrad_bf16,
         "hand-MFMA split-K weight-grad GEMM dW = x^T dy",
         py::arg("x"), py::arg("dy"), py::arg("splits") = 0);
+  m.def("wgrad_bias_bf16", &wgrad_bias_bf16,
+        "hand-MFMA wgrad with fused fp32 bias grad",
+        py::arg("x"), py::arg("dy"), py::arg("splits") = 0,
+        py::arg("want_db") = true);
   m.def("lt_linear_act", &lt_linear_act,
         "hipBLASLt GEMM with fused bias(+ReLU) epilogue");
   m.def("lt_dwt_bgrad", &lt_dwt_bgrad,

@@ -83,6 +83,7 @@ __global__ __launch_bounds__(256)
 void wgrad_kernel(const __hip_bfloat16* __restrict__ X,
                   const __hip_bfloat16* __restrict__ DY,
                   void* __restrict__ OUT,  // bf16 [KD,N] or fp32 [KD,N]
+                  float* __restrict__ DB,  // fp32 [N] bias grad, or null
                   long M, long KD, long N, long m_per_split,
                   int tx, int ty, int S) {
   __shared__ __hip_bfloat16 lx[BM][LDS_STRIDE];   // x^T image [kd][m]
@@ -134,6 +135,12 @@ void wgrad_kernel(const __hip_bfloat16* __restrict__ X,
     for (int b = 0; b < 2; ++b) acc[a][b] = (f32x16_w)0.f;
 
   bf16x8_w xr[4], yr[4];  // staged rows (m, m+1, m+2, m+3)
+  // fused bias gradient: kd-tile-0 blocks see every (n, m) exactly once;
+  // each thread accumulates its 8 dy columns across the m-slab — the dy
+  // values are already in registers for staging, so db costs no traffic
+  float dbacc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) dbacc[j] = 0.f;
 
   // interior tiles take the unconditional path: a per-lane
   // "in-range ? load : zero" select makes hipcc branch around each load and
@@ -188,13 +195,22 @@ void wgrad_kernel(const __hip_bfloat16* __restrict__ X,
     }
   };
 
+  const bool want_db = (DB != nullptr) && (kdt == 0);
   issue_loads(m_lo);
   write_tiles();
+  if (want_db) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dbacc[j] += (float)yr[r][j];
+  }
   __syncthreads();
 
   for (long m0 = m_lo; m0 < m_hi; m0 += BK) {
     const bool last = (m0 + BK >= m_hi);
-    if (!last) issue_loads(m0 + BK);  // in flight under this tile's MFMA
+    if (!last) {
+      issue_loads(m0 + BK);  // in flight under this tile's MFMA
+    }
 
 #pragma unroll
     for (int ks = 0; ks < BK / 16; ++ks) {
@@ -213,7 +229,20 @@ void wgrad_kernel(const __hip_bfloat16* __restrict__ X,
     if (!last) {
       __syncthreads();       // MFMA done reading the current images
       write_tiles();         // stage the prefetched tile
+      if (want_db) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+#pragma unroll
+          for (int j = 0; j < 8; ++j) dbacc[j] += (float)yr[r][j];
+      }
       __syncthreads();
+    }
+  }
+  if (want_db) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const long n = n0 + s_col + j;
+      if (n < N) atomicAdd(&DB[n], dbacc[j]);
     }
   }
 
@@ -244,9 +273,10 @@ void wgrad_kernel(const __hip_bfloat16* __restrict__ X,
 
 }  // namespace
 
-// dW [KD, N] bf16 = x^T @ dy. `splits` 0 = auto.
-torch::Tensor wgrad_bf16(torch::Tensor x, torch::Tensor dy,
-                         int64_t splits) {
+// dW [KD, N] bf16 = x^T @ dy (+ fp32 db = colsum(dy) fused, free since
+// dy is already in registers for staging). `splits` 0 = auto.
+std::vector<torch::Tensor> wgrad_bias_bf16(torch::Tensor x, torch::Tensor dy,
+                                           int64_t splits, bool want_db) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && dy.dim() == 2 &&
               x.scalar_type() == torch::kBFloat16 &&
               dy.scalar_type() == torch::kBFloat16 &&
@@ -265,20 +295,32 @@ torch::Tensor wgrad_bf16(torch::Tensor x, torch::Tensor dy,
   const long m_per = ((M + S - 1) / S + BK - 1) / BK * BK;
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid((unsigned)(tx * ty * S));
+  torch::Tensor db;
+  float* dbp = nullptr;
+  if (want_db) {
+    db = torch::zeros({N}, xc.options().dtype(torch::kFloat32));
+    dbp = db.data_ptr<float>();
+  } else {
+    db = torch::empty({0}, xc.options());
+  }
   if (S == 1) {
     auto out = torch::empty({KD, N}, xc.options());
     hipLaunchKernelGGL((wgrad_kernel<false>), grid, dim3(256), 0, stream,
                        reinterpret_cast<const __hip_bfloat16*>(xc.data_ptr()),
                        reinterpret_cast<const __hip_bfloat16*>(dyc.data_ptr()),
-                       out.data_ptr(), M, KD, N, m_per,
+                       out.data_ptr(), dbp, M, KD, N, m_per,
                        (int)tx, (int)ty, (int)S);
-    return out;
+    return {out, db};
   }
   auto ws = torch::zeros({KD, N}, xc.options().dtype(torch::kFloat32));
   hipLaunchKernelGGL((wgrad_kernel<true>), grid, dim3(256), 0, stream,
                      reinterpret_cast<const __hip_bfloat16*>(xc.data_ptr()),
                      reinterpret_cast<const __hip_bfloat16*>(dyc.data_ptr()),
-                     ws.data_ptr(), M, KD, N, m_per,
+                     ws.data_ptr(), dbp, M, KD, N, m_per,
                      (int)tx, (int)ty, (int)S);
-  return ws.to(torch::kBFloat16);
+  return {ws.to(torch::kBFloat16), db};
+}
+
+torch::Tensor wgrad_bf16(torch::Tensor x, torch::Tensor dy, int64_t splits) {
+  return wgrad_bias_bf16(x, dy, splits, false)[0];
 }
