@@ -451,10 +451,15 @@ class SyncGraphGroup:
 
         def run(batch) -> LossValue:
             src, trg = batch.source, batch.target
-            static.source.indices.copy_(src.indices, non_blocking=True)
-            static.source.mask.copy_(src.mask, non_blocking=True)
-            static.target.indices.copy_(trg.indices, non_blocking=True)
-            static.target.mask.copy_(trg.mask, non_blocking=True)
+            # identical tensors (e.g. replaying the example batch) must
+            # NOT self-copy: an overlapping same-pointer DtoD memcpy is
+            # undefined and faults intermittently on ROCm
+            if static.source.indices is not src.indices:
+                static.source.indices.copy_(src.indices, non_blocking=True)
+                static.source.mask.copy_(src.mask, non_blocking=True)
+            if static.target.indices is not trg.indices:
+                static.target.indices.copy_(trg.indices, non_blocking=True)
+                static.target.mask.copy_(trg.mask, non_blocking=True)
             n_labels = trg.n_tokens if getattr(trg, "n_tokens", None) \
                 is not None else int(trg.mask.sum().item())
             self._g_counts[0] = float(n_labels)
