@@ -97,85 +97,6 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
 #pragma unroll
   for (int i = 0; i < MAXC; ++i) dg[i] = db[i] = 0.f;
 
-  if constexpr (MAXC <= 16) {
-    // 2-row interleave: each wave carries TWO rows through the pass so
-    // the four wave_sum shuffle trees overlap (the 1-row variant is
-    // latency-bound on the reduce chain, 3.3 TB/s; loads/stores stay
-    // register-cached as before — the re-read variant ran at 1.3 TB/s)
-    for (int row = wave_id; row < rows; row += 2 * n_waves) {
-      const int rowB = row + n_waves;
-      const bool has2 = rowB < rows;
-      const T* xr0 = x + (size_t)row * d;
-      const T* dyr0 = dy + (size_t)row * d;
-      T* dxr0 = dx + (size_t)row * d;
-      const T* xr1 = has2 ? x + (size_t)rowB * d : xr0;
-      const T* dyr1 = has2 ? dy + (size_t)rowB * d : dyr0;
-      const float mu0 = mean[row], rs0 = rstd[row];
-      const float mu1 = has2 ? mean[rowB] : 0.f;
-      const float rs1 = has2 ? rstd[rowB] : 0.f;
-      float h1a = 0.f, h2a = 0.f, h1b = 0.f, h2b = 0.f;
-      float xh0[MAXC], gg0[MAXC], xh1[MAXC], gg1[MAXC];
-#pragma unroll
-      for (int i = 0; i < MAXC / VEC; ++i) {
-        const int c = lane * VEC + i * WAVE * VEC;
-        if (c < d) {
-          float x4[VEC], g4[VEC], gm4[VEC], x4b[VEC], g4b[VEC];
-          load4_f32(xr0 + c, x4);
-          load4_f32(dyr0 + c, g4);
-          load4_f32(gamma + c, gm4);
-          if (has2) {
-            load4_f32(xr1 + c, x4b);
-            load4_f32(dyr1 + c, g4b);
-          }
-#pragma unroll
-          for (int j = 0; j < VEC; ++j) {
-            const float xhv = (x4[j] - mu0) * rs0;
-            const float ggv = g4[j] * gm4[j];
-            xh0[i * VEC + j] = xhv;
-            gg0[i * VEC + j] = ggv;
-            h1a += ggv;
-            h2a += ggv * xhv;
-            dg[i * VEC + j] += g4[j] * xhv;
-            db[i * VEC + j] += g4[j];
-            if (has2) {
-              const float xhw = (x4b[j] - mu1) * rs1;
-              const float ggw = g4b[j] * gm4[j];
-              xh1[i * VEC + j] = xhw;
-              gg1[i * VEC + j] = ggw;
-              h1b += ggw;
-              h2b += ggw * xhw;
-              dg[i * VEC + j] += g4b[j] * xhw;
-              db[i * VEC + j] += g4b[j];
-            }
-          }
-        }
-      }
-      h1a = wave_sum(h1a) / d;
-      h2a = wave_sum(h2a) / d;
-      if (has2) {
-        h1b = wave_sum(h1b) / d;
-        h2b = wave_sum(h2b) / d;
-      }
-#pragma unroll
-      for (int i = 0; i < MAXC / VEC; ++i) {
-        const int c = lane * VEC + i * WAVE * VEC;
-        if (c < d) {
-          float o4[VEC];
-#pragma unroll
-          for (int j = 0; j < VEC; ++j)
-            o4[j] = rs0 * (gg0[i * VEC + j] - h1a - xh0[i * VEC + j] * h2a);
-          store4_f32(dxr0 + c, o4);
-          if (has2) {
-#pragma unroll
-            for (int j = 0; j < VEC; ++j)
-              o4[j] = rs1 * (gg1[i * VEC + j] - h1b
-                             - xh1[i * VEC + j] * h2b);
-            store4_f32(dx + (size_t)rowB * d + c, o4);
-          }
-        }
-      }
-    }
-  } else {
   for (int row = wave_id; row < rows; row += n_waves) {
     const T* xr = x + (size_t)row * d;
     const T* dyr = dy + (size_t)row * d;
@@ -183,7 +104,45 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
     const float mu = mean[row], rs = rstd[row];
 
     float h1 = 0.f, h2 = 0.f;
-    {
+    if constexpr (MAXC <= 16) {
+      // register-cache xhat and dy*gamma across the two passes (d <= 1024
+      // per 64-lane wave) - the re-read variant ran at 1.3 TB/s
+      float xh[MAXC], gg[MAXC];
+#pragma unroll
+      for (int i = 0; i < MAXC / VEC; ++i) {
+        const int c = lane * VEC + i * WAVE * VEC;
+        if (c < d) {
+          float x4[VEC], g4[VEC], gm4[VEC];
+          load4_f32(xr + c, x4);
+          load4_f32(dyr + c, g4);
+          load4_f32(gamma + c, gm4);
+#pragma unroll
+          for (int j = 0; j < VEC; ++j) {
+            const float xhv = (x4[j] - mu) * rs;
+            const float ggv = g4[j] * gm4[j];
+            xh[i * VEC + j] = xhv;
+            gg[i * VEC + j] = ggv;
+            h1 += ggv;
+            h2 += ggv * xhv;
+            dg[i * VEC + j] += g4[j] * xhv;
+            db[i * VEC + j] += g4[j];
+          }
+        }
+      }
+      h1 = wave_sum(h1) / d;
+      h2 = wave_sum(h2) / d;
+#pragma unroll
+      for (int i = 0; i < MAXC / VEC; ++i) {
+        const int c = lane * VEC + i * WAVE * VEC;
+        if (c < d) {
+          float o4[VEC];
+#pragma unroll
+          for (int j = 0; j < VEC; ++j)
+            o4[j] = rs * (gg[i * VEC + j] - h1 - xh[i * VEC + j] * h2);
+          store4_f32(dxr + c, o4);
+        }
+      }
+    } else {
       for (int c = lane * VEC, i = 0; c < d; c += WAVE * VEC) {
 #pragma unroll
         for (int j = 0; j < VEC; ++j, ++i) {
@@ -207,7 +166,6 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
         }
       }
     }
-  }
   }
   float* dgrow = dg_scratch + (size_t)wave_id * d;
   float* dbrow = db_scratch + (size_t)wave_id * d;
