@@ -100,21 +100,22 @@ def _quantize_w(W: torch.Tensor):
 
 
 
-# Full-fp8 backward is OPT-IN: numerically validated (grad error <8% vs
-# fp32) but measured -14% end-to-end on config #5 — the wgrad
+# Backward recipes (r2): the default is fp8 DGRAD + bf16 WGRAD — dgrad's
+# [M,N]rm x [N,K]cm layout has fast fp8 solutions, while the wgrad
 # ([K,M] rm x [M,N] cm) fp8 layout hits hipBLASLt fallback kernels and
-# the byte-transposes add passes. Revisit when hipBLASLt gains fast fp8
-# TN solutions.
+# byte-transposes (full-fp8 bwd measured -14% end-to-end in r1; it stays
+# available via MAMD_FP8_BWD=1). MAMD_FP8_DGRAD=0 restores the r1
+# all-bf16 backward.
 _FP8_BWD = os.environ.get("MAMD_FP8_BWD", "0") == "1"
+_FP8_DGRAD = os.environ.get("MAMD_FP8_DGRAD", "1") == "1"
 
 
 class Fp8LinearFn(torch.autograd.Function):
-    """fp8 forward AND backward GEMMs (delayed-scaled e4m3):
+    """fp8 forward (+ fp8 dgrad) GEMMs (delayed-scaled e4m3):
       y  = x8 @ w8_cm                       (fwd)
       dx = dy8_rm @ w8_rm^T                 (dgrad; B = rm weight .t())
-      dW = (x8_cm)^T @ dy8_cm               (wgrad; byte-transposed copies)
-    Only x8 (+scale) is saved — LESS activation memory than the bf16
-    path. db stays a bf16 column reduce."""
+      dW = x^T @ dy  in bf16                (wgrad: fp8 TN layout is slow)
+    db stays a bf16 column reduce."""
 
     @staticmethod
     def forward(ctx, x, W, b):
@@ -136,9 +137,25 @@ class Fp8LinearFn(torch.autograd.Function):
         dyc = dy.contiguous()
         if not _FP8_BWD:
             x, W = ctx.saved_tensors
-            dx = dyc @ W.t()
-            dW = x.t() @ dyc
-            db = dyc.sum(0) if ctx.has_bias else None
+            if _FP8_DGRAD:
+                # dgrad at the fp8 MFMA rate; dy delayed-scaled like the
+                # forward activations (e4m3 saturating cast)
+                _, sw, w8_rm = _quantize_w(W)
+                dy8, sdy = _quantize_x(dyc, ctx.wkey ^ 0x5F5F)
+                dx = torch._scaled_mm(dy8, w8_rm.t(), scale_a=sdy,
+                                      scale_b=sw, out_dtype=torch.bfloat16)
+            else:
+                dx = dyc @ W.t()
+            from . import _use_hand_wgrad, hip_ext
+            if x.dtype == torch.bfloat16 and dyc.dtype == torch.bfloat16 \
+                    and _use_hand_wgrad(x.shape[1], dyc.shape[1],
+                                        x.shape[0]):
+                dW, dbf = hip_ext(required=True).wgrad_bias_bf16(
+                    x, dyc, 0, ctx.has_bias)
+                db = dbf.to(dyc.dtype) if ctx.has_bias else None
+            else:
+                dW = x.t() @ dyc
+                db = dyc.sum(0) if ctx.has_bias else None
             return dx, dW, db
         x8, sx, W = ctx.saved_tensors
         w8_cm, sw, w8_rm = _quantize_w(W)
