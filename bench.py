@@ -71,6 +71,8 @@ def build_options(args, world):
     ]
     opts = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
     opts.set("dim-vocabs", [args.vocab, args.vocab])
+    if args.comm_bucket_mb:
+        opts.set("amd-comm-bucket-mb", args.comm_bucket_mb)
     return opts
 
 
@@ -84,6 +86,9 @@ def main():
                    help="sentences per GPU (0 = auto)")
     p.add_argument("--seq-len", type=int, default=64)
     p.add_argument("--vocab", type=int, default=32000)
+    p.add_argument("--comm-bucket-mb", type=int, default=0,
+                   help="gradient reduce-scatter bucket size (MB); "
+                        "0 = default 96 (xGMI per-link sweep knob)")
     p.add_argument("--dtype", default="bf16",
                    choices=["bf16", "fp16", "fp32", "float32", "bfloat16",
                             "fp8", "float8"])
