@@ -41,8 +41,47 @@ class Rescorer:
         picked = picked * batch.target.mask
         return picked.sum(dim=1).cpu(), picked.cpu(), batch.target.mask.cpu()
 
+    def run_nbest(self, out=sys.stdout):
+        """--n-best: train-sets = [src, nbest-file]. Each hypothesis line
+        `id ||| text ||| feats ||| score` is scored against src[id] and
+        re-emitted with the model score reinserted into the feature field
+        (ref: rescorer.h + score_collector.cpp OutputRescore)."""
+        opts = self.opt
+        src_path, nbest_path = opts.list("train-sets")[:2]
+        with open(src_path) as f:
+            src_lines = [ln.rstrip("\n") for ln in f]
+        entries = []
+        with open(nbest_path) as f:
+            for ln in f:
+                parts = ln.rstrip("\n").split(" ||| ")
+                entries.append((int(parts[0]), parts))
+        from ..data.batch import CorpusBatch, SubBatch
+        mb = int(opts.get("mini-batch", 64) or 64)
+        normalize = bool(opts.get("normalize", False))
+        feat = opts.get("n-best-feature", "Score") or "Score"
+        for lo in range(0, len(entries), mb):
+            chunk = entries[lo:lo + mb]
+            src_ids = [self.vocabs[0].encode(src_lines[sid])
+                       for sid, _ in chunk]
+            trg_ids = [self.vocabs[-1].encode(p[1]) for _, p in chunk]
+            batch = CorpusBatch([SubBatch.from_sentences(src_ids),
+                                 SubBatch.from_sentences(trg_ids)])
+            sums, _, mask = self.score_batch(batch)
+            for i, (sid, parts) in enumerate(chunk):
+                score = float(sums[i])
+                if normalize:
+                    score /= max(1, int(mask[i].sum()))
+                fields = list(parts)
+                while len(fields) < 4:
+                    fields.append("")
+                fields[2] = f"{feat}= {score:.4f}"
+                fields[3] = f"{score:.4f}"
+                print(" ||| ".join(fields[:4]), file=out)
+
     def run(self, out=sys.stdout):
         opts = self.opt
+        if opts.get("n-best", False):
+            return self.run_nbest(out)
         train_sets = opts.list("train-sets")
         corpus = Corpus(train_sets, self.vocabs,
                         max_length=int(opts.get("max-length", 1000)),

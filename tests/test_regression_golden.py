@@ -134,3 +134,213 @@ def test_nbest_format_golden(tmp_path):
         "line0_fields": len(lines[0].split(" ||| ")),
         "top_score": scores[0],
     }, tol=5e-3)
+
+
+# ---------------------------------------------------------------------------
+# Round-2 expansion toward MRT breadth: one shared
+# fixed-seed trained model + per-variant training curves; every scenario
+# locks numbers/tokens against committed goldens.
+# ---------------------------------------------------------------------------
+
+@pytest.fixture(scope="module")
+def golden_model(tmp_path_factory):
+    tr, d = _tiny_train(tmp_path_factory.mktemp("gm"))
+    tr.model.eval()
+    return tr, d
+
+
+def _bs(tr, **kw):
+    from marian_amd.translator.beam_search import BeamSearch, ModelScorer
+    args = dict(beam_size=4, normalize=0.6, max_length=12, n_best=1)
+    args.update(kw)
+    return BeamSearch([ModelScorer(tr.model, torch.device("cpu"))], **args)
+
+
+def _mb(sents):
+    from tests.test_decode import make_batch
+    return make_batch(sents)
+
+
+SENTS = [[5, 6, 7, 0], [9, 10, 11, 12, 0]]
+
+
+def test_beam1_greedy_golden(golden_model):
+    tr, d = golden_model
+    res = _bs(tr, beam_size=1).search(_mb(SENTS), torch.device("cpu"))
+    check_golden("decode_beam1", {
+        "tokens_0": res[0].best.tokens, "tokens_1": res[1].best.tokens,
+        "score_0": res[0].best.normalized}, tol=5e-3)
+
+
+def test_beam8_no_lennorm_golden(golden_model):
+    tr, d = golden_model
+    res = _bs(tr, beam_size=8, normalize=0.0).search(
+        _mb(SENTS), torch.device("cpu"))
+    check_golden("decode_beam8_nonorm", {
+        "tokens_0": res[0].best.tokens, "tokens_1": res[1].best.tokens,
+        "score_0": res[0].best.normalized}, tol=5e-3)
+
+
+def test_word_penalty_golden(golden_model):
+    tr, d = golden_model
+    res = _bs(tr, word_penalty=0.4).search(_mb(SENTS), torch.device("cpu"))
+    resnp = _bs(tr).search(_mb(SENTS), torch.device("cpu"))
+    # penalty reduces normalized score for same-length output
+    check_golden("decode_word_penalty", {
+        "tokens_0": res[0].best.tokens,
+        "score_delta": resnp[0].best.normalized - res[0].best.normalized,
+    }, tol=5e-3)
+
+
+def test_ensemble_golden(golden_model):
+    """Two copies of the model with weights == single model scores."""
+    tr, d = golden_model
+    from marian_amd.translator.beam_search import BeamSearch, ModelScorer
+    sc = ModelScorer(tr.model, torch.device("cpu"))
+    ens = BeamSearch([sc, sc], weights=[0.5, 0.5], beam_size=4,
+                     normalize=0.6, max_length=12)
+    single = _bs(tr).search(_mb(SENTS), torch.device("cpu"))
+    both = ens.search(_mb(SENTS), torch.device("cpu"))
+    assert both[0].best.tokens == single[0].best.tokens
+    check_golden("decode_ensemble", {
+        "tokens_0": both[0].best.tokens,
+        "score_0": both[0].best.normalized}, tol=5e-3)
+
+
+def test_force_decode_golden(golden_model):
+    """--force-decode through the CLI: forced prefix respected, output
+    line locked (ref: the reference's force-decoding on TSV input)."""
+    tr, d = golden_model
+    inp = d / "force_in.txt"
+    with open(d / "c.src") as f:
+        src_line = f.readline().strip()
+    inp.write_text(f"{src_line}\tw5 w6\n")
+    out = d / "force_out.txt"
+    from marian_amd.translator.translate import main_decoder
+    main_decoder(["--models", str(d / "model.npz"),
+                  "--vocabs", str(d / "v.yml"), str(d / "v.yml"),
+                  "--input", str(inp), "--output", str(out),
+                  "--beam-size", "2", "--cpu-threads", "1",
+                  "--force-decode", "--max-length", "20", "--quiet"])
+    line = out.read_text().splitlines()[0]
+    assert line.startswith("w5 w6"), line
+    check_golden("decode_force", {"line": line})
+
+
+def test_output_sampling_topk_golden(golden_model):
+    tr, d = golden_model
+    torch.manual_seed(1234)
+    res = _bs(tr, beam_size=1,
+              output_sampling=["topk", 5]).search(_mb(SENTS),
+                                                  torch.device("cpu"))
+    check_golden("decode_sampling_topk", {
+        "tokens_0": res[0].best.tokens, "tokens_1": res[1].best.tokens,
+    }, tol=5e-3)
+
+
+def test_shortlist_decode_golden(golden_model):
+    tr, d = golden_model
+    from marian_amd.data.shortlist import LexicalShortlistGenerator
+    lex = d / "lex.dic"
+    lex.write_text("".join(f"w{i}\tw{i}\t1.0\n" for i in range(3, 30)))
+    gen = LexicalShortlistGenerator(str(lex), tr.vocabs[0], tr.vocabs[-1],
+                                    first_num=12, best_num=6)
+    res = _bs(tr, shortlist_generator=gen).search(_mb(SENTS),
+                                                  torch.device("cpu"))
+    check_golden("decode_shortlist", {
+        "tokens_0": res[0].best.tokens,
+        "score_0": res[0].best.normalized}, tol=5e-3)
+
+
+def test_scorer_nbest_reinsertion_golden(golden_model):
+    tr, d = golden_model
+    from marian_amd.common.config_parser import ConfigParser, MODE_SCORING
+    from marian_amd.translator.rescorer import Rescorer
+    import io
+    nbest = d / "hyps.nbest"
+    with open(d / "c.src") as f:
+        first_src = f.readline().strip()
+    with open(nbest, "w") as f:
+        f.write(f"0 ||| {first_src} ||| F0= -1.0 ||| -1.0\n")
+        f.write("0 ||| w3 w4 ||| F0= -2.0 ||| -2.0\n")
+    opts = ConfigParser(MODE_SCORING).parse_options(
+        ["--models", str(d / "model.npz"),
+         "--vocabs", str(d / "v.yml"), str(d / "v.yml"),
+         "--train-sets", str(d / "c.src"), str(nbest),
+         "--n-best", "--cpu-threads", "1", "--quiet"], validate=False)
+    buf = io.StringIO()
+    Rescorer(opts).run(out=buf)
+    lines = buf.getvalue().splitlines()
+    assert len(lines) == 2 and all(" ||| " in ln for ln in lines)
+    scores = [float(ln.rsplit(" ||| ", 1)[-1]) for ln in lines]
+    assert all("Score= " in ln.split(" ||| ")[2] for ln in lines)
+    check_golden("scorer_nbest_reinsert", {
+        "n_lines": len(lines), "score_0": scores[0]}, tol=5e-3)
+
+
+def test_scorer_word_scores_golden(golden_model):
+    tr, d = golden_model
+    from marian_amd.common.config_parser import ConfigParser, MODE_SCORING
+    from marian_amd.translator.rescorer import Rescorer
+    import io
+    opts = ConfigParser(MODE_SCORING).parse_options(
+        ["--models", str(d / "model.npz"),
+         "--vocabs", str(d / "v.yml"), str(d / "v.yml"),
+         "--train-sets", str(d / "c.src"), str(d / "c.src"),
+         "--word-scores", "--cpu-threads", "1", "--quiet"], validate=False)
+    buf = io.StringIO()
+    Rescorer(opts).run(out=buf)
+    first = buf.getvalue().splitlines()[0]
+    total, words = first.split(" ||| ")
+    wlist = [float(w) for w in words.replace("WordScores= ", "").split()]
+    assert abs(sum(wlist) - float(total)) < 1e-3 * max(1, abs(float(total)))
+    check_golden("scorer_word_scores", {
+        "total_0": float(total), "n_words_0": len(wlist)}, tol=5e-3)
+
+
+def test_prenorm_training_golden(tmp_path):
+    tr, d = _tiny_train(tmp_path, extra=[
+        "--transformer-preprocess", "n", "--transformer-postprocess", "da",
+        "--transformer-postprocess-top", "n", "--after-batches", "25"])
+    from marian_amd.data.synthetic import synthetic_batch
+    probe = synthetic_batch(8, 8, 8, len(tr.vocabs[0]), seed=77)
+    probe.sub[-1].indices.copy_(probe.sub[0].indices)
+    with torch.no_grad():
+        lv = tr.graph_group.criterion(tr.model(probe), probe)
+    check_golden("training_prenorm_cost", {
+        "probe_ce_per_word": float(lv.ce_sum) / lv.label_count}, tol=5e-3)
+
+
+def test_s2s_training_golden(tmp_path):
+    tr, d = _tiny_train(tmp_path, extra=[
+        "--type", "s2s", "--dim-rnn", "48", "--after-batches", "25"])
+    from marian_amd.data.synthetic import synthetic_batch
+    probe = synthetic_batch(8, 8, 8, len(tr.vocabs[0]), seed=78)
+    probe.sub[-1].indices.copy_(probe.sub[0].indices)
+    with torch.no_grad():
+        lv = tr.graph_group.criterion(tr.model(probe), probe)
+    check_golden("training_s2s_cost", {
+        "probe_ce_per_word": float(lv.ce_sum) / lv.label_count}, tol=5e-3)
+
+
+def test_server_roundtrip_golden(golden_model):
+    tr, d = golden_model
+    from starlette.testclient import TestClient
+    from marian_amd.common.config_parser import ConfigParser, MODE_SERVER
+    from marian_amd.translator.server import build_app
+    from marian_amd.translator.translate import TranslateService
+    opts = ConfigParser(MODE_SERVER).parse_options(
+        ["--models", str(d / "model.npz"),
+         "--vocabs", str(d / "v.yml"), str(d / "v.yml"),
+         "--beam-size", "2", "--cpu-threads", "1", "--quiet"],
+        validate=False)
+    app = build_app(TranslateService(opts))
+    client = TestClient(app)
+    with open(d / "c.src") as f:
+        line = f.readline().strip()
+    r = client.post("/translate", json={"text": line})
+    assert r.status_code == 200
+    out = r.json()
+    key = "translation" if "translation" in out else list(out)[0]
+    check_golden("server_roundtrip", {"n_chars": len(str(out[key]))},
+                 tol=5e-3)
