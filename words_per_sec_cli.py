@@ -1,0 +1,59 @@
+"""Whole-job words/s through the REAL marian-train CLI:
+text corpus on disk -> vocab -> BatchGenerator (maxi-batch sort, ragged
+batches, H2D) -> training with scheduler logging. Reports the Scheduler's
+own words/s next to bench.py's synthetic step rate.
+
+Run (GPU box): python words_per_sec_cli.py
+"""
+import os
+import re
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def main():
+    import torch
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    d = os.path.abspath("gpurun_out/wps")
+    os.makedirs(d, exist_ok=True)
+    src = os.path.join(d, "c.src")
+    trg = os.path.join(d, "c.trg")
+    if not os.path.exists(src):
+        # ~60k ragged sentence pairs, T in [8, 64], 32k-word vocab surface
+        write_synthetic_corpus(src, trg, 60000, 31990, 8, 64, seed=5)
+    log = os.path.join(d, "train.log")
+    if os.path.exists(log):
+        os.remove(log)
+    from marian_amd.training.train import main_train
+    main_train([
+        "--train-sets", src, trg,
+        "--vocabs", os.path.join(d, "v1.yml"), os.path.join(d, "v2.yml"),
+        "--model", os.path.join(d, "model.npz"),
+        "--task", "transformer-big",
+        "--transformer-dropout", "0.0", "--transformer-dropout-attention",
+        "0.0", "--transformer-dropout-ffn", "0.0",
+        "--mini-batch-words", "65536", "--mini-batch", "3000",
+        "--maxi-batch", "100", "--maxi-batch-sort", "src",
+        "--precision", "bfloat16", "float32",
+        "--learn-rate", "0.0002", "--lr-warmup", "8000",
+        "--after-batches", "120", "--disp-freq", "20u",
+        "--save-freq", "100000u", "--log", log,
+        "--seed", "3", "--overwrite", "--shuffle-in-ram",
+    ])
+    rates = []
+    with open(log) as f:
+        for ln in f:
+            m = re.search(r": ([0-9.]+) words/s", ln)
+            if m:
+                rates.append(float(m.group(1)))
+    # skip the first disp window (warmup/compile); report the rest
+    steady = rates[1:] if len(rates) > 1 else rates
+    print(f"CLI words/s per disp window: {rates}")
+    if steady:
+        print(f"CLI steady words/s (mean of {len(steady)}): "
+              f"{sum(steady)/len(steady):,.0f}")
+
+
+if __name__ == "__main__":
+    main()
