@@ -17,7 +17,7 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--workdir", default="gpurun_out/convergence")
+    ap.add_argument("--workdir", default="/tmp/convergence")
     ap.add_argument("--target", type=float, default=50.0)
     ap.add_argument("--after-batches", type=int, default=1500)
     args = ap.parse_args()
@@ -78,6 +78,9 @@ def main():
                     best = max(best, val)
                 except (IndexError, ValueError):
                     pass
+    with open(os.path.join(d, "valid.log")) as f:
+        for ln in f.read().splitlines()[-8:]:
+            print("VALID:", ln)
     print(f"CONVERGENCE: best bleu-detok = {best:.2f} "
           f"(pre-registered target {args.target}) -> "
           f"{'PASS' if best >= args.target else 'FAIL'}")

@@ -15,7 +15,7 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 def main():
     import torch
     from marian_amd.data.synthetic import write_synthetic_corpus
-    d = os.path.abspath("gpurun_out/wps")
+    d = os.environ.get("WPS_DIR", "/tmp/wps")
     os.makedirs(d, exist_ok=True)
     src = os.path.join(d, "c.src")
     trg = os.path.join(d, "c.trg")
