@@ -45,6 +45,7 @@ def main():
         seq_len = args.src_len
         vocab = args.vocab
         dtype = "bf16" if on_gpu else "fp32"
+        comm_bucket_mb = 0
     opts = build_options(A, 1)
     model = create_model(opts).to(device)
     if on_gpu:

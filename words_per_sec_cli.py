@@ -40,6 +40,7 @@ def main():
         "--after-batches", "120", "--disp-freq", "20u",
         "--save-freq", "100000u", "--log", log,
         "--seed", "3", "--overwrite", "--shuffle-in-ram",
+        "--workspace", "250000",   # 250 GB: let mini-batch-fit use the HBM
     ])
     rates = []
     with open(log) as f:
