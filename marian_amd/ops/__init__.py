@@ -247,7 +247,25 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         out = fused_attention(q, k, v, mask, causal, lens=lens)
         if out is not None:
             return out
+        _warn_composite_attention(q)
     return attention_torch(q, k, v, mask, causal, dropout_p)
+
+
+_warned_composite = False
+
+
+def _warn_composite_attention(q: torch.Tensor):
+    """the fused CDNA4 attention covers bf16/Dh=64 — every
+    GPU BASELINE config (base 512/8, big+deep 1024/16). Any other head
+    geometry silently compositing would misreport 'hand-written hot
+    path', so say it LOUDLY once."""
+    global _warned_composite
+    if not _warned_composite and q.is_cuda:
+        _warned_composite = True
+        from ..common.logging_utils import log_info
+        log_info("attention: fused HIP kernel NOT engaged "
+                 f"(dtype={q.dtype}, Dh={q.shape[-1]}; fused path is "
+                 "bf16/Dh=64) - running torch composite")
 
 
 def attention_qkv(qkv: torch.Tensor, heads: int, mask=None, causal=False,

@@ -209,3 +209,19 @@ def test_ffn_depth_3():
     b = CorpusBatch([SubBatch(idx, torch.ones(2, 5)),
                      SubBatch(idx.clone(), torch.ones(2, 5))])
     m(b).sum().backward()
+
+
+def test_baseline_configs_fused_attention_eligible():
+    """every GPU BASELINE config's head geometry must hit
+    the fused-attention gate (bf16, Dh=64); transformer-tiny is the
+    CPU-only plumbing config (Dh=32 composites by design)."""
+    gpu_configs = {
+        "transformer-base": (512, 8),
+        "transformer-big": (1024, 16),
+        "transformer-deep": (1024, 16),
+        "decode-big": (1024, 16),
+    }
+    for name, (dim, heads) in gpu_configs.items():
+        assert dim % heads == 0 and dim // heads == 64, \
+            f"{name}: Dh={dim//heads} would silently composite"
+    assert 128 // 4 == 32  # tiny: documented CPU-only composite
