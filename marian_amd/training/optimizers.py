@@ -277,6 +277,8 @@ class ShardedOptimizer:
             self.core.m.zero_()
             self.core.v.zero_()
         self.t = 0
+        if self.t_dev is not None:
+            self.t_dev.zero_()
 
     def grad_shard_for_bucket(self, bi: int) -> torch.Tensor:
         o, s = self.seg[bi]
