@@ -344,3 +344,49 @@ def test_server_roundtrip_golden(golden_model):
     key = "translation" if "translation" in out else list(out)[0]
     check_golden("server_roundtrip", {"n_chars": len(str(out[key]))},
                  tol=5e-3)
+
+
+def test_guided_alignment_training_golden(tmp_path):
+    """Training with --guided-alignment locks its cost curve."""
+    d = tmp_path
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    src = str(d / "c.src")
+    write_synthetic_corpus(src, str(d / "c.trg"), 120, 40, 4, 7, seed=31)
+    # diagonal alignments i-i for the first 4 positions
+    with open(d / "c.aln", "w") as f:
+        for _ in range(120):
+            f.write("0-0 1-1 2-2 3-3\n")
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    argv = ["--train-sets", src, src,
+            "--vocabs", str(d / "v.yml"), str(d / "v.yml"),
+            "--model", str(d / "model.npz"),
+            "--guided-alignment", str(d / "c.aln"),
+            "--guided-alignment-weight", "0.2",
+            "--dim-emb", "48", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "96",
+            "--mini-batch", "20", "--learn-rate", "0.01",
+            "--after-batches", "15", "--seed", "5", "--cpu-threads", "1",
+            "--tied-embeddings-all", "--overwrite", "--quiet",
+            "--shuffle", "none"]
+    opts = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    tr = Train(opts)
+    tr.run()
+    from marian_amd.data.synthetic import synthetic_batch
+    probe = synthetic_batch(6, 6, 6, len(tr.vocabs[0]), seed=44)
+    with torch.no_grad():
+        lv = tr.graph_group.criterion(tr.model(probe), probe)
+    check_golden("training_guided_alignment_cost", {
+        "probe_ce_per_word": float(lv.ce_sum) / lv.label_count}, tol=5e-3)
+
+
+def test_right_left_training_golden(tmp_path):
+    """--right-left target reversal locks its curve (ref: --right-left)."""
+    tr, d = _tiny_train(tmp_path, extra=["--right-left",
+                                         "--after-batches", "20"])
+    from marian_amd.data.synthetic import synthetic_batch
+    probe = synthetic_batch(6, 6, 6, len(tr.vocabs[0]), seed=45)
+    with torch.no_grad():
+        lv = tr.graph_group.criterion(tr.model(probe), probe)
+    check_golden("training_right_left_cost", {
+        "probe_ce_per_word": float(lv.ce_sum) / lv.label_count}, tol=5e-3)
