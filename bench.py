@@ -151,8 +151,8 @@ def main():
                   file=sys.stderr)
             step_fn = None
     elif on_gpu:
-        print(f"# step-graph: not capturable (can={gg.can_capture()})",
-              file=sys.stderr)
+        print("# step-graph: disabled (MAMD_STEP_GRAPH=1 to enable; "
+              f"capturable={gg.can_capture()})", file=sys.stderr)
     if step_fn is None:
         step_fn = gg.update
 

@@ -68,7 +68,7 @@ def _worker(rank, world, port, result_dir, n_steps, optimizer="adam"):
 
 
 @pytest.mark.timeout(300)
-@pytest.mark.parametrize("world,port", [(2, 29531), (4, 29537)])
+@pytest.mark.parametrize("world,port", [(2, 29531), (4, 29537), (8, 29561)])
 def test_nrank_matches_single_process(tmp_path, world, port):
     n_steps = 4
     ctx = mp.get_context("spawn")
@@ -404,3 +404,58 @@ def test_reshard_across_bucket_groupings():
         off += sh2[bi]
     got = np.concatenate(got)
     np.testing.assert_array_equal(got, np.arange(total, dtype=np.float32))
+
+
+def test_reshard_4_to_2():
+    """Resume 4-shard optimizer state at world 2 (N->M, both >1)."""
+    import numpy as np
+    import types
+    from marian_amd.training.optimizers import ParameterStore
+    from marian_amd.training.graph_group import SyncGraphGroup
+
+    def make_model():
+        torch.manual_seed(17)
+        return torch.nn.Sequential(torch.nn.Linear(600, 600),
+                                   torch.nn.Linear(600, 400))
+
+    s4 = ParameterStore(make_model(), world_size=4, bucket_mb=1)
+    total = sum(b.used for b in s4.buckets)
+    used4 = [b.used for b in s4.buckets]
+
+    def layout(used, world):
+        pad = [((max(u, 256 * world) + 256 * world - 1)
+                // (256 * world)) * (256 * world) for u in used]
+        return [p // world for p in pad]
+
+    sh4 = layout(used4, 4)
+    glob = np.arange(total, dtype=np.float32)
+    # build per-rank saved vectors per the world-4 layout
+    oitems = {"bucket_used": np.array(used4, dtype=np.int64)}
+    off = 0
+    padded = []
+    for bi, u in enumerate(used4):
+        g = np.full(sh4[bi] * 4, -1.0, dtype=np.float32)
+        g[:u] = glob[off:off + u]
+        off += u
+        padded.append(g)
+    for r in range(4):
+        parts = [p[r * sh4[bi]:(r + 1) * sh4[bi]]
+                 for bi, p in enumerate(padded)]
+        oitems[f"rank{r}_master_parameters"] = np.concatenate(parts)
+
+    s2 = ParameterStore(make_model(), world_size=2, bucket_mb=1)
+    used2 = [b.used for b in s2.buckets]
+    sh2 = layout(used2, 2)
+    shards = []
+    for rank in (0, 1):
+        fake = types.SimpleNamespace(store=s2, rank=rank, world=2)
+        out = SyncGraphGroup._reshard_optimizer_items(fake, oitems, 4, used4)
+        shards.append(out["master_parameters"])
+    got = []
+    off = 0
+    for bi, u in enumerate(used2):
+        g = np.concatenate([shards[0][off:off + sh2[bi]],
+                            shards[1][off:off + sh2[bi]]])[:u]
+        got.append(g)
+        off += sh2[bi]
+    np.testing.assert_array_equal(np.concatenate(got), glob)
