@@ -343,6 +343,12 @@ class ConfigParser:
         if self.mode in (D, S, E, MODE_SERVER):
             a("precision", str, ["float32"], "Inference precision", "translation", nargs="+")
             a("fp16", "bool", False, "Shortcut for float16 inference", "translation")
+            a("gemm-precision", str, "float32",
+              "CPU decode GEMM precision (float32|int8: fbgemm packed "
+              "int8, ref: the reference's intgemm/--optimize CPU backend)",
+              "translation", choices=["float32", "int8"])
+            a("optimize", "bool", False,
+              "Shortcut for --gemm-precision int8 on CPU", "translation")
         if self.mode == T:
             a("fp16", "bool", False, "Shortcut for mixed-precision training", "training")
 

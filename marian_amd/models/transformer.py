@@ -382,6 +382,11 @@ class TransformerModel(nn.Module):
             b = self.out_b.to(dt)
             if shortlist is not None:
                 b = b.index_select(0, shortlist)
+            if shortlist is None and not x.is_cuda:
+                # stable weight tensor: eligible for the CPU int8 decode
+                # path (packed-weight cache keys on the data pointer)
+                from .. import ops
+                return ops.linear(x, W, b)
             # bias folded into the GEMM epilogue (a separate broadcast add
             # costs a full extra read+write of the [N, V] logits tensor)
             shp = x.shape[:-1] + (W.shape[1],)

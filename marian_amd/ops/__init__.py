@@ -183,6 +183,13 @@ def linear(x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor] = None
     from .fp8 import fp8_enabled, fp8_linear
     if fp8_enabled() and x.is_cuda and x.dtype == torch.bfloat16:
         return fp8_linear(x, W.to(x.dtype), b.to(x.dtype) if b is not None else None)
+    if not x.is_cuda:
+        from .cpu_int8 import int8_enabled, int8_linear
+        if int8_enabled():
+            y = int8_linear(x, W.float(),
+                            b.float() if b is not None else None)
+            if y is not None:
+                return y.to(x.dtype)
     W = W.to(x.dtype)
     shp = x.shape[:-1] + (W.shape[1],)
     if _on_gpu(x) and torch.is_grad_enabled() \

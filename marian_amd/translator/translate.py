@@ -40,6 +40,13 @@ def load_model_for_inference(path: str, options: Options, device):
         model = model.to(torch.bfloat16)
         model.set_compute_dtype(torch.bfloat16)
     model.eval()
+    # CPU int8 production decode (fbgemm packed linears,
+    # ref: tensors/cpu/ intgemm backend + --optimize)
+    if device.type == "cpu" and (
+            options.get("gemm-precision", "float32") == "int8"
+            or options.get("optimize", False)):
+        from ..ops.cpu_int8 import enable_int8
+        enable_int8(True)
     knn = [int(x) for x in (options.list("output-approx-knn") or [])]
     if len(knn) >= 2 and hasattr(model, "enable_lsh"):
         model.enable_lsh(knn[0], knn[1])
