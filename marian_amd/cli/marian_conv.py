@@ -82,7 +82,30 @@ def main(argv=None):
                    help="binarize a lexical shortlist: "
                         "--shortlist DIC SRC_VOCAB TRG_VOCAB "
                         "(--from/--to are the dic in/out paths)")
+    p.add_argument("--export-as", default="",
+                   help="onnx: export encoder+decoder graphs "
+                        "(<to>.encoder.onnx / <to>.decoder.onnx, "
+                        "ref: marian-conv --export-as onnx)")
     args = p.parse_args(argv)
+
+    if args.export_as:
+        if args.export_as not in ("onnx", "onnx-encode"):
+            raise SystemExit(f"unknown --export-as {args.export_as}")
+        from ..common.options import Options
+        from ..models.model_factory import create_model
+        items, cfg = io_npz.load_items(args.src)
+        if not cfg:
+            raise SystemExit("model has no embedded config "
+                             "(special:model.yml) - cannot rebuild")
+        opts = Options(dict(cfg))
+        model = create_model(opts)
+        model.load_marian_items(items, strict=False)
+        from ..translator.onnx_export import export_onnx
+        vocab = (opts.list("dim-vocabs") or [32000])[-1]
+        prefix = args.dst[:-5] if args.dst.endswith(".onnx") else args.dst
+        files = export_onnx(model, prefix, int(vocab))
+        print("Exported " + ", ".join(files))
+        return 0
 
     if args.shortlist:
         # text .dic -> id-resolved .blex.npz (ref: marian-conv --shortlist)

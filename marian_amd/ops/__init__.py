@@ -216,8 +216,11 @@ def layer_norm(x: torch.Tensor, scale: torch.Tensor, bias: torch.Tensor,
     if _on_gpu(x):
         from ._layernorm import LayerNormFn
         return LayerNormFn.apply(x, scale, bias, eps)
+    # normalized_shape from the WEIGHT (static at ONNX-trace time; using
+    # x.shape[-1] traces as a dynamic Gather the LN symbolic rejects)
     return torch.nn.functional.layer_norm(
-        x.float(), (x.shape[-1],), scale.float(), bias.float(), eps).to(x.dtype)
+        x.float(), (scale.shape[-1],), scale.float(), bias.float(),
+        eps).to(x.dtype)
 
 
 def layer_norm_residual(x: torch.Tensor, res: torch.Tensor,
