@@ -78,6 +78,92 @@ class GRUCellM(nn.Module):
             h = cell(torch.zeros_like(h), h)
         return h
 
+    # packed-state protocol (shared by LSTM/SSRU variants)
+    state_mult = 1
+
+    def out(self, state):
+        return state
+
+
+class LSTMCellM(nn.Module):
+    """LSTM cell with marian-style [in, out] weights (ref: rnn/cells.h
+    LSTM). State is PACKED [h, c] along the feature dim so the s2s
+    plumbing (masked updates, beam reordering) stays single-tensor."""
+
+    state_mult = 2
+
+    def __init__(self, dim_in: int, dim: int, layer_norm: bool = False,
+                 transition_depth: int = 1):
+        super().__init__()
+        self.W = nn.Parameter(torch.empty(dim_in, 4 * dim))  # i,f,o,c
+        self.U = nn.Parameter(torch.empty(dim, 4 * dim))
+        self.b = nn.Parameter(torch.zeros(4 * dim))
+        self.dim = dim
+        for w in (self.W, self.U):
+            nn.init.xavier_uniform_(w)
+        self.ln = layer_norm
+        if layer_norm:
+            self.ln_g = nn.Parameter(torch.ones(4 * dim))
+            self.ln_gb = nn.Parameter(torch.zeros(4 * dim))
+
+    def out(self, state):
+        return state[..., :self.dim]
+
+    def forward(self, x, state):
+        dt = x.dtype
+        h, c = state[..., :self.dim], state[..., self.dim:]
+        gates = x @ self.W.to(dt) + h @ self.U.to(dt) + self.b.to(dt)
+        if self.ln:
+            mu = gates.mean(-1, keepdim=True)
+            sd = gates.var(-1, unbiased=False, keepdim=True).add(1e-9).sqrt()
+            gates = (gates - mu) / sd * self.ln_g.to(dt) + self.ln_gb.to(dt)
+        i, f, o, g = gates.chunk(4, dim=-1)
+        i, f, o = torch.sigmoid(i), torch.sigmoid(f), torch.sigmoid(o)
+        c = f * c + i * torch.tanh(g)
+        h = o * torch.tanh(c)
+        return torch.cat([h, c], dim=-1)
+
+
+class SSRUCellM(nn.Module):
+    """SSRU — Simpler Simple Recurrent Unit (ref: rnn/cells.h SSRU, the
+    cell marian's production student decoders use):
+        f = sigmoid(Wf x + bf);  c = f*c + (1-f)*(W x);  h = ReLU(c)
+    State is the cell value c (single tensor)."""
+
+    state_mult = 1
+
+    def __init__(self, dim_in: int, dim: int, layer_norm: bool = False,
+                 transition_depth: int = 1):
+        super().__init__()
+        self.W = nn.Parameter(torch.empty(dim_in, dim))
+        self.Wf = nn.Parameter(torch.empty(dim_in, dim))
+        self.bf = nn.Parameter(torch.zeros(dim))
+        self.dim = dim
+        for w in (self.W, self.Wf):
+            nn.init.xavier_uniform_(w)
+        self.ln = layer_norm
+
+    def out(self, state):
+        return torch.relu(state)
+
+    def forward(self, x, c):
+        dt = x.dtype
+        f = torch.sigmoid(x @ self.Wf.to(dt) + self.bf.to(dt))
+        return f * c + (1.0 - f) * (x @ self.W.to(dt))
+
+
+_CELLS = {"gru": GRUCellM, "lstm": LSTMCellM, "ssru": SSRUCellM}
+
+
+def make_cell(kind: str, dim_in: int, dim: int, layer_norm: bool = False,
+              transition_depth: int = 1):
+    """--enc-cell / --dec-cell factory (ref: rnn/constructors.h)."""
+    kind = (kind or "gru").lower()
+    if kind not in _CELLS:
+        raise ValueError(f"unknown RNN cell type: {kind} "
+                         f"(supported: {sorted(_CELLS)})")
+    return _CELLS[kind](dim_in, dim, layer_norm, transition_depth)
+
 
 class MLPAttention(nn.Module):
     """Bahdanau attention: score = v^T tanh(W s + U h) (rnn/attention.cpp,
@@ -167,10 +253,13 @@ class S2SModel(nn.Module):
         ecd = int(options.get("enc-cell-depth", 1) or 1)
         dbd = int(options.get("dec-cell-base-depth", 2) or 2)
         dhd = int(options.get("dec-cell-high-depth", 1) or 1)
-        self.enc_fwd = GRUCellM(e, d, ln, ecd)
-        self.enc_bwd = GRUCellM(e, d, ln, ecd)
+        enc_cell = str(options.get("enc-cell", "gru") or "gru")
+        dec_cell = str(options.get("dec-cell", "gru") or "gru")
+        self.enc_cell_kind, self.dec_cell_kind = enc_cell, dec_cell
+        self.enc_fwd = make_cell(enc_cell, e, d, ln, ecd)
+        self.enc_bwd = make_cell(enc_cell, e, d, ln, ecd)
         self.enc_deep = nn.ModuleList(
-            [GRUCellM(2 * d if i == 0 else d, d, ln, ecd)
+            [make_cell(enc_cell, 2 * d if i == 0 else d, d, ln, ecd)
              for i in range(max(0, self.enc_depth - 1))])
         ctx_dim = 2 * d if not self.enc_deep else d
         self.ctx_dim = ctx_dim
@@ -178,11 +267,12 @@ class S2SModel(nn.Module):
         self.dec_init = Dense(ctx_dim, d)        # mean-pooled init state
         # marian's conditional-GRU base depth counts cell1+cell2; extra
         # base transitions attach to cell2 (ref: s2s.h decoder cell)
-        self.cell1 = GRUCellM(e, d, ln)
+        self.cell1 = make_cell(dec_cell, e, d, ln)
         self.att = MLPAttention(d, ctx_dim, d)
-        self.cell2 = GRUCellM(ctx_dim, d, ln, max(1, dbd - 1))
+        self.cell2 = make_cell(dec_cell, ctx_dim, d, ln, max(1, dbd - 1))
         self.dec_deep = nn.ModuleList(
-            [GRUCellM(d, d, ln, dhd) for _ in range(max(0, self.dec_depth - 1))])
+            [make_cell(dec_cell, d, d, ln, dhd)
+             for _ in range(max(0, self.dec_depth - 1))])
         self.dropout_rnn = float(options.get("dropout-rnn", 0.0) or 0.0)
         self.skip_connections = bool(options.get("skip", False))
 
@@ -211,28 +301,30 @@ class S2SModel(nn.Module):
         dt = emb.dtype
         B, T = emb.shape[:2]
         mask = src_mask.to(dt)
-        hf = torch.zeros(B, self.dim_rnn, device=emb.device, dtype=dt)
+        hf = torch.zeros(B, self.dim_rnn * self.enc_fwd.state_mult,
+                         device=emb.device, dtype=dt)
         hb = torch.zeros_like(hf)
         fwd_states, bwd_states = [], [None] * T
         for t in range(T):
             hf_new = self.enc_fwd(emb[:, t], hf)
             hf = mask[:, t:t + 1] * hf_new + (1 - mask[:, t:t + 1]) * hf
-            fwd_states.append(hf)
+            fwd_states.append(self.enc_fwd.out(hf))
         for t in range(T - 1, -1, -1):
             hb_new = self.enc_bwd(emb[:, t], hb)
             hb = mask[:, t:t + 1] * hb_new + (1 - mask[:, t:t + 1]) * hb
-            bwd_states[t] = hb
+            bwd_states[t] = self.enc_bwd.out(hb)
         out = torch.stack([torch.cat([f, b], dim=-1)
                            for f, b in zip(fwd_states, bwd_states)], dim=1)
         for cell in self.enc_deep:
             if self.dropout_rnn and self.training:   # --dropout-rnn
                 out = torch.nn.functional.dropout(out, self.dropout_rnn,
                                                   True)
-            h = torch.zeros(B, self.dim_rnn, device=emb.device, dtype=dt)
+            h = torch.zeros(B, self.dim_rnn * cell.state_mult,
+                            device=emb.device, dtype=dt)
             states = []
             for t in range(T):
                 h = cell(out[:, t], h)
-                states.append(h)
+                states.append(cell.out(h))
             new_out = torch.stack(states, dim=1)
             # --skip: residual connections between deep RNN layers
             out = new_out + out if (self.skip_connections
@@ -245,14 +337,21 @@ class S2SModel(nn.Module):
         dt = enc_out.dtype
         m = src_mask.to(dt).unsqueeze(-1)
         mean = (enc_out * m).sum(1) / m.sum(1).clamp_min(1.0)
-        return torch.tanh(self.dec_init(mean))
+        h0 = torch.tanh(self.dec_init(mean))
+        if self.cell1.state_mult > 1:   # LSTM: start with zero cell state
+            h0 = torch.cat([h0] + [torch.zeros_like(h0)]
+                           * (self.cell1.state_mult - 1), dim=-1)
+        return h0
 
     def _step_cells(self, emb_t, state, enc_proj, enc_out, src_mask):
         s1 = self.cell1(emb_t, state)
-        ctx, align = self.att(s1, enc_proj, enc_out, src_mask)
+        ctx, align = self.att(self.cell1.out(s1), enc_proj, enc_out,
+                              src_mask)
         s = self.cell2(ctx, s1)
         for cell in self.dec_deep:
-            s = cell(s, s)
+            # all decoder cells share one kind -> equal packed widths;
+            # input = previous level's OUTPUT, recurrent state = packed s
+            s = cell(self.cell2.out(s), s)
         return s, ctx, align
 
     def _logits(self, state, ctx, emb_t):
@@ -278,7 +377,8 @@ class S2SModel(nn.Module):
         for t in range(T):
             state, ctx, _ = self._step_cells(emb[:, t], state, enc_proj,
                                              enc_out, sm)
-            logits.append(self._logits(state, ctx, emb[:, t]))
+            logits.append(self._logits(self.cell2.out(state), ctx,
+                                       emb[:, t]))
         return torch.stack(logits, dim=1)
 
     def forward(self, batch):
@@ -303,7 +403,7 @@ class S2SModel(nn.Module):
                                              cache["enc_proj"], enc_out,
                                              src_mask.to(dt))
         cache["state"] = state
-        logits = self._logits(state, ctx, emb)
+        logits = self._logits(self.cell2.out(state), ctx, emb)
         if shortlist is not None:
             logits = logits.index_select(-1, shortlist)
         return logits
@@ -317,7 +417,20 @@ class S2SModel(nn.Module):
             m["encoder_Wemb"] = self.src_embed
             m["decoder_Wemb"] = self.trg_embed
 
-        def gru(prefix, cell: GRUCellM):
+        def gru(prefix, cell):
+            if isinstance(cell, LSTMCellM):
+                m[prefix + "_W"] = cell.W
+                m[prefix + "_U"] = cell.U
+                m[prefix + "_b"] = cell.b
+                if cell.ln:
+                    m[prefix + "_gamma1"] = cell.ln_g
+                    m[prefix + "_beta1"] = cell.ln_gb
+                return
+            if isinstance(cell, SSRUCellM):
+                m[prefix + "_W"] = cell.W
+                m[prefix + "_Wf"] = cell.Wf
+                m[prefix + "_bf"] = cell.bf
+                return
             m[prefix + "_W"] = cell.W
             m[prefix + "_U"] = cell.U
             m[prefix + "_b"] = cell.b
