@@ -35,9 +35,12 @@ def _packed_linear(W: torch.Tensor, b: Optional[torch.Tensor]):
     """Per-channel int8 packed linear for weight W [K, N] (+ bias [N]),
     cached by (data_ptr, shape) — decode weights are static."""
     key = (W.data_ptr(), tuple(W.shape), 0 if b is None else b.data_ptr())
-    ql = _qcache.get(key)
-    if ql is not None:
-        return ql
+    hit = _qcache.get(key)
+    # fingerprint guards against allocator pointer reuse after a model
+    # reload (stale packed weights would silently decode garbage)
+    fp = (float(W.reshape(-1)[0]), float(W.reshape(-1)[-1]))
+    if hit is not None and hit[1] == fp:
+        return hit[0]
     K, N = W.shape
     lin = torch.nn.Linear(K, N, bias=b is not None)
     with torch.no_grad():
@@ -51,7 +54,7 @@ def _packed_linear(W: torch.Tensor, b: Optional[torch.Tensor]):
     ql = qm[0]
     if len(_qcache) > 512:
         _qcache.clear()
-    _qcache[key] = ql
+    _qcache[key] = (ql, fp)
     return ql
 
 

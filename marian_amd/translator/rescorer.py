@@ -54,6 +54,8 @@ class Rescorer:
         with open(nbest_path) as f:
             for ln in f:
                 parts = ln.rstrip("\n").split(" ||| ")
+                if len(parts) < 2:
+                    continue   # malformed n-best line: skip, keep going
                 entries.append((int(parts[0]), parts))
         from ..data.batch import CorpusBatch, SubBatch
         mb = int(opts.get("mini-batch", 64) or 64)

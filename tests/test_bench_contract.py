@@ -52,3 +52,23 @@ def test_graft_entry_build_and_smoke_importable():
     sys.path.insert(0, ROOT)
     ge = importlib.import_module("__graft_entry__")
     assert callable(ge.build) and callable(ge.smoke)
+
+
+@pytest.mark.timeout(900)
+def test_bench_torchrun_world2_cpu():
+    """The driver's exact multi-GPU launch shape: torchrun --nnodes=1
+    --nproc-per-node 2 bench.py --gpus 2 ... must rendezvous (gloo on
+    CPU) and print one whole-job JSON line from rank 0."""
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29771", "bench.py", "--gpus", "2",
+         "--model", "transformer-tiny", "--steps", "2", "--warmup", "1"],
+        cwd=ROOT, timeout=600, capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, r.stdout   # exactly one JSON line (rank 0)
+    j = json.loads(lines[0])
+    assert j["n_gpus"] == 2
+    assert j["config"]["parallelism"] == "dp2"
+    assert j["config"]["global_batch"] == 16  # 8 per rank * 2
