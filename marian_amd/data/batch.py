@@ -60,6 +60,15 @@ class SubBatch:
                         self.mask.to(device, non_blocking=non_blocking),
                         n_tokens=self.n_tokens)
 
+    def pin_memory(self) -> "SubBatch":
+        """Pin the host buffers so .to(cuda, non_blocking=True) is a true
+        async DMA (pageable-source copies serialize the stream); called
+        from the prefetch thread, off the training hot path."""
+        if self.indices.is_cuda or self.indices.is_pinned():
+            return self
+        return SubBatch(self.indices.pin_memory(), self.mask.pin_memory(),
+                        n_tokens=self.n_tokens)
+
     def narrow(self, start: int, size: int) -> "SubBatch":
         return SubBatch(self.indices[start:start + size].clone(),
                         self.mask[start:start + size].clone())
@@ -104,6 +113,14 @@ class CorpusBatch:
 
     def target_words(self) -> int:
         return self.sub[-1].batch_words()
+
+    def pin_memory(self) -> "CorpusBatch":
+        return CorpusBatch([sb.pin_memory() for sb in self.sub],
+                           None if self.guided_alignment is None
+                           else self.guided_alignment.pin_memory(),
+                           None if self.data_weights is None
+                           else self.data_weights.pin_memory(),
+                           self.sentence_ids)
 
     def to(self, device, non_blocking: bool = True) -> "CorpusBatch":
         b = CorpusBatch([sb.to(device, non_blocking) for sb in self.sub],

@@ -11,6 +11,8 @@ Streams sentence tuples into padded CorpusBatches:
 from __future__ import annotations
 
 import queue
+
+import torch
 import threading
 from typing import Iterable, Iterator, List, Optional
 
@@ -125,10 +127,14 @@ class BatchGenerator:
         SENTINEL = object()
         err: List[BaseException] = []
 
+        pin = torch.cuda.is_available()
+
         def worker():
             try:
                 for b in self._batches():
-                    q.put(b)
+                    # pin in the prefetch thread: pageable-source H2D
+                    # copies serialize the stream; pinned ones are async
+                    q.put(b.pin_memory() if pin else b)
             except BaseException as e:  # propagate to consumer
                 err.append(e)
             finally:
