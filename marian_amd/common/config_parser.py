@@ -214,6 +214,10 @@ class ConfigParser:
               "training", nargs="+")
             a("optimizer-delay", float, 1.0, "SGD update delay (gradient accumulation)", "training")
             a("sync-sgd", "bool", True, "Use synchronous SGD", "training")
+            a("amd-async-period", int, 10,
+              "Async mode (--sync-sgd=false): updates between parameter "
+              "averagings (MI355X-native stale-update analog of the "
+              "reference's Hogwild async graph group)", "training")
             a("learn-rate", float, 0.0001, "Learning rate", "training")
             a("lr-report", "bool", False, "Report learning rate", "training")
             a("lr-decay", float, 0.0, "Per-decay factor", "training")

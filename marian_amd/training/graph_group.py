@@ -630,3 +630,71 @@ class SyncGraphGroup:
         if "rank0_optimizer_step" in oitems:
             out["optimizer_step"] = oitems["rank0_optimizer_step"]
         return out
+
+
+class AsyncGraphGroup(SyncGraphGroup):
+    """Asynchronous data parallelism, MI355X-native redesign
+    (ref: training/graph_group_async.{h,cpp} — the reference's Hogwild
+    worker threads pushing/pulling stale shards against an in-process
+    parameter server).
+
+    Under one PROCESS per GPU there is no shared-memory parameter server;
+    the faithful analog of Hogwild's staleness is LOCAL SGD: each rank
+    runs independent fwd/bwd/Adam updates on its own full parameter copy
+    (zero per-step communication — updates are stale w.r.t. peers exactly
+    as in the reference's async mode) and every --amd-async-period
+    updates the ranks average parameters over RCCL/xGMI (the
+    "pull from the server" moment). Selected by --sync-sgd=false.
+
+    Optimizer moments stay rank-local (as each Hogwild worker's optimizer
+    state was); the fp32 master is refreshed from the averaged params at
+    each exchange. Rank 0's checkpoint holds post-average state.
+    """
+
+    def __init__(self, options: Options, model: torch.nn.Module,
+                 device: torch.device, rank: int = 0, world: int = 1,
+                 comm: Optional[Communicator] = None):
+        self.data_rank = rank
+        self.data_world = world
+        # local-only internals: full (unsharded) params + optimizer
+        super().__init__(options, model, device, rank=0, world=1)
+        self.pcomm = comm or create_communicator(device, world)
+        self.period = max(1, int(
+            options.get("amd-async-period", 10) or 10))
+        self._updates_since_avg = 0
+        if world > 1:
+            self.pcomm.broadcast_params(self.store.param_buffers())
+            with torch.no_grad():
+                for b, (o, s) in zip(self.store.buckets,
+                                     self.optimizer.seg):
+                    self.optimizer.master[o:o + s].copy_(
+                        b.flat_param.reshape(-1)[:s].float())
+
+    def _average_params(self):
+        with torch.no_grad():
+            for b in self.store.buckets:
+                self.pcomm.all_reduce_scalar(b.flat_param)
+                b.flat_param.div_(self.data_world)
+            for b, (o, s) in zip(self.store.buckets, self.optimizer.seg):
+                self.optimizer.master[o:o + s].copy_(
+                    b.flat_param.reshape(-1)[:s].float())
+
+    def update(self, batch):
+        lv = super().update(batch)
+        if lv is not None and self.data_world > 1:
+            self._updates_since_avg += 1
+            if self._updates_since_avg >= self.period:
+                self._updates_since_avg = 0
+                self._average_params()
+        return lv
+
+    def save_checkpoint(self, prefix: str, model_config: dict,
+                        training_state=None, extra_model_paths=()) -> None:
+        # checkpoints hold the CONSENSUS parameters
+        if self.data_world > 1:
+            self._average_params()
+        if self.data_rank == 0:
+            super().save_checkpoint(prefix, model_config, training_state,
+                                    extra_model_paths)
+        if self.data_world > 1:
+            self.pcomm.barrier()

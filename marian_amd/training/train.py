@@ -169,8 +169,17 @@ class Train:
                     self.model.enable_factored_vocab(self.vocabs[0], "src")
                 if trg_f:
                     self.model.enable_factored_vocab(self.vocabs[-1], "trg")
-        self.graph_group = SyncGraphGroup(options, self.model, self.device,
-                                          self.rank, self.world)
+        if bool(options.get("sync-sgd", True)):
+            self.graph_group = SyncGraphGroup(options, self.model,
+                                              self.device, self.rank,
+                                              self.world)
+        else:
+            # --sync-sgd=false: asynchronous mode (local SGD with
+            # periodic parameter averaging — ref: graph_group_async)
+            from .graph_group import AsyncGraphGroup
+            self.graph_group = AsyncGraphGroup(options, self.model,
+                                               self.device, self.rank,
+                                               self.world)
         self.state = TrainingState()
         self.model_path = options.get("model", "model.npz")
 

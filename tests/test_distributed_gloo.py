@@ -459,3 +459,58 @@ def test_reshard_4_to_2():
         got.append(g)
         off += sh2[bi]
     np.testing.assert_array_equal(np.concatenate(got), glob)
+
+
+def _async_worker(rank, world, port, result_dir, n_steps):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from marian_amd.data.synthetic import synthetic_batch
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.training.graph_group import AsyncGraphGroup
+
+    torch.manual_seed(123 + rank)   # DIFFERENT init; broadcast aligns it
+    opts = Options(dict(tiny_options(), **{"amd-async-period": 3}))
+    model = create_model(opts)
+    gg = AsyncGraphGroup(opts, model, torch.device("cpu"), rank, world)
+    gg.set_lr(0.05)
+    first = last = None
+    for step in range(n_steps):
+        batch = synthetic_batch(6, 9, 9, 43, seed=500 + step * world + rank)
+        batch.sub[-1].indices.copy_(batch.sub[0].indices)  # copy task
+        lv = gg.update(batch)
+        ce = float(lv.ce_sum) / lv.label_count
+        first = first if first is not None else ce
+        last = ce
+    gg._average_params()
+    params = torch.cat([b.flat_param[:b.used]
+                        for b in gg.store.buckets]).numpy()
+    with open(os.path.join(result_dir, f"async{rank}.pkl"), "wb") as f:
+        pickle.dump({"params": params, "first": first, "last": last}, f)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_async_graph_group_world2(tmp_path):
+    """--sync-sgd=false (AsyncGraphGroup, local SGD + periodic parameter
+    averaging — the one-process-per-GPU analog of the reference's Hogwild
+    async mode): world-2 training converges on a copy task; initial
+    broadcast aligns different seeds; a final average leaves every rank
+    with identical parameters."""
+    import numpy as np
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_async_worker,
+                         args=(r, 2, 29573, str(tmp_path), 12))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    r0 = pickle.load(open(tmp_path / "async0.pkl", "rb"))
+    r1 = pickle.load(open(tmp_path / "async1.pkl", "rb"))
+    np.testing.assert_allclose(r0["params"], r1["params"], atol=1e-6)
+    assert r0["last"] < r0["first"] - 0.3, (r0["first"], r0["last"])
+    assert r1["last"] < r1["first"] - 0.3
