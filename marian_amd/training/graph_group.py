@@ -504,10 +504,10 @@ class SyncGraphGroup:
                 io_npz.save_items(p, items, config=model_config)
         self.swap_smoothed(False)
         # every rank contributes its optimizer shard; single-process path
-        # gathers locally. (multi-proc: each rank writes its shard file,
-        # rank 0 merges on load; simple + correct for single node)
-        opt_items = {f"rank{self.rank}_{k}": v
-                     for k, v in self.optimizer.state_items().items()}
+        # serializes locally, multi-process gathers tensors below
+        opt_items = ({f"rank{self.rank}_{k}": v
+                      for k, v in self.optimizer.state_items().items()}
+                     if self.world == 1 else {})
         if self.rank == 0:
             # persist the bucket layout so a resume under a different
             # world size OR bucket grouping (bucketing is disabled at
@@ -517,14 +517,23 @@ class SyncGraphGroup:
             opt_items["bucket_used"] = np.array(
                 [b.used for b in self.store.buckets], dtype=np.int64)
         if self.world > 1:
+            # gather the state VECTORS as tensors (RCCL all_gather over
+            # xGMI) instead of pickling ~GBs through gather_object —
+            # the r1 path hitched every --save-freq at scale
             import torch.distributed as dist
-            gathered: List[Optional[dict]] = [None] * self.world
-            dist.gather_object(opt_items, gathered if self.rank == 0 else None,
-                               dst=0)
+            merged = dict(opt_items) if self.rank == 0 else {}
+            for k, t in self.optimizer.state_tensors().items():
+                tc = t.contiguous()
+                bufs = [torch.empty_like(tc) for _ in range(self.world)]
+                dist.all_gather(bufs, tc)
+                if self.rank == 0:
+                    for r, bt in enumerate(bufs):
+                        merged[f"rank{r}_{k}"] = bt.cpu().numpy()
             if self.rank == 0:
-                merged = {}
-                for d in gathered:
-                    merged.update(d or {})
+                import numpy as np
+                for r in range(self.world):
+                    merged[f"rank{r}_optimizer_step"] = np.array(
+                        [self.optimizer.t], dtype=np.int64)
                 io_npz.save_items(prefix + ".optimizer.npz", merged)
         else:
             io_npz.save_items(prefix + ".optimizer.npz", opt_items)

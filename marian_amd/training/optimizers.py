@@ -447,6 +447,18 @@ class ShardedOptimizer:
 
     # ------------------------------------------------------------------
     # checkpoint (ref: model.npz.optimizer.npz via gatherState/scatterState)
+    def state_tensors(self) -> Dict[str, torch.Tensor]:
+        """Device-resident state vectors (equal-sized across ranks) for
+        the tensor-gather checkpoint path — the pickle-based
+        gather_object of ~GB state hitched every --save-freq."""
+        self.sync_t_from_device()
+        out = {"master_parameters": self.master}
+        if self.smooth is not None:
+            out["exp_smoothing"] = self.smooth
+        for k, v in self.core.state_items().items():
+            out[k] = v
+        return out
+
     def state_items(self) -> Dict[str, np.ndarray]:
         self.sync_t_from_device()
         out = {"master_parameters": self.master.cpu().numpy(),
