@@ -185,8 +185,10 @@ def linear(x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor] = None
         return fp8_linear(x, W.to(x.dtype), b.to(x.dtype) if b is not None else None)
     if not x.is_cuda:
         from .cpu_int8 import int8_enabled, int8_linear
-        if int8_enabled():
-            y = int8_linear(x, W.float(),
+        # fp32 weights only: .float() on other dtypes would allocate a
+        # fresh tensor per call and defeat the packed-weight cache
+        if int8_enabled() and W.dtype == torch.float32:
+            y = int8_linear(x.float(), W,
                             b.float() if b is not None else None)
             if y is not None:
                 return y.to(x.dtype)
