@@ -69,7 +69,7 @@ def main():
                                         True, D ** -0.5),
               iters, flops=int(2.5 * f2))
 
-    if which in ("layernorm", "all"):
+    if which in ("layernorm", "ln", "all"):
         R, d = 32768, 1024
         x = torch.randn(R, d, device=dev, dtype=torch.bfloat16)
         g = torch.rand(d, device=dev, dtype=torch.bfloat16) + 0.5
