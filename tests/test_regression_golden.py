@@ -390,3 +390,65 @@ def test_right_left_training_golden(tmp_path):
         lv = tr.graph_group.criterion(tr.model(probe), probe)
     check_golden("training_right_left_cost", {
         "probe_ce_per_word": float(lv.ce_sum) / lv.label_count}, tol=5e-3)
+
+
+def test_data_weighting_training_golden(tmp_path):
+    """--data-weighting (per-sentence) locks its cost curve."""
+    d = tmp_path
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    src = str(d / "c.src")
+    write_synthetic_corpus(src, str(d / "c.trg"), 120, 40, 4, 7, seed=33)
+    with open(d / "c.w", "w") as f:
+        for i in range(120):
+            f.write(f"{1.0 + (i % 3) * 0.5}\n")
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    argv = ["--train-sets", src, src,
+            "--vocabs", str(d / "v.yml"), str(d / "v.yml"),
+            "--model", str(d / "model.npz"),
+            "--data-weighting", str(d / "c.w"),
+            "--data-weighting-type", "sentence",
+            "--dim-emb", "48", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "96",
+            "--mini-batch", "20", "--learn-rate", "0.01",
+            "--after-batches", "15", "--seed", "6", "--cpu-threads", "1",
+            "--tied-embeddings-all", "--overwrite", "--quiet",
+            "--shuffle", "none"]
+    opts = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    tr = Train(opts)
+    tr.run()
+    from marian_amd.data.synthetic import synthetic_batch
+    probe = synthetic_batch(6, 6, 6, len(tr.vocabs[0]), seed=46)
+    with torch.no_grad():
+        lv = tr.graph_group.criterion(tr.model(probe), probe)
+    check_golden("training_data_weighting_cost", {
+        "probe_ce_per_word": float(lv.ce_sum) / lv.label_count}, tol=5e-3)
+
+
+def test_lm_training_golden(tmp_path):
+    """--type lm-transformer decoder-only curve locked."""
+    d = tmp_path
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    src = str(d / "c.src")
+    write_synthetic_corpus(src, str(d / "c.trg"), 150, 40, 4, 9, seed=35)
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    argv = ["--train-sets", src,
+            "--vocabs", str(d / "v.yml"),
+            "--model", str(d / "model.npz"), "--type", "lm-transformer",
+            "--dim-emb", "48", "--dec-depth", "2",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "96",
+            "--mini-batch", "20", "--learn-rate", "0.01",
+            "--after-batches", "20", "--seed", "8", "--cpu-threads", "1",
+            "--tied-embeddings-all", "--overwrite", "--quiet",
+            "--shuffle", "none"]
+    opts = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    tr = Train(opts)
+    tr.run()
+    from marian_amd.data.synthetic import synthetic_batch
+    probe = synthetic_batch(6, 6, 6, len(tr.vocabs[0]), seed=47)
+    probe.sub = [probe.sub[0], probe.sub[0]]
+    with torch.no_grad():
+        lv = tr.graph_group.criterion(tr.model(probe), probe)
+    check_golden("training_lm_cost", {
+        "probe_ce_per_word": float(lv.ce_sum) / lv.label_count}, tol=5e-3)
