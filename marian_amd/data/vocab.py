@@ -105,15 +105,19 @@ class Vocab:
 
     # -- persistence -------------------------------------------------------
     def save(self, path: str) -> None:
+        # atomic (tmp + rename): a concurrent reader (another rank whose
+        # existence check raced the creation) must never see a partial file
         ext = os.path.splitext(path)[1].lower()
+        tmp = path + ".tmp"
         if ext in (".yml", ".yaml", ".json"):
-            with open(path, "w", encoding="utf-8") as f:
+            with open(tmp, "w", encoding="utf-8") as f:
                 yaml.safe_dump(self.token2id, f, default_flow_style=False,
                                allow_unicode=True, sort_keys=False)
         else:
-            with open(path, "w", encoding="utf-8") as f:
+            with open(tmp, "w", encoding="utf-8") as f:
                 for i in range(len(self)):
                     f.write(self.id2token.get(i, UNK_STR) + "\n")
+        os.replace(tmp, path)
 
     # -- encode/decode -----------------------------------------------------
     def __len__(self) -> int:

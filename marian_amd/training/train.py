@@ -76,9 +76,13 @@ def load_or_create_vocabs(options: Options) -> List[Vocab]:
     from ..parallel.communicator import dist_env
     rank, _lr, world = dist_env()
     missing = [p for p in paths if not os.path.exists(p)]
-    if world > 1 and missing:
+    # the barrier must be entered by EVERY rank unconditionally: deciding
+    # on a per-rank `missing` check raced (a rank whose check ran after
+    # rank 0 created the files skipped the barrier and could read a file
+    # mid-write while rank 0's peers waited — flaky startup crash)
+    if world > 1:
         import torch.distributed as dist
-        if rank == 0:
+        if rank == 0 and missing:
             for i, p in enumerate(paths):
                 if os.path.exists(p):
                     continue

@@ -514,3 +514,60 @@ def test_async_graph_group_world2(tmp_path):
     np.testing.assert_allclose(r0["params"], r1["params"], atol=1e-6)
     assert r0["last"] < r0["first"] - 0.3, (r0["first"], r0["last"])
     assert r1["last"] < r1["first"] - 0.3
+
+
+def _async_ckpt_worker(rank, world, port, workdir):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    argv = ["--train-sets", f"{workdir}/c.src", f"{workdir}/c.src",
+            "--vocabs", f"{workdir}/v.yml", f"{workdir}/v.yml",
+            "--model", f"{workdir}/m.npz", "--sync-sgd", "false",
+            "--amd-async-period", "2",
+            "--dim-emb", "16", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "32",
+            "--mini-batch", "4", "--after-batches", "6",
+            "--cpu-threads", "1", "--seed", "3", "--overwrite", "--quiet"]
+    o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    t = Train(o)
+    t.run()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_async_mode_checkpoint_resume(tmp_path):
+    """marian-train --sync-sgd=false (AsyncGraphGroup) end-to-end at
+    world 2: trains, saves consensus checkpoint, and a SINGLE-process
+    resume loads it (async rank-0 state == world-1 layout)."""
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.training.train import Train
+    write_synthetic_corpus(str(tmp_path / "c.src"), str(tmp_path / "c.trg"),
+                           24, 25, 3, 7, seed=13)
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_async_ckpt_worker,
+                         args=(r, 2, 29583, str(tmp_path)))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    assert os.path.exists(tmp_path / "m.npz")
+    assert os.path.exists(str(tmp_path / "m.npz") + ".optimizer.npz")
+    # resume single-process, sync mode
+    argv = ["--train-sets", str(tmp_path / "c.src"), str(tmp_path / "c.src"),
+            "--vocabs", str(tmp_path / "v.yml"), str(tmp_path / "v.yml"),
+            "--model", str(tmp_path / "m.npz"),
+            "--dim-emb", "16", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "32",
+            "--mini-batch", "4", "--after-batches", "8",
+            "--cpu-threads", "1", "--seed", "3", "--overwrite", "--quiet"]
+    o = ConfigParser(MODE_TRAINING).parse_options(argv, validate=False)
+    t = Train(o)
+    assert t.graph_group.optimizer.t > 0   # optimizer state restored
+    t.run()
