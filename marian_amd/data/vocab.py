@@ -34,6 +34,17 @@ class Vocab:
 
     # -- construction ------------------------------------------------------
     @classmethod
+    def from_token2id(cls, mapping: Dict[str, int]) -> "Vocab":
+        """Build from an in-memory token->id dict (e.g. the `dictionary`
+        embedded in a LASER checkpoint)."""
+        v = cls()
+        for tok, idx in mapping.items():
+            v.token2id[str(tok)] = int(idx)
+            v.id2token[int(idx)] = str(tok)
+        v._require_special()
+        return v
+
+    @classmethod
     def load(cls, path: str, max_size: int = 0) -> "Vocab":
         v = cls()
         ext = os.path.splitext(path)[1].lower()

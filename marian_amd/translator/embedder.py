@@ -25,9 +25,23 @@ class Embedder:
         self.device = (torch.device("cuda", 0) if torch.cuda.is_available()
                        and int(options.get("cpu-threads", 0) or 0) == 0
                        else torch.device("cpu"))
+        model_path = options.list("models")[0]
+        from ..models.laser import LaserEncoder, is_laser_checkpoint
+        self.laser = None
+        if is_laser_checkpoint(model_path):
+            # LASER BiLSTM checkpoint (ref: models/laser.h): vocab may be
+            # embedded in the checkpoint's `dictionary`
+            self.laser, vdict = LaserEncoder.load(model_path, self.device)
+            if options.list("vocabs"):
+                self.vocab = Vocab.load(options.list("vocabs")[0])
+            else:
+                assert vdict is not None, \
+                    "LASER checkpoint has no dictionary; pass --vocabs"
+                self.vocab = Vocab.from_token2id(vdict)
+            return
         self.vocab = Vocab.load(options.list("vocabs")[0])
-        self.model = load_model_for_inference(
-            options.list("models")[0], options, self.device)
+        self.model = load_model_for_inference(model_path, options,
+                                              self.device)
 
     @torch.no_grad()
     def embed_lines(self, lines):
@@ -40,9 +54,12 @@ class Embedder:
         out = {}
         for batch in gen:
             b = batch.to(self.device)
-            enc = self.model.encode(b.source.indices, b.source.mask)
-            m = b.source.mask.unsqueeze(-1).to(enc.dtype)
-            vec = (enc * m).sum(1) / m.sum(1).clamp_min(1.0)
+            if self.laser is not None:
+                vec = self.laser(b.source.indices, b.source.mask)
+            else:
+                enc = self.model.encode(b.source.indices, b.source.mask)
+                m = b.source.mask.unsqueeze(-1).to(enc.dtype)
+                vec = (enc * m).sum(1) / m.sum(1).clamp_min(1.0)
             for i, sid in enumerate(batch.sentence_ids):
                 out[sid] = vec[i].float().cpu().numpy()
         return [out[i] for i in range(len(out))]

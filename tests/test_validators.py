@@ -99,3 +99,44 @@ def test_valid_script_path(tmp_path):
     t.run()
     assert "translation" in t.state.validators
     assert abs(t.state.validators["translation"]["best"] - 41.5) < 1e-6
+
+
+def test_laser_checkpoint_embedding(tmp_path):
+    """LASER-layout .pt checkpoint (public facebookresearch/LASER
+    encoder.pt format: params + fairseq BiLSTM state dict + dictionary,
+    ref: models/laser.h) loads through marian-embedder and produces
+    max-pooled bidirectional embeddings matching a direct nn.LSTM
+    reference."""
+    import numpy as np
+    import torch
+    V, E, H, L = 30, 8, 12, 2
+    torch.manual_seed(3)
+    lstm = torch.nn.LSTM(E, H, num_layers=L, bidirectional=True)
+    emb = torch.nn.Embedding(V, E, padding_idx=1)
+    sd = {"embed_tokens.weight": emb.weight.detach()}
+    for k, v in lstm.state_dict().items():
+        sd[f"lstm.{k}"] = v
+    tok2id = {"</s>": 0, "<unk>": 1}
+    for i in range(2, V):
+        tok2id[f"w{i}"] = i
+    ckpt = {"params": {"num_embeddings": V, "embed_dim": E,
+                       "hidden_size": H, "num_layers": L,
+                       "bidirectional": True, "padding_idx": 1},
+            "model": sd, "dictionary": tok2id}
+    path = tmp_path / "laser.pt"
+    torch.save(ckpt, path)
+
+    from marian_amd.common.options import Options
+    from marian_amd.translator.embedder import Embedder
+    emb_tool = Embedder(Options({"models": [str(path)], "vocabs": [],
+                                 "mini-batch": 4, "cpu-threads": 1}))
+    lines = ["w5 w6 w7", "w9 w10"]
+    vecs = emb_tool.embed_lines(lines)
+    assert len(vecs) == 2 and vecs[0].shape == (2 * H,)
+
+    # reference: same tokens through nn.LSTM directly, max-pool
+    ids = emb_tool.vocab.encode("w5 w6 w7")
+    x = emb(torch.tensor(ids)).unsqueeze(1)
+    out, _ = lstm(x)
+    ref = out.squeeze(1).max(dim=0).values.detach().numpy()
+    np.testing.assert_allclose(vecs[0], ref, atol=1e-5)
