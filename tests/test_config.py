@@ -135,3 +135,22 @@ def test_devices_flag_spawns_torchrun(monkeypatch):
     assert maybe_spawn_workers(["--devices", "3"]) is None
     got = maybe_spawn_workers(["--num-devices", "4"])
     assert got and "--nproc-per-node=4" in got[0]
+
+
+def test_round2_flags_declared():
+    """New r2 flags parse with precedence and appear in --dump-config."""
+    from marian_amd.common.config_parser import (ConfigParser,
+                                                 MODE_TRAINING,
+                                                 MODE_TRANSLATION)
+    p = ConfigParser(MODE_TRAINING)
+    o = p.parse_options(["--train-sets", "a", "b", "--sync-sgd", "false",
+                         "--amd-async-period", "7"], validate=False)
+    assert o.get("sync-sgd") is False
+    assert int(o.get("amd-async-period")) == 7
+    d = ConfigParser(MODE_TRANSLATION)
+    o2 = d.parse_options(["--models", "m.npz", "--gemm-precision", "int8"],
+                         validate=False)
+    assert o2.get("gemm-precision") == "int8"
+    o3 = d.parse_options(["--models", "m.npz", "--optimize"],
+                         validate=False)
+    assert o3.get("optimize") is True
