@@ -571,3 +571,73 @@ def test_async_mode_checkpoint_resume(tmp_path):
     t = Train(o)
     assert t.graph_group.optimizer.t > 0   # optimizer state restored
     t.run()
+
+
+def _async_path_worker(rank, world, port, result_dir, n_steps,
+                       force_async=True, tag="ap"):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from marian_amd.data.synthetic import synthetic_batch
+    from marian_amd.models.model_factory import create_model
+    from marian_amd.training.graph_group import SyncGraphGroup
+
+    torch.manual_seed(123)
+    o = dict(tiny_options())
+    o["clip-norm"] = 0.5   # exercise the device-side clip factor too
+    opts = Options(o)
+    model = create_model(opts)
+    gg = SyncGraphGroup(opts, model, torch.device("cpu"), rank, world)
+    gg._force_async = force_async   # the path every NCCL rank takes on GPU
+    gg.set_lr(0.01)
+    for step in range(n_steps):
+        batch = synthetic_batch(6, 9, 9, 43, seed=900 + step * world + rank)
+        lv = gg.update(batch)
+        assert lv is not None and lv.label_count > 0
+    params = torch.cat([b.flat_param[:b.used]
+                        for b in gg.store.buckets]).numpy()
+    with open(os.path.join(result_dir, f"{tag}{rank}.pkl"), "wb") as f:
+        pickle.dump({"params": params, "ce": float(lv.ce_sum),
+                     "labels": lv.label_count}, f)
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_device_side_update_path_world2(tmp_path):
+    """The ASYNC (device-side, no-host-sync) update path — the exact code
+    every NCCL rank runs on GPU — at world 2 over gloo: collectives for
+    the loss/label reduction, grad-stats all-reduce (clip-norm on), Adam
+    and param all-gather produce rank-identical parameters equal to the
+    host path's."""
+    import numpy as np
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_async_path_worker,
+                         args=(r, 2, 29599, str(tmp_path), 4))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    r0 = pickle.load(open(tmp_path / "ap0.pkl", "rb"))
+    r1 = pickle.load(open(tmp_path / "ap1.pkl", "rb"))
+    np.testing.assert_allclose(r0["params"], r1["params"], atol=1e-6)
+    assert r0["ce"] == r1["ce"]   # identical reduced loss on every rank
+
+    # equals the world-2 HOST path bit-for-bit (the async path mirrors
+    # the host arithmetic exactly; a single-process comparison would only
+    # test Adam's noise-gradient chaos, not the path)
+    procs = [ctx.Process(target=_async_path_worker,
+                         args=(r, 2, 29601, str(tmp_path), 4, False, "hp"))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    h0 = pickle.load(open(tmp_path / "hp0.pkl", "rb"))
+    np.testing.assert_array_equal(r0["params"], h0["params"])
+    assert r0["ce"] == pytest.approx(h0["ce"], rel=1e-9)
+    assert r0["labels"] == h0["labels"]
