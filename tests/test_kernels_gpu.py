@@ -239,7 +239,7 @@ def test_attention_fused_dispatch_inference():
 
 @pytest.mark.parametrize("causal", [False, True])
 @pytest.mark.parametrize("shape", [(2, 4, 32, 32), (2, 3, 64, 64),
-                                   (1, 2, 70, 50)])
+                                   (1, 2, 70, 50), (8, 4, 16, 16)])
 def test_attention_bwd_fused(ext, causal, shape):
     """Fused attention backward vs fp32 autograd reference."""
     torch.manual_seed(21)
