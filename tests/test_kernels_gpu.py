@@ -181,8 +181,7 @@ def test_tiny_train_step_bf16(ext):
 # ---------------------------------------------------------- fused attention
 @pytest.mark.parametrize("causal", [False, True])
 @pytest.mark.parametrize("shape", [(2, 4, 32, 40), (3, 2, 64, 64),
-                                   (2, 16, 70, 70), (1, 2, 33, 95),
-                                   (8, 4, 16, 64)])
+                                   (2, 16, 70, 70), (1, 2, 33, 95)])
 def test_attention_fwd_fused(ext, causal, shape):
     """Fused kernel vs fp32 torch reference, with prefix kv lengths."""
     torch.manual_seed(11)
@@ -239,7 +238,7 @@ def test_attention_fused_dispatch_inference():
 
 @pytest.mark.parametrize("causal", [False, True])
 @pytest.mark.parametrize("shape", [(2, 4, 32, 32), (2, 3, 64, 64),
-                                   (1, 2, 70, 50), (8, 4, 16, 16)])
+                                   (1, 2, 70, 50)])
 def test_attention_bwd_fused(ext, causal, shape):
     """Fused attention backward vs fp32 autograd reference."""
     torch.manual_seed(21)
