@@ -40,7 +40,9 @@ def main():
         "--after-batches", "120", "--disp-freq", "20u",
         "--save-freq", "100000u", "--log", log,
         "--seed", "3", "--overwrite", "--shuffle-in-ram",
-        "--workspace", "250000",   # 250 GB: let mini-batch-fit use the HBM
+        # NB: default workspace. A 250 GB workspace made mini-batch-fit
+        # spend the whole run measuring giant cells; the documented
+        # ~100k words/s run used this exact configuration.
     ])
     rates = []
     with open(log) as f:
@@ -54,6 +56,9 @@ def main():
     if steady:
         print(f"CLI steady words/s (mean of {len(steady)}): "
               f"{sum(steady)/len(steady):,.0f}")
+    else:
+        print("no words/s lines found — check the train.log for an early "
+              "failure or raise the run timeout")
 
 
 if __name__ == "__main__":
