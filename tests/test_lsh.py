@@ -45,3 +45,38 @@ def test_lsh_decode_consistency():
     got = approx.argmax(-1)
     want = full.argmax(-1)
     assert (got == want).float().mean() >= 0.5
+
+
+def test_lsh_cli_decode(tmp_path):
+    """--output-approx-knn end-to-end through the marian-decoder CLI
+    (ref: --output-approx-knn K NBITS, layers/lsh.cpp)."""
+    import os
+    from marian_amd.common.config_parser import ConfigParser, MODE_TRAINING
+    from marian_amd.data.synthetic import write_synthetic_corpus
+    from marian_amd.training.train import Train
+    src = str(tmp_path / "c.src")
+    write_synthetic_corpus(src, str(tmp_path / "c.trg"), 60, 30, 3, 7,
+                           seed=11)
+    argv = ["--train-sets", src, src,
+            "--vocabs", str(tmp_path / "v.yml"), str(tmp_path / "v.yml"),
+            "--model", str(tmp_path / "m.npz"),
+            "--dim-emb", "32", "--enc-depth", "1", "--dec-depth", "1",
+            "--transformer-heads", "2", "--transformer-dim-ffn", "64",
+            "--mini-batch", "10", "--after-batches", "10", "--seed", "4",
+            "--cpu-threads", "1", "--tied-embeddings-all",
+            "--overwrite", "--quiet"]
+    Train(ConfigParser(MODE_TRAINING).parse_options(argv,
+                                                    validate=False)).run()
+    inp = tmp_path / "in.txt"
+    inp.write_text("w5 w6 w7\nw9 w10\n")
+    from marian_amd.translator.translate import main_decoder
+    out = tmp_path / "out.txt"
+    main_decoder(["--models", str(tmp_path / "m.npz"),
+                  "--vocabs", str(tmp_path / "v.yml"),
+                  str(tmp_path / "v.yml"),
+                  "--input", str(inp), "--output", str(out),
+                  "--beam-size", "2", "--cpu-threads", "1",
+                  "--output-approx-knn", "16", "128",
+                  "--max-length", "12", "--quiet"])
+    lines = out.read_text().splitlines()
+    assert len(lines) == 2
