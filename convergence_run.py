@@ -20,6 +20,9 @@ def main():
     ap.add_argument("--workdir", default="/tmp/convergence")
     ap.add_argument("--target", type=float, default=50.0)
     ap.add_argument("--after-batches", type=int, default=1500)
+    ap.add_argument("--task", choices=["copy", "reverse"], default="copy",
+                    help="reverse: target = reversed source (needs real "
+                         "attention structure, not identity)")
     args = ap.parse_args()
     os.makedirs(args.workdir, exist_ok=True)
     d = os.path.abspath(args.workdir)
@@ -33,16 +36,22 @@ def main():
                            6150, 150, 3, 10, seed=99)
     with open(full_src) as f:
         lines = f.read().splitlines()
-    with open(os.path.join(d, "train.src"), "w") as f:
-        f.write("\n".join(lines[:6000]) + "\n")
-    with open(os.path.join(d, "valid.src"), "w") as f:
-        f.write("\n".join(lines[6000:]) + "\n")
+    if args.task == "reverse":
+        trg_lines = [" ".join(reversed(l.split())) for l in lines]
+    else:
+        trg_lines = lines
+    for name, data in (("train.src", lines[:6000]),
+                       ("train.trg", trg_lines[:6000]),
+                       ("valid.src", lines[6000:]),
+                       ("valid.trg", trg_lines[6000:])):
+        with open(os.path.join(d, name), "w") as f:
+            f.write("\n".join(data) + "\n")
 
     from marian_amd.training.train import main_train as train_main
     on_gpu = torch.cuda.is_available()
     argv = [
         "--train-sets", os.path.join(d, "train.src"),
-        os.path.join(d, "train.src"),
+        os.path.join(d, "train.trg"),
         "--vocabs", os.path.join(d, "v.yml"), os.path.join(d, "v.yml"),
         "--model", os.path.join(d, "model.npz"),
         "--dim-emb", "256", "--enc-depth", "2", "--dec-depth", "2",
@@ -56,7 +65,7 @@ def main():
         "--after-batches", str(args.after_batches),
         "--disp-freq", "50u", "--save-freq", "100000u",
         "--valid-sets", os.path.join(d, "valid.src"),
-        os.path.join(d, "valid.src"),
+        os.path.join(d, "valid.trg"),
         "--valid-freq", "100u", "--valid-metrics", "bleu-detok",
         "--valid-mini-batch", "32", "--beam-size", "4",
         "--early-stopping", "4",
