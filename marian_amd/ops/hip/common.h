@@ -76,7 +76,7 @@ __device__ __forceinline__ float block_sum(float v, float* smem) {
   if (lane == 0) smem[wid] = v;
   __syncthreads();
   const int nw = blockDim.x / WAVE;
-  float r = (threadIdx.x < nw) ? smem[threadIdx.x] : 0.f;
+  float r = ((int)threadIdx.x < nw) ? smem[threadIdx.x] : 0.f;
   r = wave_sum(r);  // only first wave holds all partials, but every wave
                     // reduces the same nw values read via broadcast below
   if (threadIdx.x == 0) smem[0] = r;
@@ -93,7 +93,7 @@ __device__ __forceinline__ float block_max(float v, float* smem) {
   if (lane == 0) smem[wid] = v;
   __syncthreads();
   const int nw = blockDim.x / WAVE;
-  float r = (threadIdx.x < nw) ? smem[threadIdx.x] : -INFINITY;
+  float r = ((int)threadIdx.x < nw) ? smem[threadIdx.x] : -INFINITY;
   r = wave_max(r);
   if (threadIdx.x == 0) smem[0] = r;
   __syncthreads();
