@@ -154,3 +154,26 @@ def test_round2_flags_declared():
     o3 = d.parse_options(["--models", "m.npz", "--optimize"],
                          validate=False)
     assert o3.get("optimize") is True
+
+
+def test_marian_vocab_cli(tmp_path, capsys):
+    """marian-vocab builds a vocab from files and from stdin-equivalent,
+    respecting --max-size (ref: command/marian_vocab.cpp)."""
+    import yaml
+    src = tmp_path / "c.txt"
+    src.write_text("a b c a a b\nd e a\n")
+    out = tmp_path / "v.yml"
+    from marian_amd.cli.marian_vocab import main as vocab_main
+    vocab_main(["--input", str(src), "--output", str(out)])
+    m = yaml.safe_load(out.read_text())
+    for tok in ("a", "b", "c", "d", "e"):
+        assert tok in m
+    # frequency order after the specials: 'a' gets the smallest id
+    content_ids = {t: i for t, i in m.items()
+                   if t not in ("</s>", "<unk>")}
+    assert content_ids["a"] == min(content_ids.values())
+    out2 = tmp_path / "v4.yml"
+    vocab_main(["--input", str(src), "--output", str(out2),
+                "--max-size", "4"])
+    m2 = yaml.safe_load(out2.read_text())
+    assert len(m2) <= 4
