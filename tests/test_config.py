@@ -177,3 +177,24 @@ def test_marian_vocab_cli(tmp_path, capsys):
                 "--max-size", "4"])
     m2 = yaml.safe_load(out2.read_text())
     assert len(m2) <= 4
+
+
+def test_marian_main_multiplexer(tmp_path):
+    """The single `marian` entry point dispatches to the tools
+    (ref: command/marian_main.cpp)."""
+    import subprocess
+    import sys as _sys
+    src = tmp_path / "c.txt"
+    src.write_text("x y z x\n")
+    out = tmp_path / "v.yml"
+    r = subprocess.run(
+        [_sys.executable, "-m", "marian_amd.cli.marian_main", "vocab",
+         "--input", str(src), "--output", str(out)],
+        capture_output=True, text=True, timeout=120,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr[-500:]
+    assert out.exists()
+    r2 = subprocess.run(
+        [_sys.executable, "-m", "marian_amd.cli.marian_main", "--help"],
+        capture_output=True, text=True, timeout=60)
+    assert "tools:" in r2.stdout
